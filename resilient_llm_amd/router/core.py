@@ -1,0 +1,233 @@
+"""Alias -> deployment routing with shuffle LB, pre-call checks, fallback
+chains and cooldown health management.
+
+This is the native replacement for the LiteLLM router the reference
+configures but never implements (SURVEY.md §2.2 X2, X3, X5, X6, X7, X9;
+reference config/config.yaml:106-114):
+
+- one alias fans out to multiple replica deployments (X2),
+- ``simple-shuffle`` = weighted random among healthy, under-quota
+  deployments (X3; weight defaults to rpm like LiteLLM's),
+- pre-call checks filter deployments that would breach their RPM/TPM
+  window BEFORE dispatch (X7),
+- when every deployment of an alias is exhausted or unhealthy the request
+  walks the alias's fallback chain (X5),
+- ``allowed_fails`` failures within a minute put a deployment in cooldown
+  for ``cooldown_time`` seconds (X6),
+- when the whole ladder is exhausted the caller maps
+  :class:`RouterRateLimit` to HTTP 429 (X9).
+
+The router is backend-agnostic: it hands out :class:`Ticket` objects; the
+gateway executes them against workers and reports success/failure back.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import random
+import threading
+import time
+from typing import Callable, Iterable, Optional
+
+from ..config import Deployment, RouterSettings
+from .token_bucket import MinuteWindowLimiter, RateSnapshot
+
+
+class NoDeploymentAvailable(Exception):
+    """No deployment of the alias (or its fallbacks) can take the request."""
+
+
+class RouterRateLimit(NoDeploymentAvailable):
+    """Every candidate was filtered by its rate window -> HTTP 429."""
+
+
+class UnknownAlias(KeyError):
+    pass
+
+
+@dataclasses.dataclass
+class DeploymentState:
+    dep: Deployment
+    limiter: MinuteWindowLimiter
+    cooldown_until: float = 0.0
+    fail_times: list = dataclasses.field(default_factory=list)
+    in_flight: int = 0
+    healthy: bool = True            # worker heartbeat / registration status
+    total_requests: int = 0
+    total_failures: int = 0
+    total_cooldowns: int = 0
+
+    def snapshot(self) -> RateSnapshot:
+        return self.limiter.snapshot()
+
+
+@dataclasses.dataclass
+class Ticket:
+    """An admitted request: which deployment, what was charged."""
+
+    state: DeploymentState
+    alias_requested: str            # what the client asked for
+    tokens_estimated: int
+    is_fallback: bool               # resolved via a fallback chain?
+    attempted: list                 # deployment model_ids tried before this one
+    done: bool = False
+
+    @property
+    def deployment(self) -> Deployment:
+        return self.state.dep
+
+
+class Router:
+    def __init__(self, deployments: Iterable[Deployment], settings: RouterSettings,
+                 clock: Callable[[], float] = time.monotonic,
+                 rng: Optional[random.Random] = None) -> None:
+        self.settings = settings
+        self._clock = clock
+        self._rng = rng or random.Random()
+        self._lock = threading.Lock()
+        self.states: list[DeploymentState] = [
+            DeploymentState(dep=d,
+                            limiter=MinuteWindowLimiter(d.rpm, d.tpm, clock=clock))
+            for d in deployments
+        ]
+        self._by_alias: dict[str, list[DeploymentState]] = {}
+        for s in self.states:
+            self._by_alias.setdefault(s.dep.model_name, []).append(s)
+
+    # ------------------------------------------------------------- lookup
+    def alias_states(self, alias: str) -> list[DeploymentState]:
+        try:
+            return self._by_alias[alias]
+        except KeyError:
+            raise UnknownAlias(alias) from None
+
+    def state_for_id(self, model_id: str) -> Optional[DeploymentState]:
+        for s in self.states:
+            if s.dep.model_id == model_id:
+                return s
+        return None
+
+    # ----------------------------------------------------------- policies
+    def _available(self, s: DeploymentState, tokens: int, exclude: set) -> bool:
+        if id(s) in exclude or not s.healthy:
+            return False
+        if s.cooldown_until > self._clock():
+            return False
+        if self.settings.enable_pre_call_checks and not s.limiter.would_admit(tokens):
+            return False
+        return True
+
+    def _shuffle_pick(self, candidates: list[DeploymentState]) -> DeploymentState:
+        if len(candidates) == 1:
+            return candidates[0]
+        strategy = self.settings.routing_strategy
+        if strategy == "round-robin":
+            # stable order by fewest total requests — deterministic spread
+            return min(candidates, key=lambda s: s.total_requests)
+        if strategy == "least-busy":
+            return min(candidates, key=lambda s: s.in_flight)
+        # simple-shuffle: weighted random (weight defaults to rpm — config.py)
+        weights = [max(1, s.dep.weight) for s in candidates]
+        return self._rng.choices(candidates, weights=weights, k=1)[0]
+
+    # ------------------------------------------------------------ acquire
+    def _try_alias(self, alias: str, tokens: int, exclude: set,
+                   attempted: list) -> Optional[DeploymentState]:
+        states = self.alias_states(alias)
+        candidates = [s for s in states if self._available(s, tokens, exclude)]
+        while candidates:
+            s = self._shuffle_pick(candidates)
+            if s.limiter.try_acquire(tokens):
+                return s
+            attempted.append(s.dep.model_id)
+            candidates.remove(s)
+        return None
+
+    def acquire(self, alias: str, tokens_estimate: int = 0,
+                exclude: Optional[set] = None) -> Ticket:
+        """Admit a request for ``alias``: pick a deployment, charge its
+        window, or walk the fallback chain.  Raises
+        :class:`RouterRateLimit` when everything is exhausted and
+        :class:`UnknownAlias` when the alias isn't configured."""
+        exclude = exclude or set()
+        attempted: list = []
+        with self._lock:
+            s = self._try_alias(alias, tokens_estimate, exclude, attempted)
+            if s is not None:
+                return self._issue(s, alias, tokens_estimate, False, attempted)
+            for fb_alias in self.settings.fallbacks.get(alias, []):
+                try:
+                    s = self._try_alias(fb_alias, tokens_estimate, exclude, attempted)
+                except UnknownAlias:
+                    continue
+                if s is not None:
+                    return self._issue(s, alias, tokens_estimate, True, attempted)
+        raise RouterRateLimit(
+            f"no deployment available for {alias!r} "
+            f"(rate limits / cooldowns exhausted; tried fallbacks "
+            f"{self.settings.fallbacks.get(alias, [])})")
+
+    def _issue(self, s: DeploymentState, alias: str, tokens: int,
+               is_fallback: bool, attempted: list) -> Ticket:
+        s.in_flight += 1
+        s.total_requests += 1
+        return Ticket(state=s, alias_requested=alias, tokens_estimated=tokens,
+                      is_fallback=is_fallback, attempted=attempted)
+
+    # ----------------------------------------------------------- complete
+    def complete(self, ticket: Ticket, actual_tokens: Optional[int] = None) -> None:
+        if ticket.done:
+            return
+        ticket.done = True
+        s = ticket.state
+        with self._lock:
+            s.in_flight = max(0, s.in_flight - 1)
+            if actual_tokens is not None:
+                s.limiter.reconcile(ticket.tokens_estimated, actual_tokens)
+
+    def fail(self, ticket: Ticket, *, charge: bool = False) -> None:
+        """Record a failure.  Unless ``charge``, the rate-window admission
+        is refunded so a retried request doesn't double-bill (SURVEY.md §7
+        hard-part 2)."""
+        if ticket.done:
+            return
+        ticket.done = True
+        s = ticket.state
+        now = self._clock()
+        with self._lock:
+            s.in_flight = max(0, s.in_flight - 1)
+            s.total_failures += 1
+            if not charge:
+                s.limiter.release(ticket.tokens_estimated)
+            s.fail_times = [t for t in s.fail_times if now - t < 60.0]
+            s.fail_times.append(now)
+            if len(s.fail_times) >= self.settings.allowed_fails:
+                s.cooldown_until = now + self.settings.cooldown_time
+                s.total_cooldowns += 1
+                s.fail_times.clear()
+
+    # -------------------------------------------------------------- admin
+    def set_healthy(self, model_id: str, healthy: bool) -> None:
+        with self._lock:
+            for s in self.states:
+                if s.dep.model_id == model_id:
+                    s.healthy = healthy
+
+    def describe(self) -> list[dict]:
+        now = self._clock()
+        out = []
+        for s in self.states:
+            snap = s.snapshot()
+            out.append({
+                "model_name": s.dep.model_name,
+                "model_id": s.dep.model_id,
+                "backend": s.dep.model,
+                "rpm": s.dep.rpm, "tpm": s.dep.tpm,
+                "rpm_used": snap.rpm_used, "tpm_used": snap.tpm_used,
+                "healthy": s.healthy,
+                "cooldown_remaining": max(0.0, s.cooldown_until - now),
+                "in_flight": s.in_flight,
+                "total_requests": s.total_requests,
+                "total_failures": s.total_failures,
+            })
+        return out
