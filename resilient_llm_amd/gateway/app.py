@@ -1,0 +1,490 @@
+"""The gateway application: OpenAI-compatible routes + dispatch loop.
+
+This is the native equivalent of the LiteLLM proxy the reference launches
+(reference bin/start-gateway.sh:56) — the full capability surface of
+SURVEY.md §2.2 in one place:
+
+- X1  POST /chat/completions (and /v1/...) with model alias, messages,
+      max_tokens, temperature, stream; `response.model` carries the
+      serving deployment id (demo_fallback.py:166 reads it).
+- X8  response header ``x-gateway-model-id`` names the deployment
+      (the reference reads ``x-litellm-model-id``, demo_load_balancing.py:132).
+- X9  429 + OpenAI error body when every bucket/ladder is exhausted.
+- X5/X6/X14  retry ladder onto fallback deployments, cooldowns, and SSE
+      streaming with hot failover mid-stream (stateless replay: the
+      replacement worker regenerates and the gateway discards the tokens
+      the client already has — SURVEY.md §5.3).
+- X10/X12  /admin/distribution — the Logs-Insights-style query over the
+      in-process ledger (no propagation wait, same table content).
+- Fault injection: POST /admin/fault {device, mode} (drives the failover
+      demo where the reference could only starve quotas).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+import uuid
+from typing import AsyncIterator, Optional
+
+from ..config import Config
+from ..obs.ledger import InvocationLedger, InvocationRecord
+from ..router.core import (
+    NoDeploymentAvailable, Router, RouterRateLimit, Ticket, UnknownAlias,
+)
+from ..router.token_bucket import MinuteWindowLimiter
+from ..workers.base import (
+    GenerationRequest, GenerationResult, Worker, WorkerDead, WorkerError,
+    WorkerRegistry, WorkerThrottled,
+)
+from ..workers.stub import estimate_tokens
+from .http import Request, Response
+
+DEFAULT_MAX_TOKENS = 128
+STREAM_CHUNK_TIMEOUT_S = 30.0
+REQUEST_TIMEOUT_S = 120.0
+
+
+class ConsumerLimiter:
+    """Optional per-consumer (api-key) buckets on top of per-deployment
+    ones — SURVEY.md §3.4 'MI355X equivalent: per-consumer token buckets
+    keyed on api_key'.  Configured via a ``consumer_limits:`` section."""
+
+    def __init__(self, raw: dict | None, clock=time.monotonic) -> None:
+        raw = raw or {}
+        self._default = raw.get("default")          # {"rpm":..,"tpm":..} | None
+        self._per_key_cfg = raw.get("keys", {})
+        self._clock = clock
+        self._limiters: dict[str, MinuteWindowLimiter] = {}
+
+    def _limiter_for(self, key: str) -> Optional[MinuteWindowLimiter]:
+        cfg = self._per_key_cfg.get(key, self._default)
+        if not cfg:
+            return None
+        if key not in self._limiters:
+            self._limiters[key] = MinuteWindowLimiter(
+                rpm=cfg.get("rpm"), tpm=cfg.get("tpm"), clock=self._clock)
+        return self._limiters[key]
+
+    def try_acquire(self, key: str, tokens: int) -> bool:
+        lim = self._limiter_for(key)
+        return True if lim is None else lim.try_acquire(tokens)
+
+    def reconcile(self, key: str, estimated: int, actual: int) -> None:
+        lim = self._limiter_for(key)
+        if lim is not None:
+            lim.reconcile(estimated, actual)
+
+
+class GatewayApp:
+    def __init__(self, config: Config, registry: WorkerRegistry,
+                 router: Optional[Router] = None,
+                 ledger: Optional[InvocationLedger] = None,
+                 health_interval_s: float = 2.0) -> None:
+        self.config = config
+        self.registry = registry
+        self.router = router or Router(config.deployments, config.router)
+        self.ledger = ledger or InvocationLedger(
+            jsonl_path=config.cluster.ledger_path)
+        self.consumers = ConsumerLimiter(config.raw.get("consumer_limits"))
+        self.health_interval_s = health_interval_s
+        self._health_task: Optional[asyncio.Task] = None
+        self.started_at = time.time()
+
+    # ---------------------------------------------------------- lifecycle
+    async def start_background(self) -> None:
+        if self._health_task is None:
+            self._health_task = asyncio.create_task(self._health_loop())
+
+    async def stop_background(self) -> None:
+        if self._health_task is not None:
+            self._health_task.cancel()
+            try:
+                await self._health_task
+            except asyncio.CancelledError:
+                pass
+            self._health_task = None
+
+    def _worker_for(self, ticket: Ticket) -> Worker:
+        d = ticket.deployment
+        return self.registry.get(d.backend_kind, d.backend_target, d.backend_model)
+
+    def _deployments_on(self, worker: Worker) -> list[str]:
+        """model_ids of deployments that resolve to this worker (for
+        health marking).  ``*`` targets are skipped — they re-resolve."""
+        kind, _, target = worker.device.partition(":")
+        out = []
+        for d in self.config.deployments:
+            if d.backend_kind == kind and d.backend_target == target:
+                out.append(d.model_id)
+        return out
+
+    async def _health_loop(self) -> None:
+        """Per-worker heartbeat -> router health flags (X6)."""
+        while True:
+            await asyncio.sleep(self.health_interval_s)
+            for worker in self.registry.all().values():
+                try:
+                    await asyncio.wait_for(worker.health(),
+                                           timeout=self.health_interval_s)
+                    ok = True
+                except Exception:
+                    ok = False
+                for model_id in self._deployments_on(worker):
+                    self.router.set_healthy(model_id, ok)
+
+    # ------------------------------------------------------------ routing
+    async def handle(self, req: Request) -> Response:
+        route = (req.method, req.path)
+        if req.path in ("/chat/completions", "/v1/chat/completions"):
+            if req.method != "POST":
+                return Response.error(405, "use POST")
+            return await self.chat_completions(req)
+        if route == ("GET", "/health"):
+            return await self.health(req)
+        if req.path in ("/models", "/v1/models") and req.method == "GET":
+            return self.list_models()
+        if route == ("GET", "/admin/distribution"):
+            return self.admin_distribution(req)
+        if route == ("GET", "/admin/router"):
+            return Response.json_response({"deployments": self.router.describe(),
+                                           "settings": {
+                                               "routing_strategy": self.config.router.routing_strategy,
+                                               "enable_pre_call_checks": self.config.router.enable_pre_call_checks,
+                                               "allowed_fails": self.config.router.allowed_fails,
+                                               "cooldown_time": self.config.router.cooldown_time,
+                                               "fallbacks": self.config.router.fallbacks,
+                                           }})
+        if route == ("POST", "/admin/fault"):
+            return await self.admin_fault(req)
+        if route == ("GET", "/metrics"):
+            return self.prometheus_metrics()
+        return Response.error(404, f"no route {req.method} {req.path}",
+                              err_type="invalid_request_error")
+
+    # ----------------------------------------------------- chat endpoint
+    @staticmethod
+    def _consumer_of(req: Request) -> str:
+        auth = req.headers.get("authorization", "")
+        if auth.lower().startswith("bearer "):
+            return auth[7:].strip() or "anonymous"
+        return req.headers.get("x-api-key", "anonymous")
+
+    @staticmethod
+    def _parse_chat_body(req: Request) -> dict:
+        body = req.json()
+        if not isinstance(body, dict):
+            raise ValueError("body must be a JSON object")
+        if not isinstance(body.get("model"), str):
+            raise ValueError("'model' is required")
+        msgs = body.get("messages")
+        if not isinstance(msgs, list) or not msgs:
+            raise ValueError("'messages' must be a non-empty list")
+        return body
+
+    def _estimate(self, body: dict) -> tuple[int, int]:
+        prompt_est = sum(estimate_tokens(str(m.get("content", "")))
+                         for m in body["messages"])
+        max_tokens = int(body.get("max_tokens") or DEFAULT_MAX_TOKENS)
+        return prompt_est, prompt_est + max_tokens
+
+    def _gen_request(self, body: dict, consumer: str, ticket: Ticket,
+                     stream: bool) -> GenerationRequest:
+        return GenerationRequest(
+            request_id=f"req-{uuid.uuid4().hex[:16]}",
+            model=ticket.deployment.backend_model,
+            messages=body["messages"],
+            max_tokens=int(body.get("max_tokens") or DEFAULT_MAX_TOKENS),
+            temperature=float(body.get("temperature", 0.0)),
+            top_p=float(body.get("top_p", 1.0)),
+            seed=body.get("seed"),
+            stream=stream,
+            consumer=consumer,
+        )
+
+    def _record(self, ticket: Ticket, greq: Optional[GenerationRequest],
+                consumer: str, status: str, t0: float,
+                result: Optional[GenerationResult] = None,
+                device: str = "") -> None:
+        self.ledger.record(InvocationRecord(
+            ts=time.time(),
+            request_id=greq.request_id if greq else "-",
+            alias=ticket.alias_requested,
+            model_id=ticket.deployment.model_id,
+            device=device,
+            consumer=consumer,
+            status=status,
+            is_fallback=ticket.is_fallback,
+            prompt_tokens=result.prompt_tokens if result else 0,
+            completion_tokens=result.completion_tokens if result else 0,
+            latency_ms=(time.monotonic() - t0) * 1000.0,
+            ttft_ms=result.ttft_ms if result else None,
+        ))
+
+    async def chat_completions(self, req: Request) -> Response:
+        try:
+            body = self._parse_chat_body(req)
+        except (ValueError, json.JSONDecodeError) as e:
+            return Response.error(400, str(e), err_type="invalid_request_error")
+        alias = body["model"]
+        consumer = self._consumer_of(req)
+        prompt_est, total_est = self._estimate(body)
+        if not self.consumers.try_acquire(consumer, total_est):
+            return Response.error(429, f"consumer {consumer!r} rate limit exceeded",
+                                  err_type="rate_limit_error", code="rate_limit_exceeded")
+        if bool(body.get("stream")):
+            return await self._chat_stream(body, alias, consumer, total_est)
+        return await self._chat_once(body, alias, consumer, total_est)
+
+    # max attempts across replicas + fallback chain before giving up
+    def _max_attempts(self, alias: str) -> int:
+        n = len(self.router.alias_states(alias))
+        for fb in self.config.router.fallbacks.get(alias, []):
+            try:
+                n += len(self.router.alias_states(fb))
+            except UnknownAlias:
+                pass
+        return max(2, min(n + 1, 8))
+
+    async def _chat_once(self, body: dict, alias: str, consumer: str,
+                         total_est: int) -> Response:
+        t0 = time.monotonic()
+        exclude: set = set()
+        last_err: Optional[Exception] = None
+        try:
+            attempts = self._max_attempts(alias)
+        except UnknownAlias:
+            return Response.error(404, f"unknown model {alias!r}",
+                                  err_type="invalid_request_error", code="model_not_found")
+        for _ in range(attempts):
+            try:
+                ticket = self.router.acquire(alias, total_est, exclude=exclude)
+            except RouterRateLimit as e:
+                return Response.error(429, str(e), err_type="rate_limit_error",
+                                      code="rate_limit_exceeded")
+            greq = self._gen_request(body, consumer, ticket, stream=False)
+            try:
+                worker = self._worker_for(ticket)
+                result = await asyncio.wait_for(worker.generate(greq),
+                                                timeout=REQUEST_TIMEOUT_S)
+            except (WorkerError, asyncio.TimeoutError) as e:
+                last_err = e
+                throttled = isinstance(e, WorkerThrottled)
+                self.router.fail(ticket, charge=throttled)
+                self._record(ticket, greq, consumer,
+                             "throttled" if throttled else "error", t0)
+                exclude.add(id(ticket.state))
+                continue
+            actual = result.prompt_tokens + result.completion_tokens
+            self.router.complete(ticket, actual_tokens=actual)
+            self.consumers.reconcile(consumer, total_est, actual)
+            self._record(ticket, greq, consumer, "ok", t0, result, worker.device)
+            return self._completion_response(greq, ticket, result, worker)
+        from ..utils.logging import sanitize_error
+        msg = sanitize_error(last_err) if last_err else "no deployment available"
+        status = 429 if isinstance(last_err, WorkerThrottled) else 502
+        return Response.error(status, f"all deployments failed: {msg}",
+                              err_type="api_error")
+
+    def _completion_response(self, greq: GenerationRequest, ticket: Ticket,
+                             result: GenerationResult, worker: Worker) -> Response:
+        payload = {
+            "id": f"chatcmpl-{greq.request_id}",
+            "object": "chat.completion",
+            "created": int(time.time()),
+            "model": ticket.deployment.model_id,
+            "choices": [{
+                "index": 0,
+                "message": {"role": "assistant", "content": result.text},
+                "finish_reason": result.finish_reason,
+            }],
+            "usage": {
+                "prompt_tokens": result.prompt_tokens,
+                "completion_tokens": result.completion_tokens,
+                "total_tokens": result.prompt_tokens + result.completion_tokens,
+            },
+        }
+        headers = {
+            "x-gateway-model-id": ticket.deployment.model_id,
+            "x-gateway-device": worker.device,
+            "x-gateway-fallback": "true" if ticket.is_fallback else "false",
+        }
+        return Response.json_response(payload, headers=headers)
+
+    # -------------------------------------------------------- streaming
+    async def _chat_stream(self, body: dict, alias: str, consumer: str,
+                           total_est: int) -> Response:
+        """SSE streaming (X14) with hot failover: a mid-stream worker
+        death/hang re-routes to the next deployment, which regenerates;
+        tokens the client already received are discarded (stateless
+        replay)."""
+        try:
+            first_ticket = self.router.acquire(alias, total_est)
+        except UnknownAlias:
+            return Response.error(404, f"unknown model {alias!r}",
+                                  err_type="invalid_request_error", code="model_not_found")
+        except RouterRateLimit as e:
+            return Response.error(429, str(e), err_type="rate_limit_error",
+                                  code="rate_limit_exceeded")
+
+        app = self
+
+        async def body_iter() -> AsyncIterator[bytes]:
+            t0 = time.monotonic()
+            sent_tokens = 0
+            exclude: set = set()
+            ticket: Optional[Ticket] = first_ticket
+            last_err: Optional[Exception] = None
+            attempts = app._max_attempts(alias)
+            comp_id = f"chatcmpl-{uuid.uuid4().hex[:16]}"
+            for attempt in range(attempts):
+                if ticket is None:
+                    try:
+                        ticket = app.router.acquire(alias, total_est, exclude=exclude)
+                    except NoDeploymentAvailable as e:
+                        last_err = e
+                        break
+                greq = app._gen_request(body, consumer, ticket, stream=True)
+                try:
+                    worker = app._worker_for(ticket)
+                except WorkerError as e:
+                    last_err = e
+                    app.router.fail(ticket)
+                    exclude.add(id(ticket.state))
+                    ticket = None
+                    continue
+                model_id = ticket.deployment.model_id
+                n_emitted = 0
+                completion_tokens = 0
+                finish = None
+                try:
+                    it = worker.generate_stream(greq).__aiter__()
+                    if attempt > 0:
+                        yield (f": failover -> {model_id}\n\n").encode()
+                    while True:
+                        try:
+                            chunk = await asyncio.wait_for(
+                                it.__anext__(), timeout=STREAM_CHUNK_TIMEOUT_S)
+                        except StopAsyncIteration:
+                            break
+                        completion_tokens += 1
+                        if completion_tokens <= sent_tokens:
+                            finish = chunk.finish_reason or finish
+                            continue   # replay skip after failover
+                        n_emitted += 1
+                        finish = chunk.finish_reason or finish
+                        delta = ({"role": "assistant", "content": chunk.text}
+                                 if sent_tokens + n_emitted == 1
+                                 else {"content": chunk.text})
+                        evt = {
+                            "id": comp_id, "object": "chat.completion.chunk",
+                            "created": int(time.time()), "model": model_id,
+                            "choices": [{"index": 0, "delta": delta,
+                                         "finish_reason": chunk.finish_reason}],
+                        }
+                        yield f"data: {json.dumps(evt)}\n\n".encode()
+                    sent_tokens += n_emitted
+                    result = GenerationResult(
+                        text="", prompt_tokens=total_est, completion_tokens=sent_tokens,
+                        finish_reason=finish or "stop")
+                    app.router.complete(ticket, actual_tokens=None)
+                    app._record(ticket, greq, consumer, "ok", t0, result, worker.device)
+                    yield b"data: [DONE]\n\n"
+                    return
+                except (WorkerError, asyncio.TimeoutError) as e:
+                    last_err = e
+                    sent_tokens += n_emitted
+                    throttled = isinstance(e, WorkerThrottled)
+                    app.router.fail(ticket, charge=throttled)
+                    app._record(ticket, greq, consumer,
+                                "throttled" if throttled else "error", t0)
+                    exclude.add(id(ticket.state))
+                    ticket = None
+                    continue
+            from ..utils.logging import sanitize_error
+            err_evt = {"error": {"message": f"stream failed: "
+                                            f"{sanitize_error(last_err) if last_err else 'exhausted'}",
+                                 "type": "api_error"}}
+            yield f"data: {json.dumps(err_evt)}\n\n".encode()
+            yield b"data: [DONE]\n\n"
+
+        headers = {
+            "x-gateway-model-id": first_ticket.deployment.model_id,
+            "x-gateway-fallback": "true" if first_ticket.is_fallback else "false",
+            "cache-control": "no-cache",
+        }
+        return Response(status=200, headers=headers,
+                        content_type="text/event-stream", body_iter=body_iter())
+
+    # ------------------------------------------------------------- admin
+    async def health(self, req: Request) -> Response:
+        workers = {}
+        for key, w in self.registry.all().items():
+            try:
+                workers[key] = await w.health()
+            except Exception as e:
+                workers[key] = {"device": key, "status": "unhealthy", "error": str(e)}
+        return Response.json_response({
+            "status": "ok",
+            "uptime_s": time.time() - self.started_at,
+            "workers": workers,
+        })
+
+    def list_models(self) -> Response:
+        return Response.json_response({
+            "object": "list",
+            "data": [{"id": a, "object": "model", "owned_by": "resilient_llm_amd"}
+                     for a in self.config.aliases],
+        })
+
+    def admin_distribution(self, req: Request) -> Response:
+        """The Logs-Insights-style aggregation (X12): counts by device /
+        model_id / consumer / alias plus serving stats — synchronously,
+        no propagation retry ladder needed."""
+        by = req.query.get("by", "device")
+        if by not in ("device", "model_id", "alias", "consumer", "status"):
+            return Response.error(400, f"bad 'by' dimension {by!r}")
+        since = None
+        if "since_s" in req.query:
+            since = time.time() - float(req.query["since_s"])
+        alias = req.query.get("alias")
+        status = req.query.get("status", "ok") or None
+        dist = self.ledger.distribution(by=by, since=since, alias=alias, status=status)
+        total = sum(dist.values())
+        return Response.json_response({
+            "by": by, "alias": alias, "total": total,
+            "distribution": dist,
+            "percentages": {k: round(100.0 * v / total, 1) for k, v in dist.items()}
+            if total else {},
+            "stats": self.ledger.stats(since=since, alias=alias),
+        })
+
+    async def admin_fault(self, req: Request) -> Response:
+        try:
+            body = req.json() or {}
+            device = body["device"]
+            mode = body.get("mode", "none")
+        except (KeyError, json.JSONDecodeError):
+            return Response.error(400, "need {device, mode}")
+        workers = self.registry.all()
+        if device not in workers:
+            return Response.error(404, f"no worker {device!r} "
+                                       f"(have {sorted(workers)})")
+        await workers[device].inject_fault(mode)
+        return Response.json_response({"device": device, "mode": mode})
+
+    def prometheus_metrics(self) -> Response:
+        lines = []
+        for row in self.router.describe():
+            labels = (f'alias="{row["model_name"]}",model_id="{row["model_id"]}"')
+            lines.append(f'gateway_requests_total{{{labels}}} {row["total_requests"]}')
+            lines.append(f'gateway_failures_total{{{labels}}} {row["total_failures"]}')
+            lines.append(f'gateway_in_flight{{{labels}}} {row["in_flight"]}')
+            lines.append(f'gateway_rpm_used{{{labels}}} {row["rpm_used"]}')
+            lines.append(f'gateway_healthy{{{labels}}} {int(row["healthy"])}')
+        st = self.ledger.stats()
+        for k in ("total", "ok", "errors", "throttled", "fallbacks"):
+            lines.append(f'gateway_ledger_{k} {st[k]}')
+        return Response(body="\n".join(lines) + "\n",
+                        content_type="text/plain; version=0.0.4")

@@ -1,0 +1,100 @@
+"""Gateway assembly + ``python -m resilient_llm_amd.gateway.server`` CLI.
+
+The native equivalent of the reference launcher (reference
+bin/start-gateway.sh: parse/validate config, then exec the proxy).  Builds
+the worker set the config names — CPU stubs in-process, one engine worker
+per GPU, TP pools — then serves the OpenAI-compatible API.
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import signal
+from typing import Optional
+
+from ..config import Config, load_config
+from ..utils.logging import log_with_timestamp
+from ..workers.base import WorkerRegistry
+from ..workers.stub import StubWorker
+from .app import GatewayApp
+from .http import HttpServer
+
+
+def build_registry(config: Config, stub_kwargs: Optional[dict] = None) -> WorkerRegistry:
+    """Instantiate one worker per distinct backend target in the config.
+
+    ``stub/...`` targets become in-process :class:`StubWorker`s.
+    ``gpu/...`` and ``pool/...`` targets are spawned as engine worker
+    processes by :mod:`resilient_llm_amd.workers.gpu` (needs a GPU) — the
+    caller registers those; this helper only auto-builds stubs so the
+    plumbing path works anywhere.
+    """
+    registry = WorkerRegistry()
+    stub_models: dict[str, set[str]] = {}
+    for d in config.deployments:
+        if d.backend_kind != "stub":
+            continue
+        targets = [d.backend_target]
+        if d.backend_target == "*":
+            # a spread alias needs every stub device that exists elsewhere
+            targets = sorted({e.backend_target for e in config.deployments
+                              if e.backend_kind == "stub" and e.backend_target != "*"}) or ["0"]
+        for t in targets:
+            stub_models.setdefault(t, set()).add(d.backend_model or d.model_id)
+    for target, models in stub_models.items():
+        registry.register("stub", target, StubWorker(target, models,
+                                                     **(stub_kwargs or {})))
+    return registry
+
+
+async def serve(config: Config, registry: Optional[WorkerRegistry] = None,
+                ready_event: Optional[asyncio.Event] = None) -> None:
+    registry = registry or build_registry(config)
+    app = GatewayApp(config, registry)
+    server = HttpServer(app.handle, host=config.cluster.host, port=config.cluster.port)
+    await server.start()
+    await app.start_background()
+    log_with_timestamp(
+        f"gateway listening on http://{config.cluster.host}:{config.cluster.port} "
+        f"({len(config.deployments)} deployments, "
+        f"{len(registry.all())} workers)", "green")
+    if ready_event is not None:
+        ready_event.set()
+    stop = asyncio.Event()
+    loop = asyncio.get_running_loop()
+    for sig in (signal.SIGINT, signal.SIGTERM):
+        try:
+            loop.add_signal_handler(sig, stop.set)
+        except NotImplementedError:
+            pass
+    try:
+        await stop.wait()
+    finally:
+        await app.stop_background()
+        await server.stop()
+        await registry.close()
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser(description="resilient_llm_amd gateway")
+    ap.add_argument("--config", default="config/config.yaml")
+    ap.add_argument("--port", type=int, default=None, help="override cluster.port")
+    ap.add_argument("--host", default=None)
+    args = ap.parse_args()
+    config = load_config(args.config)
+    if args.port is not None:
+        config.cluster.port = args.port
+    if args.host is not None:
+        config.cluster.host = args.host
+
+    kinds = {d.backend_kind for d in config.deployments}
+    registry = build_registry(config)
+    if kinds & {"gpu", "pool"}:
+        from ..workers.gpu import register_gpu_workers
+        register_gpu_workers(config, registry)
+    asyncio.run(serve(config, registry))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,5 @@
+from .base import (  # noqa: F401
+    GenerationRequest, GenerationChunk, GenerationResult, Worker,
+    WorkerError, WorkerThrottled, WorkerDead, WorkerRegistry,
+)
+from .stub import StubWorker  # noqa: F401

@@ -1,0 +1,120 @@
+"""Test harness: run the gateway (with stub workers) in a background
+thread, yield a sync client pointed at it."""
+
+from __future__ import annotations
+
+import asyncio
+import contextlib
+import socket
+import threading
+
+from resilient_llm_amd.client import OpenAIClient
+from resilient_llm_amd.config import load_config
+from resilient_llm_amd.gateway.server import build_registry, serve
+
+
+def free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def stub_config_dict(port: int, **overrides) -> dict:
+    data = {
+        "cluster": {"port": port, "host": "127.0.0.1"},
+        "model_list": [
+            {"model_name": "llama-fallback-demo",
+             "litellm_params": {"model": "stub/0/llama-3-8b"},
+             "model_info": {"id": "stub0/llama-3-8b-primary"},
+             "rpm": 3, "tpm": 100000},
+            {"model_name": "llama-loadbalance-demo",
+             "litellm_params": {"model": "stub/0/llama-3-8b"},
+             "model_info": {"id": "stub0/llama-3-8b-r0"},
+             "rpm": 3, "tpm": 100000},
+            {"model_name": "llama-loadbalance-demo",
+             "litellm_params": {"model": "stub/1/llama-3-8b"},
+             "model_info": {"id": "stub1/llama-3-8b-r1"},
+             "rpm": 3, "tpm": 100000},
+            {"model_name": "llama-fallback-loadbalance",
+             "litellm_params": {"model": "stub/1/llama-3-8b"},
+             "model_info": {"id": "stub1/llama-3-8b-fb"},
+             "rpm": 25, "tpm": 250000},
+            {"model_name": "llama-fallback-quota",
+             "litellm_params": {"model": "stub/1/llama-3-8b"},
+             "model_info": {"id": "stub1/llama-3-8b-fbq"},
+             "rpm": 25, "tpm": 250000},
+            {"model_name": "consumer-a-model",
+             "litellm_params": {"model": "stub/0/llama-3-8b"},
+             "rpm": 3, "tpm": 30000},
+            {"model_name": "consumer-b-model",
+             "litellm_params": {"model": "stub/0/llama-3-8b"},
+             "rpm": 10, "tpm": 100000},
+            {"model_name": "consumer-c-model",
+             "litellm_params": {"model": "stub/1/llama-3-8b"},
+             "rpm": 10, "tpm": 100000},
+            {"model_name": "llama-cris-demo",
+             "litellm_params": {"model": "stub/*/llama-3-8b"},
+             "rpm": 1000, "tpm": 10000000},
+        ],
+        "cris": {"model_id": "llama-cris-demo"},
+        "router_settings": {
+            "routing_strategy": "simple-shuffle",
+            "enable_pre_call_checks": True,
+            "allowed_fails": 2,
+            "cooldown_time": 15,
+            "fallbacks": [
+                {"llama-fallback-demo": ["llama-fallback-quota"]},
+                {"llama-loadbalance-demo": ["llama-fallback-loadbalance"]},
+            ],
+        },
+    }
+    data.update(overrides)
+    return data
+
+
+@contextlib.contextmanager
+def run_gateway(config_dict=None, stub_kwargs=None, health_interval_s=0.2):
+    port = None
+    if config_dict is None:
+        port = free_port()
+        config_dict = stub_config_dict(port)
+    else:
+        port = config_dict["cluster"]["port"]
+    config = load_config(data=config_dict)
+    registry = build_registry(config, stub_kwargs=stub_kwargs)
+
+    loop = asyncio.new_event_loop()
+    ready = threading.Event()
+    stop_future: dict = {}
+
+    async def _main():
+        ev = asyncio.Event()
+        stop_future["stop"] = asyncio.Event()
+
+        async def _serve():
+            from resilient_llm_amd.gateway.app import GatewayApp
+            from resilient_llm_amd.gateway.http import HttpServer
+            app = GatewayApp(config, registry, health_interval_s=health_interval_s)
+            stop_future["app"] = app
+            server = HttpServer(app.handle, host=config.cluster.host,
+                                port=config.cluster.port)
+            await server.start()
+            await app.start_background()
+            ev.set()
+            ready.set()
+            await stop_future["stop"].wait()
+            await app.stop_background()
+            await server.stop()
+
+        await _serve()
+
+    th = threading.Thread(target=lambda: loop.run_until_complete(_main()), daemon=True)
+    th.start()
+    assert ready.wait(timeout=10), "gateway failed to start"
+    client = OpenAIClient(f"http://127.0.0.1:{port}")
+    try:
+        yield client, registry, config
+    finally:
+        loop.call_soon_threadsafe(stop_future["stop"].set)
+        th.join(timeout=10)
+        loop.close()

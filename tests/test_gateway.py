@@ -1,0 +1,221 @@
+"""End-to-end gateway tests over HTTP with the CPU stub backend —
+BASELINE.json config 1 (plumbing, no GPU)."""
+
+import concurrent.futures as cf
+import time
+
+import pytest
+
+from resilient_llm_amd.client import APIError, RateLimitError
+from tests.gateway_harness import run_gateway
+
+MSGS = [{"role": "user", "content": "What is resilient inference?"}]
+
+
+def test_basic_completion_and_headers():
+    with run_gateway() as (client, registry, config):
+        r = client.chat.completions.create(model="llama-cris-demo", messages=MSGS,
+                                           max_tokens=8)
+        assert r.choices[0].message.content
+        assert r.usage.completion_tokens == 8
+        assert r.model_id_header == "llama-cris-demo".replace("llama-cris-demo", r.model)
+        assert r.device_header.startswith("stub:")
+        assert r.object == "chat.completion"
+
+
+def test_unknown_model_404():
+    with run_gateway() as (client, *_):
+        with pytest.raises(APIError) as ei:
+            client.chat.completions.create(model="nope", messages=MSGS)
+        assert ei.value.status == 404
+
+
+def test_bad_body_400():
+    with run_gateway() as (client, *_):
+        with pytest.raises(APIError) as ei:
+            client.chat.completions.create(model="llama-cris-demo", messages=[])
+        assert ei.value.status == 400
+
+
+def test_rate_limit_429_openai_body():
+    """consumer-a rpm=3, no fallback: 4th+ request in the minute -> 429
+    mapped to RateLimitError (X9, reference README.md:255-256)."""
+    with run_gateway() as (client, *_):
+        ok = 0
+        limited = 0
+        for _ in range(5):
+            try:
+                client.chat.completions.create(model="consumer-a-model",
+                                               messages=MSGS, max_tokens=4)
+                ok += 1
+            except RateLimitError as e:
+                assert e.body["error"]["type"] == "rate_limit_error"
+                limited += 1
+        assert ok == 3 and limited == 2
+
+
+def test_fallback_transparent():
+    """rpm=3 primary + fallback chain: 10 requests -> 3 primary, 7 on the
+    fallback deployment, all successful (reference README.md:167-180)."""
+    with run_gateway() as (client, *_):
+        models = []
+        for _ in range(10):
+            r = client.chat.completions.create(model="llama-fallback-demo",
+                                               messages=MSGS, max_tokens=4)
+            models.append((r.model, r.was_fallback))
+        primary = [m for m, fb in models if not fb]
+        fallback = [m for m, fb in models if fb]
+        assert len(primary) == 3 and len(fallback) == 7
+        assert set(m for m, fb in models if fb) == {"stub1/llama-3-8b-fbq"}
+
+
+def test_load_balancing_spreads():
+    with run_gateway() as (client, *_):
+        seen = set()
+        for _ in range(6):
+            r = client.chat.completions.create(model="llama-loadbalance-demo",
+                                               messages=MSGS, max_tokens=4)
+            seen.add(r.model_id_header)
+        assert len(seen) >= 2   # 2 replicas + fallback pool once rpm spent
+
+
+def test_concurrent_burst_cris_spread():
+    """10 concurrent requests to the spread alias land on multiple stub
+    devices and the distribution API reports them (X10/X12)."""
+    with run_gateway() as (client, *_):
+        def one(i):
+            return client.chat.completions.create(
+                model="llama-cris-demo", messages=MSGS, max_tokens=4).device_header
+        with cf.ThreadPoolExecutor(10) as ex:
+            devices = list(ex.map(one, range(10)))
+        assert all(d.startswith("stub:") for d in devices)
+        dist = client.distribution(by="device", alias="llama-cris-demo")
+        assert dist["total"] == 10
+        assert sum(dist["distribution"].values()) == 10
+        assert set(dist["distribution"]) == set(devices)
+
+
+def test_distribution_percentages_and_stats():
+    with run_gateway() as (client, *_):
+        for _ in range(4):
+            client.chat.completions.create(model="llama-cris-demo",
+                                           messages=MSGS, max_tokens=4)
+        d = client.distribution(by="model_id")
+        assert abs(sum(d["percentages"].values()) - 100.0) < 1.0
+        st = d["stats"]
+        assert st["ok"] == 4 and st["success_rate"] == 1.0
+        assert st["latency_ms"]["p50"] is not None
+
+
+def test_health_and_models_endpoints():
+    with run_gateway() as (client, *_):
+        h = client.health()
+        assert h["status"] == "ok"
+        assert all(w.get("status") == "ok" for w in h["workers"].values())
+        models = client._get("/v1/models")
+        ids = [m["id"] for m in models["data"]]
+        assert "llama-fallback-demo" in ids
+
+
+def test_router_admin_state():
+    with run_gateway() as (client, *_):
+        client.chat.completions.create(model="llama-fallback-demo",
+                                       messages=MSGS, max_tokens=4)
+        state = client.router_state()
+        rows = {r["model_id"]: r for r in state["deployments"]}
+        assert rows["stub0/llama-3-8b-primary"]["rpm_used"] == 1
+        assert state["settings"]["routing_strategy"] == "simple-shuffle"
+
+
+def test_fault_injection_reroutes():
+    """Kill the primary worker: requests flow to the fallback deployment
+    on the healthy device (X6 hot failover, no GPU needed)."""
+    with run_gateway() as (client, registry, _):
+        client.inject_fault("stub:0", "kill")
+        r = client.chat.completions.create(model="llama-fallback-demo",
+                                           messages=MSGS, max_tokens=4)
+        assert r.was_fallback
+        assert r.device_header == "stub:1"
+        client.inject_fault("stub:0", "none")
+
+
+def test_health_loop_marks_unhealthy_then_recovers():
+    with run_gateway() as (client, registry, _):
+        client.inject_fault("stub:0", "kill")
+        time.sleep(0.6)   # health loop interval 0.2s
+        state = client.router_state()
+        rows = {r["model_id"]: r for r in state["deployments"]}
+        assert rows["stub0/llama-3-8b-primary"]["healthy"] is False
+        client.inject_fault("stub:0", "none")
+        time.sleep(0.6)
+        rows = {r["model_id"]: r
+                for r in client.router_state()["deployments"]}
+        assert rows["stub0/llama-3-8b-primary"]["healthy"] is True
+
+
+def test_streaming_basic():
+    with run_gateway() as (client, *_):
+        stream = client.chat.completions.create(model="llama-cris-demo",
+                                                messages=MSGS, max_tokens=6,
+                                                stream=True)
+        text, model = stream.collect_text()
+        assert len(text.split()) == 6
+        assert model
+
+
+def test_streaming_midstream_failover():
+    """Mid-stream worker death -> gateway replays on another deployment;
+    client receives the full text exactly once (BASELINE config 4 shape)."""
+    with run_gateway(stub_kwargs={"token_delay_ms": 20}) as (client, registry, _):
+        stream = client.chat.completions.create(model="llama-loadbalance-demo",
+                                                messages=MSGS, max_tokens=10,
+                                                stream=True, timeout=30)
+        first_device = stream.headers["x-gateway-model-id"]
+        collected = []
+        killed = False
+        for evt in stream:
+            for c in evt.get("choices", []):
+                content = (c.get("delta") or {}).get("content")
+                if content:
+                    collected.append(content)
+            if len(collected) == 3 and not killed:
+                killed = True
+                # kill whichever stub is serving
+                dev = "stub:0" if "stub0" in first_device else "stub:1"
+                client.inject_fault(dev, "kill")
+        text = "".join(collected)
+        assert len(text.split()) == 10, text
+        assert killed
+        client.inject_fault("stub:0", "none")
+        client.inject_fault("stub:1", "none")
+
+
+def test_metrics_endpoint():
+    with run_gateway() as (client, *_):
+        client.chat.completions.create(model="llama-cris-demo", messages=MSGS,
+                                       max_tokens=2)
+        conn = client._connect(None)
+        conn.request("GET", "/metrics")
+        body = conn.getresponse().read().decode()
+        conn.close()
+        assert "gateway_requests_total" in body
+        assert "gateway_ledger_ok 1" in body
+
+
+def test_throttled_worker_typed_status():
+    """Worker queue full -> typed Throttled (X13) -> retry on another
+    deployment rather than hard failure."""
+    with run_gateway(stub_kwargs={"max_concurrency": 1, "token_delay_ms": 30}) as \
+            (client, registry, _):
+        def one(i):
+            try:
+                r = client.chat.completions.create(model="llama-loadbalance-demo",
+                                                   messages=MSGS, max_tokens=10,
+                                                   timeout=30)
+                return ("ok", r.device_header)
+            except APIError as e:
+                return ("err", e.status)
+        with cf.ThreadPoolExecutor(4) as ex:
+            results = list(ex.map(one, range(4)))
+        oks = [r for r in results if r[0] == "ok"]
+        assert len(oks) >= 2
