@@ -15,6 +15,7 @@ Bedrock's ThrottlingException vs other failures (demo_cris.py:261-283).
 from __future__ import annotations
 
 import dataclasses
+import random
 import time
 from typing import AsyncIterator, Optional
 
@@ -110,7 +111,11 @@ class WorkerRegistry:
                           if k.startswith(f"{kind}:") and model in w.models]
             if not candidates:
                 raise WorkerDead(f"no {kind} worker holds model {model!r}")
-            return min(candidates, key=lambda w: w.in_flight)
+            # capacity-driven spread: least in flight, random among ties —
+            # mirrors CRIS's "distribution is capacity-driven, not
+            # client-controlled" 
+            lowest = min(w.in_flight for w in candidates)
+            return random.choice([w for w in candidates if w.in_flight == lowest])
         try:
             w = self._workers[f"{kind}:{target}"]
         except KeyError:
