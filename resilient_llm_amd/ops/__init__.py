@@ -1,0 +1,115 @@
+"""Op dispatch: hand-written gfx950 HIP kernels on GPU, fp32 torch
+references on CPU.
+
+On a CUDA/ROCm device the HIP extension is REQUIRED — a missing .so
+raises immediately rather than silently falling back to eager PyTorch
+(the round-end check records which native libraries the GPU tests
+actually loaded).  On CPU the ops/ref.py implementations run, which is
+what the no-GPU plumbing tests and the gloo multi-process tests use.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+
+from . import ref  # noqa: F401
+
+_SO_PATH = os.path.join(os.path.dirname(os.path.abspath(__file__)), "_rlli_hip.so")
+_loaded = False
+
+
+class ExtensionMissing(RuntimeError):
+    pass
+
+
+def load_extension(required: bool = False) -> bool:
+    """Load the in-tree .so (idempotent).  ``required=True`` (or any GPU
+    call) raises if it is absent — no silent eager fallback on GPU."""
+    global _loaded
+    if _loaded:
+        return True
+    if os.path.exists(_SO_PATH):
+        torch.ops.load_library(_SO_PATH)
+        _loaded = True
+        return True
+    if required:
+        raise ExtensionMissing(
+            f"HIP extension not built: {_SO_PATH} missing. "
+            f"Run `python -m resilient_llm_amd.ops.build`.")
+    return False
+
+
+def extension_loaded() -> bool:
+    return _loaded
+
+
+def _gpu() -> None:
+    load_extension(required=True)
+
+
+# ------------------------------------------------------------------ ops
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if x.is_cuda:
+        _gpu()
+        return torch.ops.rlli.rmsnorm(x, w, eps)
+    return ref.rmsnorm(x, w, eps)
+
+
+def rmsnorm_residual_(x: torch.Tensor, residual: torch.Tensor, w: torch.Tensor,
+                      eps: float = 1e-5) -> torch.Tensor:
+    if x.is_cuda:
+        _gpu()
+        return torch.ops.rlli.rmsnorm_residual_(x, residual, w, eps)
+    return ref.rmsnorm_residual_(x, residual, w, eps)
+
+
+def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:
+    if gate_up.is_cuda:
+        _gpu()
+        return torch.ops.rlli.silu_mul(gate_up)
+    return ref.silu_mul(gate_up)
+
+
+def rope_kv_append_(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                    positions: torch.Tensor, cos_sin: torch.Tensor,
+                    k_cache: torch.Tensor, v_cache: torch.Tensor,
+                    slot_mapping: torch.Tensor) -> None:
+    if q.is_cuda:
+        _gpu()
+        torch.ops.rlli.rope_kv_append_(q, k, v, positions, cos_sin,
+                                       k_cache, v_cache, slot_mapping)
+    else:
+        ref.rope_kv_append_(q, k, v, positions, cos_sin, k_cache, v_cache,
+                            slot_mapping)
+
+
+def decode_attn(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
+                block_table: torch.Tensor, seq_lens: torch.Tensor,
+                scale: float) -> torch.Tensor:
+    if q.is_cuda:
+        _gpu()
+        return torch.ops.rlli.decode_attn(q, k_cache, v_cache, block_table,
+                                          seq_lens, scale)
+    return ref.decode_attn(q, k_cache, v_cache, block_table, seq_lens, scale)
+
+
+def prefill_attn(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                 cu_seqlens: torch.Tensor, scale: float) -> torch.Tensor:
+    if q.is_cuda:
+        _gpu()
+        return torch.ops.rlli.prefill_attn(q, k, v, cu_seqlens, scale)
+    return ref.prefill_attn(q, k, v, cu_seqlens, scale)
+
+
+def sample(logits: torch.Tensor, temperatures: torch.Tensor,
+           seed: int = 0) -> torch.Tensor:
+    if logits.is_cuda:
+        _gpu()
+        return torch.ops.rlli.sample(logits, temperatures, seed)
+    return ref.sample(logits, temperatures, seed)
+
+
+build_cos_sin = ref.build_cos_sin

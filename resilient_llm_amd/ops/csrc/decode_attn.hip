@@ -1,0 +1,199 @@
+// Paged GQA decode attention (one new query token per sequence).
+//
+// Memory-bound: the cost is streaming each sequence's K/V exactly once
+// from HBM.  Geometry per workgroup = (sequence, kv-head), 256 threads =
+// 4 waves:
+//   - the GROUP (= n_q / n_kv, Llama-8B: 4) query heads that share this
+//     kv-head ride along in registers, so K/V bytes are read ONCE for
+//     all of them;
+//   - a wave64 splits into 64/GW lane-groups of GW = head_dim/8 lanes;
+//     each lane holds 8 dims (one bf16x8 = 16 B load — G13), each
+//     lane-group walks its share of the tokens of a cache block, each
+//     wave walks every 4th cache block (split-KV);
+//   - each lane-group keeps a private online-softmax partial
+//     (m, l, acc[GROUP][8]); the 4*(64/GW) partials merge through LDS at
+//     the end (log-sum-exp combine).
+// Lane-group token validity is group-uniform, so the tail branch never
+// diverges within a lane-group's reduction.
+
+#include "common.h"
+
+namespace rlli {
+
+namespace {
+
+constexpr float kNegInf = -1e30f;
+
+template <int GW, int GROUP>
+__global__ __launch_bounds__(256)
+void decode_attn_kernel(const uint16_t* __restrict__ q,
+                        const uint16_t* __restrict__ k_cache,
+                        const uint16_t* __restrict__ v_cache,
+                        const int32_t* __restrict__ block_table,
+                        const int32_t* __restrict__ seq_lens,
+                        uint16_t* __restrict__ out,
+                        int n_kv_heads, int block_size, int max_blocks,
+                        float scale) {
+  constexpr int D = GW * 8;
+  constexpr int GPW = 64 / GW;                 // lane-groups per wave
+  const int seq = blockIdx.x / n_kv_heads;
+  const int kvh = blockIdx.x % n_kv_heads;
+  const int len = seq_lens[seq];
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wave = threadIdx.x >> 6;
+  const int n_waves = blockDim.x >> 6;
+  const int group = lane / GW;
+  const int gl = lane % GW;
+  const int d0 = gl * 8;
+  const int n_part = n_waves * GPW;
+  const int pid = wave * GPW + group;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* accs = reinterpret_cast<float*>(smem_raw);       // [n_part][GROUP][D]
+  float* ml = accs + n_part * GROUP * D;                  // [n_part][GROUP][2]
+
+  // ---- load the GROUP query heads that map to this kv head ----
+  float qv[GROUP][8];
+#pragma unroll
+  for (int h = 0; h < GROUP; ++h) {
+    bf16x8 qh;
+    qh.u = *reinterpret_cast<const uint4*>(
+        q + (int64_t(seq) * n_kv_heads * GROUP + kvh * GROUP + h) * D + d0);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) qv[h][i] = bf16_to_f32(qh.s[i]);
+  }
+
+  float m[GROUP], l[GROUP], acc[GROUP][8];
+#pragma unroll
+  for (int h = 0; h < GROUP; ++h) {
+    m[h] = kNegInf;
+    l[h] = 0.f;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) acc[h][i] = 0.f;
+  }
+
+  const int n_blocks = (len + block_size - 1) / block_size;
+  const int tok_per_grp = block_size / GPW;
+  for (int b = wave; b < n_blocks; b += n_waves) {
+    const int phys = block_table[int64_t(seq) * max_blocks + b];
+    const int64_t kv_base =
+        (int64_t(phys) * n_kv_heads + kvh) * block_size * D;
+    for (int i = 0; i < tok_per_grp; ++i) {
+      const int tok = group + GPW * i;         // token within the block
+      if (b * block_size + tok >= len) continue;   // group-uniform tail
+      bf16x8 kv;
+      kv.u = *reinterpret_cast<const uint4*>(
+          k_cache + kv_base + int64_t(tok) * D + d0);
+      float kf[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) kf[j] = bf16_to_f32(kv.s[j]);
+
+      bf16x8 vv;
+      vv.u = *reinterpret_cast<const uint4*>(
+          v_cache + kv_base + int64_t(tok) * D + d0);
+      float vf[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vf[j] = bf16_to_f32(vv.s[j]);
+
+#pragma unroll
+      for (int h = 0; h < GROUP; ++h) {
+        float s = 0.f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) s += qv[h][j] * kf[j];
+        s = group_sum<GW>(s);
+        s *= scale;
+        const float m_new = fmaxf(m[h], s);
+        const float corr = __expf(m[h] - m_new);
+        const float p = __expf(s - m_new);
+        m[h] = m_new;
+        l[h] = l[h] * corr + p;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[h][j] = acc[h][j] * corr + p * vf[j];
+      }
+    }
+  }
+
+  // ---- publish per-lane-group partials ----
+#pragma unroll
+  for (int h = 0; h < GROUP; ++h) {
+    float* dst = accs + ((int64_t(pid) * GROUP + h) * D) + d0;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dst[j] = acc[h][j];
+    if (gl == 0) {
+      ml[(pid * GROUP + h) * 2 + 0] = m[h];
+      ml[(pid * GROUP + h) * 2 + 1] = l[h];
+    }
+  }
+  __syncthreads();
+
+  // ---- merge: each thread owns (head, dim) output elements ----
+  for (int idx = threadIdx.x; idx < GROUP * D; idx += blockDim.x) {
+    const int h = idx / D;
+    const int d = idx % D;
+    float M = kNegInf;
+    for (int p = 0; p < n_part; ++p)
+      M = fmaxf(M, ml[(p * GROUP + h) * 2 + 0]);
+    float num = 0.f, den = 0.f;
+    for (int p = 0; p < n_part; ++p) {
+      const float w = __expf(ml[(p * GROUP + h) * 2 + 0] - M);
+      num += w * accs[(int64_t(p) * GROUP + h) * D + d];
+      den += w * ml[(p * GROUP + h) * 2 + 1];
+    }
+    out[(int64_t(seq) * n_kv_heads * GROUP + kvh * GROUP + h) * D + d] =
+        f32_to_bf16(den > 0.f ? num / den : 0.f);
+  }
+}
+
+template <int GW, int GROUP>
+void dispatch_decode(const uint16_t* q, const uint16_t* k_cache,
+                     const uint16_t* v_cache, const int32_t* block_table,
+                     const int32_t* seq_lens, uint16_t* out, int batch,
+                     int n_kv_heads, int block_size, int max_blocks,
+                     float scale, hipStream_t stream) {
+  constexpr int D = GW * 8;
+  constexpr int GPW = 64 / GW;
+  const int n_part = 4 * GPW;
+  const size_t smem = size_t(n_part) * GROUP * (D + 2) * sizeof(float);
+  hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP>),
+                     dim3(batch * n_kv_heads), dim3(256), smem, stream,
+                     q, k_cache, v_cache, block_table, seq_lens, out,
+                     n_kv_heads, block_size, max_blocks, scale);
+}
+
+}  // namespace
+
+void launch_decode_attn(const uint16_t* q, const uint16_t* k_cache,
+                        const uint16_t* v_cache, const int32_t* block_table,
+                        const int32_t* seq_lens, uint16_t* out, int batch,
+                        int n_q_heads, int n_kv_heads, int head_dim,
+                        int block_size, int max_blocks, float scale,
+                        hipStream_t stream) {
+  if (batch == 0) return;
+  const int group = n_q_heads / n_kv_heads;
+  auto run = [&](auto gw_tag, auto group_tag) {
+    dispatch_decode<decltype(gw_tag)::value, decltype(group_tag)::value>(
+        q, k_cache, v_cache, block_table, seq_lens, out, batch, n_kv_heads,
+        block_size, max_blocks, scale, stream);
+  };
+  using I8 = std::integral_constant<int, 8>;
+  using I16 = std::integral_constant<int, 16>;
+  using G1 = std::integral_constant<int, 1>;
+  using G2 = std::integral_constant<int, 2>;
+  using G4 = std::integral_constant<int, 4>;
+  using G8 = std::integral_constant<int, 8>;
+  if (head_dim == 128) {
+    if (group == 1) run(I16{}, G1{});
+    else if (group == 2) run(I16{}, G2{});
+    else if (group == 4) run(I16{}, G4{});
+    else if (group == 8) run(I16{}, G8{});
+    else return;   // caller validates
+  } else if (head_dim == 64) {
+    if (group == 1) run(I8{}, G1{});
+    else if (group == 2) run(I8{}, G2{});
+    else if (group == 4) run(I8{}, G4{});
+    else if (group == 8) run(I8{}, G8{});
+    else return;
+  }
+}
+
+}  // namespace rlli

@@ -1,0 +1,208 @@
+// Torch bindings for the gfx950 kernels (TORCH_LIBRARY, not pybind — the
+// ops dispatch through torch.ops.rlli.* and stay graph-capturable).
+// This is the only translation unit that includes torch headers; the
+// kernels live in pure-HIP .hip files behind kernels.h.
+
+#include <torch/library.h>
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+#include <c10/hip/HIPGuard.h>
+
+#include "kernels.h"
+
+namespace {
+
+using at::Tensor;
+
+hipStream_t current_stream(const Tensor& t) {
+  return c10::hip::getCurrentHIPStream(t.device().index()).stream();
+}
+
+uint16_t* bf16_ptr(const Tensor& t) {
+  return reinterpret_cast<uint16_t*>(t.data_ptr());
+}
+
+void check_bf16_contig(const Tensor& t, const char* name) {
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.is_cuda(), name, " must be on the GPU");
+}
+
+// ------------------------------------------------------------- rmsnorm
+Tensor rmsnorm(const Tensor& x, const Tensor& w, double eps) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(w, "w");
+  const int dim = int(x.size(-1));
+  TORCH_CHECK(dim % 8 == 0, "dim must be a multiple of 8");
+  TORCH_CHECK(w.numel() == dim, "weight/dim mismatch");
+  const int rows = int(x.numel() / dim);
+  Tensor y = at::empty_like(x);
+  c10::hip::HIPGuard guard(x.device());
+  rlli::launch_rmsnorm(bf16_ptr(x), nullptr, bf16_ptr(w), bf16_ptr(y), rows,
+                       dim, float(eps), current_stream(x));
+  return y;
+}
+
+Tensor rmsnorm_residual_(const Tensor& x, Tensor residual, const Tensor& w,
+                         double eps) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(residual, "residual");
+  check_bf16_contig(w, "w");
+  TORCH_CHECK(x.sizes() == residual.sizes(), "x/residual shape mismatch");
+  const int dim = int(x.size(-1));
+  TORCH_CHECK(dim % 8 == 0 && w.numel() == dim, "bad dim/weight");
+  const int rows = int(x.numel() / dim);
+  Tensor y = at::empty_like(x);
+  c10::hip::HIPGuard guard(x.device());
+  rlli::launch_rmsnorm(bf16_ptr(x), bf16_ptr(residual), bf16_ptr(w),
+                       bf16_ptr(y), rows, dim, float(eps), current_stream(x));
+  return y;
+}
+
+// ------------------------------------------------------------ silu_mul
+Tensor silu_mul(const Tensor& gate_up) {
+  check_bf16_contig(gate_up, "gate_up");
+  const int twoi = int(gate_up.size(-1));
+  TORCH_CHECK(twoi % 16 == 0, "2*intermediate must be a multiple of 16");
+  const int inter = twoi / 2;
+  const int rows = int(gate_up.numel() / twoi);
+  auto sizes = gate_up.sizes().vec();
+  sizes.back() = inter;
+  Tensor y = at::empty(sizes, gate_up.options());
+  c10::hip::HIPGuard guard(gate_up.device());
+  rlli::launch_silu_mul(bf16_ptr(gate_up), bf16_ptr(y), rows, inter,
+                        current_stream(gate_up));
+  return y;
+}
+
+// ------------------------------------------------------- rope_kv_append
+void rope_kv_append_(Tensor q, Tensor k, const Tensor& v,
+                     const Tensor& positions, const Tensor& cos_sin,
+                     Tensor k_cache, Tensor v_cache,
+                     const Tensor& slot_mapping) {
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  check_bf16_contig(v, "v");
+  check_bf16_contig(k_cache, "k_cache");
+  check_bf16_contig(v_cache, "v_cache");
+  TORCH_CHECK(positions.scalar_type() == at::kInt && positions.is_contiguous());
+  TORCH_CHECK(slot_mapping.scalar_type() == at::kInt && slot_mapping.is_contiguous());
+  TORCH_CHECK(cos_sin.scalar_type() == at::kFloat && cos_sin.is_contiguous());
+  const int tokens = int(q.size(0));
+  const int n_q = int(q.size(1));
+  const int n_kv = int(k.size(1));
+  const int D = int(q.size(2));
+  TORCH_CHECK(k.size(2) == D && v.size(1) == n_kv && v.size(2) == D);
+  TORCH_CHECK(D % 16 == 0, "head_dim must be a multiple of 16");
+  TORCH_CHECK(cos_sin.size(1) == D, "cos_sin must be [max_pos, head_dim]");
+  TORCH_CHECK(k_cache.size(1) == n_kv && k_cache.size(3) == D);
+  const int block_size = int(k_cache.size(2));
+  TORCH_CHECK(positions.numel() == tokens && slot_mapping.numel() == tokens);
+  c10::hip::HIPGuard guard(q.device());
+  rlli::launch_rope_kv_append(
+      bf16_ptr(q), bf16_ptr(k), bf16_ptr(v),
+      positions.data_ptr<int32_t>(), cos_sin.data_ptr<float>(),
+      bf16_ptr(k_cache), bf16_ptr(v_cache), slot_mapping.data_ptr<int32_t>(),
+      tokens, n_q, n_kv, D, block_size, current_stream(q));
+}
+
+// ---------------------------------------------------------- decode_attn
+Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
+                   const Tensor& v_cache, const Tensor& block_table,
+                   const Tensor& seq_lens, double scale) {
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k_cache, "k_cache");
+  check_bf16_contig(v_cache, "v_cache");
+  TORCH_CHECK(block_table.scalar_type() == at::kInt && block_table.is_contiguous());
+  TORCH_CHECK(seq_lens.scalar_type() == at::kInt && seq_lens.is_contiguous());
+  const int batch = int(q.size(0));
+  const int n_q = int(q.size(1));
+  const int D = int(q.size(2));
+  const int n_kv = int(k_cache.size(1));
+  const int block_size = int(k_cache.size(2));
+  const int max_blocks = int(block_table.size(1));
+  const int group = n_q / n_kv;
+  TORCH_CHECK(n_q % n_kv == 0 && (group == 1 || group == 2 || group == 4 || group == 8),
+              "GQA group must be 1/2/4/8, got ", group);
+  TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128, got ", D);
+  const int gpw = 64 / (D / 8);
+  TORCH_CHECK(block_size % gpw == 0, "block_size must be a multiple of ", gpw);
+  TORCH_CHECK(k_cache.size(3) == D && v_cache.sizes() == k_cache.sizes());
+  TORCH_CHECK(block_table.size(0) == batch && seq_lens.numel() == batch);
+  Tensor out = at::empty_like(q);
+  c10::hip::HIPGuard guard(q.device());
+  rlli::launch_decode_attn(
+      bf16_ptr(q), bf16_ptr(k_cache), bf16_ptr(v_cache),
+      block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
+      bf16_ptr(out), batch, n_q, n_kv, D, block_size, max_blocks,
+      float(scale), current_stream(q));
+  return out;
+}
+
+// --------------------------------------------------------- prefill_attn
+Tensor prefill_attn(const Tensor& q, const Tensor& k, const Tensor& v,
+                    const Tensor& cu_seqlens, double scale) {
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  check_bf16_contig(v, "v");
+  TORCH_CHECK(cu_seqlens.scalar_type() == at::kInt && cu_seqlens.is_contiguous());
+  const int T = int(q.size(0));
+  const int n_q = int(q.size(1));
+  const int D = int(q.size(2));
+  const int n_kv = int(k.size(1));
+  const int group = n_q / n_kv;
+  TORCH_CHECK(n_q % n_kv == 0 && (group == 1 || group == 2 || group == 4 || group == 8),
+              "GQA group must be 1/2/4/8, got ", group);
+  TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128, got ", D);
+  TORCH_CHECK(k.size(0) == T && v.size(0) == T && k.size(2) == D);
+  const int n_seqs = int(cu_seqlens.numel()) - 1;
+  Tensor out = at::empty_like(q);
+  c10::hip::HIPGuard guard(q.device());
+  rlli::launch_prefill_attn(
+      bf16_ptr(q), bf16_ptr(k), bf16_ptr(v), cu_seqlens.data_ptr<int32_t>(),
+      bf16_ptr(out), n_seqs, T, n_q, n_kv, D, float(scale),
+      current_stream(q));
+  return out;
+}
+
+// --------------------------------------------------------------- sample
+Tensor sample(const Tensor& logits, const Tensor& temperatures, int64_t seed) {
+  check_bf16_contig(logits, "logits");
+  TORCH_CHECK(temperatures.scalar_type() == at::kFloat &&
+              temperatures.is_contiguous() && temperatures.is_cuda());
+  const int batch = int(logits.size(0));
+  const int vocab = int(logits.size(1));
+  TORCH_CHECK(temperatures.numel() == batch);
+  Tensor out = at::empty({batch}, logits.options().dtype(at::kInt));
+  c10::hip::HIPGuard guard(logits.device());
+  rlli::launch_sample(bf16_ptr(logits), temperatures.data_ptr<float>(),
+                      uint64_t(seed), out.data_ptr<int32_t>(), batch, vocab,
+                      current_stream(logits));
+  return out;
+}
+
+}  // namespace
+
+TORCH_LIBRARY(rlli, m) {
+  m.def("rmsnorm(Tensor x, Tensor w, float eps) -> Tensor");
+  m.def("rmsnorm_residual_(Tensor x, Tensor(a!) residual, Tensor w, float eps) -> Tensor");
+  m.def("silu_mul(Tensor gate_up) -> Tensor");
+  m.def("rope_kv_append_(Tensor(a!) q, Tensor(b!) k, Tensor v, Tensor positions, "
+        "Tensor cos_sin, Tensor(c!) k_cache, Tensor(d!) v_cache, "
+        "Tensor slot_mapping) -> ()");
+  m.def("decode_attn(Tensor q, Tensor k_cache, Tensor v_cache, "
+        "Tensor block_table, Tensor seq_lens, float scale) -> Tensor");
+  m.def("prefill_attn(Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, "
+        "float scale) -> Tensor");
+  m.def("sample(Tensor logits, Tensor temperatures, int seed) -> Tensor");
+}
+
+TORCH_LIBRARY_IMPL(rlli, CUDA, m) {
+  m.impl("rmsnorm", &rmsnorm);
+  m.impl("rmsnorm_residual_", &rmsnorm_residual_);
+  m.impl("silu_mul", &silu_mul);
+  m.impl("rope_kv_append_", &rope_kv_append_);
+  m.impl("decode_attn", &decode_attn);
+  m.impl("prefill_attn", &prefill_attn);
+  m.impl("sample", &sample);
+}

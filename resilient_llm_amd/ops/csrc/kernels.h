@@ -1,0 +1,60 @@
+// Launch-function declarations: pure-HIP translation units (one per
+// kernel family) export these; ext.cpp (the only file that includes
+// torch headers) validates tensors and calls them.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+namespace rlli {
+
+// y = rmsnorm(x) * w           (residual == nullptr)
+// r += x; y = rmsnorm(r) * w   (fused residual-add form)
+void launch_rmsnorm(const uint16_t* x, uint16_t* residual, const uint16_t* w,
+                    uint16_t* y, int rows, int dim, float eps,
+                    hipStream_t stream);
+
+// out[t, i] = silu(gu[t, i]) * gu[t, I + i]   for packed gate|up rows
+void launch_silu_mul(const uint16_t* gate_up, uint16_t* out, int rows,
+                     int inter, hipStream_t stream);
+
+// In-place RoPE (NeoX-interleaved-halves style) on q and k, then scatter
+// k/v into the paged cache at slot_mapping[t].
+void launch_rope_kv_append(
+    uint16_t* q, uint16_t* k, uint16_t* v,
+    const int32_t* positions, const float* cos_sin,   // [max_pos, head_dim]
+    uint16_t* k_cache, uint16_t* v_cache,
+    const int32_t* slot_mapping,
+    int tokens, int n_q_heads, int n_kv_heads, int head_dim,
+    int block_size, hipStream_t stream);
+
+// Paged GQA decode attention: one new q token per sequence.
+void launch_decode_attn(
+    const uint16_t* q,                 // [batch, n_q_heads, head_dim]
+    const uint16_t* k_cache,           // [blocks, n_kv, block_size, head_dim]
+    const uint16_t* v_cache,
+    const int32_t* block_table,        // [batch, max_blocks]
+    const int32_t* seq_lens,           // [batch]
+    uint16_t* out,                     // [batch, n_q_heads, head_dim]
+    int batch, int n_q_heads, int n_kv_heads, int head_dim,
+    int block_size, int max_blocks, float scale, hipStream_t stream);
+
+// Varlen causal prefill attention over in-batch q/k/v.
+void launch_prefill_attn(
+    const uint16_t* q, const uint16_t* k, const uint16_t* v,
+    const int32_t* cu_seqlens,         // [n_seqs + 1]
+    uint16_t* out,
+    int n_seqs, int total_tokens, int n_q_heads, int n_kv_heads,
+    int head_dim, float scale, hipStream_t stream);
+
+// Fused sampling: greedy argmax when temperature[i] == 0, else Gumbel-max
+// sampling of softmax(logits / temperature[i]) with an in-kernel counter
+// hash RNG keyed on (seed, row, column) — no 32 MB random tensor per step.
+void launch_sample(
+    const uint16_t* logits,            // [batch, vocab] bf16
+    const float* temperatures,         // [batch]
+    uint64_t seed,
+    int32_t* out_tokens,               // [batch]
+    int batch, int vocab, hipStream_t stream);
+
+}  // namespace rlli

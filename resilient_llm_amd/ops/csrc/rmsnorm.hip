@@ -1,0 +1,165 @@
+// Fused (residual-add +) RMSNorm, bf16 I/O, f32 accumulation.
+//
+// Memory-bound: the lever on gfx950 is bf16x8 (16 B/lane) vectorized
+// loads — scalar bf16 loads are ~2x slower (guide G13; measured 2.35 vs
+// 4.89 TB/s on this op class).  One workgroup per row, one bf16x8 per
+// lane when dim <= 8192 (Llama-8B dim 4096 -> 512 threads, 70B 8192 ->
+// 1024), two-pass re-read (L2-hot) for larger dims.
+//
+// Replaces: no reference kernel exists (the reference runs no model code,
+// SURVEY.md §2.1 native-code census) — greenfield per BASELINE.json.
+
+#include "common.h"
+
+namespace rlli {
+
+namespace {
+
+// Cross-wave reduction of one f32 per wave through LDS; every thread
+// returns the block total.
+template <int MAX_WAVES>
+DEV_INLINE float block_sum(float v, float* lds) {
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wave = threadIdx.x / kWave;
+  v = wave_sum(v);
+  if (lane == 0) lds[wave] = v;
+  __syncthreads();
+  const int n_waves = (blockDim.x + kWave - 1) / kWave;
+  float total = 0.f;
+#pragma unroll
+  for (int w = 0; w < MAX_WAVES; ++w)
+    if (w < n_waves) total += lds[w];
+  return total;
+}
+
+// dim/8 <= blockDim: one bf16x8 per lane, data stays in registers.
+template <bool FUSE_RESIDUAL>
+__global__ void rmsnorm_one_vec(const uint16_t* __restrict__ x,
+                                uint16_t* __restrict__ residual,
+                                const uint16_t* __restrict__ w,
+                                uint16_t* __restrict__ y,
+                                int dim, float eps) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* lds = reinterpret_cast<float*>(smem_raw);
+  const int row = blockIdx.x;
+  const int64_t base = int64_t(row) * dim;
+  const int v8 = threadIdx.x;            // which bf16x8 of the row
+  const bool active = v8 * 8 < dim;
+
+  float vals[8];
+  if (active) {
+    bf16x8 xv;
+    xv.u = *reinterpret_cast<const uint4*>(x + base + v8 * 8);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) vals[i] = bf16_to_f32(xv.s[i]);
+    if constexpr (FUSE_RESIDUAL) {
+      bf16x8 rv;
+      rv.u = *reinterpret_cast<const uint4*>(residual + base + v8 * 8);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) vals[i] += bf16_to_f32(rv.s[i]);
+      bf16x8 out;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) out.s[i] = f32_to_bf16(vals[i]);
+      *reinterpret_cast<uint4*>(residual + base + v8 * 8) = out.u;
+      // re-read the rounded residual so y == rmsnorm(stored residual)
+#pragma unroll
+      for (int i = 0; i < 8; ++i) vals[i] = bf16_to_f32(out.s[i]);
+    }
+  } else {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) vals[i] = 0.f;
+  }
+
+  float ss = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) ss += vals[i] * vals[i];
+  ss = block_sum<16>(ss, lds);
+  const float inv = rsqrtf(ss / dim + eps);
+
+  if (active) {
+    bf16x8 wv, out;
+    wv.u = *reinterpret_cast<const uint4*>(w + v8 * 8);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      out.s[i] = f32_to_bf16(vals[i] * inv * bf16_to_f32(wv.s[i]));
+    *reinterpret_cast<uint4*>(y + base + v8 * 8) = out.u;
+  }
+}
+
+// Large dims: pass 1 accumulates sumsq (and stores fused residual),
+// pass 2 re-reads (L2-hot) and writes.
+template <bool FUSE_RESIDUAL>
+__global__ void rmsnorm_two_pass(const uint16_t* __restrict__ x,
+                                 uint16_t* __restrict__ residual,
+                                 const uint16_t* __restrict__ w,
+                                 uint16_t* __restrict__ y,
+                                 int dim, float eps) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* lds = reinterpret_cast<float*>(smem_raw);
+  const int row = blockIdx.x;
+  const int64_t base = int64_t(row) * dim;
+  const int nvec = dim / 8;
+
+  float ss = 0.f;
+  for (int v8 = threadIdx.x; v8 < nvec; v8 += blockDim.x) {
+    bf16x8 xv;
+    xv.u = *reinterpret_cast<const uint4*>(x + base + v8 * 8);
+    float vals[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) vals[i] = bf16_to_f32(xv.s[i]);
+    if constexpr (FUSE_RESIDUAL) {
+      bf16x8 rv;
+      rv.u = *reinterpret_cast<const uint4*>(residual + base + v8 * 8);
+      bf16x8 out;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        vals[i] += bf16_to_f32(rv.s[i]);
+        out.s[i] = f32_to_bf16(vals[i]);
+        vals[i] = bf16_to_f32(out.s[i]);
+      }
+      *reinterpret_cast<uint4*>(residual + base + v8 * 8) = out.u;
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) ss += vals[i] * vals[i];
+  }
+  ss = block_sum<16>(ss, lds);
+  const float inv = rsqrtf(ss / dim + eps);
+
+  const uint16_t* src = FUSE_RESIDUAL ? residual : x;
+  for (int v8 = threadIdx.x; v8 < nvec; v8 += blockDim.x) {
+    bf16x8 sv, wv, out;
+    sv.u = *reinterpret_cast<const uint4*>(src + base + v8 * 8);
+    wv.u = *reinterpret_cast<const uint4*>(w + v8 * 8);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      out.s[i] = f32_to_bf16(bf16_to_f32(sv.s[i]) * inv * bf16_to_f32(wv.s[i]));
+    *reinterpret_cast<uint4*>(y + base + v8 * 8) = out.u;
+  }
+}
+
+}  // namespace
+
+void launch_rmsnorm(const uint16_t* x, uint16_t* residual, const uint16_t* w,
+                    uint16_t* y, int rows, int dim, float eps,
+                    hipStream_t stream) {
+  if (rows == 0) return;
+  const int smem = 16 * sizeof(float);
+  if (dim <= 8192) {
+    int threads = ((dim / 8 + kWave - 1) / kWave) * kWave;
+    if (residual)
+      hipLaunchKernelGGL((rmsnorm_one_vec<true>), dim3(rows), dim3(threads),
+                         smem, stream, x, residual, w, y, dim, eps);
+    else
+      hipLaunchKernelGGL((rmsnorm_one_vec<false>), dim3(rows), dim3(threads),
+                         smem, stream, x, nullptr, w, y, dim, eps);
+  } else {
+    if (residual)
+      hipLaunchKernelGGL((rmsnorm_two_pass<true>), dim3(rows), dim3(1024),
+                         smem, stream, x, residual, w, y, dim, eps);
+    else
+      hipLaunchKernelGGL((rmsnorm_two_pass<false>), dim3(rows), dim3(1024),
+                         smem, stream, x, nullptr, w, y, dim, eps);
+  }
+}
+
+}  // namespace rlli

@@ -1,0 +1,154 @@
+"""Numerics tests: each gfx950 HIP kernel vs its plain-PyTorch fp32
+reference (tolerances sized for bf16 I/O).  All gpu-marked."""
+
+import math
+
+import pytest
+import torch
+
+import resilient_llm_amd.ops as ops
+from resilient_llm_amd.ops import ref
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def assert_close_bf16(a, b, atol=2e-2, rtol=2e-2, msg=""):
+    af, bf = a.float(), b.float()
+    torch.testing.assert_close(af, bf, atol=atol, rtol=rtol, msg=msg)
+
+
+@pytest.fixture(autouse=True)
+def _seed():
+    torch.manual_seed(1234)
+
+
+@pytest.mark.parametrize("rows,dim", [(1, 256), (64, 4096), (17, 4096),
+                                      (8, 8192), (3, 16384)])
+def test_rmsnorm(rows, dim):
+    x = torch.randn(rows, dim, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(dim, dtype=torch.bfloat16, device=DEV)
+    y = ops.rmsnorm(x, w, 1e-5)
+    y_ref = ref.rmsnorm(x.cpu(), w.cpu(), 1e-5)
+    assert_close_bf16(y.cpu(), y_ref)
+
+
+def test_rmsnorm_residual_fused():
+    x = torch.randn(32, 4096, dtype=torch.bfloat16, device=DEV)
+    r = torch.randn(32, 4096, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(4096, dtype=torch.bfloat16, device=DEV)
+    r_cpu = r.cpu().clone()
+    x_cpu = x.cpu().clone()
+    y = ops.rmsnorm_residual_(x, r, w, 1e-5)
+    y_ref = ref.rmsnorm_residual_(x_cpu, r_cpu, w.cpu(), 1e-5)
+    assert_close_bf16(y.cpu(), y_ref)
+    assert_close_bf16(r.cpu(), r_cpu)   # updated residual matches
+
+
+@pytest.mark.parametrize("rows,inter", [(4, 128), (64, 14336), (7, 1024)])
+def test_silu_mul(rows, inter):
+    gu = torch.randn(rows, 2 * inter, dtype=torch.bfloat16, device=DEV)
+    y = ops.silu_mul(gu)
+    y_ref = ref.silu_mul(gu.cpu())
+    assert_close_bf16(y.cpu(), y_ref)
+
+
+@pytest.mark.parametrize("n_q,n_kv,D", [(32, 8, 128), (8, 8, 128), (4, 2, 64)])
+def test_rope_kv_append(n_q, n_kv, D):
+    T, blocks, bs = 10, 8, 16
+    q = torch.randn(T, n_q, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, n_kv, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, n_kv, D, dtype=torch.bfloat16, device=DEV)
+    pos = torch.randint(0, 500, (T,), dtype=torch.int32, device=DEV)
+    cs = ops.build_cos_sin(512, D, device=DEV)
+    kc = torch.zeros(blocks, n_kv, bs, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    slots = torch.arange(3, 3 + T, dtype=torch.int32, device=DEV) * 7 % (blocks * bs)
+    q2, k2, v2 = q.cpu().clone(), k.cpu().clone(), v.cpu().clone()
+    kc2, vc2 = kc.cpu().clone(), vc.cpu().clone()
+
+    ops.rope_kv_append_(q, k, v, pos, cs, kc, vc, slots)
+    ref.rope_kv_append_(q2, k2, v2, pos.cpu(), cs.cpu(), kc2, vc2, slots.cpu())
+    assert_close_bf16(q.cpu(), q2)
+    assert_close_bf16(k.cpu(), k2)
+    assert_close_bf16(kc.cpu(), kc2)
+    assert_close_bf16(vc.cpu(), vc2)
+
+
+@pytest.mark.parametrize("batch,n_q,n_kv,D,lens", [
+    (4, 32, 8, 128, [1, 17, 100, 256]),
+    (2, 8, 8, 128, [33, 64]),
+    (3, 16, 2, 64, [5, 90, 41]),
+    (1, 8, 1, 128, [300]),
+])
+def test_decode_attn(batch, n_q, n_kv, D, lens):
+    bs = 16
+    max_blocks = (max(lens) + bs - 1) // bs + 1
+    n_blocks_total = batch * max_blocks + 2
+    kc = torch.randn(n_blocks_total, n_kv, bs, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    q = torch.randn(batch, n_q, D, dtype=torch.bfloat16, device=DEV)
+    perm = torch.randperm(n_blocks_total, device=DEV)[:batch * max_blocks]
+    bt = perm.reshape(batch, max_blocks).int().contiguous()
+    seq_lens = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.decode_attn(q, kc, vc, bt, seq_lens, scale)
+    out_ref = ref.decode_attn(q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
+                              seq_lens.cpu(), scale)
+    assert_close_bf16(out.cpu(), out_ref)
+
+
+@pytest.mark.parametrize("n_q,n_kv,D,lens", [
+    (32, 8, 128, [5, 128, 63]),
+    (8, 8, 128, [200]),
+    (16, 2, 64, [1, 300, 37]),
+])
+def test_prefill_attn(n_q, n_kv, D, lens):
+    T = sum(lens)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32, device=DEV)
+    q = torch.randn(T, n_q, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, n_kv, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, n_kv, D, dtype=torch.bfloat16, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.prefill_attn(q, k, v, cu, scale)
+    out_ref = ref.prefill_attn(q.cpu(), k.cpu(), v.cpu(), cu.cpu(), scale)
+    assert_close_bf16(out.cpu(), out_ref)
+
+
+def test_sample_greedy_matches_argmax():
+    torch.manual_seed(7)
+    logits = torch.randn(16, 128256, dtype=torch.bfloat16, device=DEV)
+    temps = torch.zeros(16, device=DEV)
+    toks = ops.sample(logits, temps, seed=1)
+    ref_toks = logits.float().argmax(-1).int()
+    assert torch.equal(toks.cpu(), ref_toks.cpu())
+
+
+def test_sample_temperature_distribution():
+    """Gumbel-max sampling follows softmax(logits/T): chi-square-ish
+    sanity on a 4-way categorical."""
+    vocab = 8
+    logits = torch.full((1, vocab), -1e4, dtype=torch.bfloat16, device=DEV)
+    logits[0, :4] = torch.tensor([2.0, 1.0, 0.0, -1.0], dtype=torch.bfloat16)
+    temps = torch.ones(1, device=DEV)
+    counts = torch.zeros(vocab)
+    n = 2000
+    for s in range(n):
+        t = ops.sample(logits, temps, seed=s)
+        counts[t.item()] += 1
+    probs = torch.softmax(logits[0, :4].float(), -1)
+    emp = counts[:4] / n
+    assert counts[4:].sum() == 0
+    assert torch.allclose(emp, probs.cpu(), atol=0.05), (emp, probs)
+
+
+def test_sample_deterministic_given_seed():
+    logits = torch.randn(4, 1000, dtype=torch.bfloat16, device=DEV)
+    temps = torch.full((4,), 0.8, device=DEV)
+    a = ops.sample(logits, temps, seed=42)
+    b = ops.sample(logits, temps, seed=42)
+    c = ops.sample(logits, temps, seed=43)
+    assert torch.equal(a, b)
+    assert not torch.equal(a, c)   # overwhelmingly likely
