@@ -1,0 +1,238 @@
+"""Continuous-batching inference engine (one per GPU worker).
+
+The serving hot loop (SURVEY.md §3.2 "MI355X equivalent"): admitted
+sequences share every decode step (one fused forward for the whole
+batch); new arrivals are prefilled in batched varlen form and join the
+running batch the same iteration.  Prefill-priority scheduling, paged KV
+with full-reservation admission (a sequence admitted never dies of OOM —
+over-capacity surfaces at admission as Throttled, X13).
+
+The decode step is deliberately shaped for hipGraph capture (static
+tensor shapes per batch-size bucket — see engine/graph.py).
+"""
+
+from __future__ import annotations
+
+import collections
+import dataclasses
+import time
+from typing import Optional
+
+import torch
+
+from .. import ops
+from .kv_cache import PagedKVCache
+
+
+@dataclasses.dataclass
+class SamplingParams:
+    max_tokens: int = 128
+    temperature: float = 0.0
+    seed: Optional[int] = None
+    stop_on_eos: bool = True
+
+
+@dataclasses.dataclass
+class StepOutput:
+    req_id: str
+    token_id: int
+    finished: bool
+    finish_reason: Optional[str] = None   # stop | length
+    n_output_tokens: int = 0
+
+
+@dataclasses.dataclass
+class SeqState:
+    req_id: str
+    prompt_ids: list
+    params: SamplingParams
+    output_ids: list = dataclasses.field(default_factory=list)
+    blocks: list = dataclasses.field(default_factory=list)
+    n_cached: int = 0
+    arrived_at: float = dataclasses.field(default_factory=time.monotonic)
+
+    @property
+    def reserved_blocks_needed(self) -> int:
+        return -(-(len(self.prompt_ids) + self.params.max_tokens) // 16)
+
+
+class CapacityExceeded(RuntimeError):
+    """Queue + KV budget cannot take this request now (-> 429/Throttled)."""
+
+
+class LLMEngine:
+    def __init__(self, model, kv_cache: PagedKVCache,
+                 max_batch_size: int = 64,
+                 max_prefill_tokens: int = 8192,
+                 max_queue: int = 256,
+                 seed: int = 0) -> None:
+        self.model = model
+        self.kv = kv_cache
+        self.device = model.device
+        self.max_batch_size = max_batch_size
+        self.max_prefill_tokens = max_prefill_tokens
+        self.max_queue = max_queue
+        self.seed = seed
+        self.eos_id = 2
+        self.waiting: collections.deque[SeqState] = collections.deque()
+        self.running: list[SeqState] = []
+        self.step_count = 0
+        self.block_size = kv_cache.block_size
+        self._aborted: set[str] = set()
+        # decode graph runner installed by engine/graph.py (GPU only)
+        self.graph_runner = None
+
+    # ------------------------------------------------------------- admin
+    @property
+    def n_active(self) -> int:
+        return len(self.waiting) + len(self.running)
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    def add_request(self, req_id: str, prompt_ids: list,
+                    params: Optional[SamplingParams] = None) -> None:
+        params = params or SamplingParams()
+        if len(self.waiting) >= self.max_queue:
+            raise CapacityExceeded(f"queue full ({self.max_queue})")
+        seq = SeqState(req_id=req_id, prompt_ids=list(prompt_ids), params=params)
+        if seq.reserved_blocks_needed > self.kv.num_blocks:
+            raise CapacityExceeded(
+                f"request needs {seq.reserved_blocks_needed} KV blocks, "
+                f"cache has {self.kv.num_blocks}")
+        self.waiting.append(seq)
+
+    def abort(self, req_id: str) -> None:
+        self._aborted.add(req_id)
+
+    # --------------------------------------------------------- scheduling
+    def _admit(self) -> list[SeqState]:
+        admitted: list[SeqState] = []
+        tokens = 0
+        while (self.waiting
+               and len(self.running) + len(admitted) < self.max_batch_size):
+            seq = self.waiting[0]
+            if seq.req_id in self._aborted:
+                self.waiting.popleft()
+                self._aborted.discard(seq.req_id)
+                continue
+            if tokens + len(seq.prompt_ids) > self.max_prefill_tokens and admitted:
+                break
+            if seq.reserved_blocks_needed > self.kv.free_blocks:
+                break
+            seq.blocks = self.kv.allocate(seq.reserved_blocks_needed)
+            tokens += len(seq.prompt_ids)
+            admitted.append(self.waiting.popleft())
+        return admitted
+
+    def _slot(self, seq: SeqState, pos: int) -> int:
+        return seq.blocks[pos // self.block_size] * self.block_size \
+            + pos % self.block_size
+
+    def _finish(self, seq: SeqState, outs: list[StepOutput], token: int) -> bool:
+        p = seq.params
+        reason = None
+        if p.stop_on_eos and token == self.eos_id:
+            reason = "stop"
+        elif len(seq.output_ids) >= p.max_tokens:
+            reason = "length"
+        outs.append(StepOutput(req_id=seq.req_id, token_id=token,
+                               finished=reason is not None,
+                               finish_reason=reason,
+                               n_output_tokens=len(seq.output_ids)))
+        if reason is not None:
+            self.kv.free(seq.blocks)
+            seq.blocks = []
+            return True
+        return False
+
+    # -------------------------------------------------------------- steps
+    def step(self) -> list[StepOutput]:
+        """One engine iteration: a prefill batch if arrivals are waiting
+        and fit, else one decode for the running batch."""
+        self.step_count += 1
+        self._drop_aborted()
+        admitted = self._admit()
+        if admitted:
+            return self._prefill_step(admitted)
+        if self.running:
+            return self._decode_step()
+        return []
+
+    def _drop_aborted(self) -> None:
+        if not self._aborted:
+            return
+        keep = []
+        for seq in self.running:
+            if seq.req_id in self._aborted:
+                self.kv.free(seq.blocks)
+                self._aborted.discard(seq.req_id)
+            else:
+                keep.append(seq)
+        self.running = keep
+
+    def _sample(self, logits: torch.Tensor, seqs: list[SeqState]) -> list[int]:
+        temps = torch.tensor([s.params.temperature for s in seqs],
+                             dtype=torch.float32, device=self.device)
+        seed = (self.seed * 0x9e3779b9 + self.step_count) & 0x7fffffffffffffff
+        toks = ops.sample(logits, temps, seed)
+        return toks.tolist()
+
+    def _prefill_step(self, admitted: list[SeqState]) -> list[StepOutput]:
+        dev = self.device
+        input_ids, positions, slots, cu = [], [], [], [0]
+        for seq in admitted:
+            plen = len(seq.prompt_ids)
+            input_ids.extend(seq.prompt_ids)
+            positions.extend(range(plen))
+            slots.extend(self._slot(seq, p) for p in range(plen))
+            cu.append(cu[-1] + plen)
+        logits = self.model.forward_prefill(
+            torch.tensor(input_ids, dtype=torch.int32, device=dev),
+            torch.tensor(positions, dtype=torch.int32, device=dev),
+            self.kv,
+            torch.tensor(slots, dtype=torch.int32, device=dev),
+            torch.tensor(cu, dtype=torch.int32, device=dev))
+        tokens = self._sample(logits, admitted)
+        outs: list[StepOutput] = []
+        for seq, tok in zip(admitted, tokens):
+            seq.n_cached = len(seq.prompt_ids)
+            seq.output_ids.append(tok)
+            if not self._finish(seq, outs, tok):
+                self.running.append(seq)
+        return outs
+
+    def _decode_step(self) -> list[StepOutput]:
+        dev = self.device
+        seqs = self.running
+        input_ids = [s.output_ids[-1] for s in seqs]
+        positions = [s.n_cached for s in seqs]
+        slots = [self._slot(s, s.n_cached) for s in seqs]
+        seq_lens = [s.n_cached + 1 for s in seqs]
+        max_blocks = max(len(s.blocks) for s in seqs)
+        bt = torch.full((len(seqs), max_blocks), 0, dtype=torch.int32)
+        for i, s in enumerate(seqs):
+            bt[i, :len(s.blocks)] = torch.tensor(s.blocks, dtype=torch.int32)
+        args = (
+            torch.tensor(input_ids, dtype=torch.int32, device=dev),
+            torch.tensor(positions, dtype=torch.int32, device=dev),
+            torch.tensor(slots, dtype=torch.int32, device=dev),
+            bt.to(dev),
+            torch.tensor(seq_lens, dtype=torch.int32, device=dev),
+        )
+        if self.graph_runner is not None:
+            logits = self.graph_runner.run(*args)
+        else:
+            ids, pos, slot_t, bt_d, lens = args
+            logits = self.model.forward_decode(ids, pos, self.kv, slot_t,
+                                               bt_d, lens)
+        tokens = self._sample(logits, seqs)
+        outs: list[StepOutput] = []
+        still_running: list[SeqState] = []
+        for seq, tok in zip(seqs, tokens):
+            seq.n_cached += 1
+            seq.output_ids.append(tok)
+            if not self._finish(seq, outs, tok):
+                still_running.append(seq)
+        self.running = still_running
+        return outs

@@ -1,0 +1,280 @@
+"""Llama-family model, MI355X-first.
+
+No nn.Module graph: a flat parameter store + an explicit forward written
+against the serving ops — fused QKV / gate-up GEMMs through hipBLASLt
+(torch.nn.functional.linear), everything between the GEMMs as hand-written
+gfx950 kernels (resilient_llm_amd.ops): fused residual-add RMSNorm,
+RoPE+paged-KV append, paged GQA decode attention, varlen prefill
+attention, fused SwiGLU.  On CPU the same forward runs on the fp32
+reference ops — that is the no-GPU test path.
+
+The reference runs no model code at all (model execution lives inside
+AWS Bedrock — SURVEY.md §0); targets here come from BASELINE.json:
+Llama-3-8B bf16 as the flagship, Llama-3-70B TP=4 per pool.
+
+Weight loading: random init (no-network default) or a safetensors
+directory with HF Llama names.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import math
+import os
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from .. import ops
+
+
+@dataclasses.dataclass
+class LlamaConfig:
+    name: str
+    hidden_size: int
+    n_layers: int
+    n_heads: int
+    n_kv_heads: int
+    head_dim: int
+    intermediate_size: int
+    vocab_size: int
+    rope_theta: float = 500000.0
+    rms_eps: float = 1e-5
+    max_position: int = 8192
+    tie_embeddings: bool = False
+
+    @property
+    def q_size(self) -> int:
+        return self.n_heads * self.head_dim
+
+    @property
+    def kv_size(self) -> int:
+        return self.n_kv_heads * self.head_dim
+
+
+MODEL_PRESETS: dict[str, LlamaConfig] = {
+    "llama-3-8b": LlamaConfig(
+        name="llama-3-8b", hidden_size=4096, n_layers=32, n_heads=32,
+        n_kv_heads=8, head_dim=128, intermediate_size=14336,
+        vocab_size=128256),
+    "llama-3-70b": LlamaConfig(
+        name="llama-3-70b", hidden_size=8192, n_layers=80, n_heads=64,
+        n_kv_heads=8, head_dim=128, intermediate_size=28672,
+        vocab_size=128256),
+    # CPU-testable models
+    "tiny": LlamaConfig(
+        name="tiny", hidden_size=256, n_layers=2, n_heads=4, n_kv_heads=2,
+        head_dim=64, intermediate_size=512, vocab_size=512,
+        max_position=2048),
+    "tiny-128": LlamaConfig(
+        name="tiny-128", hidden_size=512, n_layers=2, n_heads=4, n_kv_heads=1,
+        head_dim=128, intermediate_size=1024, vocab_size=512,
+        max_position=2048),
+}
+
+
+def get_config(name: str) -> LlamaConfig:
+    try:
+        return MODEL_PRESETS[name]
+    except KeyError:
+        raise KeyError(f"unknown model {name!r}; have {sorted(MODEL_PRESETS)}")
+
+
+class LlamaForCausalLM:
+    """Inference-only Llama.  TP sharding plugs in via `shard` (see
+    resilient_llm_amd.parallel): rank/world split q/kv heads and the
+    intermediate dim column-wise, o/down row-wise, with an all-reduce
+    after o_proj and down_proj."""
+
+    def __init__(self, config: LlamaConfig, device: str = "cpu",
+                 dtype: torch.dtype = torch.bfloat16,
+                 tp_rank: int = 0, tp_world: int = 1,
+                 tp_group=None, seed: int = 0) -> None:
+        self.config = config
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self.tp_rank = tp_rank
+        self.tp_world = tp_world
+        self.tp_group = tp_group
+        c = config
+        assert c.n_heads % tp_world == 0 and c.n_kv_heads % tp_world == 0, \
+            f"TP={tp_world} must divide heads ({c.n_heads}/{c.n_kv_heads})"
+        assert c.intermediate_size % tp_world == 0
+        self.n_heads = c.n_heads // tp_world
+        self.n_kv_heads = c.n_kv_heads // tp_world
+        self.inter = c.intermediate_size // tp_world
+        self.q_size = self.n_heads * c.head_dim
+        self.kv_size = self.n_kv_heads * c.head_dim
+        self.scale = 1.0 / math.sqrt(c.head_dim)
+        self.params: dict[str, torch.Tensor] = {}
+        self._init_weights(seed)
+        self.cos_sin = ops.build_cos_sin(c.max_position, c.head_dim,
+                                         c.rope_theta, device=self.device)
+
+    # ------------------------------------------------------------ weights
+    def _mk(self, name: str, shape: tuple, gen: torch.Generator,
+            std: float = 0.02) -> None:
+        t = torch.empty(shape, dtype=self.dtype, device=self.device)
+        t.normal_(0.0, std, generator=gen)
+        self.params[name] = t
+
+    def _full(self, shape: tuple, gen: torch.Generator,
+              std: float = 0.02) -> torch.Tensor:
+        t = torch.empty(shape, dtype=self.dtype, device=self.device)
+        t.normal_(0.0, std, generator=gen)
+        return t
+
+    def _init_weights(self, seed: int) -> None:
+        """Every rank draws the FULL weight with the same seed and keeps
+        its shard — so TP=N computes the same function as TP=1 (the
+        gloo/CPU TP tests rely on this)."""
+        c = self.config
+        r, w = self.tp_rank, self.tp_world
+        gen = torch.Generator(device=self.device)
+        gen.manual_seed(seed + 1)
+        self._mk("embed", (c.vocab_size, c.hidden_size), gen)
+        std_o = 0.02 / math.sqrt(2 * c.n_layers)
+        for i in range(c.n_layers):
+            gen.manual_seed(seed * 1000003 + i * 1009)
+            qkv = self._full((c.q_size + 2 * c.kv_size, c.hidden_size), gen)
+            qf, kf, vf = torch.split(qkv, [c.q_size, c.kv_size, c.kv_size], dim=0)
+            self.params[f"l{i}.qkv"] = torch.cat([
+                qf[r * self.q_size:(r + 1) * self.q_size],
+                kf[r * self.kv_size:(r + 1) * self.kv_size],
+                vf[r * self.kv_size:(r + 1) * self.kv_size]], dim=0).contiguous()
+            o = self._full((c.hidden_size, c.q_size), gen, std_o)
+            self.params[f"l{i}.o"] = o[:, r * self.q_size:(r + 1) * self.q_size].contiguous()
+            gu = self._full((2 * c.intermediate_size, c.hidden_size), gen)
+            gf, uf = torch.split(gu, [c.intermediate_size, c.intermediate_size], dim=0)
+            self.params[f"l{i}.gate_up"] = torch.cat([
+                gf[r * self.inter:(r + 1) * self.inter],
+                uf[r * self.inter:(r + 1) * self.inter]], dim=0).contiguous()
+            down = self._full((c.hidden_size, c.intermediate_size), gen, std_o)
+            self.params[f"l{i}.down"] = down[:, r * self.inter:(r + 1) * self.inter].contiguous()
+            self.params[f"l{i}.ln1"] = torch.ones(c.hidden_size, dtype=self.dtype,
+                                                  device=self.device)
+            self.params[f"l{i}.ln2"] = torch.ones_like(self.params[f"l{i}.ln1"])
+        gen.manual_seed(seed + 2)
+        self.params["final_ln"] = torch.ones(c.hidden_size, dtype=self.dtype,
+                                             device=self.device)
+        if c.tie_embeddings:
+            self.params["lm_head"] = self.params["embed"]
+        else:
+            self._mk("lm_head", (c.vocab_size, c.hidden_size), gen)
+
+    def load_safetensors(self, path: str) -> int:
+        """Load HF-Llama-named safetensors shards (tp_world == 1 only for
+        now).  Returns the number of tensors consumed."""
+        import safetensors.torch as st
+        assert self.tp_world == 1, "safetensors load implemented for TP=1"
+        c = self.config
+        loaded = 0
+        files = sorted(f for f in os.listdir(path) if f.endswith(".safetensors"))
+        raw: dict[str, torch.Tensor] = {}
+        for f in files:
+            raw.update(st.load_file(os.path.join(path, f)))
+        def take(name):
+            nonlocal loaded
+            t = raw[name].to(self.dtype).to(self.device)
+            loaded += 1
+            return t
+        self.params["embed"] = take("model.embed_tokens.weight")
+        for i in range(c.n_layers):
+            p = f"model.layers.{i}."
+            q = take(p + "self_attn.q_proj.weight")
+            k = take(p + "self_attn.k_proj.weight")
+            v = take(p + "self_attn.v_proj.weight")
+            self.params[f"l{i}.qkv"] = torch.cat([q, k, v], dim=0).contiguous()
+            self.params[f"l{i}.o"] = take(p + "self_attn.o_proj.weight")
+            g = take(p + "mlp.gate_proj.weight")
+            u = take(p + "mlp.up_proj.weight")
+            self.params[f"l{i}.gate_up"] = torch.cat([g, u], dim=0).contiguous()
+            self.params[f"l{i}.down"] = take(p + "mlp.down_proj.weight")
+            self.params[f"l{i}.ln1"] = take(p + "input_layernorm.weight")
+            self.params[f"l{i}.ln2"] = take(p + "post_attention_layernorm.weight")
+        self.params["final_ln"] = take("model.norm.weight")
+        if "lm_head.weight" in raw:
+            self.params["lm_head"] = take("lm_head.weight")
+        else:
+            self.params["lm_head"] = self.params["embed"]
+        return loaded
+
+    def param_bytes(self) -> int:
+        seen = set()
+        total = 0
+        for t in self.params.values():
+            if id(t) not in seen:
+                seen.add(id(t))
+                total += t.numel() * t.element_size()
+        return total
+
+    # ------------------------------------------------------------ compute
+    def _all_reduce(self, x: torch.Tensor) -> torch.Tensor:
+        if self.tp_world > 1:
+            import torch.distributed as dist
+            dist.all_reduce(x, group=self.tp_group)
+        return x
+
+    def _layer(self, i: int, x: torch.Tensor, residual: Optional[torch.Tensor],
+               positions: torch.Tensor, kv_cache, slot_mapping: torch.Tensor,
+               attn_fn) -> tuple[torch.Tensor, torch.Tensor]:
+        p = self.params
+        c = self.config
+        if residual is None:
+            residual = x.clone()
+            h = ops.rmsnorm(x, p[f"l{i}.ln1"], c.rms_eps)
+        else:
+            h = ops.rmsnorm_residual_(x, residual, p[f"l{i}.ln1"], c.rms_eps)
+        qkv = F.linear(h, p[f"l{i}.qkv"])
+        T = qkv.shape[0]
+        q = qkv[:, :self.q_size].reshape(T, self.n_heads, c.head_dim).contiguous()
+        k = qkv[:, self.q_size:self.q_size + self.kv_size] \
+            .reshape(T, self.n_kv_heads, c.head_dim).contiguous()
+        v = qkv[:, self.q_size + self.kv_size:] \
+            .reshape(T, self.n_kv_heads, c.head_dim).contiguous()
+        k_cache, v_cache = kv_cache.layer(i)
+        ops.rope_kv_append_(q, k, v, positions, self.cos_sin,
+                            k_cache, v_cache, slot_mapping)
+        attn_out = attn_fn(i, q, k, v, k_cache, v_cache)
+        h = self._all_reduce(F.linear(attn_out.reshape(T, self.q_size),
+                                      p[f"l{i}.o"]))
+        h2 = ops.rmsnorm_residual_(h, residual, p[f"l{i}.ln2"], c.rms_eps)
+        gu = F.linear(h2, p[f"l{i}.gate_up"])
+        mlp = self._all_reduce(F.linear(ops.silu_mul(gu), p[f"l{i}.down"]))
+        return mlp, residual
+
+    def forward_prefill(self, input_ids: torch.Tensor, positions: torch.Tensor,
+                        kv_cache, slot_mapping: torch.Tensor,
+                        cu_seqlens: torch.Tensor) -> torch.Tensor:
+        """Returns logits for the LAST token of each sequence:
+        [n_seqs, vocab]."""
+        def attn(i, q, k, v, k_cache, v_cache):
+            return ops.prefill_attn(q, k, v, cu_seqlens, self.scale)
+        return self._forward(input_ids, positions, kv_cache, slot_mapping,
+                             attn, last_idx=cu_seqlens[1:].long() - 1)
+
+    def forward_decode(self, input_ids: torch.Tensor, positions: torch.Tensor,
+                       kv_cache, slot_mapping: torch.Tensor,
+                       block_tables: torch.Tensor,
+                       seq_lens: torch.Tensor) -> torch.Tensor:
+        """One token per sequence; returns [batch, vocab] logits."""
+        def attn(i, q, k, v, k_cache, v_cache):
+            return ops.decode_attn(q, k_cache, v_cache, block_tables,
+                                   seq_lens, self.scale)
+        return self._forward(input_ids, positions, kv_cache, slot_mapping,
+                             attn, last_idx=None)
+
+    def _forward(self, input_ids, positions, kv_cache, slot_mapping, attn_fn,
+                 last_idx) -> torch.Tensor:
+        p = self.params
+        x = F.embedding(input_ids.long(), p["embed"])
+        residual = None
+        for i in range(self.config.n_layers):
+            x, residual = self._layer(i, x, residual, positions, kv_cache,
+                                      slot_mapping, attn_fn)
+        h = ops.rmsnorm_residual_(x, residual, p["final_ln"],
+                                  self.config.rms_eps)
+        if last_idx is not None:
+            h = h[last_idx]
+        return F.linear(h, p["lm_head"])   # bf16 logits feed ops.sample
