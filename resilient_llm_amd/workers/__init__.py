@@ -3,3 +3,4 @@ from .base import (  # noqa: F401
     WorkerError, WorkerThrottled, WorkerDead, WorkerRegistry,
 )
 from .stub import StubWorker  # noqa: F401
+from .engine_worker import EngineWorker  # noqa: F401

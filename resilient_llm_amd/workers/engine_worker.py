@@ -1,0 +1,245 @@
+"""Engine-backed worker: the per-GPU serving unit.
+
+Owns one :class:`LLMEngine` (model + paged KV on one device) and exposes
+the async Worker interface the gateway dispatches onto.  The engine's
+blocking step loop runs on a dedicated thread; step outputs stream back
+to per-request asyncio queues (time-to-first-token measured here).
+
+Capacity pressure maps to the typed taxonomy (X13): CapacityExceeded ->
+WorkerThrottled.  Fault injection supports 'error' / 'kill' / 'hang'
+in-process; the subprocess wrapper (workers/gpu.py) escalates 'kill' to
+killing the actual GPU process.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import threading
+import time
+from typing import AsyncIterator, Optional
+
+import torch
+
+from ..engine import LLMEngine, PagedKVCache, SamplingParams
+from ..engine.engine import CapacityExceeded
+from ..models import LlamaForCausalLM, get_config
+from ..utils.tokenizer import ByteTokenizer
+from .base import (
+    GenerationChunk, GenerationRequest, GenerationResult, Worker,
+    WorkerDead, WorkerError, WorkerThrottled,
+)
+
+DEFAULT_KV_GB = 24.0
+
+
+def default_num_blocks(config, kv_gb: float = DEFAULT_KV_GB,
+                       block_size: int = 16, tp_world: int = 1) -> int:
+    per_block = (config.n_layers * 2 * (config.n_kv_heads // tp_world)
+                 * block_size * config.head_dim * 2)
+    # cap: tiny test models would otherwise turn a GB budget into
+    # millions of blocks (the allocator free list + zeros get silly)
+    return max(64, min(65536, int(kv_gb * (1 << 30) / per_block)))
+
+
+def render_prompt(messages: list) -> str:
+    parts = []
+    for m in messages:
+        parts.append(f"{m.get('role', 'user')}: {m.get('content', '')}")
+    parts.append("assistant:")
+    return "\n".join(parts)
+
+
+class EngineWorker(Worker):
+    def __init__(self, device: str, model_name: str = "llama-3-8b",
+                 device_label: Optional[str] = None,
+                 kv_gb: float = DEFAULT_KV_GB,
+                 max_batch_size: int = 64,
+                 max_queue: int = 256,
+                 dtype: Optional[torch.dtype] = None,
+                 num_blocks: Optional[int] = None,
+                 use_graphs: bool = False,
+                 tp_rank: int = 0, tp_world: int = 1, tp_group=None,
+                 seed: int = 0) -> None:
+        super().__init__(device=device_label or f"gpu:{device}",
+                         models={model_name})
+        self.model_name = model_name
+        torch_device = device if ":" in str(device) or device == "cpu" \
+            else f"cuda:{device}"
+        config = get_config(model_name)
+        if dtype is None:
+            dtype = torch.bfloat16 if torch_device != "cpu" else torch.float32
+        self.config = config
+        self.model = LlamaForCausalLM(config, device=torch_device, dtype=dtype,
+                                      tp_rank=tp_rank, tp_world=tp_world,
+                                      tp_group=tp_group, seed=seed)
+        nb = num_blocks or default_num_blocks(config, kv_gb, tp_world=tp_world)
+        kv = PagedKVCache.for_model(config, nb, device=torch_device,
+                                    tp_world=tp_world)
+        if dtype != torch.bfloat16:
+            kv.k = kv.k.to(dtype)
+            kv.v = kv.v.to(dtype)
+        self.engine = LLMEngine(self.model, kv, max_batch_size=max_batch_size,
+                                max_queue=max_queue, seed=seed)
+        if use_graphs and torch_device != "cpu":
+            from ..engine.graph import install_graph_runner
+            install_graph_runner(self.engine)
+        self.tokenizer = ByteTokenizer(config.vocab_size)
+        self.fault_mode = "none"
+        self.total_served = 0
+        self._req_counter = 0
+        self._in_flight = 0
+        self._lock = threading.Lock()
+        # req_id -> (asyncio queue, loop)
+        self._sinks: dict[str, tuple[asyncio.Queue, asyncio.AbstractEventLoop]] = {}
+        self._work_event = threading.Event()
+        self._stop = False
+        self._thread = threading.Thread(target=self._engine_loop, daemon=True,
+                                        name=f"engine-{self.device}")
+        self._thread.start()
+
+    # --------------------------------------------------------- engine loop
+    def _engine_loop(self) -> None:
+        while not self._stop:
+            if self.fault_mode == "hang":
+                time.sleep(0.05)
+                continue
+            with self._lock:
+                has_work = self.engine.has_work()
+            if not has_work:
+                self._work_event.wait(timeout=0.01)
+                self._work_event.clear()
+                continue
+            try:
+                with self._lock:
+                    outputs = self.engine.step()
+            except Exception as e:  # engine-level failure -> fail all in flight
+                self._broadcast_error(e)
+                continue
+            for out in outputs:
+                sink = self._sinks.get(out.req_id)
+                if sink is None:
+                    continue
+                q, loop = sink
+                loop.call_soon_threadsafe(q.put_nowait, out)
+
+    def _broadcast_error(self, e: Exception) -> None:
+        for q, loop in list(self._sinks.values()):
+            loop.call_soon_threadsafe(q.put_nowait, e)
+
+    # ----------------------------------------------------------- helpers
+    def _check_fault(self) -> None:
+        if self.fault_mode == "kill":
+            raise WorkerDead(f"{self.device} injected fault: killed")
+        if self.fault_mode == "error":
+            raise WorkerError(f"{self.device} injected fault: error")
+
+    def _enqueue(self, req: GenerationRequest) -> tuple[str, asyncio.Queue, int]:
+        self._check_fault()
+        prompt_ids = self.tokenizer.encode(render_prompt(req.messages))
+        params = SamplingParams(max_tokens=req.max_tokens,
+                                temperature=req.temperature,
+                                seed=req.seed)
+        self._req_counter += 1
+        rid = f"{req.request_id}-{self._req_counter}"
+        q: asyncio.Queue = asyncio.Queue()
+        loop = asyncio.get_running_loop()
+        self._sinks[rid] = (q, loop)
+        try:
+            with self._lock:
+                self.engine.add_request(rid, prompt_ids, params)
+        except CapacityExceeded as e:
+            del self._sinks[rid]
+            raise WorkerThrottled(str(e)) from e
+        self._work_event.set()
+        return rid, q, len(prompt_ids)
+
+    def _cleanup(self, rid: str) -> None:
+        self._sinks.pop(rid, None)
+
+    @property
+    def in_flight(self) -> int:
+        return self._in_flight
+
+    # --------------------------------------------------------------- API
+    async def generate(self, req: GenerationRequest) -> GenerationResult:
+        t0 = time.monotonic()
+        rid, q, n_prompt = self._enqueue(req)
+        self._in_flight += 1
+        try:
+            token_ids: list[int] = []
+            ttft = None
+            finish = "stop"
+            while True:
+                out = await q.get()
+                self._check_fault()
+                if isinstance(out, Exception):
+                    raise WorkerError(f"engine error: {out}") from out
+                if ttft is None:
+                    ttft = (time.monotonic() - t0) * 1000.0
+                token_ids.append(out.token_id)
+                if out.finished:
+                    finish = out.finish_reason or "stop"
+                    break
+            self.total_served += 1
+            return GenerationResult(
+                text=self.tokenizer.decode(token_ids),
+                prompt_tokens=n_prompt,
+                completion_tokens=len(token_ids),
+                finish_reason=finish,
+                ttft_ms=ttft)
+        finally:
+            self._in_flight -= 1
+            self._cleanup(rid)
+            with self._lock:
+                self.engine.abort(rid)
+
+    async def _stream_impl(self, req: GenerationRequest) -> AsyncIterator[GenerationChunk]:
+        rid, q, n_prompt = self._enqueue(req)
+        self._in_flight += 1
+        try:
+            emitted = ""
+            token_ids: list[int] = []
+            while True:
+                out = await q.get()
+                self._check_fault()
+                if isinstance(out, Exception):
+                    raise WorkerError(f"engine error: {out}") from out
+                token_ids.append(out.token_id)
+                full = self.tokenizer.decode(token_ids)
+                delta = full[len(emitted):]
+                emitted = full
+                yield GenerationChunk(
+                    text=delta, token_id=out.token_id,
+                    finish_reason=out.finish_reason if out.finished else None)
+                if out.finished:
+                    break
+            self.total_served += 1
+        finally:
+            self._in_flight -= 1
+            self._cleanup(rid)
+            with self._lock:
+                self.engine.abort(rid)
+
+    def generate_stream(self, req: GenerationRequest) -> AsyncIterator[GenerationChunk]:
+        return self._stream_impl(req)
+
+    async def health(self) -> dict:
+        if self.fault_mode in ("kill", "hang"):
+            raise WorkerDead(f"{self.device} unhealthy (fault={self.fault_mode})")
+        return {"device": self.device, "status": "ok",
+                "models": sorted(self.models),
+                "model": self.model_name,
+                "in_flight": self._in_flight,
+                "queued": len(self.engine.waiting),
+                "running": len(self.engine.running),
+                "kv_free_blocks": self.engine.kv.free_blocks,
+                "total_served": self.total_served}
+
+    async def inject_fault(self, mode: str) -> None:
+        assert mode in ("none", "kill", "hang", "error")
+        self.fault_mode = mode
+
+    async def close(self) -> None:
+        self._stop = True
+        self._work_event.set()
+        self._thread.join(timeout=5)
