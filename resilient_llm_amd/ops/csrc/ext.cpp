@@ -5,8 +5,8 @@
 
 #include <torch/library.h>
 #include <ATen/ATen.h>
-#include <c10/hip/HIPStream.h>
-#include <c10/hip/HIPGuard.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+#include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
 
 #include "kernels.h"
 
@@ -15,7 +15,8 @@ namespace {
 using at::Tensor;
 
 hipStream_t current_stream(const Tensor& t) {
-  return c10::hip::getCurrentHIPStream(t.device().index()).stream();
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA(
+      t.device().index()).stream();
 }
 
 uint16_t* bf16_ptr(const Tensor& t) {
@@ -37,7 +38,7 @@ Tensor rmsnorm(const Tensor& x, const Tensor& w, double eps) {
   TORCH_CHECK(w.numel() == dim, "weight/dim mismatch");
   const int rows = int(x.numel() / dim);
   Tensor y = at::empty_like(x);
-  c10::hip::HIPGuard guard(x.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   rlli::launch_rmsnorm(bf16_ptr(x), nullptr, bf16_ptr(w), bf16_ptr(y), rows,
                        dim, float(eps), current_stream(x));
   return y;
@@ -53,7 +54,7 @@ Tensor rmsnorm_residual_(const Tensor& x, Tensor residual, const Tensor& w,
   TORCH_CHECK(dim % 8 == 0 && w.numel() == dim, "bad dim/weight");
   const int rows = int(x.numel() / dim);
   Tensor y = at::empty_like(x);
-  c10::hip::HIPGuard guard(x.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   rlli::launch_rmsnorm(bf16_ptr(x), bf16_ptr(residual), bf16_ptr(w),
                        bf16_ptr(y), rows, dim, float(eps), current_stream(x));
   return y;
@@ -69,7 +70,7 @@ Tensor silu_mul(const Tensor& gate_up) {
   auto sizes = gate_up.sizes().vec();
   sizes.back() = inter;
   Tensor y = at::empty(sizes, gate_up.options());
-  c10::hip::HIPGuard guard(gate_up.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(gate_up.device());
   rlli::launch_silu_mul(bf16_ptr(gate_up), bf16_ptr(y), rows, inter,
                         current_stream(gate_up));
   return y;
@@ -98,7 +99,7 @@ void rope_kv_append_(Tensor q, Tensor k, const Tensor& v,
   TORCH_CHECK(k_cache.size(1) == n_kv && k_cache.size(3) == D);
   const int block_size = int(k_cache.size(2));
   TORCH_CHECK(positions.numel() == tokens && slot_mapping.numel() == tokens);
-  c10::hip::HIPGuard guard(q.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(q.device());
   rlli::launch_rope_kv_append(
       bf16_ptr(q), bf16_ptr(k), bf16_ptr(v),
       positions.data_ptr<int32_t>(), cos_sin.data_ptr<float>(),
@@ -130,7 +131,7 @@ Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
   TORCH_CHECK(k_cache.size(3) == D && v_cache.sizes() == k_cache.sizes());
   TORCH_CHECK(block_table.size(0) == batch && seq_lens.numel() == batch);
   Tensor out = at::empty_like(q);
-  c10::hip::HIPGuard guard(q.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(q.device());
   rlli::launch_decode_attn(
       bf16_ptr(q), bf16_ptr(k_cache), bf16_ptr(v_cache),
       block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
@@ -157,7 +158,7 @@ Tensor prefill_attn(const Tensor& q, const Tensor& k, const Tensor& v,
   TORCH_CHECK(k.size(0) == T && v.size(0) == T && k.size(2) == D);
   const int n_seqs = int(cu_seqlens.numel()) - 1;
   Tensor out = at::empty_like(q);
-  c10::hip::HIPGuard guard(q.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(q.device());
   rlli::launch_prefill_attn(
       bf16_ptr(q), bf16_ptr(k), bf16_ptr(v), cu_seqlens.data_ptr<int32_t>(),
       bf16_ptr(out), n_seqs, T, n_q, n_kv, D, float(scale),
@@ -174,7 +175,7 @@ Tensor sample(const Tensor& logits, const Tensor& temperatures, int64_t seed) {
   const int vocab = int(logits.size(1));
   TORCH_CHECK(temperatures.numel() == batch);
   Tensor out = at::empty({batch}, logits.options().dtype(at::kInt));
-  c10::hip::HIPGuard guard(logits.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(logits.device());
   rlli::launch_sample(bf16_ptr(logits), temperatures.data_ptr<float>(),
                       uint64_t(seed), out.data_ptr<int32_t>(), batch, vocab,
                       current_stream(logits));
