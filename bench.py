@@ -1,0 +1,314 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark — the driver contract.
+
+Measures BASELINE.json's metric: sustained reqs/sec + success-rate + p50
+latency @ 64 concurrent, Llama-3-8B bf16, synthetic prompts, random-init
+weights, on N GPUs of one node (weak scaling: 64 concurrent clients per
+GPU).
+
+Full-stack path: every rank hosts one EngineWorker on its GPU behind the
+unix-socket RPC; rank 0 additionally runs the OpenAI-compatible gateway
+(router + ledger) and the closed-loop load generator over HTTP.  One
+"step" = every concurrency slot completes exactly one request
+(prompt 128 tokens -> 64 new tokens, greedy).  W untimed warmup steps,
+then K timed steps bracketed by gloo barrier + torch.cuda.synchronize on
+both sides; elapsed is the MAX over ranks; rank 0 prints ONE JSON line.
+
+Launch (driver):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import random
+import socket
+import string
+import sys
+import tempfile
+import threading
+import time
+
+REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
+if REPO_ROOT not in sys.path:
+    sys.path.insert(0, REPO_ROOT)
+
+PROMPT_TOKENS = 128
+OUTPUT_TOKENS = 64
+CONCURRENCY_PER_GPU = 64
+
+
+def log(msg):
+    print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def make_prompt(rng: random.Random, n_tokens: int) -> str:
+    # ByteTokenizer: 1 token per byte (+bos +"user: ...\nassistant:" framing)
+    n = max(8, n_tokens - 18)
+    return "".join(rng.choice(string.ascii_lowercase + " ") for _ in range(n))
+
+
+class LoopThread:
+    """An asyncio loop on a daemon thread; run() schedules coroutines."""
+
+    def __init__(self) -> None:
+        self.loop = asyncio.new_event_loop()
+        self._thread = threading.Thread(
+            target=self.loop.run_forever, daemon=True)
+        self._thread.start()
+
+    def run(self, coro, timeout=None):
+        return asyncio.run_coroutine_threadsafe(coro, self.loop).result(timeout)
+
+
+def start_worker(loopth: LoopThread, device: str, label: str, sock: str,
+                 model: str, max_batch: int, num_blocks: int,
+                 use_graphs: bool):
+    from resilient_llm_amd.workers.engine_worker import EngineWorker
+    from resilient_llm_amd.workers.rpc import WorkerRpcServer
+
+    async def _start():
+        worker = EngineWorker(device=device, model_name=model,
+                              device_label=label,
+                              max_batch_size=max_batch,
+                              max_queue=max_batch * 4,
+                              num_blocks=num_blocks,
+                              use_graphs=use_graphs)
+        server = WorkerRpcServer(worker, sock)
+        await server.start()
+        return worker
+
+    return loopth.run(_start(), timeout=600)
+
+
+def start_gateway(loopth: LoopThread, n_gpus: int, model: str, port: int,
+                  run_dir: str):
+    from resilient_llm_amd.config import load_config
+    from resilient_llm_amd.gateway.app import GatewayApp
+    from resilient_llm_amd.gateway.http import HttpServer
+    from resilient_llm_amd.workers.base import WorkerRegistry
+    from resilient_llm_amd.workers.rpc import RpcWorkerClient
+
+    data = {
+        "cluster": {"port": port},
+        "model_list": [
+            {"model_name": "bench-model",
+             "litellm_params": {"model": f"gpu/*/{model}"},
+             "model_info": {"id": "bench-spread"}},
+        ],
+        "router_settings": {"routing_strategy": "simple-shuffle",
+                            "enable_pre_call_checks": False},
+    }
+    config = load_config(data=data)
+
+    async def _start():
+        registry = WorkerRegistry()
+        for r in range(n_gpus):
+            client = RpcWorkerClient(f"gpu:{r}", {model},
+                                     os.path.join(run_dir, f"w{r}.sock"))
+            await client.connect(timeout=900)
+            registry.register("gpu", str(r), client)
+        app = GatewayApp(config, registry, health_interval_s=5.0)
+        server = HttpServer(app.handle, host="127.0.0.1", port=port)
+        await server.start()
+        await app.start_background()
+        return app
+
+    return loopth.run(_start(), timeout=1200)
+
+
+async def one_request(session, url: str, prompt: str, results: list):
+    t0 = time.monotonic()
+    body = {"model": "bench-model",
+            "messages": [{"role": "user", "content": prompt}],
+            "max_tokens": OUTPUT_TOKENS, "temperature": 0.0}
+    try:
+        async with session.post(url, json=body) as resp:
+            data = await resp.json()
+            ok = resp.status == 200
+            usage = data.get("usage", {}) if ok else {}
+            results.append({
+                "ok": ok,
+                "latency": time.monotonic() - t0,
+                "completion_tokens": usage.get("completion_tokens", 0),
+                "prompt_tokens": usage.get("prompt_tokens", 0),
+            })
+    except Exception as e:
+        results.append({"ok": False, "latency": time.monotonic() - t0,
+                        "error": str(e), "completion_tokens": 0,
+                        "prompt_tokens": 0})
+
+
+def run_load_round(loopth: LoopThread, port: int, prompts: list[str]) -> list:
+    import aiohttp
+
+    async def _round():
+        results: list = []
+        url = f"http://127.0.0.1:{port}/chat/completions"
+        conn = aiohttp.TCPConnector(limit=0)
+        timeout = aiohttp.ClientTimeout(total=600)
+        async with aiohttp.ClientSession(connector=conn, timeout=timeout) as s:
+            await asyncio.gather(*[one_request(s, url, p, results)
+                                   for p in prompts])
+        return results
+
+    return loopth.run(_round(), timeout=900)
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=4)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--model", default="llama-3-8b")
+    ap.add_argument("--concurrency", type=int, default=CONCURRENCY_PER_GPU,
+                    help="concurrent clients per GPU (weak scaling)")
+    ap.add_argument("--prompt-tokens", type=int, default=PROMPT_TOKENS)
+    ap.add_argument("--output-tokens", type=int, default=OUTPUT_TOKENS)
+    ap.add_argument("--no-graphs", action="store_true")
+    ap.add_argument("--device", default=None,
+                    help="override torch device (tests: cpu)")
+    args = ap.parse_args()
+    globals()["OUTPUT_TOKENS"] = args.output_tokens
+
+    import torch
+    import torch.distributed as dist
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = args.gpus
+    assert world == n_gpus or world == 1, \
+        f"WORLD_SIZE {world} must equal --gpus {n_gpus}"
+    distributed = world > 1
+    if distributed:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group(backend="gloo", rank=rank, world_size=world)
+
+    on_gpu = args.device is None and torch.cuda.is_available()
+    device = args.device or (f"cuda:{local_rank}" if on_gpu else "cpu")
+    if on_gpu:
+        torch.cuda.set_device(local_rank)
+
+    def sync():
+        if on_gpu:
+            torch.cuda.synchronize()
+        if distributed:
+            dist.barrier()
+
+    # shared run dir for worker sockets (rank 0 creates, broadcasts)
+    if distributed:
+        holder = [tempfile.mkdtemp(prefix="rlli-bench-")] if rank == 0 else [None]
+        dist.broadcast_object_list(holder, src=0)
+        run_dir = holder[0]
+    else:
+        run_dir = tempfile.mkdtemp(prefix="rlli-bench-")
+
+    conc_total = args.concurrency * n_gpus
+    max_seq_blocks = -(-(args.prompt_tokens + args.output_tokens + 8) // 16)
+    num_blocks = args.concurrency * (max_seq_blocks + 1) + 64
+
+    loopth = LoopThread()
+    t_init = time.time()
+    worker = start_worker(loopth, device, f"gpu:{rank}",
+                          os.path.join(run_dir, f"w{rank}.sock"),
+                          args.model, max_batch=args.concurrency,
+                          num_blocks=num_blocks,
+                          use_graphs=on_gpu and not args.no_graphs)
+    log(f"rank {rank}: worker up on {device} in {time.time() - t_init:.1f}s "
+        f"({num_blocks} KV blocks)")
+    sync()
+
+    port = None
+    results_timed: list = []
+    if rank == 0:
+        port = free_port()
+        start_gateway(loopth, n_gpus if distributed else 1, args.model, port,
+                      run_dir)
+        log(f"gateway on :{port}, driving {conc_total} concurrent clients")
+
+    rng = random.Random(1234)
+
+    def round_prompts():
+        return [make_prompt(rng, args.prompt_tokens) for _ in range(conc_total)]
+
+    # ---- warmup ----
+    for w in range(args.warmup):
+        if rank == 0:
+            res = run_load_round(loopth, port, round_prompts())
+            ok = sum(1 for r in res if r["ok"])
+            log(f"warmup {w}: {ok}/{len(res)} ok")
+        sync()
+
+    # ---- timed ----
+    sync()
+    t0 = time.monotonic()
+    for k in range(args.steps):
+        if rank == 0:
+            results_timed.extend(run_load_round(loopth, port, round_prompts()))
+    sync()
+    elapsed = time.monotonic() - t0
+    if distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        n_req = len(results_timed)
+        ok = [r for r in results_timed if r["ok"]]
+        lats = sorted(r["latency"] for r in ok)
+        p50 = lats[len(lats) // 2] if lats else None
+        p99 = lats[min(len(lats) - 1, int(0.99 * len(lats)))] if lats else None
+        out_toks = sum(r["completion_tokens"] for r in ok)
+        total_toks = out_toks + sum(r["prompt_tokens"] for r in ok)
+        reqs_per_s = len(ok) / elapsed
+        result = {
+            "metric": "sustained reqs/sec + success-rate + p50 latency "
+                      "@ 64 concurrent, Llama-3-8B",
+            "value": round(reqs_per_s, 3),
+            "unit": "reqs/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 1),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": conc_total,
+                "seq_len": args.prompt_tokens + args.output_tokens,
+                "parallelism": f"dp{n_gpus}",
+                "concurrency_per_gpu": args.concurrency,
+                "prompt_tokens": args.prompt_tokens,
+                "output_tokens": args.output_tokens,
+                "success_rate": round(len(ok) / n_req, 4) if n_req else None,
+                "p50_latency_s": round(p50, 3) if p50 else None,
+                "p99_latency_s": round(p99, 3) if p99 else None,
+                "output_tokens_per_s": round(out_toks / elapsed, 1),
+                "total_tokens_per_s": round(total_toks / elapsed, 1),
+                "temperature": 0.0,
+            },
+        }
+        print(json.dumps(result), flush=True)
+    if distributed:
+        dist.barrier()
+        dist.destroy_process_group()
+    os._exit(0)   # daemon loops/threads: exit hard once results are printed
+
+
+if __name__ == "__main__":
+    main()

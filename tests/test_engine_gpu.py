@@ -87,3 +87,22 @@ def test_llama8b_short_generation():
     e.add_request("a", list(range(5, 133)), SamplingParams(max_tokens=4))
     outs = drain(e)
     assert len(outs["a"]) == 4
+
+
+def test_graph_runner_matches_eager():
+    """hipGraph-captured decode produces the same greedy tokens as the
+    eager path."""
+    from resilient_llm_amd.engine.graph import install_graph_runner
+    prompts = {f"g{i}": list(range(4 + i, 45 + 2 * i)) for i in range(5)}
+
+    eager = make_engine("tiny-128", num_blocks=256, max_batch_size=8, seed=5)
+    for rid, p in prompts.items():
+        eager.add_request(rid, p, SamplingParams(max_tokens=12))
+    eager_out = drain(eager)
+
+    graphed = make_engine("tiny-128", num_blocks=256, max_batch_size=8, seed=5)
+    install_graph_runner(graphed)
+    for rid, p in prompts.items():
+        graphed.add_request(rid, p, SamplingParams(max_tokens=12))
+    graph_out = drain(graphed)
+    assert eager_out == graph_out
