@@ -79,6 +79,10 @@ class LLMEngine:
         self.step_count = 0
         self.block_size = kv_cache.block_size
         self._aborted: set[str] = set()
+        self._batch_dirty = True
+        self.stats = {"prefill_steps": 0, "decode_steps": 0,
+                      "prefill_time": 0.0, "decode_time": 0.0,
+                      "prefill_tokens": 0, "decode_tokens": 0}
         # decode graph runner installed by engine/graph.py (GPU only)
         self.graph_runner = None
 
@@ -154,9 +158,19 @@ class LLMEngine:
         self._drop_aborted()
         admitted = self._admit()
         if admitted:
-            return self._prefill_step(admitted)
+            t0 = time.monotonic()
+            out = self._prefill_step(admitted)
+            self.stats["prefill_steps"] += 1
+            self.stats["prefill_time"] += time.monotonic() - t0
+            self.stats["prefill_tokens"] += sum(len(s.prompt_ids) for s in admitted)
+            return out
         if self.running:
-            return self._decode_step()
+            t0 = time.monotonic()
+            out = self._decode_step()
+            self.stats["decode_steps"] += 1
+            self.stats["decode_time"] += time.monotonic() - t0
+            self.stats["decode_tokens"] += len(out)
+            return out
         return []
 
     def _drop_aborted(self) -> None:
@@ -169,6 +183,8 @@ class LLMEngine:
                 self._aborted.discard(seq.req_id)
             else:
                 keep.append(seq)
+        if len(keep) != len(self.running):
+            self._batch_dirty = True
         self.running = keep
 
     def _sample(self, logits: torch.Tensor, seqs: list[SeqState]) -> list[int]:
@@ -200,33 +216,46 @@ class LLMEngine:
             seq.output_ids.append(tok)
             if not self._finish(seq, outs, tok):
                 self.running.append(seq)
+                self._batch_dirty = True
         return outs
 
-    def _decode_step(self) -> list[StepOutput]:
+    def _rebuild_batch(self) -> None:
+        """(Re)build persistent device tensors for the running batch.
+        Block tables are complete at admission (full reservation), so
+        between composition changes every decode step is device-side."""
         dev = self.device
         seqs = self.running
-        input_ids = [s.output_ids[-1] for s in seqs]
-        positions = [s.n_cached for s in seqs]
-        slots = [self._slot(s, s.n_cached) for s in seqs]
-        seq_lens = [s.n_cached + 1 for s in seqs]
         max_blocks = max(len(s.blocks) for s in seqs)
-        bt = torch.full((len(seqs), max_blocks), 0, dtype=torch.int32)
+        bt = torch.zeros((len(seqs), max_blocks), dtype=torch.int32)
         for i, s in enumerate(seqs):
             bt[i, :len(s.blocks)] = torch.tensor(s.blocks, dtype=torch.int32)
-        args = (
-            torch.tensor(input_ids, dtype=torch.int32, device=dev),
-            torch.tensor(positions, dtype=torch.int32, device=dev),
-            torch.tensor(slots, dtype=torch.int32, device=dev),
-            bt.to(dev),
-            torch.tensor(seq_lens, dtype=torch.int32, device=dev),
-        )
+        self._b_ids = torch.tensor([s.output_ids[-1] for s in seqs],
+                                   dtype=torch.int32, device=dev)
+        self._b_pos = torch.tensor([s.n_cached for s in seqs],
+                                   dtype=torch.int32, device=dev)
+        self._b_bt = bt.to(dev)
+        self._b_temps = torch.tensor([s.params.temperature for s in seqs],
+                                     dtype=torch.float32, device=dev)
+        self._batch_dirty = False
+
+    def _decode_step(self) -> list[StepOutput]:
+        seqs = self.running
+        if getattr(self, "_batch_dirty", True) or self._b_ids.shape[0] != len(seqs):
+            self._rebuild_batch()
+        pos = self._b_pos
+        bt = self._b_bt
+        block_idx = (pos // self.block_size).long()
+        slots = bt.gather(1, block_idx.unsqueeze(1)).squeeze(1) \
+            * self.block_size + (pos - block_idx.int() * self.block_size)
+        seq_lens = pos + 1
         if self.graph_runner is not None:
-            logits = self.graph_runner.run(*args)
+            logits = self.graph_runner.run(self._b_ids, pos, slots, bt, seq_lens)
         else:
-            ids, pos, slot_t, bt_d, lens = args
-            logits = self.model.forward_decode(ids, pos, self.kv, slot_t,
-                                               bt_d, lens)
-        tokens = self._sample(logits, seqs)
+            logits = self.model.forward_decode(self._b_ids, pos, self.kv,
+                                               slots, bt, seq_lens)
+        seed = (self.seed * 0x9e3779b9 + self.step_count) & 0x7fffffffffffffff
+        tok_dev = ops.sample(logits, self._b_temps, seed)
+        tokens = tok_dev.tolist()
         outs: list[StepOutput] = []
         still_running: list[SeqState] = []
         for seq, tok in zip(seqs, tokens):
@@ -234,5 +263,11 @@ class LLMEngine:
             seq.output_ids.append(tok)
             if not self._finish(seq, outs, tok):
                 still_running.append(seq)
+        if len(still_running) == len(seqs):
+            # same batch: advance device state in place
+            self._b_ids.copy_(tok_dev)
+            self._b_pos += 1
+        else:
+            self._batch_dirty = True
         self.running = still_running
         return outs

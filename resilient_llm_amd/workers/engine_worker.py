@@ -233,7 +233,8 @@ class EngineWorker(Worker):
                 "queued": len(self.engine.waiting),
                 "running": len(self.engine.running),
                 "kv_free_blocks": self.engine.kv.free_blocks,
-                "total_served": self.total_served}
+                "total_served": self.total_served,
+                "engine_stats": dict(self.engine.stats)}
 
     async def inject_fault(self, mode: str) -> None:
         assert mode in ("none", "kill", "hang", "error")
