@@ -259,6 +259,11 @@ def main() -> None:
             results_timed.extend(run_load_round(loopth, port, round_prompts()))
     sync()
     elapsed = time.monotonic() - t0
+    try:
+        h = loopth.run(worker.health(), timeout=30)
+        log(f"rank {rank} engine stats: {h.get('engine_stats')}")
+    except Exception:
+        pass
     if distributed:
         t = torch.tensor([elapsed], dtype=torch.float64)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
