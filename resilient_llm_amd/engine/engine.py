@@ -80,6 +80,8 @@ class LLMEngine:
         self.block_size = kv_cache.block_size
         self._aborted: set[str] = set()
         self._batch_dirty = True
+        self._pending = None          # (seqs, event, host_tokens) 1-deep
+        self._pinned = None
         self.stats = {"prefill_steps": 0, "decode_steps": 0,
                       "prefill_time": 0.0, "decode_time": 0.0,
                       "prefill_tokens": 0, "decode_tokens": 0}
@@ -92,7 +94,7 @@ class LLMEngine:
         return len(self.waiting) + len(self.running)
 
     def has_work(self) -> bool:
-        return bool(self.waiting or self.running)
+        return bool(self.waiting or self.running or self._pending)
 
     def add_request(self, req_id: str, prompt_ids: list,
                     params: Optional[SamplingParams] = None) -> None:
@@ -152,26 +154,56 @@ class LLMEngine:
 
     # -------------------------------------------------------------- steps
     def step(self) -> list[StepOutput]:
-        """One engine iteration: a prefill batch if arrivals are waiting
-        and fit, else one decode for the running batch."""
+        """One engine iteration: drain the async sampling pipeline, then a
+        prefill batch if arrivals are waiting and fit, else one decode for
+        the running batch.
+
+        On GPU the decode loop is pipelined one step deep: the forward
+        consumes the previous step's sampled tokens directly on device,
+        while their host copy (needed only for EOS/finish bookkeeping)
+        arrives asynchronously and is processed at the NEXT step() — so
+        steady-state decode has no host<->device synchronization at all."""
         self.step_count += 1
+        outs = self._process_pending()
         self._drop_aborted()
         admitted = self._admit()
         if admitted:
             t0 = time.monotonic()
-            out = self._prefill_step(admitted)
+            outs += self._prefill_step(admitted)
             self.stats["prefill_steps"] += 1
             self.stats["prefill_time"] += time.monotonic() - t0
             self.stats["prefill_tokens"] += sum(len(s.prompt_ids) for s in admitted)
-            return out
+            return outs
         if self.running:
             t0 = time.monotonic()
-            out = self._decode_step()
+            outs += self._decode_step()
             self.stats["decode_steps"] += 1
             self.stats["decode_time"] += time.monotonic() - t0
-            self.stats["decode_tokens"] += len(out)
-            return out
-        return []
+            return outs
+        return outs
+
+    def _process_pending(self) -> list[StepOutput]:
+        if self._pending is None:
+            return []
+        seqs, event, buf = self._pending
+        self._pending = None
+        if event is not None:
+            event.synchronize()
+        tokens = buf.tolist() if isinstance(buf, torch.Tensor) else buf
+        outs: list[StepOutput] = []
+        still: list[SeqState] = []
+        running_ids = {id(s) for s in self.running}
+        for seq, tok in zip(seqs, tokens):
+            if id(seq) not in running_ids:
+                continue               # aborted between forward and landing
+            seq.output_ids.append(tok)
+            if not self._finish(seq, outs, tok):
+                still.append(seq)
+        if len(still) != len(seqs):
+            self._batch_dirty = True
+        self.running = still
+        self.stats["decode_tokens"] += len(outs)
+        return outs
 
     def _drop_aborted(self) -> None:
         if not self._aborted:
@@ -180,6 +212,7 @@ class LLMEngine:
         for seq in self.running:
             if seq.req_id in self._aborted:
                 self.kv.free(seq.blocks)
+                seq.blocks = []
                 self._aborted.discard(seq.req_id)
             else:
                 keep.append(seq)
@@ -255,19 +288,23 @@ class LLMEngine:
                                                slots, bt, seq_lens)
         seed = (self.seed * 0x9e3779b9 + self.step_count) & 0x7fffffffffffffff
         tok_dev = ops.sample(logits, self._b_temps, seed)
-        tokens = tok_dev.tolist()
-        outs: list[StepOutput] = []
-        still_running: list[SeqState] = []
-        for seq, tok in zip(seqs, tokens):
-            seq.n_cached += 1
-            seq.output_ids.append(tok)
-            if not self._finish(seq, outs, tok):
-                still_running.append(seq)
-        if len(still_running) == len(seqs):
-            # same batch: advance device state in place
-            self._b_ids.copy_(tok_dev)
-            self._b_pos += 1
-        else:
-            self._batch_dirty = True
-        self.running = still_running
-        return outs
+        for seq in seqs:
+            seq.n_cached += 1          # KV of the fed token was appended
+        # advance device state in place: the NEXT forward feeds tok_dev
+        # without the host ever seeing it (a rebuild overrides if the
+        # batch composition changes when the pending tokens land)
+        self._b_ids.copy_(tok_dev)
+        self._b_pos += 1
+        B = len(seqs)
+        if self.device.type == "cuda":
+            if self._pinned is None or self._pinned.shape[0] < self.max_batch_size:
+                self._pinned = torch.empty(self.max_batch_size,
+                                           dtype=torch.int32, pin_memory=True)
+            host = self._pinned[:B]
+            host.copy_(tok_dev, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record()
+            self._pending = (list(seqs), ev, host)
+            return []
+        self._pending = (list(seqs), None, tok_dev.tolist())
+        return self._process_pending()

@@ -227,18 +227,11 @@ class LlamaForCausalLM:
         else:
             h = ops.rmsnorm_residual_(x, residual, p[f"l{i}.ln1"], c.rms_eps)
         qkv = F.linear(h, p[f"l{i}.qkv"])
-        T = qkv.shape[0]
-        q = qkv[:, :self.q_size].reshape(T, self.n_heads, c.head_dim).contiguous()
-        k = qkv[:, self.q_size:self.q_size + self.kv_size] \
-            .reshape(T, self.n_kv_heads, c.head_dim).contiguous()
-        v = qkv[:, self.q_size + self.kv_size:] \
-            .reshape(T, self.n_kv_heads, c.head_dim).contiguous()
         k_cache, v_cache = kv_cache.layer(i)
-        ops.rope_kv_append_(q, k, v, positions, self.cos_sin,
-                            k_cache, v_cache, slot_mapping)
-        attn_out = attn_fn(i, q, k, v, k_cache, v_cache)
-        h = self._all_reduce(F.linear(attn_out.reshape(T, self.q_size),
-                                      p[f"l{i}.o"]))
+        ops.rope_kv_append_qkv_(qkv, positions, self.cos_sin,
+                                k_cache, v_cache, slot_mapping, self.n_heads)
+        attn_out = attn_fn(i, qkv, k_cache, v_cache)   # [T, q_size]
+        h = self._all_reduce(F.linear(attn_out, p[f"l{i}.o"]))
         h2 = ops.rmsnorm_residual_(h, residual, p[f"l{i}.ln2"], c.rms_eps)
         gu = F.linear(h2, p[f"l{i}.gate_up"])
         mlp = self._all_reduce(F.linear(ops.silu_mul(gu), p[f"l{i}.down"]))
@@ -249,8 +242,10 @@ class LlamaForCausalLM:
                         cu_seqlens: torch.Tensor) -> torch.Tensor:
         """Returns logits for the LAST token of each sequence:
         [n_seqs, vocab]."""
-        def attn(i, q, k, v, k_cache, v_cache):
-            return ops.prefill_attn(q, k, v, cu_seqlens, self.scale)
+        def attn(i, qkv, k_cache, v_cache):
+            return ops.prefill_attn_qkv(qkv, cu_seqlens, self.scale,
+                                        self.n_heads, self.n_kv_heads,
+                                        self.config.head_dim)
         return self._forward(input_ids, positions, kv_cache, slot_mapping,
                              attn, last_idx=cu_seqlens[1:].long() - 1)
 
@@ -259,9 +254,9 @@ class LlamaForCausalLM:
                        block_tables: torch.Tensor,
                        seq_lens: torch.Tensor) -> torch.Tensor:
         """One token per sequence; returns [batch, vocab] logits."""
-        def attn(i, q, k, v, k_cache, v_cache):
-            return ops.decode_attn(q, k_cache, v_cache, block_tables,
-                                   seq_lens, self.scale)
+        def attn(i, qkv, k_cache, v_cache):
+            return ops.decode_attn_qkv(qkv, k_cache, v_cache, block_tables,
+                                       seq_lens, self.scale, self.n_heads)
         return self._forward(input_ids, positions, kv_cache, slot_mapping,
                              attn, last_idx=None)
 

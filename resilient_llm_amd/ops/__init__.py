@@ -86,6 +86,56 @@ def rope_kv_append_(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                             slot_mapping)
 
 
+def rope_kv_append_qkv_(qkv: torch.Tensor, positions: torch.Tensor,
+                        cos_sin: torch.Tensor, k_cache: torch.Tensor,
+                        v_cache: torch.Tensor, slot_mapping: torch.Tensor,
+                        n_q: int) -> None:
+    """Fused-QKV form: rope + cache append on strided views of the raw
+    qkv GEMM output (no .contiguous() copies)."""
+    if qkv.is_cuda:
+        _gpu()
+        torch.ops.rlli.rope_kv_append_qkv_(qkv, positions, cos_sin, k_cache,
+                                           v_cache, slot_mapping, n_q)
+        return
+    T = qkv.shape[0]
+    n_kv, _, D = k_cache.shape[1], k_cache.shape[2], k_cache.shape[3]
+    q = qkv[:, :n_q * D].view(T, n_q, D)
+    k = qkv[:, n_q * D:(n_q + n_kv) * D].view(T, n_kv, D)
+    v = qkv[:, (n_q + n_kv) * D:].view(T, n_kv, D)
+    ref.rope_kv_append_(q, k, v, positions, cos_sin, k_cache, v_cache,
+                        slot_mapping)
+
+
+def decode_attn_qkv(qkv: torch.Tensor, k_cache: torch.Tensor,
+                    v_cache: torch.Tensor, block_table: torch.Tensor,
+                    seq_lens: torch.Tensor, scale: float,
+                    n_q: int) -> torch.Tensor:
+    if qkv.is_cuda:
+        _gpu()
+        return torch.ops.rlli.decode_attn_qkv(qkv, k_cache, v_cache,
+                                              block_table, seq_lens, scale, n_q)
+    B = qkv.shape[0]
+    D = k_cache.shape[3]
+    q = qkv[:, :n_q * D].reshape(B, n_q, D)
+    out = ref.decode_attn(q, k_cache, v_cache, block_table, seq_lens, scale)
+    return out.reshape(B, n_q * D)
+
+
+def prefill_attn_qkv(qkv: torch.Tensor, cu_seqlens: torch.Tensor,
+                     scale: float, n_q: int, n_kv: int,
+                     head_dim: int) -> torch.Tensor:
+    if qkv.is_cuda:
+        _gpu()
+        return torch.ops.rlli.prefill_attn_qkv(qkv, cu_seqlens, scale, n_q,
+                                               n_kv, head_dim)
+    T = qkv.shape[0]
+    D = head_dim
+    q = qkv[:, :n_q * D].reshape(T, n_q, D)
+    k = qkv[:, n_q * D:(n_q + n_kv) * D].reshape(T, n_kv, D)
+    v = qkv[:, (n_q + n_kv) * D:].reshape(T, n_kv, D)
+    return ref.prefill_attn(q, k, v, cu_seqlens, scale).reshape(T, n_q * D)
+
+
 def decode_attn(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
                 block_table: torch.Tensor, seq_lens: torch.Tensor,
                 scale: float) -> torch.Tensor:

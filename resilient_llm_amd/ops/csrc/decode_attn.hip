@@ -33,7 +33,7 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
                         const int32_t* __restrict__ seq_lens,
                         uint16_t* __restrict__ out,
                         int n_kv_heads, int block_size, int max_blocks,
-                        float scale) {
+                        float scale, int q_stride) {
   constexpr int D = GW * 8;
   constexpr int GPW = 64 / GW;                 // lane-groups per wave
   const int seq = blockIdx.x / n_kv_heads;
@@ -58,7 +58,7 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
   for (int h = 0; h < GROUP; ++h) {
     bf16x8 qh;
     qh.u = *reinterpret_cast<const uint4*>(
-        q + (int64_t(seq) * n_kv_heads * GROUP + kvh * GROUP + h) * D + d0);
+        q + int64_t(seq) * q_stride + (kvh * GROUP + h) * D + d0);
 #pragma unroll
     for (int i = 0; i < 8; ++i) qv[h][i] = bf16_to_f32(qh.s[i]);
   }
@@ -149,7 +149,7 @@ void dispatch_decode(const uint16_t* q, const uint16_t* k_cache,
                      const uint16_t* v_cache, const int32_t* block_table,
                      const int32_t* seq_lens, uint16_t* out, int batch,
                      int n_kv_heads, int block_size, int max_blocks,
-                     float scale, hipStream_t stream) {
+                     float scale, int q_stride, hipStream_t stream) {
   constexpr int D = GW * 8;
   constexpr int GPW = 64 / GW;
   const int n_part = 4 * GPW;
@@ -157,7 +157,7 @@ void dispatch_decode(const uint16_t* q, const uint16_t* k_cache,
   hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP>),
                      dim3(batch * n_kv_heads), dim3(256), smem, stream,
                      q, k_cache, v_cache, block_table, seq_lens, out,
-                     n_kv_heads, block_size, max_blocks, scale);
+                     n_kv_heads, block_size, max_blocks, scale, q_stride);
 }
 
 }  // namespace
@@ -167,13 +167,13 @@ void launch_decode_attn(const uint16_t* q, const uint16_t* k_cache,
                         const int32_t* seq_lens, uint16_t* out, int batch,
                         int n_q_heads, int n_kv_heads, int head_dim,
                         int block_size, int max_blocks, float scale,
-                        hipStream_t stream) {
+                        int q_stride, hipStream_t stream) {
   if (batch == 0) return;
   const int group = n_q_heads / n_kv_heads;
   auto run = [&](auto gw_tag, auto group_tag) {
     dispatch_decode<decltype(gw_tag)::value, decltype(group_tag)::value>(
         q, k_cache, v_cache, block_table, seq_lens, out, batch, n_kv_heads,
-        block_size, max_blocks, scale, stream);
+        block_size, max_blocks, scale, q_stride, stream);
   };
   using I8 = std::integral_constant<int, 8>;
   using I16 = std::integral_constant<int, 16>;

@@ -104,7 +104,34 @@ void rope_kv_append_(Tensor q, Tensor k, const Tensor& v,
       bf16_ptr(q), bf16_ptr(k), bf16_ptr(v),
       positions.data_ptr<int32_t>(), cos_sin.data_ptr<float>(),
       bf16_ptr(k_cache), bf16_ptr(v_cache), slot_mapping.data_ptr<int32_t>(),
-      tokens, n_q, n_kv, D, block_size, current_stream(q));
+      tokens, n_q, n_kv, D, block_size, n_q * D, n_kv * D, current_stream(q));
+}
+
+// Fused-QKV form: qkv is the raw [T, (n_q + 2*n_kv) * D] GEMM output;
+// rope + cache append run on strided views, zero copies.
+void rope_kv_append_qkv_(Tensor qkv, const Tensor& positions,
+                         const Tensor& cos_sin, Tensor k_cache,
+                         Tensor v_cache, const Tensor& slot_mapping,
+                         int64_t n_q) {
+  check_bf16_contig(qkv, "qkv");
+  check_bf16_contig(k_cache, "k_cache");
+  check_bf16_contig(v_cache, "v_cache");
+  const int n_kv = int(k_cache.size(1));
+  const int D = int(k_cache.size(3));
+  const int block_size = int(k_cache.size(2));
+  const int stride = int(qkv.size(1));
+  TORCH_CHECK(stride == (n_q + 2 * n_kv) * D, "qkv width mismatch");
+  const int tokens = int(qkv.size(0));
+  TORCH_CHECK(positions.numel() == tokens && slot_mapping.numel() == tokens);
+  TORCH_CHECK(cos_sin.size(1) == D && D % 16 == 0);
+  uint16_t* base = bf16_ptr(qkv);
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
+  rlli::launch_rope_kv_append(
+      base, base + n_q * D, base + (n_q + n_kv) * D,
+      positions.data_ptr<int32_t>(), cos_sin.data_ptr<float>(),
+      bf16_ptr(k_cache), bf16_ptr(v_cache), slot_mapping.data_ptr<int32_t>(),
+      tokens, int(n_q), n_kv, D, block_size, stride, stride,
+      current_stream(qkv));
 }
 
 // ---------------------------------------------------------- decode_attn
@@ -136,7 +163,32 @@ Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
       bf16_ptr(q), bf16_ptr(k_cache), bf16_ptr(v_cache),
       block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       bf16_ptr(out), batch, n_q, n_kv, D, block_size, max_blocks,
-      float(scale), current_stream(q));
+      float(scale), n_q * D, current_stream(q));
+  return out;
+}
+
+// Fused-QKV decode attention: q read strided from the qkv GEMM output.
+Tensor decode_attn_qkv(const Tensor& qkv, const Tensor& k_cache,
+                       const Tensor& v_cache, const Tensor& block_table,
+                       const Tensor& seq_lens, double scale, int64_t n_q) {
+  check_bf16_contig(qkv, "qkv");
+  const int batch = int(qkv.size(0));
+  const int n_kv = int(k_cache.size(1));
+  const int D = int(k_cache.size(3));
+  const int block_size = int(k_cache.size(2));
+  const int max_blocks = int(block_table.size(1));
+  const int stride = int(qkv.size(1));
+  TORCH_CHECK(stride == (n_q + 2 * n_kv) * D, "qkv width mismatch");
+  const int group = int(n_q) / n_kv;
+  TORCH_CHECK(group == 1 || group == 2 || group == 4 || group == 8);
+  TORCH_CHECK(D == 64 || D == 128);
+  Tensor out = at::empty({batch, n_q * D}, qkv.options());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
+  rlli::launch_decode_attn(
+      bf16_ptr(qkv), bf16_ptr(k_cache), bf16_ptr(v_cache),
+      block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
+      bf16_ptr(out), batch, int(n_q), n_kv, D, block_size, max_blocks,
+      float(scale), stride, current_stream(qkv));
   return out;
 }
 
@@ -162,7 +214,32 @@ Tensor prefill_attn(const Tensor& q, const Tensor& k, const Tensor& v,
   rlli::launch_prefill_attn(
       bf16_ptr(q), bf16_ptr(k), bf16_ptr(v), cu_seqlens.data_ptr<int32_t>(),
       bf16_ptr(out), n_seqs, T, n_q, n_kv, D, float(scale),
-      current_stream(q));
+      n_q * D, n_kv * D, current_stream(q));
+  return out;
+}
+
+// Fused-QKV prefill attention: q/k/v all read strided from qkv (k was
+// rope-rotated in place by rope_kv_append_qkv_).
+Tensor prefill_attn_qkv(const Tensor& qkv, const Tensor& cu_seqlens,
+                        double scale, int64_t n_q, int64_t n_kv,
+                        int64_t head_dim) {
+  check_bf16_contig(qkv, "qkv");
+  TORCH_CHECK(cu_seqlens.scalar_type() == at::kInt && cu_seqlens.is_contiguous());
+  const int T = int(qkv.size(0));
+  const int D = int(head_dim);
+  const int stride = int(qkv.size(1));
+  TORCH_CHECK(stride == (n_q + 2 * n_kv) * D, "qkv width mismatch");
+  const int group = int(n_q) / int(n_kv);
+  TORCH_CHECK(group == 1 || group == 2 || group == 4 || group == 8);
+  TORCH_CHECK(D == 64 || D == 128);
+  const int n_seqs = int(cu_seqlens.numel()) - 1;
+  Tensor out = at::empty({T, n_q * D}, qkv.options());
+  uint16_t* base = bf16_ptr(qkv);
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
+  rlli::launch_prefill_attn(
+      base, base + n_q * D, base + (n_q + n_kv) * D,
+      cu_seqlens.data_ptr<int32_t>(), bf16_ptr(out), n_seqs, T, int(n_q),
+      int(n_kv), D, float(scale), stride, stride, current_stream(qkv));
   return out;
 }
 
@@ -193,6 +270,13 @@ TORCH_LIBRARY(rlli, m) {
         "Tensor slot_mapping) -> ()");
   m.def("decode_attn(Tensor q, Tensor k_cache, Tensor v_cache, "
         "Tensor block_table, Tensor seq_lens, float scale) -> Tensor");
+  m.def("rope_kv_append_qkv_(Tensor(a!) qkv, Tensor positions, Tensor cos_sin, "
+        "Tensor(b!) k_cache, Tensor(c!) v_cache, Tensor slot_mapping, "
+        "int n_q) -> ()");
+  m.def("decode_attn_qkv(Tensor qkv, Tensor k_cache, Tensor v_cache, "
+        "Tensor block_table, Tensor seq_lens, float scale, int n_q) -> Tensor");
+  m.def("prefill_attn_qkv(Tensor qkv, Tensor cu_seqlens, float scale, "
+        "int n_q, int n_kv, int head_dim) -> Tensor");
   m.def("prefill_attn(Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, "
         "float scale) -> Tensor");
   m.def("sample(Tensor logits, Tensor temperatures, int seed) -> Tensor");
@@ -204,6 +288,9 @@ TORCH_LIBRARY_IMPL(rlli, CUDA, m) {
   m.impl("silu_mul", &silu_mul);
   m.impl("rope_kv_append_", &rope_kv_append_);
   m.impl("decode_attn", &decode_attn);
+  m.impl("rope_kv_append_qkv_", &rope_kv_append_qkv_);
+  m.impl("decode_attn_qkv", &decode_attn_qkv);
+  m.impl("prefill_attn_qkv", &prefill_attn_qkv);
   m.impl("prefill_attn", &prefill_attn);
   m.impl("sample", &sample);
 }

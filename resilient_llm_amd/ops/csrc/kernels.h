@@ -20,13 +20,15 @@ void launch_silu_mul(const uint16_t* gate_up, uint16_t* out, int rows,
 
 // In-place RoPE (NeoX-interleaved-halves style) on q and k, then scatter
 // k/v into the paged cache at slot_mapping[t].
+// q/k/v may be strided slices of one fused [T, (n_q+2*n_kv)*D] qkv
+// tensor: *_stride is the per-token element stride of each view.
 void launch_rope_kv_append(
     uint16_t* q, uint16_t* k, uint16_t* v,
     const int32_t* positions, const float* cos_sin,   // [max_pos, head_dim]
     uint16_t* k_cache, uint16_t* v_cache,
     const int32_t* slot_mapping,
     int tokens, int n_q_heads, int n_kv_heads, int head_dim,
-    int block_size, hipStream_t stream);
+    int block_size, int q_stride, int kv_stride, hipStream_t stream);
 
 // Paged GQA decode attention: one new q token per sequence.
 void launch_decode_attn(
@@ -37,7 +39,8 @@ void launch_decode_attn(
     const int32_t* seq_lens,           // [batch]
     uint16_t* out,                     // [batch, n_q_heads, head_dim]
     int batch, int n_q_heads, int n_kv_heads, int head_dim,
-    int block_size, int max_blocks, float scale, hipStream_t stream);
+    int block_size, int max_blocks, float scale, int q_stride,
+    hipStream_t stream);
 
 // Varlen causal prefill attention over in-batch q/k/v.
 void launch_prefill_attn(
@@ -45,7 +48,8 @@ void launch_prefill_attn(
     const int32_t* cu_seqlens,         // [n_seqs + 1]
     uint16_t* out,
     int n_seqs, int total_tokens, int n_q_heads, int n_kv_heads,
-    int head_dim, float scale, hipStream_t stream);
+    int head_dim, float scale, int q_stride, int kv_stride,
+    hipStream_t stream);
 
 // Fused sampling: greedy argmax when temperature[i] == 0, else Gumbel-max
 // sampling of softmax(logits / temperature[i]) with an in-kernel counter

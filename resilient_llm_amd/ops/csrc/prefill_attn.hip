@@ -26,7 +26,8 @@ void prefill_attn_kernel(const uint16_t* __restrict__ q,
                          const uint16_t* __restrict__ v,
                          const int32_t* __restrict__ cu_seqlens,
                          uint16_t* __restrict__ out,
-                         int n_seqs, int n_kv_heads, float scale) {
+                         int n_seqs, int n_kv_heads, float scale,
+                         int q_stride, int kv_stride) {
   constexpr int D = GW * 8;
   constexpr int GPW = 64 / GW;
   const int t = blockIdx.x / n_kv_heads;       // global query token
@@ -59,7 +60,7 @@ void prefill_attn_kernel(const uint16_t* __restrict__ q,
   for (int h = 0; h < GROUP; ++h) {
     bf16x8 qh;
     qh.u = *reinterpret_cast<const uint4*>(
-        q + (int64_t(t) * n_q_heads + kvh * GROUP + h) * D + d0);
+        q + int64_t(t) * q_stride + (kvh * GROUP + h) * D + d0);
 #pragma unroll
     for (int i = 0; i < 8; ++i) qv[h][i] = bf16_to_f32(qh.s[i]);
   }
@@ -75,14 +76,14 @@ void prefill_attn_kernel(const uint16_t* __restrict__ q,
 
   // lane-group pid takes kv tokens pid, pid + n_part, ...
   for (int j = pid; j < kv_len; j += n_part) {
-    const int64_t row = int64_t(seq_start + j) * n_kv_heads + kvh;
+    const int64_t row = int64_t(seq_start + j) * kv_stride + kvh * D;
     bf16x8 kv8;
-    kv8.u = *reinterpret_cast<const uint4*>(k + row * D + d0);
+    kv8.u = *reinterpret_cast<const uint4*>(k + row + d0);
     float kf[8];
 #pragma unroll
     for (int i = 0; i < 8; ++i) kf[i] = bf16_to_f32(kv8.s[i]);
     bf16x8 vv8;
-    vv8.u = *reinterpret_cast<const uint4*>(v + row * D + d0);
+    vv8.u = *reinterpret_cast<const uint4*>(v + row + d0);
     float vf[8];
 #pragma unroll
     for (int i = 0; i < 8; ++i) vf[i] = bf16_to_f32(vv8.s[i]);
@@ -137,14 +138,15 @@ template <int GW, int GROUP>
 void dispatch_prefill(const uint16_t* q, const uint16_t* k, const uint16_t* v,
                       const int32_t* cu_seqlens, uint16_t* out, int n_seqs,
                       int total_tokens, int n_kv_heads, float scale,
-                      hipStream_t stream) {
+                      int q_stride, int kv_stride, hipStream_t stream) {
   constexpr int D = GW * 8;
   constexpr int GPW = 64 / GW;
   const int n_part = 4 * GPW;
   const size_t smem = size_t(n_part) * GROUP * (D + 2) * sizeof(float);
   hipLaunchKernelGGL((prefill_attn_kernel<GW, GROUP>),
                      dim3(total_tokens * n_kv_heads), dim3(256), smem, stream,
-                     q, k, v, cu_seqlens, out, n_seqs, n_kv_heads, scale);
+                     q, k, v, cu_seqlens, out, n_seqs, n_kv_heads, scale,
+                     q_stride, kv_stride);
 }
 
 }  // namespace
@@ -153,13 +155,14 @@ void launch_prefill_attn(const uint16_t* q, const uint16_t* k,
                          const uint16_t* v, const int32_t* cu_seqlens,
                          uint16_t* out, int n_seqs, int total_tokens,
                          int n_q_heads, int n_kv_heads, int head_dim,
-                         float scale, hipStream_t stream) {
+                         float scale, int q_stride, int kv_stride,
+                         hipStream_t stream) {
   if (total_tokens == 0) return;
   const int group = n_q_heads / n_kv_heads;
   auto run = [&](auto gw_tag, auto group_tag) {
     dispatch_prefill<decltype(gw_tag)::value, decltype(group_tag)::value>(
         q, k, v, cu_seqlens, out, n_seqs, total_tokens, n_kv_heads, scale,
-        stream);
+        q_stride, kv_stride, stream);
   };
   using I8 = std::integral_constant<int, 8>;
   using I16 = std::integral_constant<int, 16>;

@@ -26,7 +26,7 @@ __global__ void rope_kv_kernel(uint16_t* __restrict__ q,
                                uint16_t* __restrict__ v_cache,
                                const int32_t* __restrict__ slot_mapping,
                                int n_q_heads, int n_kv_heads, int head_dim,
-                               int block_size) {
+                               int block_size, int q_stride, int kv_stride) {
   const int t = blockIdx.x;
   const int pos = positions[t];
   const int half = head_dim / 2;
@@ -48,8 +48,8 @@ __global__ void rope_kv_kernel(uint16_t* __restrict__ q,
     const int p8 = (w % vecs_per_head) * 8;    // first pair index of this vec
     const bool is_q = h < n_q_heads;
     uint16_t* base = is_q
-        ? q + (int64_t(t) * n_q_heads + h) * head_dim
-        : k + (int64_t(t) * n_kv_heads + (h - n_q_heads)) * head_dim;
+        ? q + int64_t(t) * q_stride + int64_t(h) * head_dim
+        : k + int64_t(t) * kv_stride + int64_t(h - n_q_heads) * head_dim;
     bf16x8 lo, hi, olo, ohi;
     lo.u = *reinterpret_cast<const uint4*>(base + p8);
     hi.u = *reinterpret_cast<const uint4*>(base + half + p8);
@@ -83,7 +83,7 @@ __global__ void rope_kv_kernel(uint16_t* __restrict__ q,
       const int h = w / (head_dim / 8);
       const int d8 = (w % (head_dim / 8)) * 8;
       const uint4 vv = *reinterpret_cast<const uint4*>(
-          v + (int64_t(t) * n_kv_heads + h) * head_dim + d8);
+          v + int64_t(t) * kv_stride + int64_t(h) * head_dim + d8);
       uint16_t* dst = v_cache + cache_base +
                       int64_t(h) * block_size * head_dim + d8;
       *reinterpret_cast<uint4*>(dst) = vv;
@@ -99,12 +99,13 @@ void launch_rope_kv_append(
     uint16_t* k_cache, uint16_t* v_cache,
     const int32_t* slot_mapping,
     int tokens, int n_q_heads, int n_kv_heads, int head_dim,
-    int block_size, hipStream_t stream) {
+    int block_size, int q_stride, int kv_stride, hipStream_t stream) {
   if (tokens == 0) return;
   const int threads = 256;
   hipLaunchKernelGGL(rope_kv_kernel, dim3(tokens), dim3(threads), 0, stream,
                      q, k, v, positions, cos_sin, k_cache, v_cache,
-                     slot_mapping, n_q_heads, n_kv_heads, head_dim, block_size);
+                     slot_mapping, n_q_heads, n_kv_heads, head_dim, block_size,
+                     q_stride, kv_stride);
 }
 
 }  // namespace rlli

@@ -152,3 +152,47 @@ def test_sample_deterministic_given_seed():
     c = ops.sample(logits, temps, seed=43)
     assert torch.equal(a, b)
     assert not torch.equal(a, c)   # overwhelmingly likely
+
+
+def test_fused_qkv_ops_match_unfused():
+    """rope_kv_append_qkv_ / decode_attn_qkv / prefill_attn_qkv on strided
+    slices of one qkv tensor match the unfused contiguous ops."""
+    import resilient_llm_amd.ops as ops
+    n_q, n_kv, D, T, bs = 8, 2, 128, 12, 16
+    width = (n_q + 2 * n_kv) * D
+    qkv = torch.randn(T, width, dtype=torch.bfloat16, device=DEV)
+    qkv2 = qkv.clone()
+    pos = torch.arange(T, dtype=torch.int32, device=DEV)
+    cs = ops.build_cos_sin(64, D, device=DEV)
+    kc = torch.zeros(4, n_kv, bs, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    kc2, vc2 = kc.clone(), vc.clone()
+    slots = torch.arange(T, dtype=torch.int32, device=DEV)
+
+    # unfused path on contiguous copies
+    q = qkv2[:, :n_q * D].reshape(T, n_q, D).contiguous()
+    k = qkv2[:, n_q * D:(n_q + n_kv) * D].reshape(T, n_kv, D).contiguous()
+    v = qkv2[:, (n_q + n_kv) * D:].reshape(T, n_kv, D).contiguous()
+    ops.rope_kv_append_(q, k, v, pos, cs, kc2, vc2, slots)
+
+    ops.rope_kv_append_qkv_(qkv, pos, cs, kc, vc, slots, n_q)
+    torch.testing.assert_close(
+        qkv[:, :n_q * D].reshape(T, n_q, D).float(), q.float())
+    torch.testing.assert_close(kc.float(), kc2.float())
+    torch.testing.assert_close(vc.float(), vc2.float())
+
+    cu = torch.tensor([0, 5, T], dtype=torch.int32, device=DEV)
+    scale = 0.088
+    out_f = ops.prefill_attn_qkv(qkv, cu, scale, n_q, n_kv, D)
+    out_u = ops.prefill_attn(q, k, v, cu, scale)
+    torch.testing.assert_close(out_f.reshape(T, n_q, D).float(),
+                               out_u.float())
+
+    bt = torch.tensor([[0, 1], [2, 3]], dtype=torch.int32, device=DEV)
+    lens = torch.tensor([5, 7], dtype=torch.int32, device=DEV)
+    qkv_d = qkv[:2].contiguous()
+    q_d = qkv_d[:, :n_q * D].reshape(2, n_q, D).contiguous()
+    out_fd = ops.decode_attn_qkv(qkv_d, kc, vc, bt, lens, scale, n_q)
+    out_ud = ops.decode_attn(q_d, kc, vc, bt, lens, scale)
+    torch.testing.assert_close(out_fd.reshape(2, n_q, D).float(),
+                               out_ud.float())
