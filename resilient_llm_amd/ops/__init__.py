@@ -162,4 +162,23 @@ def sample(logits: torch.Tensor, temperatures: torch.Tensor,
     return ref.sample(logits, temperatures, seed)
 
 
+def skinny_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Decode-projection GEMM (M<=64): custom MFMA weight-streaming
+    kernel on GPU, F.linear on CPU."""
+    if x.is_cuda:
+        _gpu()
+        return torch.ops.rlli.skinny_linear(x, w)
+    return torch.nn.functional.linear(x, w)
+
+
+def linear_auto(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Pick the skinny-M kernel when the shape qualifies, else
+    hipBLASLt via F.linear."""
+    if (x.is_cuda and x.dim() == 2 and 1 <= x.shape[0] <= 64
+            and x.shape[1] % 256 == 0 and w.shape[0] % 64 == 0):
+        _gpu()
+        return torch.ops.rlli.skinny_linear(x, w)
+    return torch.nn.functional.linear(x, w)
+
+
 build_cos_sin = ref.build_cos_sin

@@ -30,6 +30,7 @@ SOURCES = [
     "decode_attn.hip",
     "prefill_attn.hip",
     "sampling.hip",
+    "skinny_gemm.hip",
 ]
 HEADERS = ["common.h", "kernels.h"]
 

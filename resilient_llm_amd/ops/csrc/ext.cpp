@@ -243,6 +243,36 @@ Tensor prefill_attn_qkv(const Tensor& qkv, const Tensor& cu_seqlens,
   return out;
 }
 
+// --------------------------------------------------------- skinny_linear
+Tensor skinny_linear(const Tensor& x, const Tensor& w) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(w, "w");
+  const int M = int(x.size(0));
+  const int K = int(x.size(1));
+  const int N = int(w.size(0));
+  TORCH_CHECK(w.size(1) == K, "x/W K mismatch");
+  TORCH_CHECK(M >= 1 && M <= 64, "skinny_linear needs 1 <= M <= 64, got ", M);
+  TORCH_CHECK(K % 256 == 0 && N % 64 == 0,
+              "skinny_linear needs K%256==0 and N%64==0, got K=", K, " N=", N);
+  const int n_panels = N / 64;
+  int splitk = 1;
+  if (n_panels < 256) {
+    splitk = std::min<int>(8, std::max<int>(1, 512 / n_panels));
+    splitk = std::min<int>(splitk, K / 256);
+  }
+  Tensor out = at::empty({M, N}, x.options());
+  Tensor ws;
+  float* ws_ptr = nullptr;
+  if (splitk > 1) {
+    ws = at::empty({splitk, M, N}, x.options().dtype(at::kFloat));
+    ws_ptr = ws.data_ptr<float>();
+  }
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  rlli::launch_skinny_gemm(bf16_ptr(x), bf16_ptr(w), ws_ptr, bf16_ptr(out),
+                           M, N, K, splitk, current_stream(x));
+  return out;
+}
+
 // --------------------------------------------------------------- sample
 Tensor sample(const Tensor& logits, const Tensor& temperatures, int64_t seed) {
   check_bf16_contig(logits, "logits");
@@ -280,6 +310,7 @@ TORCH_LIBRARY(rlli, m) {
   m.def("prefill_attn(Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, "
         "float scale) -> Tensor");
   m.def("sample(Tensor logits, Tensor temperatures, int seed) -> Tensor");
+  m.def("skinny_linear(Tensor x, Tensor w) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(rlli, CUDA, m) {
@@ -293,4 +324,5 @@ TORCH_LIBRARY_IMPL(rlli, CUDA, m) {
   m.impl("prefill_attn_qkv", &prefill_attn_qkv);
   m.impl("prefill_attn", &prefill_attn);
   m.impl("sample", &sample);
+  m.impl("skinny_linear", &skinny_linear);
 }

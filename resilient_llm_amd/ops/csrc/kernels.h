@@ -51,6 +51,12 @@ void launch_prefill_attn(
     int head_dim, float scale, int q_stride, int kv_stride,
     hipStream_t stream);
 
+// Skinny-M GEMM (decode projections): out[M,N] = x[M,K] @ W[N,K]^T.
+// ws is a [splitk, M, N] f32 workspace (unused when splitk == 1).
+void launch_skinny_gemm(const uint16_t* x, const uint16_t* w, float* ws,
+                        uint16_t* out, int M, int N, int K, int splitk,
+                        hipStream_t stream);
+
 // Fused sampling: greedy argmax when temperature[i] == 0, else Gumbel-max
 // sampling of softmax(logits / temperature[i]) with an in-kernel counter
 // hash RNG keyed on (seed, row, column) — no 32 MB random tensor per step.

@@ -196,3 +196,19 @@ def test_fused_qkv_ops_match_unfused():
     out_ud = ops.decode_attn(q_d, kc, vc, bt, lens, scale)
     torch.testing.assert_close(out_fd.reshape(2, n_q, D).float(),
                                out_ud.float())
+
+
+@pytest.mark.parametrize("M,N,K", [
+    (64, 5120, 4096),    # 8B qkv
+    (64, 4096, 4096),    # 8B o-proj
+    (64, 28672, 4096),   # 8B gate_up
+    (64, 4096, 14336),   # 8B down
+    (64, 128256, 4096),  # 8B lm_head
+    (1, 4096, 4096), (17, 4096, 4096), (33, 1024, 512), (48, 64, 256),
+])
+def test_skinny_linear_matches_blaslt(M, N, K):
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.1
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.1
+    y = ops.skinny_linear(x, w)
+    y_ref = torch.nn.functional.linear(x.float(), w.float())
+    torch.testing.assert_close(y.float(), y_ref, atol=0.05, rtol=0.05)
