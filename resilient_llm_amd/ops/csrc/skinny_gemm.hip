@@ -41,7 +41,7 @@ void skinny_gemm_kernel(const uint16_t* __restrict__ x,
                         int M, int N, int K,
                         int chunks_per_slice, int splitk) {
   constexpr int MROWS = M_TILES * 16;
-  __shared__ __attribute__((aligned(16))) uint16_t xs[MROWS * KC];
+  __shared__ __attribute__((aligned(16))) uint16_t xs[2 * MROWS * KC];
 
   const int n_panels = N / BN;
   const int panel = blockIdx.x % n_panels;
@@ -61,11 +61,15 @@ void skinny_gemm_kernel(const uint16_t* __restrict__ x,
 #pragma unroll
   for (int mt = 0; mt < M_TILES; ++mt) acc[mt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  for (int c = chunk0; c < chunk1; ++c) {
+  using u32x4 = __attribute__((ext_vector_type(4))) unsigned int;
+  constexpr int KSTEPS = KC / 32;
+  // W row base for this lane's B fragments
+  const uint16_t* wrow = w + int64_t(n0 + jcol) * K + koct * 8;
+
+  // ---- prologue: stage x chunk0 ----
+  constexpr int PIECES = MROWS * KC / 8 / 256;
+  auto stage_x = [&](int c) {
     const int kbase = c * KC;
-    // ---- stage x[0:MROWS, kbase:kbase+KC] into swizzled LDS ----
-    // 256 threads x (MROWS*KC/8/256) bf16x8 pieces
-    constexpr int PIECES = MROWS * KC / 8 / 256;
 #pragma unroll
     for (int p = 0; p < PIECES; ++p) {
       const int idx = p * 256 + threadIdx.x;   // which bf16x8 of the tile
@@ -76,26 +80,43 @@ void skinny_gemm_kernel(const uint16_t* __restrict__ x,
         piece = *reinterpret_cast<const uint4*>(
             x + int64_t(row) * K + kbase + col8 * 8);
       *reinterpret_cast<uint4*>(
-          reinterpret_cast<char*>(xs) + swz(row, (row * KC + col8 * 8) * 2)) = piece;
+          reinterpret_cast<char*>(xs) +
+          ((c & 1) ? MROWS * KC * 2 : 0) +
+          swz(row, (row * KC + col8 * 8) * 2)) = piece;
     }
-    __syncthreads();
-
-    // ---- 8 MFMA k-steps over the chunk ----
+  };
+  // preload the whole chunk's B fragments: KSTEPS outstanding nt loads
+  u32x4 braw[KSTEPS];
+  auto load_b = [&](int c) {
 #pragma unroll
-    for (int ks = 0; ks < KC / 32; ++ks) {
+    for (int ks = 0; ks < KSTEPS; ++ks)
+      braw[ks] = __builtin_nontemporal_load(
+          reinterpret_cast<const u32x4*>(wrow + c * KC + ks * 32));
+  };
+
+  stage_x(chunk0);
+  load_b(chunk0);
+  __syncthreads();
+
+  for (int c = chunk0; c < chunk1; ++c) {
+    const char* xbase = reinterpret_cast<const char*>(xs) +
+                        ((c & 1) ? MROWS * KC * 2 : 0);
+    u32x4 bcur[KSTEPS];
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) bcur[ks] = braw[ks];
+    if (c + 1 < chunk1) {
+      stage_x(c + 1);        // issues into the other LDS buffer
+      load_b(c + 1);         // next chunk's B stream in flight
+    }
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
       const int k0 = ks * 32;
-      // B fragment: 16 contiguous bf16 of W row (n0 + jcol)
-      const uint16_t* wp = w + int64_t(n0 + jcol) * K + kbase + k0 + koct * 8;
-      using u32x4 = __attribute__((ext_vector_type(4))) unsigned int;
-      u32x4 raw = __builtin_nontemporal_load(
-          reinterpret_cast<const u32x4*>(wp));
-      bf16x8_vec bfrag = *reinterpret_cast<bf16x8_vec*>(&raw);
+      bf16x8_vec bfrag = *reinterpret_cast<bf16x8_vec*>(&bcur[ks]);
 #pragma unroll
       for (int mt = 0; mt < M_TILES; ++mt) {
         const int row = mt * 16 + jcol;
         uint4 araw = *reinterpret_cast<const uint4*>(
-            reinterpret_cast<const char*>(xs) +
-            swz(row, (row * KC + k0 + koct * 8) * 2));
+            xbase + swz(row, (row * KC + k0 + koct * 8) * 2));
         bf16x8_vec afrag = *reinterpret_cast<bf16x8_vec*>(&araw);
         acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag, bfrag, acc[mt], 0, 0, 0);

@@ -59,6 +59,7 @@ class EngineWorker(Worker):
                  num_blocks: Optional[int] = None,
                  use_graphs: bool = False,
                  tp_rank: int = 0, tp_world: int = 1, tp_group=None,
+                 tp_control=None,
                  seed: int = 0) -> None:
         super().__init__(device=device_label or f"gpu:{device}",
                          models={model_name})
@@ -83,6 +84,7 @@ class EngineWorker(Worker):
         if use_graphs and torch_device != "cpu":
             from ..engine.graph import install_graph_runner
             install_graph_runner(self.engine)
+        self.tp_control = tp_control
         self.tokenizer = ByteTokenizer(config.vocab_size)
         self.fault_mode = "none"
         self.total_served = 0
@@ -93,9 +95,13 @@ class EngineWorker(Worker):
         self._sinks: dict[str, tuple[asyncio.Queue, asyncio.AbstractEventLoop]] = {}
         self._work_event = threading.Event()
         self._stop = False
-        self._thread = threading.Thread(target=self._engine_loop, daemon=True,
-                                        name=f"engine-{self.device}")
-        self._thread.start()
+        self._thread = None
+        # TP followers run the loop in their own main thread (pool_main)
+        if tp_control is None or tp_control.is_leader:
+            self._thread = threading.Thread(target=self._engine_loop,
+                                            daemon=True,
+                                            name=f"engine-{self.device}")
+            self._thread.start()
 
     # --------------------------------------------------------- engine loop
     def _engine_loop(self) -> None:
@@ -104,11 +110,17 @@ class EngineWorker(Worker):
                 time.sleep(0.05)
                 continue
             with self._lock:
-                has_work = self.engine.has_work()
+                has_work = self.engine.has_work() or (
+                    self.tp_control is not None and self.tp_control.pending())
             if not has_work:
                 self._work_event.wait(timeout=0.01)
                 self._work_event.clear()
                 continue
+            if self.tp_control is not None:
+                with self._lock:
+                    self.tp_control.sync(self.engine)
+                    if not self.engine.has_work():
+                        continue
             try:
                 with self._lock:
                     outputs = self.engine.step()
@@ -145,8 +157,23 @@ class EngineWorker(Worker):
         loop = asyncio.get_running_loop()
         self._sinks[rid] = (q, loop)
         try:
-            with self._lock:
-                self.engine.add_request(rid, prompt_ids, params)
+            if self.tp_control is not None:
+                # leader: validate locally, then lockstep-broadcast the op
+                blocks_needed = -(-(len(prompt_ids) + params.max_tokens) // 16)
+                if blocks_needed > self.engine.kv.num_blocks:
+                    raise CapacityExceeded(
+                        f"request needs {blocks_needed} KV blocks, cache has "
+                        f"{self.engine.kv.num_blocks}")
+                if len(self.engine.waiting) >= self.engine.max_queue:
+                    raise CapacityExceeded(f"queue full ({self.engine.max_queue})")
+                self.tp_control.submit(
+                    ("add", rid, prompt_ids,
+                     {"max_tokens": params.max_tokens,
+                      "temperature": params.temperature,
+                      "seed": params.seed}))
+            else:
+                with self._lock:
+                    self.engine.add_request(rid, prompt_ids, params)
         except CapacityExceeded as e:
             del self._sinks[rid]
             raise WorkerThrottled(str(e)) from e
@@ -190,8 +217,12 @@ class EngineWorker(Worker):
         finally:
             self._in_flight -= 1
             self._cleanup(rid)
-            with self._lock:
-                self.engine.abort(rid)
+            if self.tp_control is not None:
+                self.tp_control.submit(("abort", rid))
+                self._work_event.set()
+            else:
+                with self._lock:
+                    self.engine.abort(rid)
 
     async def _stream_impl(self, req: GenerationRequest) -> AsyncIterator[GenerationChunk]:
         rid, q, n_prompt = self._enqueue(req)
@@ -217,8 +248,12 @@ class EngineWorker(Worker):
         finally:
             self._in_flight -= 1
             self._cleanup(rid)
-            with self._lock:
-                self.engine.abort(rid)
+            if self.tp_control is not None:
+                self.tp_control.submit(("abort", rid))
+                self._work_event.set()
+            else:
+                with self._lock:
+                    self.engine.abort(rid)
 
     def generate_stream(self, req: GenerationRequest) -> AsyncIterator[GenerationChunk]:
         return self._stream_impl(req)
