@@ -80,12 +80,21 @@ def gateway_session(args, stub_kwargs: Optional[dict] = None):
         app = GatewayApp(config, registry, health_interval_s=0.5)
         server = HttpServer(app.handle, host="127.0.0.1", port=config.cluster.port)
         await server.start()
+        # wait for process-backed workers to come up before serving
+        for w in registry.all().values():
+            if hasattr(w, "connect"):
+                await w.connect(timeout=900)
         await app.start_background()
         ready.set()
         await holder["stop"].wait()
         await app.stop_background()
         await server.stop()
         await registry.close()
+        # let in-flight connection handlers unwind before the loop dies
+        tasks = [t for t in asyncio.all_tasks() if t is not asyncio.current_task()]
+        for t in tasks:
+            t.cancel()
+        await asyncio.gather(*tasks, return_exceptions=True)
 
     th = threading.Thread(target=lambda: loop.run_until_complete(_main()), daemon=True)
     th.start()

@@ -54,6 +54,9 @@ async def serve(config: Config, registry: Optional[WorkerRegistry] = None,
     app = GatewayApp(config, registry)
     server = HttpServer(app.handle, host=config.cluster.host, port=config.cluster.port)
     await server.start()
+    for w in registry.all().values():
+        if hasattr(w, "connect"):
+            await w.connect(timeout=900)
     await app.start_background()
     log_with_timestamp(
         f"gateway listening on http://{config.cluster.host}:{config.cluster.port} "

@@ -193,7 +193,7 @@ class RpcWorkerClient(Worker):
 
     async def _call(self, op: str, data) -> tuple[int, asyncio.Queue]:
         if self._writer is None:
-            await self.connect(timeout=5.0)
+            await self.connect(timeout=600.0)
         self._next_id += 1
         mid = self._next_id
         q: asyncio.Queue = asyncio.Queue()
