@@ -98,7 +98,7 @@ def gateway_session(args, stub_kwargs: Optional[dict] = None):
 
     th = threading.Thread(target=lambda: loop.run_until_complete(_main()), daemon=True)
     th.start()
-    if not ready.wait(timeout=120):
+    if not ready.wait(timeout=900):
         raise RuntimeError("self-hosted gateway failed to start")
     log_with_timestamp(
         f"self-hosted gateway on http://127.0.0.1:{config.cluster.port}", "grey")
