@@ -72,26 +72,42 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
     for (int i = 0; i < 8; ++i) acc[h][i] = 0.f;
   }
 
+  // Latency discipline: batch-issue ALL of a cache block's K/V vectors
+  // for this lane-group (TPG k-loads + TPG v-loads back to back, plus
+  // the NEXT block's K prefetched) before any arithmetic touches them —
+  // one lane then has ~2*TPG+ loads in flight instead of 2, which is
+  // what moves this kernel from latency-bound to bandwidth-bound
+  // (guide Guideline 7/15: waits belong at the first consumer).
   const int n_blocks = (len + block_size - 1) / block_size;
+  constexpr int TPG = 16 / GPW >= 1 ? 16 / GPW : 1;   // tokens per group @bs16
   const int tok_per_grp = block_size / GPW;
   for (int b = wave; b < n_blocks; b += n_waves) {
     const int phys = block_table[int64_t(seq) * max_blocks + b];
     const int64_t kv_base =
         (int64_t(phys) * n_kv_heads + kvh) * block_size * D;
-    for (int i = 0; i < tok_per_grp; ++i) {
-      const int tok = group + GPW * i;         // token within the block
+    for (int bi = 0; bi < tok_per_grp; bi += TPG) {
+    uint4 kraw[TPG], vraw[TPG];
+#pragma unroll
+    for (int i = 0; i < TPG; ++i) {
+      if (bi + i < tok_per_grp) {
+        const int tok = group + GPW * (bi + i);
+        kraw[i] = *reinterpret_cast<const uint4*>(
+            k_cache + kv_base + int64_t(tok) * D + d0);
+        vraw[i] = *reinterpret_cast<const uint4*>(
+            v_cache + kv_base + int64_t(tok) * D + d0);
+      }
+    }
+#pragma unroll
+    for (int i = 0; i < TPG; ++i) {
+      if (bi + i >= tok_per_grp) break;
+      const int tok = group + GPW * (bi + i);  // token within the block
       if (b * block_size + tok >= len) continue;   // group-uniform tail
-      bf16x8 kv;
-      kv.u = *reinterpret_cast<const uint4*>(
-          k_cache + kv_base + int64_t(tok) * D + d0);
-      float kf[8];
+      bf16x8 kv, vv;
+      kv.u = kraw[i];
+      vv.u = vraw[i];
+      float kf[8], vf[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) kf[j] = bf16_to_f32(kv.s[j]);
-
-      bf16x8 vv;
-      vv.u = *reinterpret_cast<const uint4*>(
-          v_cache + kv_base + int64_t(tok) * D + d0);
-      float vf[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) vf[j] = bf16_to_f32(vv.s[j]);
 
@@ -110,6 +126,7 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
 #pragma unroll
         for (int j = 0; j < 8; ++j) acc[h][j] = acc[h][j] * corr + p * vf[j];
       }
+    }
     }
   }
 
