@@ -29,6 +29,7 @@ SOURCES = [
     "rope_kv.hip",
     "decode_attn.hip",
     "prefill_attn.hip",
+    "prefill_mfma.hip",
     "sampling.hip",
     "skinny_gemm.hip",
 ]

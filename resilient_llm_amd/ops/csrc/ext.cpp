@@ -10,6 +10,8 @@
 
 #include "kernels.h"
 
+#include <vector>
+
 namespace {
 
 using at::Tensor;
@@ -234,12 +236,26 @@ Tensor prefill_attn_qkv(const Tensor& qkv, const Tensor& cu_seqlens,
   TORCH_CHECK(D == 64 || D == 128);
   const int n_seqs = int(cu_seqlens.numel()) - 1;
   Tensor out = at::empty({T, n_q * D}, qkv.options());
-  uint16_t* base = bf16_ptr(qkv);
   c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
-  rlli::launch_prefill_attn(
-      base, base + n_q * D, base + (n_q + n_kv) * D,
-      cu_seqlens.data_ptr<int32_t>(), bf16_ptr(out), n_seqs, T, int(n_q),
-      int(n_kv), D, float(scale), stride, stride, current_stream(qkv));
+  // chunk tables for the MFMA kernel (host-side; prefill is not captured)
+  Tensor cu_cpu = cu_seqlens.to(at::kCPU);
+  const int32_t* cu = cu_cpu.data_ptr<int32_t>();
+  std::vector<int32_t> t0s, starts, ends;
+  for (int s = 0; s < n_seqs; ++s)
+    for (int t = cu[s]; t < cu[s + 1]; t += 32) {
+      t0s.push_back(t);
+      starts.push_back(cu[s]);
+      ends.push_back(cu[s + 1]);
+    }
+  const int n_chunks = int(t0s.size());
+  auto iopt = at::TensorOptions().dtype(at::kInt);
+  Tensor t0_d = at::tensor(t0s, iopt).to(qkv.device(), /*nb=*/true);
+  Tensor st_d = at::tensor(starts, iopt).to(qkv.device(), true);
+  Tensor en_d = at::tensor(ends, iopt).to(qkv.device(), true);
+  rlli::launch_prefill_mfma(
+      bf16_ptr(qkv), t0_d.data_ptr<int32_t>(), st_d.data_ptr<int32_t>(),
+      en_d.data_ptr<int32_t>(), bf16_ptr(out), n_chunks, int(n_kv),
+      group, D, stride, float(scale), current_stream(qkv));
   return out;
 }
 

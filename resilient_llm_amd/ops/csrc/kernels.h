@@ -51,6 +51,14 @@ void launch_prefill_attn(
     int head_dim, float scale, int q_stride, int kv_stride,
     hipStream_t stream);
 
+// MFMA varlen causal prefill attention over a fused qkv tensor.
+void launch_prefill_mfma(const uint16_t* qkv, const int32_t* chunk_t0,
+                         const int32_t* chunk_seq_start,
+                         const int32_t* chunk_seq_end, uint16_t* out,
+                         int n_chunks, int n_kv_heads, int group,
+                         int head_dim, int qkv_stride, float scale,
+                         hipStream_t stream);
+
 // Skinny-M GEMM (decode projections): out[M,N] = x[M,K] @ W[N,K]^T.
 // ws is a [splitk, M, N] f32 workspace (unused when splitk == 1).
 void launch_skinny_gemm(const uint16_t* x, const uint16_t* w, float* ws,

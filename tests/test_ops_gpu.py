@@ -212,3 +212,30 @@ def test_skinny_linear_matches_blaslt(M, N, K):
     y = ops.skinny_linear(x, w)
     y_ref = torch.nn.functional.linear(x.float(), w.float())
     torch.testing.assert_close(y.float(), y_ref, atol=0.05, rtol=0.05)
+
+
+@pytest.mark.parametrize("n_q,n_kv,D,lens", [
+    (32, 8, 128, [5, 128, 63, 200]),     # 8B GQA shape, ragged
+    (8, 8, 128, [33]),                   # group 1
+    (16, 8, 128, [100, 17]),             # group 2
+    (16, 2, 128, [70, 31]),              # group 8 (two head-split WGs)
+    (8, 2, 64, [129, 64, 1]),            # tiny D=64
+])
+def test_prefill_mfma_vs_valu_and_ref(n_q, n_kv, D, lens):
+    """The MFMA fused prefill matches both the VALU v1 kernel and the
+    fp32 CPU reference on ragged causal batches."""
+    T = sum(lens)
+    width = (n_q + 2 * n_kv) * D
+    qkv = torch.randn(T, width, dtype=torch.bfloat16, device=DEV)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    q = qkv[:, :n_q * D].reshape(T, n_q, D).contiguous()
+    k = qkv[:, n_q * D:(n_q + n_kv) * D].reshape(T, n_kv, D).contiguous()
+    v = qkv[:, (n_q + n_kv) * D:].reshape(T, n_kv, D).contiguous()
+
+    out_mfma = ops.prefill_attn_qkv(qkv, cu, scale, n_q, n_kv, D)
+    out_valu = ops.prefill_attn(q, k, v, cu, scale)
+    assert_close_bf16(out_mfma.reshape(T, n_q, D).cpu(), out_valu.cpu())
+    out_ref = ref.prefill_attn(q.cpu(), k.cpu(), v.cpu(), cu.cpu(), scale)
+    assert_close_bf16(out_mfma.reshape(T, n_q, D).cpu(), out_ref)
