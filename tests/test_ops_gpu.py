@@ -185,8 +185,8 @@ def test_fused_qkv_ops_match_unfused():
     scale = 0.088
     out_f = ops.prefill_attn_qkv(qkv, cu, scale, n_q, n_kv, D)
     out_u = ops.prefill_attn(q, k, v, cu, scale)
-    torch.testing.assert_close(out_f.reshape(T, n_q, D).float(),
-                               out_u.float())
+    # MFMA vs VALU kernels: same math, different f32 summation orders
+    assert_close_bf16(out_f.reshape(T, n_q, D).cpu(), out_u.cpu())
 
     bt = torch.tensor([[0, 1], [2, 3]], dtype=torch.int32, device=DEV)
     lens = torch.tensor([5, 7], dtype=torch.int32, device=DEV)
