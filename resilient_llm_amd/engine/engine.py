@@ -15,6 +15,7 @@ from __future__ import annotations
 
 import collections
 import dataclasses
+import os
 import time
 from typing import Optional
 
@@ -22,6 +23,10 @@ import torch
 
 from .. import ops
 from .kv_cache import PagedKVCache
+
+# roctx phase markers for rocprofv3 --marker-trace:
+# torch.cuda.nvtx lowers to roctx on ROCm.  Opt-in: RLLI_ROCTX=1.
+_ROCTX = os.environ.get("RLLI_ROCTX") == "1"
 
 
 @dataclasses.dataclass
@@ -169,14 +174,23 @@ class LLMEngine:
         admitted = self._admit()
         if admitted:
             t0 = time.monotonic()
+            if _ROCTX:
+                torch.cuda.nvtx.range_push(
+                    f"prefill b{len(admitted)}")
             outs += self._prefill_step(admitted)
+            if _ROCTX:
+                torch.cuda.nvtx.range_pop()
             self.stats["prefill_steps"] += 1
             self.stats["prefill_time"] += time.monotonic() - t0
             self.stats["prefill_tokens"] += sum(len(s.prompt_ids) for s in admitted)
             return outs
         if self.running:
             t0 = time.monotonic()
+            if _ROCTX:
+                torch.cuda.nvtx.range_push(f"decode b{len(self.running)}")
             outs += self._decode_step()
+            if _ROCTX:
+                torch.cuda.nvtx.range_pop()
             self.stats["decode_steps"] += 1
             self.stats["decode_time"] += time.monotonic() - t0
             return outs

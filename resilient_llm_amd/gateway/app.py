@@ -77,6 +77,34 @@ class ConsumerLimiter:
             lim.reconcile(estimated, actual)
 
 
+class AccessPolicy:
+    """API-key -> alias ACL — the least-privilege IAM policy analogue
+    (reference iam/policy.json, C8 in SURVEY.md §2.1): which consumers
+    may invoke which model aliases, and who may touch admin routes.
+    Configured via an ``auth:`` section; absent -> allow everything
+    (matching the reference demos' cosmetic api keys)."""
+
+    def __init__(self, raw: dict | None) -> None:
+        raw = raw or {}
+        self.enforce = bool(raw.get("enforce", False))
+        self.keys: dict = raw.get("keys", {})
+        self.default = raw.get("default", {"allow": ["*"]})
+
+    def _rule(self, key: str) -> dict:
+        return self.keys.get(key, self.default)
+
+    def allows(self, key: str, alias: str) -> bool:
+        if not self.enforce:
+            return True
+        allow = self._rule(key).get("allow", [])
+        return "*" in allow or alias in allow
+
+    def is_admin(self, key: str) -> bool:
+        if not self.enforce:
+            return True
+        return bool(self._rule(key).get("admin", False))
+
+
 class GatewayApp:
     def __init__(self, config: Config, registry: WorkerRegistry,
                  router: Optional[Router] = None,
@@ -88,6 +116,7 @@ class GatewayApp:
         self.ledger = ledger or InvocationLedger(
             jsonl_path=config.cluster.ledger_path)
         self.consumers = ConsumerLimiter(config.raw.get("consumer_limits"))
+        self.policy = AccessPolicy(config.raw.get("auth"))
         self.health_interval_s = health_interval_s
         self._health_task: Optional[asyncio.Task] = None
         self.started_at = time.time()
@@ -145,6 +174,10 @@ class GatewayApp:
             return await self.health(req)
         if req.path in ("/models", "/v1/models") and req.method == "GET":
             return self.list_models()
+        if req.path.startswith("/admin/") and not self.policy.is_admin(
+                self._consumer_of(req)):
+            return Response.error(403, "admin access denied",
+                                  err_type="permission_error")
         if route == ("GET", "/admin/distribution"):
             return self.admin_distribution(req)
         if route == ("GET", "/admin/router"):
@@ -229,6 +262,10 @@ class GatewayApp:
             return Response.error(400, str(e), err_type="invalid_request_error")
         alias = body["model"]
         consumer = self._consumer_of(req)
+        if not self.policy.allows(consumer, alias):
+            return Response.error(403, f"api key not permitted for model "
+                                       f"{alias!r}", err_type="permission_error",
+                                  code="model_access_denied")
         prompt_est, total_est = self._estimate(body)
         if not self.consumers.try_acquire(consumer, total_est):
             return Response.error(429, f"consumer {consumer!r} rate limit exceeded",

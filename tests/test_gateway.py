@@ -219,3 +219,38 @@ def test_throttled_worker_typed_status():
             results = list(ex.map(one, range(4)))
         oks = [r for r in results if r[0] == "ok"]
         assert len(oks) >= 2
+
+
+def test_auth_policy_acl():
+    """IAM-analogue ACL (C8): per-key alias allow-lists + admin gating."""
+    from tests.gateway_harness import free_port, stub_config_dict
+    port = free_port()
+    cfg = stub_config_dict(port)
+    cfg["auth"] = {
+        "enforce": True,
+        "keys": {
+            "sk-a": {"allow": ["consumer-a-model"]},
+            "sk-admin": {"allow": ["*"], "admin": True},
+        },
+        "default": {"allow": ["llama-cris-demo"], "admin": False},
+    }
+    with run_gateway(cfg) as (client, *_):
+        from resilient_llm_amd.client import OpenAIClient
+        base = f"http://127.0.0.1:{port}"
+        a = OpenAIClient(base, api_key="sk-a")
+        a.chat.completions.create(model="consumer-a-model", messages=MSGS,
+                                  max_tokens=2)
+        with pytest.raises(APIError) as ei:
+            a.chat.completions.create(model="consumer-b-model", messages=MSGS)
+        assert ei.value.status == 403
+        # default key: open alias ok, admin denied
+        anon = OpenAIClient(base, api_key="sk-unknown")
+        anon.chat.completions.create(model="llama-cris-demo", messages=MSGS,
+                                     max_tokens=2)
+        with pytest.raises(APIError) as ei:
+            anon.distribution()
+        assert ei.value.status == 403
+        admin = OpenAIClient(base, api_key="sk-admin")
+        assert "distribution" in admin.distribution()
+        # health stays open
+        assert anon.health()["status"] == "ok"
