@@ -128,42 +128,36 @@ def start_gateway(loopth: LoopThread, n_gpus: int, model: str, port: int,
     return loopth.run(_start(), timeout=1200)
 
 
-async def one_request(session, url: str, prompt: str, results: list):
-    t0 = time.monotonic()
-    body = {"model": "bench-model",
-            "messages": [{"role": "user", "content": prompt}],
-            "max_tokens": OUTPUT_TOKENS, "temperature": 0.0}
-    try:
-        async with session.post(url, json=body) as resp:
-            data = await resp.json()
-            ok = resp.status == 200
-            usage = data.get("usage", {}) if ok else {}
-            results.append({
-                "ok": ok,
-                "latency": time.monotonic() - t0,
-                "completion_tokens": usage.get("completion_tokens", 0),
-                "prompt_tokens": usage.get("prompt_tokens", 0),
-            })
-    except Exception as e:
-        results.append({"ok": False, "latency": time.monotonic() - t0,
-                        "error": str(e), "completion_tokens": 0,
-                        "prompt_tokens": 0})
+class LoadGen:
+    """Driver for the loadgen subprocess (scripts/loadgen.py) — the
+    client side runs in its own process so rank 0's gateway loop does
+    not share a GIL with it at 8-GPU request rates."""
 
+    def __init__(self, port: int, concurrency: int, prompt_tokens: int,
+                 output_tokens: int) -> None:
+        import subprocess
+        self.proc = subprocess.Popen(
+            [sys.executable, os.path.join(REPO_ROOT, "scripts", "loadgen.py"),
+             "--port", str(port), "--concurrency", str(concurrency),
+             "--prompt-tokens", str(prompt_tokens),
+             "--output-tokens", str(output_tokens)],
+            stdin=subprocess.PIPE, stdout=subprocess.PIPE, text=True)
+        ready = self.proc.stdout.readline().strip()
+        assert ready == "READY", ready
 
-def run_load_round(loopth: LoopThread, port: int, prompts: list[str]) -> list:
-    import aiohttp
+    def round(self) -> dict:
+        self.proc.stdin.write("ROUND\n")
+        self.proc.stdin.flush()
+        line = self.proc.stdout.readline()
+        return json.loads(line)
 
-    async def _round():
-        results: list = []
-        url = f"http://127.0.0.1:{port}/chat/completions"
-        conn = aiohttp.TCPConnector(limit=0)
-        timeout = aiohttp.ClientTimeout(total=600)
-        async with aiohttp.ClientSession(connector=conn, timeout=timeout) as s:
-            await asyncio.gather(*[one_request(s, url, p, results)
-                                   for p in prompts])
-        return results
-
-    return loopth.run(_round(), timeout=900)
+    def close(self) -> None:
+        try:
+            self.proc.stdin.write("QUIT\n")
+            self.proc.stdin.flush()
+            self.proc.wait(timeout=10)
+        except Exception:
+            self.proc.kill()
 
 
 def main() -> None:
@@ -231,24 +225,22 @@ def main() -> None:
     sync()
 
     port = None
-    results_timed: list = []
+    loadgen = None
+    rounds_timed: list = []
     if rank == 0:
         port = free_port()
         start_gateway(loopth, n_gpus if distributed else 1, args.model, port,
                       run_dir)
-        log(f"gateway on :{port}, driving {conc_total} concurrent clients")
-
-    rng = random.Random(1234)
-
-    def round_prompts():
-        return [make_prompt(rng, args.prompt_tokens) for _ in range(conc_total)]
+        loadgen = LoadGen(port, conc_total, args.prompt_tokens,
+                          args.output_tokens)
+        log(f"gateway on :{port}, driving {conc_total} concurrent clients "
+            f"(loadgen subprocess)")
 
     # ---- warmup ----
     for w in range(args.warmup):
         if rank == 0:
-            res = run_load_round(loopth, port, round_prompts())
-            ok = sum(1 for r in res if r["ok"])
-            log(f"warmup {w}: {ok}/{len(res)} ok")
+            res = loadgen.round()
+            log(f"warmup {w}: {res['ok']}/{res['total']} ok")
         sync()
 
     # ---- timed ----
@@ -256,7 +248,7 @@ def main() -> None:
     t0 = time.monotonic()
     for k in range(args.steps):
         if rank == 0:
-            results_timed.extend(run_load_round(loopth, port, round_prompts()))
+            rounds_timed.append(loadgen.round())
     sync()
     elapsed = time.monotonic() - t0
     try:
@@ -270,14 +262,15 @@ def main() -> None:
         elapsed = float(t.item())
 
     if rank == 0:
-        n_req = len(results_timed)
-        ok = [r for r in results_timed if r["ok"]]
-        lats = sorted(r["latency"] for r in ok)
+        loadgen.close()
+        n_req = sum(r["total"] for r in rounds_timed)
+        n_ok = sum(r["ok"] for r in rounds_timed)
+        lats = sorted(l for r in rounds_timed for l in r["latencies"])
         p50 = lats[len(lats) // 2] if lats else None
         p99 = lats[min(len(lats) - 1, int(0.99 * len(lats)))] if lats else None
-        out_toks = sum(r["completion_tokens"] for r in ok)
-        total_toks = out_toks + sum(r["prompt_tokens"] for r in ok)
-        reqs_per_s = len(ok) / elapsed
+        out_toks = sum(r["completion_tokens"] for r in rounds_timed)
+        total_toks = out_toks + sum(r["prompt_tokens"] for r in rounds_timed)
+        reqs_per_s = n_ok / elapsed
         result = {
             "metric": "sustained reqs/sec + success-rate + p50 latency "
                       "@ 64 concurrent, Llama-3-8B",
@@ -300,7 +293,7 @@ def main() -> None:
                 "concurrency_per_gpu": args.concurrency,
                 "prompt_tokens": args.prompt_tokens,
                 "output_tokens": args.output_tokens,
-                "success_rate": round(len(ok) / n_req, 4) if n_req else None,
+                "success_rate": round(n_ok / n_req, 4) if n_req else None,
                 "p50_latency_s": round(p50, 3) if p50 else None,
                 "p99_latency_s": round(p99, 3) if p99 else None,
                 "output_tokens_per_s": round(out_toks / elapsed, 1),
