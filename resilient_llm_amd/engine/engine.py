@@ -70,6 +70,7 @@ class LLMEngine:
                  max_batch_size: int = 64,
                  max_prefill_tokens: int = 8192,
                  max_queue: int = 256,
+                 max_model_len: Optional[int] = None,
                  seed: int = 0) -> None:
         self.model = model
         self.kv = kv_cache
@@ -77,6 +78,8 @@ class LLMEngine:
         self.max_batch_size = max_batch_size
         self.max_prefill_tokens = max_prefill_tokens
         self.max_queue = max_queue
+        self.max_model_len = max_model_len or getattr(
+            model.config, "max_position", 8192)
         self.seed = seed
         self.eos_id = 2
         self.waiting: collections.deque[SeqState] = collections.deque()
@@ -107,6 +110,11 @@ class LLMEngine:
         if len(self.waiting) >= self.max_queue:
             raise CapacityExceeded(f"queue full ({self.max_queue})")
         seq = SeqState(req_id=req_id, prompt_ids=list(prompt_ids), params=params)
+        if len(seq.prompt_ids) + params.max_tokens > self.max_model_len:
+            raise CapacityExceeded(
+                f"prompt+max_tokens = "
+                f"{len(seq.prompt_ids) + params.max_tokens} exceeds "
+                f"max_model_len {self.max_model_len}")
         if seq.reserved_blocks_needed > self.kv.num_blocks:
             raise CapacityExceeded(
                 f"request needs {seq.reserved_blocks_needed} KV blocks, "
