@@ -118,6 +118,7 @@ class GatewayApp:
         self.consumers = ConsumerLimiter(config.raw.get("consumer_limits"))
         self.policy = AccessPolicy(config.raw.get("auth"))
         self.health_interval_s = health_interval_s
+        self.respawn_cooldown_s = 10.0
         self._health_task: Optional[asyncio.Task] = None
         self.started_at = time.time()
 
@@ -150,10 +151,14 @@ class GatewayApp:
         return out
 
     async def _health_loop(self) -> None:
-        """Per-worker heartbeat -> router health flags (X6)."""
+        """Per-worker heartbeat -> router health flags (X6) and elastic
+        recovery: a dead worker PROCESS with a respawn hook is restarted
+        (with a cooldown) and rejoins rotation once healthy
+        (SURVEY.md §5.3)."""
+        respawning: set = set()
         while True:
             await asyncio.sleep(self.health_interval_s)
-            for worker in self.registry.all().values():
+            for key, worker in self.registry.all().items():
                 try:
                     await asyncio.wait_for(worker.health(),
                                            timeout=self.health_interval_s)
@@ -162,6 +167,29 @@ class GatewayApp:
                     ok = False
                 for model_id in self._deployments_on(worker):
                     self.router.set_healthy(model_id, ok)
+                if (not ok and key not in respawning
+                        and getattr(worker, "respawn", None) is not None
+                        and getattr(worker, "proc", None) is not None
+                        and worker.proc.poll() is not None
+                        and time.monotonic() - worker.last_respawn
+                        > self.respawn_cooldown_s):
+                    respawning.add(key)
+
+                    async def _do(worker=worker, key=key):
+                        from ..utils.logging import log_with_timestamp
+                        try:
+                            log_with_timestamp(
+                                f"respawning dead worker {key}", "yellow")
+                            await worker.respawn_now()
+                            log_with_timestamp(
+                                f"worker {key} back online", "green")
+                        except Exception as e:
+                            log_with_timestamp(
+                                f"respawn of {key} failed: {e}", "red")
+                        finally:
+                            respawning.discard(key)
+
+                    asyncio.ensure_future(_do())
 
     # ------------------------------------------------------------ routing
     async def handle(self, req: Request) -> Response:
@@ -508,7 +536,11 @@ class GatewayApp:
         if device not in workers:
             return Response.error(404, f"no worker {device!r} "
                                        f"(have {sorted(workers)})")
-        await workers[device].inject_fault(mode)
+        try:
+            await workers[device].inject_fault(mode)
+        except WorkerError as e:
+            return Response.error(409, f"worker {device} cannot take fault "
+                                       f"command: {e}")
         return Response.json_response({"device": device, "mode": mode})
 
     def prometheus_metrics(self) -> Response:

@@ -88,6 +88,10 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
                                 max_batch=max_batch, use_graphs=use_graphs)
         client = RpcWorkerClient(f"gpu:{target}", {model_name}, sock)
         client.proc = proc
+        client.respawn = (lambda t=target, m=model_name, s=sock:
+                          spawn_gpu_worker(t, m, s, kv_gb=kv_gb,
+                                           max_batch=max_batch,
+                                           use_graphs=use_graphs))
         registry.register("gpu", target, client)
 
     for pool_name, model_name in sorted(pools.items()):

@@ -147,6 +147,8 @@ class RpcWorkerClient(Worker):
         self._reader_task: Optional[asyncio.Task] = None
         self._in_flight = 0
         self.proc = None              # set by the spawner (subprocess handle)
+        self.respawn = None           # callable -> new Popen (elastic recovery)
+        self.last_respawn = 0.0
 
     @property
     def in_flight(self) -> int:
@@ -258,6 +260,28 @@ class RpcWorkerClient(Worker):
         except (asyncio.TimeoutError, WorkerDead):
             if mode != "kill":
                 raise
+
+    async def respawn_now(self) -> bool:
+        """Elastic recovery: restart a dead worker
+        process and reconnect.  Returns True once serving again."""
+        import time as _time
+        if self.respawn is None:
+            return False
+        if self.proc is not None and self.proc.poll() is None:
+            return True
+        self.last_respawn = _time.monotonic()
+        if self._reader_task:
+            self._reader_task.cancel()
+            self._reader_task = None
+        self._connection_lost()
+        try:
+            import os as _os
+            _os.unlink(self.socket_path)
+        except OSError:
+            pass
+        self.proc = self.respawn()
+        await self.connect(timeout=900)
+        return True
 
     async def close(self) -> None:
         if self._reader_task:

@@ -1,0 +1,105 @@
+"""Elastic recovery: a dead worker process is respawned by the gateway
+health loop and rejoins rotation."""
+
+import asyncio
+import os
+import subprocess
+import sys
+import tempfile
+import threading
+import time
+
+import pytest
+
+from resilient_llm_amd.client import OpenAIClient, APIError
+from resilient_llm_amd.config import load_config
+from resilient_llm_amd.gateway.app import GatewayApp
+from resilient_llm_amd.gateway.http import HttpServer
+from resilient_llm_amd.workers.base import WorkerRegistry
+from resilient_llm_amd.workers.rpc import RpcWorkerClient
+from tests.gateway_harness import free_port
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+pytestmark = pytest.mark.timeout(180)
+
+
+def spawn_cpu_worker(sock):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    return subprocess.Popen(
+        [sys.executable, "-m", "resilient_llm_amd.workers.gpu_main",
+         "--device-label", "gpu:0", "--model", "tiny", "--socket", sock,
+         "--device", "cpu", "--num-blocks", "64"], env=env)
+
+
+def test_worker_respawn_after_kill():
+    sock = os.path.join(tempfile.mkdtemp(prefix="rlli-respawn-"), "w.sock")
+    port = free_port()
+    cfg = load_config(data={
+        "cluster": {"port": port},
+        "model_list": [
+            {"model_name": "tiny-serve",
+             "litellm_params": {"model": "gpu/0/tiny"},
+             "model_info": {"id": "gpu0/tiny"}}],
+        "router_settings": {"routing_strategy": "simple-shuffle"},
+    })
+    loop = asyncio.new_event_loop()
+    ready = threading.Event()
+    holder = {}
+
+    async def main():
+        holder["stop"] = asyncio.Event()
+        registry = WorkerRegistry()
+        client = RpcWorkerClient("gpu:0", {"tiny"}, sock)
+        client.proc = spawn_cpu_worker(sock)
+        client.respawn = lambda: spawn_cpu_worker(sock)
+        await client.connect(timeout=120)
+        registry.register("gpu", "0", client)
+        holder["client"] = client
+        app = GatewayApp(cfg, registry, health_interval_s=0.3)
+        app.respawn_cooldown_s = 0.0
+        server = HttpServer(app.handle, port=port)
+        await server.start()
+        await app.start_background()
+        ready.set()
+        await holder["stop"].wait()
+        await app.stop_background()
+        await server.stop()
+        await registry.close()
+
+    th = threading.Thread(target=lambda: loop.run_until_complete(main()),
+                          daemon=True)
+    th.start()
+    assert ready.wait(120)
+    try:
+        http = OpenAIClient(f"http://127.0.0.1:{port}")
+        msgs = [{"role": "user", "content": "hello"}]
+        r = http.chat.completions.create(model="tiny-serve", messages=msgs,
+                                         max_tokens=3)
+        assert r.usage.completion_tokens == 3
+
+        # real process kill
+        http.inject_fault("gpu:0", "kill")
+        first_proc = holder["client"].proc
+        t0 = time.time()
+        while first_proc.poll() is None and time.time() - t0 < 15:
+            time.sleep(0.1)
+        assert first_proc.poll() is not None
+
+        # the health loop respawns; serving resumes
+        deadline = time.time() + 90
+        ok = False
+        while time.time() < deadline:
+            try:
+                r = http.chat.completions.create(model="tiny-serve",
+                                                 messages=msgs, max_tokens=3)
+                ok = True
+                break
+            except APIError:
+                time.sleep(1.0)
+        assert ok, "worker never came back after respawn"
+        assert holder["client"].proc is not first_proc
+    finally:
+        loop.call_soon_threadsafe(holder["stop"].set)
+        th.join(timeout=15)
