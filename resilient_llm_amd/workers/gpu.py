@@ -31,8 +31,12 @@ def spawn_gpu_worker(device_index: str, model_name: str,
                      socket_path: str, kv_gb: float = 24.0,
                      max_batch: int = 64, use_graphs: bool = True,
                      extra_env: dict | None = None) -> subprocess.Popen:
+    """``device_index`` may be a sub-device replica like ``0.1``: several
+    worker processes co-located on physical GPU 0 — 288 GB of HBM3E holds
+    many 8B replicas, and a co-located warm replica makes single-GPU hot
+    failover real (SURVEY.md §5.4 warm-standby)."""
     env = dict(os.environ)
-    env["HIP_VISIBLE_DEVICES"] = str(device_index)
+    env["HIP_VISIBLE_DEVICES"] = str(device_index).split(".")[0]
     env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
     env.update(extra_env or {})
