@@ -28,6 +28,16 @@ import torch.nn.functional as F
 
 from .. import ops
 
+# Opt-in: route decode-shaped GEMMs (M<=64) through the custom skinny-M
+# MFMA kernel instead of hipBLASLt (A/B lever; see ops/csrc/skinny_gemm.hip)
+_USE_SKINNY = os.environ.get("RLLI_SKINNY") == "1"
+
+
+def _linear(x, w):
+    if _USE_SKINNY:
+        return ops.linear_auto(x, w)
+    return F.linear(x, w)
+
 
 @dataclasses.dataclass
 class LlamaConfig:
@@ -226,15 +236,15 @@ class LlamaForCausalLM:
             h = ops.rmsnorm(x, p[f"l{i}.ln1"], c.rms_eps)
         else:
             h = ops.rmsnorm_residual_(x, residual, p[f"l{i}.ln1"], c.rms_eps)
-        qkv = F.linear(h, p[f"l{i}.qkv"])
+        qkv = _linear(h, p[f"l{i}.qkv"])
         k_cache, v_cache = kv_cache.layer(i)
         ops.rope_kv_append_qkv_(qkv, positions, self.cos_sin,
                                 k_cache, v_cache, slot_mapping, self.n_heads)
         attn_out = attn_fn(i, qkv, k_cache, v_cache)   # [T, q_size]
-        h = self._all_reduce(F.linear(attn_out, p[f"l{i}.o"]))
+        h = self._all_reduce(_linear(attn_out, p[f"l{i}.o"]))
         h2 = ops.rmsnorm_residual_(h, residual, p[f"l{i}.ln2"], c.rms_eps)
-        gu = F.linear(h2, p[f"l{i}.gate_up"])
-        mlp = self._all_reduce(F.linear(ops.silu_mul(gu), p[f"l{i}.down"]))
+        gu = _linear(h2, p[f"l{i}.gate_up"])
+        mlp = self._all_reduce(_linear(ops.silu_mul(gu), p[f"l{i}.down"]))
         return mlp, residual
 
     def forward_prefill(self, input_ids: torch.Tensor, positions: torch.Tensor,
@@ -272,4 +282,4 @@ class LlamaForCausalLM:
                                   self.config.rms_eps)
         if last_idx is not None:
             h = h[last_idx]
-        return F.linear(h, p["lm_head"])   # bf16 logits feed ops.sample
+        return _linear(h, p["lm_head"])   # bf16 logits feed ops.sample
