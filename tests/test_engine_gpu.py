@@ -106,3 +106,17 @@ def test_graph_runner_matches_eager():
         graphed.add_request(rid, p, SamplingParams(max_tokens=12))
     graph_out = drain(graphed)
     assert eager_out == graph_out
+
+
+def test_long_context_beyond_graph_envelope():
+    """A sequence longer than the graph runner's block-table width falls
+    back to the eager decode path and stays correct."""
+    from resilient_llm_amd.engine.graph import install_graph_runner
+    cfg = get_config("llama-3-8b")
+    model = LlamaForCausalLM(cfg, device="cuda:0", dtype=torch.bfloat16, seed=1)
+    kv = PagedKVCache.for_model(cfg, 256, device="cuda:0")
+    e = LLMEngine(model, kv, max_batch_size=2, max_prefill_tokens=4096)
+    install_graph_runner(e, max_blocks_per_seq=8)   # envelope: 128 tokens
+    e.add_request("long", list(range(5, 1505)), SamplingParams(max_tokens=6))
+    outs = drain(e)
+    assert len(outs["long"]) == 6
