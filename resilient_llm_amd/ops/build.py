@@ -21,6 +21,10 @@ CSRC = os.path.join(HERE, "csrc")
 BUILD = os.path.join(HERE, "build")
 OUT_SO = os.path.join(HERE, "_rlli_hip.so")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
+# Host-side sanitizers for the extension: RLLI_ASAN=1 /
+# RLLI_TSAN=1 add -fsanitize to host code (device code unaffected).
+SAN_FLAGS = (["-fsanitize=address"] if os.environ.get("RLLI_ASAN") == "1" else
+             ["-fsanitize=thread"] if os.environ.get("RLLI_TSAN") == "1" else [])
 
 SOURCES = [
     "ext.cpp",
@@ -60,9 +64,9 @@ def _compile_one(src: str, includes: list[str], force: bool) -> str:
     obj = os.path.join(BUILD, src.replace("/", "_") + ".o")
     if not force and not _needs_build(src, obj):
         return obj
-    cmd = ["hipcc", f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-fPIC",
-           "-DNDEBUG", "-c", os.path.join(CSRC, src), "-o", obj,
-           f"-I{CSRC}"] + includes
+    cmd = (["hipcc", f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-fPIC",
+            "-DNDEBUG", "-c", os.path.join(CSRC, src), "-o", obj,
+            f"-I{CSRC}"] + SAN_FLAGS + includes)
     print("  hipcc -c", src, flush=True)
     r = subprocess.run(cmd, capture_output=True, text=True)
     if r.returncode != 0:
@@ -79,7 +83,7 @@ def build(force: bool = False, verbose: bool = True) -> str:
     if (force or not os.path.exists(OUT_SO)
             or any(os.path.getmtime(o) > os.path.getmtime(OUT_SO) for o in objs)):
         cmd = (["hipcc", f"--offload-arch={ARCH}", "-shared", "-fPIC",
-                "-o", OUT_SO] + objs + libs)
+                "-o", OUT_SO] + SAN_FLAGS + objs + libs)
         if verbose:
             print("  hipcc -shared ->", os.path.basename(OUT_SO), flush=True)
         r = subprocess.run(cmd, capture_output=True, text=True)
