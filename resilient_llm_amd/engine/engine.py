@@ -71,6 +71,7 @@ class LLMEngine:
                  max_prefill_tokens: int = 8192,
                  max_queue: int = 256,
                  max_model_len: Optional[int] = None,
+                 chunk_size: int = 2048,
                  seed: int = 0) -> None:
         self.model = model
         self.kv = kv_cache
@@ -83,7 +84,9 @@ class LLMEngine:
         self.seed = seed
         self.eos_id = 2
         self.waiting: collections.deque[SeqState] = collections.deque()
+        self.prefilling: list[SeqState] = []   # admitted, prompt not fully cached
         self.running: list[SeqState] = []
+        self.chunk_size = chunk_size
         self.step_count = 0
         self.block_size = kv_cache.block_size
         self._aborted: set[str] = set()
@@ -99,10 +102,11 @@ class LLMEngine:
     # ------------------------------------------------------------- admin
     @property
     def n_active(self) -> int:
-        return len(self.waiting) + len(self.running)
+        return len(self.waiting) + len(self.prefilling) + len(self.running)
 
     def has_work(self) -> bool:
-        return bool(self.waiting or self.running or self._pending)
+        return bool(self.waiting or self.prefilling or self.running
+                    or self._pending)
 
     def add_request(self, req_id: str, prompt_ids: list,
                     params: Optional[SamplingParams] = None) -> None:
@@ -129,7 +133,8 @@ class LLMEngine:
         admitted: list[SeqState] = []
         tokens = 0
         while (self.waiting
-               and len(self.running) + len(admitted) < self.max_batch_size):
+               and len(self.running) + len(self.prefilling) + len(admitted)
+               < self.max_batch_size):
             seq = self.waiting[0]
             if seq.req_id in self._aborted:
                 self.waiting.popleft()
@@ -179,18 +184,19 @@ class LLMEngine:
         self.step_count += 1
         outs = self._process_pending()
         self._drop_aborted()
-        admitted = self._admit()
-        if admitted:
+        self.prefilling.extend(self._admit())
+        if self.prefilling:
             t0 = time.monotonic()
             if _ROCTX:
                 torch.cuda.nvtx.range_push(
-                    f"prefill b{len(admitted)}")
-            outs += self._prefill_step(admitted)
+                    f"mixed d{len(self.running)}+p{len(self.prefilling)}")
+            n_chunk_tokens, new_outs = self._mixed_step()
+            outs += new_outs
             if _ROCTX:
                 torch.cuda.nvtx.range_pop()
             self.stats["prefill_steps"] += 1
             self.stats["prefill_time"] += time.monotonic() - t0
-            self.stats["prefill_tokens"] += sum(len(s.prompt_ids) for s in admitted)
+            self.stats["prefill_tokens"] += n_chunk_tokens
             return outs
         if self.running:
             t0 = time.monotonic()
@@ -230,17 +236,18 @@ class LLMEngine:
     def _drop_aborted(self) -> None:
         if not self._aborted:
             return
-        keep = []
-        for seq in self.running:
-            if seq.req_id in self._aborted:
-                self.kv.free(seq.blocks)
-                seq.blocks = []
-                self._aborted.discard(seq.req_id)
-            else:
-                keep.append(seq)
-        if len(keep) != len(self.running):
-            self._batch_dirty = True
-        self.running = keep
+        for attr in ("running", "prefilling"):
+            keep = []
+            for seq in getattr(self, attr):
+                if seq.req_id in self._aborted:
+                    self.kv.free(seq.blocks)
+                    seq.blocks = []
+                    self._aborted.discard(seq.req_id)
+                else:
+                    keep.append(seq)
+            if len(keep) != len(getattr(self, attr)):
+                self._batch_dirty = True
+            setattr(self, attr, keep)
 
     def _sample(self, logits: torch.Tensor, seqs: list[SeqState]) -> list[int]:
         temps = torch.tensor([s.params.temperature for s in seqs],
@@ -249,30 +256,91 @@ class LLMEngine:
         toks = ops.sample(logits, temps, seed)
         return toks.tolist()
 
-    def _prefill_step(self, admitted: list[SeqState]) -> list[StepOutput]:
+    def _mixed_step(self) -> tuple[int, list[StepOutput]]:
+        """One forward over [decode rows | prefill-chunk rows]: running
+        sequences decode while prefilling sequences advance by up to
+        chunk_size prompt tokens each (budget max_prefill_tokens/step).
+        Chunk rows attend over the paged cache (prefix + the chunk
+        itself), so a long prompt never blocks the decode batch for more
+        than one bounded step."""
         dev = self.device
-        input_ids, positions, slots, cu = [], [], [], [0]
-        for seq in admitted:
-            plen = len(seq.prompt_ids)
-            input_ids.extend(seq.prompt_ids)
-            positions.extend(range(plen))
-            slots.extend(self._slot(seq, p) for p in range(plen))
-            cu.append(cu[-1] + plen)
-        logits = self.model.forward_prefill(
-            torch.tensor(input_ids, dtype=torch.int32, device=dev),
-            torch.tensor(positions, dtype=torch.int32, device=dev),
-            self.kv,
-            torch.tensor(slots, dtype=torch.int32, device=dev),
-            torch.tensor(cu, dtype=torch.int32, device=dev))
-        tokens = self._sample(logits, admitted)
+        seqs_d = list(self.running)
+        B_d = len(seqs_d)
+        input_ids = [s.output_ids[-1] for s in seqs_d]
+        positions = [s.n_cached for s in seqs_d]
+        slots = [self._slot(s, s.n_cached) for s in seqs_d]
+        seq_lens_d = [s.n_cached + 1 for s in seqs_d]
+
+        budget = self.max_prefill_tokens
+        chunk_plan: list = []
+        for seq in self.prefilling:
+            if budget <= 0:
+                break
+            n = min(len(seq.prompt_ids) - seq.n_cached, self.chunk_size, budget)
+            if n <= 0:
+                continue
+            chunk_plan.append((seq, seq.n_cached, n))
+            budget -= n
+
+        c_row0, c_pos0, c_nrows, c_btrow = [], [], [], []
+        bt_rows: list[list[int]] = []
+        sample_idx = list(range(B_d))
+        sampled_seqs = list(seqs_d)
+        row = B_d
+        for seq, start, n in chunk_plan:
+            btr = len(bt_rows)
+            bt_rows.append(seq.blocks)
+            for q0 in range(0, n, 32):
+                c_row0.append(row + q0)
+                c_pos0.append(start + q0)
+                c_nrows.append(min(32, n - q0))
+                c_btrow.append(btr)
+            input_ids.extend(seq.prompt_ids[start:start + n])
+            positions.extend(range(start, start + n))
+            slots.extend(self._slot(seq, p) for p in range(start, start + n))
+            if start + n == len(seq.prompt_ids):
+                sample_idx.append(row + n - 1)
+                sampled_seqs.append(seq)
+            row += n
+
+        def i32(x):
+            return torch.tensor(x, dtype=torch.int32, device=dev)
+
+        def bt_tensor(rows):
+            w = max((len(r) for r in rows), default=1)
+            t = torch.zeros((max(len(rows), 1), w), dtype=torch.int32)
+            for i, r in enumerate(rows):
+                t[i, :len(r)] = torch.tensor(r, dtype=torch.int32)
+            return t.to(dev)
+
+        logits = self.model.forward_mixed(
+            i32(input_ids), i32(positions), self.kv, i32(slots), B_d,
+            bt_tensor([s.blocks for s in seqs_d]) if B_d else None,
+            i32(seq_lens_d) if B_d else None,
+            i32(c_row0), i32(c_pos0), i32(c_nrows), i32(c_btrow),
+            bt_tensor(bt_rows),
+            torch.tensor(sample_idx, dtype=torch.long, device=dev))
+        tokens = self._sample(logits, sampled_seqs)
+
+        n_chunk_tokens = 0
+        for seq, start, n in chunk_plan:
+            seq.n_cached = start + n
+            n_chunk_tokens += n
         outs: list[StepOutput] = []
-        for seq, tok in zip(admitted, tokens):
-            seq.n_cached = len(seq.prompt_ids)
+        still_running: list[SeqState] = []
+        decode_set = {id(s) for s in seqs_d}
+        for seq, tok in zip(sampled_seqs, tokens):
+            if id(seq) in decode_set:
+                seq.n_cached += 1
             seq.output_ids.append(tok)
             if not self._finish(seq, outs, tok):
-                self.running.append(seq)
-                self._batch_dirty = True
-        return outs
+                still_running.append(seq)
+        self.running = still_running
+        done = {id(s) for s, start, n in chunk_plan
+                if start + n == len(s.prompt_ids)}
+        self.prefilling = [s for s in self.prefilling if id(s) not in done]
+        self._batch_dirty = True
+        return n_chunk_tokens, outs
 
     def _rebuild_batch(self) -> None:
         """(Re)build persistent device tensors for the running batch.

@@ -259,6 +259,27 @@ class LlamaForCausalLM:
         return self._forward(input_ids, positions, kv_cache, slot_mapping,
                              attn, last_idx=cu_seqlens[1:].long() - 1)
 
+    def forward_mixed(self, input_ids: torch.Tensor, positions: torch.Tensor,
+                      kv_cache, slot_mapping: torch.Tensor, n_decode: int,
+                      block_tables_dec, seq_lens_dec,
+                      chunk_row0, chunk_pos0, chunk_nrows, chunk_btrow,
+                      block_tables_pre, sample_idx) -> torch.Tensor:
+        """Mixed batch: rows [0, n_decode) are decode tokens, the rest
+        are prefill-chunk tokens attending over the paged cache
+        (chunked prefill).  Returns logits for ``sample_idx`` rows."""
+        def attn(i, qkv, k_cache, v_cache):
+            out = ops.prefill_paged_attn(
+                qkv, k_cache, v_cache, chunk_row0, chunk_pos0, chunk_nrows,
+                chunk_btrow, block_tables_pre, self.scale, self.n_heads)
+            if n_decode:
+                out_d = ops.decode_attn_qkv(
+                    qkv[:n_decode], k_cache, v_cache, block_tables_dec,
+                    seq_lens_dec, self.scale, self.n_heads)
+                out[:n_decode] = out_d
+            return out
+        return self._forward(input_ids, positions, kv_cache, slot_mapping,
+                             attn, last_idx=sample_idx)
+
     def forward_decode(self, input_ids: torch.Tensor, positions: torch.Tensor,
                        kv_cache, slot_mapping: torch.Tensor,
                        block_tables: torch.Tensor,

@@ -162,6 +162,19 @@ def sample(logits: torch.Tensor, temperatures: torch.Tensor,
     return ref.sample(logits, temperatures, seed)
 
 
+def prefill_paged_attn(qkv, k_cache, v_cache, chunk_row0, chunk_pos0,
+                       chunk_nrows, chunk_btrow, block_tables,
+                       scale: float, n_q: int) -> torch.Tensor:
+    if qkv.is_cuda:
+        _gpu()
+        return torch.ops.rlli.prefill_paged_attn(
+            qkv, k_cache, v_cache, chunk_row0, chunk_pos0, chunk_nrows,
+            chunk_btrow, block_tables, scale, n_q)
+    return ref.prefill_paged_attn(qkv, k_cache, v_cache, chunk_row0,
+                                  chunk_pos0, chunk_nrows, chunk_btrow,
+                                  block_tables, scale, n_q)
+
+
 def skinny_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """Decode-projection GEMM (M<=64): custom MFMA weight-streaming
     kernel on GPU, F.linear on CPU."""

@@ -34,6 +34,7 @@ SOURCES = [
     "decode_attn.hip",
     "prefill_attn.hip",
     "prefill_mfma.hip",
+    "prefill_paged.hip",
     "sampling.hip",
     "skinny_gemm.hip",
 ]

@@ -259,6 +259,45 @@ Tensor prefill_attn_qkv(const Tensor& qkv, const Tensor& cu_seqlens,
   return out;
 }
 
+// ------------------------------------------------------ prefill (paged)
+Tensor prefill_paged_attn(const Tensor& qkv, const Tensor& k_cache,
+                          const Tensor& v_cache, const Tensor& chunk_row0,
+                          const Tensor& chunk_pos0, const Tensor& chunk_nrows,
+                          const Tensor& chunk_btrow,
+                          const Tensor& block_tables, double scale,
+                          int64_t n_q) {
+  check_bf16_contig(qkv, "qkv");
+  check_bf16_contig(k_cache, "k_cache");
+  check_bf16_contig(v_cache, "v_cache");
+  for (const Tensor* t : {&chunk_row0, &chunk_pos0, &chunk_nrows,
+                          &chunk_btrow, &block_tables}) {
+    TORCH_CHECK(t->scalar_type() == at::kInt && t->is_contiguous() &&
+                t->is_cuda(), "chunk/table tensors must be contiguous "
+                "int32 on the GPU");
+  }
+  const int T = int(qkv.size(0));
+  const int n_kv = int(k_cache.size(1));
+  const int block_size = int(k_cache.size(2));
+  const int D = int(k_cache.size(3));
+  const int stride = int(qkv.size(1));
+  TORCH_CHECK(stride >= n_q * D, "qkv too narrow");
+  const int group = int(n_q) / n_kv;
+  TORCH_CHECK(group == 1 || group == 2 || group == 4 || group == 8);
+  TORCH_CHECK(D == 64 || D == 128);
+  const int n_chunks = int(chunk_row0.numel());
+  const int max_blocks = int(block_tables.size(1));
+  Tensor out = at::empty({T, n_q * D}, qkv.options());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
+  rlli::launch_prefill_paged(
+      bf16_ptr(qkv), bf16_ptr(k_cache), bf16_ptr(v_cache),
+      chunk_row0.data_ptr<int32_t>(), chunk_pos0.data_ptr<int32_t>(),
+      chunk_nrows.data_ptr<int32_t>(), chunk_btrow.data_ptr<int32_t>(),
+      block_tables.data_ptr<int32_t>(), bf16_ptr(out), n_chunks, n_kv,
+      group, D, stride, max_blocks, block_size, float(scale),
+      current_stream(qkv));
+  return out;
+}
+
 // --------------------------------------------------------- skinny_linear
 Tensor skinny_linear(const Tensor& x, const Tensor& w) {
   check_bf16_contig(x, "x");
@@ -327,6 +366,10 @@ TORCH_LIBRARY(rlli, m) {
         "float scale) -> Tensor");
   m.def("sample(Tensor logits, Tensor temperatures, int seed) -> Tensor");
   m.def("skinny_linear(Tensor x, Tensor w) -> Tensor");
+  m.def("prefill_paged_attn(Tensor qkv, Tensor k_cache, Tensor v_cache, "
+        "Tensor chunk_row0, Tensor chunk_pos0, Tensor chunk_nrows, "
+        "Tensor chunk_btrow, Tensor block_tables, float scale, int n_q) "
+        "-> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(rlli, CUDA, m) {
@@ -341,4 +384,5 @@ TORCH_LIBRARY_IMPL(rlli, CUDA, m) {
   m.impl("prefill_attn", &prefill_attn);
   m.impl("sample", &sample);
   m.impl("skinny_linear", &skinny_linear);
+  m.impl("prefill_paged_attn", &prefill_paged_attn);
 }

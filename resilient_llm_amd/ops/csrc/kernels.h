@@ -59,6 +59,17 @@ void launch_prefill_mfma(const uint16_t* qkv, const int32_t* chunk_t0,
                          int head_dim, int qkv_stride, float scale,
                          hipStream_t stream);
 
+// MFMA chunked-prefill attention over the paged cache (mixed batches).
+void launch_prefill_paged(const uint16_t* qkv, const uint16_t* k_cache,
+                          const uint16_t* v_cache, const int32_t* chunk_row0,
+                          const int32_t* chunk_pos0,
+                          const int32_t* chunk_nrows,
+                          const int32_t* chunk_btrow,
+                          const int32_t* block_tables, uint16_t* out,
+                          int n_chunks, int n_kv_heads, int group,
+                          int head_dim, int qkv_stride, int max_blocks,
+                          int block_size, float scale, hipStream_t stream);
+
 // Skinny-M GEMM (decode projections): out[M,N] = x[M,K] @ W[N,K]^T.
 // ws is a [splitk, M, N] f32 workspace (unused when splitk == 1).
 void launch_skinny_gemm(const uint16_t* x, const uint16_t* w, float* ws,

@@ -118,6 +118,36 @@ def prefill_attn(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return out
 
 
+def prefill_paged_attn(qkv: torch.Tensor, k_cache: torch.Tensor,
+                       v_cache: torch.Tensor, chunk_row0, chunk_pos0,
+                       chunk_nrows, chunk_btrow, block_tables,
+                       scale: float, n_q: int) -> torch.Tensor:
+    """Chunked prefill over the paged cache: each chunk's q rows attend
+    causally over their sequence's cached prefix (which includes the
+    chunk itself — rope_kv_append ran first)."""
+    T = qkv.shape[0]
+    n_kv, block_size, D = k_cache.shape[1], k_cache.shape[2], k_cache.shape[3]
+    group = n_q // n_kv
+    out = torch.zeros(T, n_q * D, dtype=qkv.dtype)
+    for c in range(len(chunk_row0)):
+        row0 = int(chunk_row0[c]); pos0 = int(chunk_pos0[c])
+        nrows = int(chunk_nrows[c]); btr = int(chunk_btrow[c])
+        kv_hi = pos0 + nrows
+        n_blocks = (kv_hi + block_size - 1) // block_size
+        blocks = block_tables[btr, :n_blocks].long()
+        k = k_cache[blocks].transpose(0, 1).reshape(n_kv, -1, D)[:, :kv_hi].float()
+        v = v_cache[blocks].transpose(0, 1).reshape(n_kv, -1, D)[:, :kv_hi].float()
+        q = qkv[row0:row0 + nrows, :n_q * D].reshape(nrows, n_kv, group, D).float()
+        s = torch.einsum("qhgd,htd->hgqt", q, k) * scale
+        qpos = torch.arange(pos0, pos0 + nrows).unsqueeze(1)
+        kvpos = torch.arange(kv_hi).unsqueeze(0)
+        s.masked_fill_((kvpos > qpos), float("-inf"))
+        p = torch.softmax(s, dim=-1)
+        o = torch.einsum("hgqt,htd->qhgd", p, v)
+        out[row0:row0 + nrows] = o.reshape(nrows, n_q * D).to(qkv.dtype)
+    return out
+
+
 def sample(logits: torch.Tensor, temperatures: torch.Tensor,
            seed: int = 0) -> torch.Tensor:
     """Greedy rows match the kernel exactly; stochastic rows use torch's

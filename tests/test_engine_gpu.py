@@ -120,3 +120,16 @@ def test_long_context_beyond_graph_envelope():
     e.add_request("long", list(range(5, 1505)), SamplingParams(max_tokens=6))
     outs = drain(e)
     assert len(outs["long"]) == 6
+
+
+def test_chunked_prefill_gpu_matches_unchunked():
+    e1 = make_engine("tiny-128", num_blocks=256, max_batch_size=4, seed=5)
+    e1.add_request("a", list(range(10, 150)), SamplingParams(max_tokens=6))
+    want = drain(e1)["a"]
+    cfg = get_config("tiny-128")
+    model = LlamaForCausalLM(cfg, device="cuda:0", dtype=torch.bfloat16, seed=5)
+    kv = PagedKVCache.for_model(cfg, 256, device="cuda:0")
+    e2 = LLMEngine(model, kv, max_batch_size=4, chunk_size=48)
+    e2.add_request("a", list(range(10, 150)), SamplingParams(max_tokens=6))
+    got = drain(e2)["a"]
+    assert got == want

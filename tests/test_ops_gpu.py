@@ -239,3 +239,37 @@ def test_prefill_mfma_vs_valu_and_ref(n_q, n_kv, D, lens):
     assert_close_bf16(out_mfma.reshape(T, n_q, D).cpu(), out_valu.cpu())
     out_ref = ref.prefill_attn(q.cpu(), k.cpu(), v.cpu(), cu.cpu(), scale)
     assert_close_bf16(out_mfma.reshape(T, n_q, D).cpu(), out_ref)
+
+
+@pytest.mark.parametrize("n_q,n_kv,D", [(32, 8, 128), (16, 2, 128), (8, 2, 64)])
+def test_prefill_paged_vs_ref(n_q, n_kv, D):
+    """Paged chunked-prefill attention vs the fp32 reference: chunks of a
+    sequence attending over cached prefix + chunk."""
+    torch.manual_seed(3)
+    bs = 16
+    width = (n_q + 2 * n_kv) * D
+    # sequence of 75 tokens split as chunks [0:40) and [40:75); cache
+    # pre-filled for the whole 75 (rope_kv_append ran first in real flow)
+    L = 75
+    nb = (L + bs - 1) // bs
+    kc = torch.randn(nb + 3, n_kv, bs, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    bt = torch.tensor([[4, 1, 3, 0, 2]], dtype=torch.int32, device=DEV)
+    qkv = torch.randn(L, width, dtype=torch.bfloat16, device=DEV)
+
+    def run(row0s, pos0s, nrowss, btrows):
+        return ops.prefill_paged_attn(
+            qkv, kc, vc,
+            torch.tensor(row0s, dtype=torch.int32, device=DEV),
+            torch.tensor(pos0s, dtype=torch.int32, device=DEV),
+            torch.tensor(nrowss, dtype=torch.int32, device=DEV),
+            torch.tensor(btrows, dtype=torch.int32, device=DEV),
+            bt, 0.088, n_q)
+
+    # engine splits chunks into <=32-row kernel chunks
+    out = run([0, 32, 40, 72], [0, 32, 40, 72], [32, 8, 32, 3], [0, 0, 0, 0])
+    out_ref = ref.prefill_paged_attn(
+        qkv.cpu(), kc.cpu(), vc.cpu(),
+        [0, 32, 40, 72], [0, 32, 40, 72], [32, 8, 32, 3], [0, 0, 0, 0],
+        bt.cpu(), 0.088, n_q)
+    assert_close_bf16(out.cpu(), out_ref)
