@@ -313,6 +313,18 @@ class LLMEngine:
                 t[i, :len(r)] = torch.tensor(r, dtype=torch.int32)
             return t.to(dev)
 
+        # fast path: no decode rows and every chunk is a whole prompt ->
+        # classic batched varlen prefill over in-batch K/V (no paged
+        # reads, no row-copy) — this is the bench/burst-arrival shape
+        if B_d == 0 and all(start == 0 and start + n == len(s.prompt_ids)
+                            for s, start, n in chunk_plan):
+            cu = [0]
+            for s, _, n in chunk_plan:
+                cu.append(cu[-1] + n)
+            logits = self.model.forward_prefill(
+                i32(input_ids), i32(positions), self.kv, i32(slots), i32(cu))
+            return self._mixed_finish(chunk_plan, sampled_seqs, logits)
+
         logits = self.model.forward_mixed(
             i32(input_ids), i32(positions), self.kv, i32(slots), B_d,
             bt_tensor([s.blocks for s in seqs_d]) if B_d else None,
@@ -320,22 +332,27 @@ class LLMEngine:
             i32(c_row0), i32(c_pos0), i32(c_nrows), i32(c_btrow),
             bt_tensor(bt_rows),
             torch.tensor(sample_idx, dtype=torch.long, device=dev))
-        tokens = self._sample(logits, sampled_seqs)
+        return self._mixed_finish(chunk_plan, sampled_seqs, logits,
+                                  decode_seqs=seqs_d)
 
+    def _mixed_finish(self, chunk_plan, sampled_seqs, logits,
+                      decode_seqs=()) -> tuple[int, list[StepOutput]]:
+        tokens = self._sample(logits, sampled_seqs)
         n_chunk_tokens = 0
         for seq, start, n in chunk_plan:
             seq.n_cached = start + n
             n_chunk_tokens += n
         outs: list[StepOutput] = []
         still_running: list[SeqState] = []
-        decode_set = {id(s) for s in seqs_d}
+        decode_set = {id(s) for s in decode_seqs}
         for seq, tok in zip(sampled_seqs, tokens):
             if id(seq) in decode_set:
                 seq.n_cached += 1
             seq.output_ids.append(tok)
             if not self._finish(seq, outs, tok):
                 still_running.append(seq)
-        self.running = still_running
+        self.running = [s for s in self.running
+                        if id(s) not in decode_set] + still_running
         done = {id(s) for s, start, n in chunk_plan
                 if start + n == len(s.prompt_ids)}
         self.prefilling = [s for s in self.prefilling if id(s) not in done]
