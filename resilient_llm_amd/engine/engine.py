@@ -72,6 +72,7 @@ class LLMEngine:
                  max_queue: int = 256,
                  max_model_len: Optional[int] = None,
                  chunk_size: int = 2048,
+                 admission_window_s: float = 0.0,
                  seed: int = 0) -> None:
         self.model = model
         self.kv = kv_cache
@@ -87,6 +88,10 @@ class LLMEngine:
         self.prefilling: list[SeqState] = []   # admitted, prompt not fully cached
         self.running: list[SeqState] = []
         self.chunk_size = chunk_size
+        # coalesce request bursts into one prefill batch: hold admission
+        # while arrivals are still landing (MUST be 0 under TP lockstep —
+        # wall-clock decisions would diverge across ranks)
+        self.admission_window_s = admission_window_s
         self.step_count = 0
         self.block_size = kv_cache.block_size
         self._aborted: set[str] = set()
@@ -131,6 +136,12 @@ class LLMEngine:
     # --------------------------------------------------------- scheduling
     def _admit(self) -> list[SeqState]:
         admitted: list[SeqState] = []
+        if (self.waiting and self.admission_window_s > 0
+                and time.monotonic() - self.waiting[-1].arrived_at
+                < self.admission_window_s
+                and len(self.waiting) + len(self.running)
+                + len(self.prefilling) < self.max_batch_size):
+            return admitted
         tokens = 0
         while (self.waiting
                and len(self.running) + len(self.prefilling) + len(admitted)

@@ -80,7 +80,9 @@ class EngineWorker(Worker):
             kv.k = kv.k.to(dtype)
             kv.v = kv.v.to(dtype)
         self.engine = LLMEngine(self.model, kv, max_batch_size=max_batch_size,
-                                max_queue=max_queue, seed=seed)
+                                max_queue=max_queue, seed=seed,
+                                admission_window_s=0.0 if tp_control is not None
+                                else 0.004)
         if use_graphs and torch_device != "cpu":
             from ..engine.graph import install_graph_runner
             install_graph_runner(self.engine)
