@@ -253,7 +253,10 @@ def main() -> None:
     elapsed = time.monotonic() - t0
     try:
         h = loopth.run(worker.health(), timeout=30)
-        log(f"rank {rank} engine stats: {h.get('engine_stats')}")
+        stats = h.get("engine_stats", {})
+        admits = stats.pop("admit_events", [])
+        log(f"rank {rank} engine stats: {stats}")
+        log(f"rank {rank} admits (t, n, still_waiting): {admits[:40]}")
     except Exception:
         pass
     if distributed:

@@ -100,7 +100,8 @@ class LLMEngine:
         self._pinned = None
         self.stats = {"prefill_steps": 0, "decode_steps": 0,
                       "prefill_time": 0.0, "decode_time": 0.0,
-                      "prefill_tokens": 0, "decode_tokens": 0}
+                      "prefill_tokens": 0, "decode_tokens": 0,
+                      "admit_events": []}
         # decode graph runner installed by engine/graph.py (GPU only)
         self.graph_runner = None
 
@@ -158,6 +159,9 @@ class LLMEngine:
             seq.blocks = self.kv.allocate(seq.reserved_blocks_needed)
             tokens += len(seq.prompt_ids)
             admitted.append(self.waiting.popleft())
+        if admitted and len(self.stats["admit_events"]) < 1000:
+            self.stats["admit_events"].append(
+                (round(time.monotonic(), 4), len(admitted), len(self.waiting)))
         return admitted
 
     def _slot(self, seq: SeqState, pos: int) -> int:

@@ -55,6 +55,8 @@ class EngineWorker(Worker):
                  kv_gb: float = DEFAULT_KV_GB,
                  max_batch_size: int = 64,
                  max_queue: int = 256,
+                 max_prefill_tokens: int = 16384,
+                 chunk_size: int = 512,
                  dtype: Optional[torch.dtype] = None,
                  num_blocks: Optional[int] = None,
                  use_graphs: bool = False,
@@ -81,8 +83,10 @@ class EngineWorker(Worker):
             kv.v = kv.v.to(dtype)
         self.engine = LLMEngine(self.model, kv, max_batch_size=max_batch_size,
                                 max_queue=max_queue, seed=seed,
+                                max_prefill_tokens=max_prefill_tokens,
+                                chunk_size=chunk_size,
                                 admission_window_s=0.0 if tp_control is not None
-                                else 0.004)
+                                else 0.006)
         if use_graphs and torch_device != "cpu":
             from ..engine.graph import install_graph_runner
             install_graph_runner(self.engine)
