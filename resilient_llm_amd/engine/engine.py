@@ -286,7 +286,11 @@ class LLMEngine:
         slots = [self._slot(s, s.n_cached) for s in seqs_d]
         seq_lens_d = [s.n_cached + 1 for s in seqs_d]
 
-        budget = self.max_prefill_tokens
+        # bound a mixed step's prefill work when decode rows ride along,
+        # so running sequences keep a reasonable token cadence during
+        # other requests' prefills; pure-prefill steps use the full budget
+        budget = self.max_prefill_tokens if not seqs_d \
+            else min(self.max_prefill_tokens, 4096)
         chunk_plan: list = []
         for seq in self.prefilling:
             if budget <= 0:

@@ -97,8 +97,11 @@ class EngineWorker(Worker):
         self._req_counter = 0
         self._in_flight = 0
         self._lock = threading.Lock()
-        # req_id -> (asyncio queue, loop)
-        self._sinks: dict[str, tuple[asyncio.Queue, asyncio.AbstractEventLoop]] = {}
+        # req_id -> [queue, loop, mode, buffer, first_token_t]
+        # mode "stream": one queue item per token; mode "final": tokens
+        # buffer engine-side, ONE queue item at finish (64x fewer loop
+        # wakeups for non-streaming traffic)
+        self._sinks: dict[str, list] = {}
         self._work_event = threading.Event()
         self._stop = False
         self._thread = None
@@ -133,15 +136,24 @@ class EngineWorker(Worker):
             except Exception as e:  # engine-level failure -> fail all in flight
                 self._broadcast_error(e)
                 continue
+            now = time.monotonic()
             for out in outputs:
                 sink = self._sinks.get(out.req_id)
                 if sink is None:
                     continue
-                q, loop = sink
-                loop.call_soon_threadsafe(q.put_nowait, out)
+                q, loop, mode, buf, first_t = sink
+                if sink[4] is None:
+                    sink[4] = now
+                if mode == "stream":
+                    loop.call_soon_threadsafe(q.put_nowait, out)
+                else:
+                    buf.append(out)
+                    if out.finished:
+                        loop.call_soon_threadsafe(q.put_nowait,
+                                                  (list(buf), sink[4]))
 
     def _broadcast_error(self, e: Exception) -> None:
-        for q, loop in list(self._sinks.values()):
+        for q, loop, *_ in list(self._sinks.values()):
             loop.call_soon_threadsafe(q.put_nowait, e)
 
     # ----------------------------------------------------------- helpers
@@ -151,7 +163,8 @@ class EngineWorker(Worker):
         if self.fault_mode == "error":
             raise WorkerError(f"{self.device} injected fault: error")
 
-    def _enqueue(self, req: GenerationRequest) -> tuple[str, asyncio.Queue, int]:
+    def _enqueue(self, req: GenerationRequest,
+                 mode: str = "stream") -> tuple[str, asyncio.Queue, int]:
         self._check_fault()
         prompt_ids = self.tokenizer.encode(render_prompt(req.messages))
         params = SamplingParams(max_tokens=req.max_tokens,
@@ -161,7 +174,7 @@ class EngineWorker(Worker):
         rid = f"{req.request_id}-{self._req_counter}"
         q: asyncio.Queue = asyncio.Queue()
         loop = asyncio.get_running_loop()
-        self._sinks[rid] = (q, loop)
+        self._sinks[rid] = [q, loop, mode, [], None]
         try:
             if self.tp_control is not None:
                 # leader: validate locally, then lockstep-broadcast the op
@@ -196,30 +209,23 @@ class EngineWorker(Worker):
     # --------------------------------------------------------------- API
     async def generate(self, req: GenerationRequest) -> GenerationResult:
         t0 = time.monotonic()
-        rid, q, n_prompt = self._enqueue(req)
+        rid, q, n_prompt = self._enqueue(req, mode="final")
         self._in_flight += 1
         try:
-            token_ids: list[int] = []
-            ttft = None
-            finish = "stop"
-            while True:
-                out = await q.get()
-                self._check_fault()
-                if isinstance(out, Exception):
-                    raise WorkerError(f"engine error: {out}") from out
-                if ttft is None:
-                    ttft = (time.monotonic() - t0) * 1000.0
-                token_ids.append(out.token_id)
-                if out.finished:
-                    finish = out.finish_reason or "stop"
-                    break
+            got = await q.get()
+            self._check_fault()
+            if isinstance(got, Exception):
+                raise WorkerError(f"engine error: {got}") from got
+            outs, first_t = got
+            token_ids = [o.token_id for o in outs]
+            finish = outs[-1].finish_reason or "stop"
             self.total_served += 1
             return GenerationResult(
                 text=self.tokenizer.decode(token_ids),
                 prompt_tokens=n_prompt,
                 completion_tokens=len(token_ids),
                 finish_reason=finish,
-                ttft_ms=ttft)
+                ttft_ms=max(0.0, (first_t - t0) * 1000.0))
         finally:
             self._in_flight -= 1
             self._cleanup(rid)
