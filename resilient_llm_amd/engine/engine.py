@@ -16,6 +16,7 @@ from __future__ import annotations
 import collections
 import dataclasses
 import os
+import threading
 import time
 from typing import Optional
 
@@ -84,6 +85,11 @@ class LLMEngine:
             model.config, "max_position", 8192)
         self.seed = seed
         self.eos_id = 2
+        # waiting/abort are the ONLY cross-thread surfaces: guarded by a
+        # small queue lock so enqueue NEVER waits on a running step (a
+        # coarse lock here serialized arrivals behind 100ms prefills and
+        # fragmented bursts into waves)
+        self._queue_lock = threading.Lock()
         self.waiting: collections.deque[SeqState] = collections.deque()
         self.prefilling: list[SeqState] = []   # admitted, prompt not fully cached
         self.running: list[SeqState] = []
@@ -129,13 +135,19 @@ class LLMEngine:
             raise CapacityExceeded(
                 f"request needs {seq.reserved_blocks_needed} KV blocks, "
                 f"cache has {self.kv.num_blocks}")
-        self.waiting.append(seq)
+        with self._queue_lock:
+            self.waiting.append(seq)
 
     def abort(self, req_id: str) -> None:
-        self._aborted.add(req_id)
+        with self._queue_lock:
+            self._aborted.add(req_id)
 
     # --------------------------------------------------------- scheduling
     def _admit(self) -> list[SeqState]:
+        with self._queue_lock:
+            return self._admit_locked()
+
+    def _admit_locked(self) -> list[SeqState]:
         admitted: list[SeqState] = []
         if (self.waiting and self.admission_window_s > 0
                 and time.monotonic() - self.waiting[-1].arrived_at
@@ -286,11 +298,7 @@ class LLMEngine:
         slots = [self._slot(s, s.n_cached) for s in seqs_d]
         seq_lens_d = [s.n_cached + 1 for s in seqs_d]
 
-        # bound a mixed step's prefill work when decode rows ride along,
-        # so running sequences keep a reasonable token cadence during
-        # other requests' prefills; pure-prefill steps use the full budget
-        budget = self.max_prefill_tokens if not seqs_d \
-            else min(self.max_prefill_tokens, 4096)
+        budget = self.max_prefill_tokens
         chunk_plan: list = []
         for seq in self.prefilling:
             if budget <= 0:

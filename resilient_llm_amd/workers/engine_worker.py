@@ -118,21 +118,18 @@ class EngineWorker(Worker):
             if self.fault_mode == "hang":
                 time.sleep(0.05)
                 continue
-            with self._lock:
-                has_work = self.engine.has_work() or (
-                    self.tp_control is not None and self.tp_control.pending())
+            has_work = self.engine.has_work() or (
+                self.tp_control is not None and self.tp_control.pending())
             if not has_work:
                 self._work_event.wait(timeout=0.01)
                 self._work_event.clear()
                 continue
             if self.tp_control is not None:
-                with self._lock:
-                    self.tp_control.sync(self.engine)
-                    if not self.engine.has_work():
-                        continue
+                self.tp_control.sync(self.engine)
+                if not self.engine.has_work():
+                    continue
             try:
-                with self._lock:
-                    outputs = self.engine.step()
+                outputs = self.engine.step()
             except Exception as e:  # engine-level failure -> fail all in flight
                 self._broadcast_error(e)
                 continue
@@ -191,8 +188,7 @@ class EngineWorker(Worker):
                       "temperature": params.temperature,
                       "seed": params.seed}))
             else:
-                with self._lock:
-                    self.engine.add_request(rid, prompt_ids, params)
+                self.engine.add_request(rid, prompt_ids, params)
         except CapacityExceeded as e:
             del self._sinks[rid]
             raise WorkerThrottled(str(e)) from e
@@ -233,8 +229,7 @@ class EngineWorker(Worker):
                 self.tp_control.submit(("abort", rid))
                 self._work_event.set()
             else:
-                with self._lock:
-                    self.engine.abort(rid)
+                self.engine.abort(rid)
 
     async def _stream_impl(self, req: GenerationRequest) -> AsyncIterator[GenerationChunk]:
         rid, q, n_prompt = self._enqueue(req)
@@ -264,8 +259,7 @@ class EngineWorker(Worker):
                 self.tp_control.submit(("abort", rid))
                 self._work_event.set()
             else:
-                with self._lock:
-                    self.engine.abort(rid)
+                self.engine.abort(rid)
 
     def generate_stream(self, req: GenerationRequest) -> AsyncIterator[GenerationChunk]:
         return self._stream_impl(req)
