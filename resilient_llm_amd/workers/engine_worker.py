@@ -96,7 +96,6 @@ class EngineWorker(Worker):
         self.total_served = 0
         self._req_counter = 0
         self._in_flight = 0
-        self._lock = threading.Lock()
         # req_id -> [queue, loop, mode, buffer, first_token_t]
         # mode "stream": one queue item per token; mode "final": tokens
         # buffer engine-side, ONE queue item at finish (64x fewer loop
