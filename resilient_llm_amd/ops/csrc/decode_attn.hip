@@ -81,26 +81,28 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
   const int n_blocks = (len + block_size - 1) / block_size;
   constexpr int TPG = 16 / GPW >= 1 ? 16 / GPW : 1;   // tokens per group @bs16
   const int tok_per_grp = block_size / GPW;
-  for (int b = wave; b < n_blocks; b += n_waves) {
+
+  auto issue_block = [&](int b, uint4 (&kraw)[TPG], uint4 (&vraw)[TPG]) {
     const int phys = block_table[int64_t(seq) * max_blocks + b];
     const int64_t kv_base =
         (int64_t(phys) * n_kv_heads + kvh) * block_size * D;
-    for (int bi = 0; bi < tok_per_grp; bi += TPG) {
-    uint4 kraw[TPG], vraw[TPG];
 #pragma unroll
     for (int i = 0; i < TPG; ++i) {
-      if (bi + i < tok_per_grp) {
-        const int tok = group + GPW * (bi + i);
+      if (i < tok_per_grp) {
+        const int tok = group + GPW * i;
         kraw[i] = *reinterpret_cast<const uint4*>(
             k_cache + kv_base + int64_t(tok) * D + d0);
         vraw[i] = *reinterpret_cast<const uint4*>(
             v_cache + kv_base + int64_t(tok) * D + d0);
       }
     }
+  };
+  auto consume_block = [&](int b, const uint4 (&kraw)[TPG],
+                           const uint4 (&vraw)[TPG]) {
 #pragma unroll
     for (int i = 0; i < TPG; ++i) {
-      if (bi + i >= tok_per_grp) break;
-      const int tok = group + GPW * (bi + i);  // token within the block
+      if (i >= tok_per_grp) break;
+      const int tok = group + GPW * i;         // token within the block
       if (b * block_size + tok >= len) continue;   // group-uniform tail
       bf16x8 kv, vv;
       kv.u = kraw[i];
@@ -127,6 +129,60 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
         for (int j = 0; j < 8; ++j) acc[h][j] = acc[h][j] * corr + p * vf[j];
       }
     }
+  };
+
+  if (tok_per_grp <= TPG) {
+    // depth-2 software pipeline over cache blocks: the NEXT block's
+    // K/V loads are in flight while this block computes (named A/B
+    // buffers, statically indexed — guide §5.4 rule 20)
+    uint4 kA[TPG], vA[TPG], kB[TPG], vB[TPG];
+    int b = wave;
+    if (b < n_blocks) issue_block(b, kA, vA);
+    while (b < n_blocks) {
+      if (b + n_waves < n_blocks) issue_block(b + n_waves, kB, vB);
+      consume_block(b, kA, vA);
+      b += n_waves;
+      if (b >= n_blocks) break;
+      if (b + n_waves < n_blocks) issue_block(b + n_waves, kA, vA);
+      consume_block(b, kB, vB);
+      b += n_waves;
+    }
+  } else {
+    // oversized cache blocks: plain per-sub-batch loop
+    for (int b = wave; b < n_blocks; b += n_waves) {
+      const int phys = block_table[int64_t(seq) * max_blocks + b];
+      const int64_t kv_base =
+          (int64_t(phys) * n_kv_heads + kvh) * block_size * D;
+      for (int bi = 0; bi < tok_per_grp; ++bi) {
+        const int tok = group + GPW * bi;
+        if (b * block_size + tok >= len) continue;
+        bf16x8 kv, vv;
+        kv.u = *reinterpret_cast<const uint4*>(
+            k_cache + kv_base + int64_t(tok) * D + d0);
+        vv.u = *reinterpret_cast<const uint4*>(
+            v_cache + kv_base + int64_t(tok) * D + d0);
+        float kf[8], vf[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) kf[j] = bf16_to_f32(kv.s[j]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vf[j] = bf16_to_f32(vv.s[j]);
+#pragma unroll
+        for (int h = 0; h < GROUP; ++h) {
+          float s = 0.f;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) s += qv[h][j] * kf[j];
+          s = group_sum<GW>(s);
+          s *= scale;
+          const float m_new = fmaxf(m[h], s);
+          const float corr = __expf(m[h] - m_new);
+          const float p = __expf(s - m_new);
+          m[h] = m_new;
+          l[h] = l[h] * corr + p;
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            acc[h][j] = acc[h][j] * corr + p * vf[j];
+        }
+      }
     }
   }
 

@@ -7,6 +7,7 @@
 #include <ATen/ATen.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
+#include <c10/hip/HIPGraphsC10Utils.h>
 
 #include "kernels.h"
 
@@ -19,6 +20,16 @@ using at::Tensor;
 hipStream_t current_stream(const Tensor& t) {
   return at::hip::getCurrentHIPStreamMasqueradingAsCUDA(
       t.device().index()).stream();
+}
+
+// after-launch error check (skipped inside hipGraph capture, where the
+// runtime reports capture-sequencing pseudo-errors)
+void check_launch(const char* op) {
+  if (c10::hip::currentStreamCaptureStatusMayInitCtx() !=
+      c10::hip::CaptureStatus::None)
+    return;
+  hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, op, " launch failed: ", hipGetErrorString(e));
 }
 
 uint16_t* bf16_ptr(const Tensor& t) {
@@ -43,6 +54,7 @@ Tensor rmsnorm(const Tensor& x, const Tensor& w, double eps) {
   c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   rlli::launch_rmsnorm(bf16_ptr(x), nullptr, bf16_ptr(w), bf16_ptr(y), rows,
                        dim, float(eps), current_stream(x));
+  check_launch("rmsnorm");
   return y;
 }
 
@@ -59,6 +71,7 @@ Tensor rmsnorm_residual_(const Tensor& x, Tensor residual, const Tensor& w,
   c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   rlli::launch_rmsnorm(bf16_ptr(x), bf16_ptr(residual), bf16_ptr(w),
                        bf16_ptr(y), rows, dim, float(eps), current_stream(x));
+  check_launch("rmsnorm_residual_");
   return y;
 }
 
@@ -75,6 +88,7 @@ Tensor silu_mul(const Tensor& gate_up) {
   c10::hip::HIPGuardMasqueradingAsCUDA guard(gate_up.device());
   rlli::launch_silu_mul(bf16_ptr(gate_up), bf16_ptr(y), rows, inter,
                         current_stream(gate_up));
+  check_launch("silu_mul");
   return y;
 }
 
@@ -107,6 +121,7 @@ void rope_kv_append_(Tensor q, Tensor k, const Tensor& v,
       positions.data_ptr<int32_t>(), cos_sin.data_ptr<float>(),
       bf16_ptr(k_cache), bf16_ptr(v_cache), slot_mapping.data_ptr<int32_t>(),
       tokens, n_q, n_kv, D, block_size, n_q * D, n_kv * D, current_stream(q));
+  check_launch("rope_kv_append_");
 }
 
 // Fused-QKV form: qkv is the raw [T, (n_q + 2*n_kv) * D] GEMM output;
@@ -134,6 +149,7 @@ void rope_kv_append_qkv_(Tensor qkv, const Tensor& positions,
       bf16_ptr(k_cache), bf16_ptr(v_cache), slot_mapping.data_ptr<int32_t>(),
       tokens, int(n_q), n_kv, D, block_size, stride, stride,
       current_stream(qkv));
+  check_launch("rope_kv_append_qkv_");
 }
 
 // ---------------------------------------------------------- decode_attn
@@ -166,6 +182,7 @@ Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
       block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       bf16_ptr(out), batch, n_q, n_kv, D, block_size, max_blocks,
       float(scale), n_q * D, current_stream(q));
+  check_launch("decode_attn");
   return out;
 }
 
@@ -191,6 +208,7 @@ Tensor decode_attn_qkv(const Tensor& qkv, const Tensor& k_cache,
       block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       bf16_ptr(out), batch, int(n_q), n_kv, D, block_size, max_blocks,
       float(scale), stride, current_stream(qkv));
+  check_launch("decode_attn_qkv");
   return out;
 }
 
@@ -217,6 +235,7 @@ Tensor prefill_attn(const Tensor& q, const Tensor& k, const Tensor& v,
       bf16_ptr(q), bf16_ptr(k), bf16_ptr(v), cu_seqlens.data_ptr<int32_t>(),
       bf16_ptr(out), n_seqs, T, n_q, n_kv, D, float(scale),
       n_q * D, n_kv * D, current_stream(q));
+  check_launch("prefill_attn");
   return out;
 }
 
@@ -256,6 +275,7 @@ Tensor prefill_attn_qkv(const Tensor& qkv, const Tensor& cu_seqlens,
       bf16_ptr(qkv), t0_d.data_ptr<int32_t>(), st_d.data_ptr<int32_t>(),
       en_d.data_ptr<int32_t>(), bf16_ptr(out), n_chunks, int(n_kv),
       group, D, stride, float(scale), current_stream(qkv));
+  check_launch("prefill_attn_qkv");
   return out;
 }
 
@@ -295,6 +315,7 @@ Tensor prefill_paged_attn(const Tensor& qkv, const Tensor& k_cache,
       block_tables.data_ptr<int32_t>(), bf16_ptr(out), n_chunks, n_kv,
       group, D, stride, max_blocks, block_size, float(scale),
       current_stream(qkv));
+  check_launch("prefill_paged_attn");
   return out;
 }
 
@@ -325,6 +346,7 @@ Tensor skinny_linear(const Tensor& x, const Tensor& w) {
   c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   rlli::launch_skinny_gemm(bf16_ptr(x), bf16_ptr(w), ws_ptr, bf16_ptr(out),
                            M, N, K, splitk, current_stream(x));
+  check_launch("skinny_linear");
   return out;
 }
 
@@ -341,6 +363,7 @@ Tensor sample(const Tensor& logits, const Tensor& temperatures, int64_t seed) {
   rlli::launch_sample(bf16_ptr(logits), temperatures.data_ptr<float>(),
                       uint64_t(seed), out.data_ptr<int32_t>(), batch, vocab,
                       current_stream(logits));
+  check_launch("sample");
   return out;
 }
 
