@@ -37,6 +37,7 @@ SOURCES = [
     "prefill_paged.hip",
     "sampling.hip",
     "skinny_gemm.hip",
+    "streamprobe.hip",
 ]
 HEADERS = ["common.h", "kernels.h"]
 

@@ -350,6 +350,17 @@ Tensor skinny_linear(const Tensor& x, const Tensor& w) {
   return out;
 }
 
+Tensor stream_probe(const Tensor& w, int64_t splitk) {
+  check_bf16_contig(w, "w");
+  Tensor sink = at::zeros({256}, w.options().dtype(at::kFloat));
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(w.device());
+  rlli::launch_stream_probe(bf16_ptr(w), sink.data_ptr<float>(),
+                            int(w.size(0)), int(w.size(1)), int(splitk),
+                            current_stream(w));
+  check_launch("stream_probe");
+  return sink;
+}
+
 // --------------------------------------------------------------- sample
 Tensor sample(const Tensor& logits, const Tensor& temperatures, int64_t seed) {
   check_bf16_contig(logits, "logits");
@@ -389,6 +400,7 @@ TORCH_LIBRARY(rlli, m) {
         "float scale) -> Tensor");
   m.def("sample(Tensor logits, Tensor temperatures, int seed) -> Tensor");
   m.def("skinny_linear(Tensor x, Tensor w) -> Tensor");
+  m.def("stream_probe(Tensor w, int splitk) -> Tensor");
   m.def("prefill_paged_attn(Tensor qkv, Tensor k_cache, Tensor v_cache, "
         "Tensor chunk_row0, Tensor chunk_pos0, Tensor chunk_nrows, "
         "Tensor chunk_btrow, Tensor block_tables, float scale, int n_q) "
@@ -407,5 +419,6 @@ TORCH_LIBRARY_IMPL(rlli, CUDA, m) {
   m.impl("prefill_attn", &prefill_attn);
   m.impl("sample", &sample);
   m.impl("skinny_linear", &skinny_linear);
+  m.impl("stream_probe", &stream_probe);
   m.impl("prefill_paged_attn", &prefill_paged_attn);
 }

@@ -76,6 +76,10 @@ void launch_skinny_gemm(const uint16_t* x, const uint16_t* w, float* ws,
                         uint16_t* out, int M, int N, int K, int splitk,
                         hipStream_t stream);
 
+// Diagnostic: pure nt-stream of W at skinny_gemm geometry.
+void launch_stream_probe(const uint16_t* w, float* sink, int N, int K,
+                         int splitk, hipStream_t stream);
+
 // Fused sampling: greedy argmax when temperature[i] == 0, else Gumbel-max
 // sampling of softmax(logits / temperature[i]) with an in-kernel counter
 // hash RNG keyed on (seed, row, column) — no 32 MB random tensor per step.

@@ -19,7 +19,20 @@ def timeit(fn, iters=50):
     torch.cuda.synchronize()
     return (time.monotonic() - t0) / iters * 1e6
 
+def probe():
+    print("— pure nt-stream ceiling at skinny geometry —")
+    for name, M, N, K in SHAPES:
+        w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
+        gb = N * K * 2 / 1e9
+        for splitk in (1, 2, 4, 8):
+            if (N // 64) * splitk > 2048 and splitk > 1:
+                continue
+            t = timeit(lambda: torch.ops.rlli.stream_probe(w, splitk))
+            print(f"  {name:8s} splitk={splitk}: {t:7.1f} us "
+                  f"({gb/t*1e6/1e3:5.2f} TB/s)")
+
 def main():
+    probe()
     tot_s = tot_b = 0
     for name, M, N, K in SHAPES:
         x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
