@@ -20,6 +20,7 @@ def timeit(fn, iters=50):
     return (time.monotonic() - t0) / iters * 1e6
 
 def probe():
+    ops.load_extension(required=True)
     print("— pure nt-stream ceiling at skinny geometry —")
     for name, M, N, K in SHAPES:
         w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
