@@ -217,6 +217,9 @@ class GatewayApp:
                                                "cooldown_time": self.config.router.cooldown_time,
                                                "fallbacks": self.config.router.fallbacks,
                                            }})
+        if route == ("GET", "/admin/requests"):
+            n = min(int(req.query.get("n", "50")), 1000)
+            return Response.json_response({"requests": self.ledger.recent(n)})
         if route == ("POST", "/admin/fault"):
             return await self.admin_fault(req)
         if route == ("GET", "/metrics"):

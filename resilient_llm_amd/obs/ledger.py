@@ -131,6 +131,12 @@ class InvocationLedger:
             "reqs_per_sec": (n_ok / span) if span else None,
         }
 
+    def recent(self, n: int = 50) -> list[dict]:
+        import dataclasses as _dc
+        with self._lock:
+            recs = list(self._records)[-n:]
+        return [_dc.asdict(r) for r in reversed(recs)]
+
     def close(self) -> None:
         if self._jsonl_file is not None:
             self._jsonl_file.close()

@@ -254,3 +254,14 @@ def test_auth_policy_acl():
         assert "distribution" in admin.distribution()
         # health stays open
         assert anon.health()["status"] == "ok"
+
+
+def test_admin_requests_trace():
+    with run_gateway() as (client, *_):
+        client.chat.completions.create(model="llama-cris-demo", messages=MSGS,
+                                       max_tokens=3)
+        recs = client._get("/admin/requests", {"n": 10})["requests"]
+        assert len(recs) == 1
+        r = recs[0]
+        assert r["alias"] == "llama-cris-demo" and r["status"] == "ok"
+        assert r["completion_tokens"] == 3 and r["latency_ms"] > 0
