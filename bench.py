@@ -254,8 +254,8 @@ def main() -> None:
     try:
         h = loopth.run(worker.health(), timeout=30)
         stats = h.get("engine_stats", {})
-        admits = stats.pop("admit_events", [])
         log(f"rank {rank} engine stats: {stats}")
+        admits = worker.engine.stats.get("admit_events", [])
         log(f"rank {rank} admits (t, n, still_waiting): {admits[:40]}")
     except Exception:
         pass

@@ -274,7 +274,8 @@ class EngineWorker(Worker):
                 "running": len(self.engine.running),
                 "kv_free_blocks": self.engine.kv.free_blocks,
                 "total_served": self.total_served,
-                "engine_stats": dict(self.engine.stats)}
+                "engine_stats": {k: v for k, v in self.engine.stats.items()
+                                 if k != "admit_events"}}
 
     async def inject_fault(self, mode: str) -> None:
         assert mode in ("none", "kill", "hang", "error")
