@@ -406,76 +406,90 @@ class GatewayApp:
             last_err: Optional[Exception] = None
             attempts = app._max_attempts(alias)
             comp_id = f"chatcmpl-{uuid.uuid4().hex[:16]}"
-            for attempt in range(attempts):
-                if ticket is None:
-                    try:
-                        ticket = app.router.acquire(alias, total_est, exclude=exclude)
-                    except NoDeploymentAvailable as e:
-                        last_err = e
-                        break
-                greq = app._gen_request(body, consumer, ticket, stream=True)
-                try:
-                    worker = app._worker_for(ticket)
-                except WorkerError as e:
-                    last_err = e
-                    app.router.fail(ticket)
-                    exclude.add(id(ticket.state))
-                    ticket = None
-                    continue
-                model_id = ticket.deployment.model_id
-                n_emitted = 0
-                completion_tokens = 0
-                finish = None
-                try:
-                    it = worker.generate_stream(greq).__aiter__()
-                    if attempt > 0:
-                        yield (f": failover -> {model_id}\n\n").encode()
-                    while True:
+            try:
+                for attempt in range(attempts):
+                    if ticket is None:
                         try:
-                            chunk = await asyncio.wait_for(
-                                it.__anext__(), timeout=STREAM_CHUNK_TIMEOUT_S)
-                        except StopAsyncIteration:
+                            ticket = app.router.acquire(alias, total_est, exclude=exclude)
+                        except NoDeploymentAvailable as e:
+                            last_err = e
                             break
-                        completion_tokens += 1
-                        if completion_tokens <= sent_tokens:
+                    greq = app._gen_request(body, consumer, ticket, stream=True)
+                    try:
+                        worker = app._worker_for(ticket)
+                    except WorkerError as e:
+                        last_err = e
+                        app.router.fail(ticket)
+                        exclude.add(id(ticket.state))
+                        ticket = None
+                        continue
+                    model_id = ticket.deployment.model_id
+                    n_emitted = 0
+                    completion_tokens = 0
+                    finish = None
+                    try:
+                        it = worker.generate_stream(greq).__aiter__()
+                        if attempt > 0:
+                            yield (f": failover -> {model_id}\n\n").encode()
+                        while True:
+                            try:
+                                chunk = await asyncio.wait_for(
+                                    it.__anext__(), timeout=STREAM_CHUNK_TIMEOUT_S)
+                            except StopAsyncIteration:
+                                break
+                            completion_tokens += 1
+                            if completion_tokens <= sent_tokens:
+                                finish = chunk.finish_reason or finish
+                                continue   # replay skip after failover
+                            n_emitted += 1
                             finish = chunk.finish_reason or finish
-                            continue   # replay skip after failover
-                        n_emitted += 1
-                        finish = chunk.finish_reason or finish
-                        delta = ({"role": "assistant", "content": chunk.text}
-                                 if sent_tokens + n_emitted == 1
-                                 else {"content": chunk.text})
-                        evt = {
-                            "id": comp_id, "object": "chat.completion.chunk",
-                            "created": int(time.time()), "model": model_id,
-                            "choices": [{"index": 0, "delta": delta,
-                                         "finish_reason": chunk.finish_reason}],
-                        }
-                        yield f"data: {json.dumps(evt)}\n\n".encode()
-                    sent_tokens += n_emitted
-                    result = GenerationResult(
-                        text="", prompt_tokens=total_est, completion_tokens=sent_tokens,
-                        finish_reason=finish or "stop")
+                            delta = ({"role": "assistant", "content": chunk.text}
+                                     if sent_tokens + n_emitted == 1
+                                     else {"content": chunk.text})
+                            evt = {
+                                "id": comp_id, "object": "chat.completion.chunk",
+                                "created": int(time.time()), "model": model_id,
+                                "choices": [{"index": 0, "delta": delta,
+                                             "finish_reason": chunk.finish_reason}],
+                            }
+                            yield f"data: {json.dumps(evt)}\n\n".encode()
+                        sent_tokens += n_emitted
+                        result = GenerationResult(
+                            text="", prompt_tokens=total_est, completion_tokens=sent_tokens,
+                            finish_reason=finish or "stop")
+                        app.router.complete(ticket, actual_tokens=None)
+                        app._record(ticket, greq, consumer, "ok", t0, result, worker.device)
+                        yield b"data: [DONE]\n\n"
+                        return
+                    except (WorkerError, asyncio.TimeoutError) as e:
+                        last_err = e
+                        sent_tokens += n_emitted
+                        throttled = isinstance(e, WorkerThrottled)
+                        app.router.fail(ticket, charge=throttled)
+                        app._record(ticket, greq, consumer,
+                                    "throttled" if throttled else "error", t0)
+                        exclude.add(id(ticket.state))
+                        ticket = None
+                        continue
+                from ..utils.logging import sanitize_error
+                err_evt = {"error": {"message": f"stream failed: "
+                                                f"{sanitize_error(last_err) if last_err else 'exhausted'}",
+                                     "type": "api_error"}}
+                yield f"data: {json.dumps(err_evt)}\n\n".encode()
+                yield b"data: [DONE]\n\n"
+            finally:
+                # a client disconnect closes this generator at a yield:
+                # settle the open ticket so in-flight counts and the
+                # ledger stay truthful (X12)
+                if ticket is not None and not ticket.done:
                     app.router.complete(ticket, actual_tokens=None)
-                    app._record(ticket, greq, consumer, "ok", t0, result, worker.device)
-                    yield b"data: [DONE]\n\n"
-                    return
-                except (WorkerError, asyncio.TimeoutError) as e:
-                    last_err = e
-                    sent_tokens += n_emitted
-                    throttled = isinstance(e, WorkerThrottled)
-                    app.router.fail(ticket, charge=throttled)
-                    app._record(ticket, greq, consumer,
-                                "throttled" if throttled else "error", t0)
-                    exclude.add(id(ticket.state))
-                    ticket = None
-                    continue
-            from ..utils.logging import sanitize_error
-            err_evt = {"error": {"message": f"stream failed: "
-                                            f"{sanitize_error(last_err) if last_err else 'exhausted'}",
-                                 "type": "api_error"}}
-            yield f"data: {json.dumps(err_evt)}\n\n".encode()
-            yield b"data: [DONE]\n\n"
+                    app.ledger.record(InvocationRecord(
+                        ts=time.time(), request_id=comp_id, alias=alias,
+                        model_id=ticket.deployment.model_id, device="",
+                        consumer=consumer, status="cancelled",
+                        is_fallback=ticket.is_fallback,
+                        completion_tokens=sent_tokens,
+                        latency_ms=(time.monotonic() - t0) * 1000.0))
 
         headers = {
             "x-gateway-model-id": first_ticket.deployment.model_id,

@@ -152,8 +152,20 @@ class HttpServer:
                 writer.write(f"{len(chunk):x}\r\n".encode() + chunk + b"\r\n")
                 await writer.drain()
         finally:
-            writer.write(b"0\r\n\r\n")
-            await writer.drain()
+            # on client disconnect the generator is abandoned mid-yield;
+            # close it NOW so its cleanup (router ticket settlement) runs
+            # promptly instead of at interpreter shutdown
+            aclose = getattr(resp.body_iter, "aclose", None)
+            if aclose is not None:
+                try:
+                    await aclose()
+                except Exception:
+                    pass
+            try:
+                writer.write(b"0\r\n\r\n")
+                await writer.drain()
+            except (ConnectionResetError, BrokenPipeError, OSError):
+                pass
 
     # --------------------------------------------------------- connection
     async def _handle_conn(self, reader: asyncio.StreamReader,

@@ -265,3 +265,26 @@ def test_admin_requests_trace():
         r = recs[0]
         assert r["alias"] == "llama-cris-demo" and r["status"] == "ok"
         assert r["completion_tokens"] == 3 and r["latency_ms"] > 0
+
+
+def test_stream_client_disconnect_settles_ticket():
+    """Dropping an SSE connection mid-stream must not leak router
+    in-flight counts; the ledger records the cancellation."""
+    with run_gateway(stub_kwargs={"token_delay_ms": 50}) as (client, registry, _):
+        stream = client.chat.completions.create(
+            model="llama-cris-demo", messages=MSGS, max_tokens=50,
+            stream=True, timeout=30)
+        it = iter(stream)
+        next(it)
+        stream._conn.close()   # hard client disconnect
+        deadline = time.time() + 10
+        settled = False
+        while time.time() < deadline:
+            rows = {r["model_id"]: r for r in client.router_state()["deployments"]}
+            if all(r["in_flight"] == 0 for r in rows.values()):
+                recs = client._get("/admin/requests", {"n": 20})["requests"]
+                if any(r["status"] == "cancelled" for r in recs):
+                    settled = True
+                    break
+            time.sleep(0.25)
+        assert settled, "ticket not settled after client disconnect"
