@@ -470,6 +470,13 @@ class GatewayApp:
                                     "throttled" if throttled else "error", t0)
                         exclude.add(id(ticket.state))
                         ticket = None
+                        # release the failed worker's stream NOW (its
+                        # cleanup decrements in-flight / aborts the
+                        # engine request) rather than at GC
+                        try:
+                            await asyncio.wait_for(it.aclose(), timeout=2.0)
+                        except Exception:
+                            pass
                         continue
                 from ..utils.logging import sanitize_error
                 err_evt = {"error": {"message": f"stream failed: "
