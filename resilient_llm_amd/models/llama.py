@@ -238,8 +238,8 @@ class LlamaForCausalLM:
             h = ops.rmsnorm_residual_(x, residual, p[f"l{i}.ln1"], c.rms_eps)
         qkv = _linear(h, p[f"l{i}.qkv"])
         k_cache, v_cache = kv_cache.layer(i)
-        ops.rope_kv_append_qkv_(qkv, positions, self.cos_sin,
-                                k_cache, v_cache, slot_mapping, self.n_heads)
+        # rope + cache append happen inside attn_fn (the decode path fuses
+        # them into the attention kernel itself)
         attn_out = attn_fn(i, qkv, k_cache, v_cache)   # [T, q_size]
         h = self._all_reduce(_linear(attn_out, p[f"l{i}.o"]))
         h2 = ops.rmsnorm_residual_(h, residual, p[f"l{i}.ln2"], c.rms_eps)
@@ -253,6 +253,8 @@ class LlamaForCausalLM:
         """Returns logits for the LAST token of each sequence:
         [n_seqs, vocab]."""
         def attn(i, qkv, k_cache, v_cache):
+            ops.rope_kv_append_qkv_(qkv, positions, self.cos_sin, k_cache,
+                                    v_cache, slot_mapping, self.n_heads)
             return ops.prefill_attn_qkv(qkv, cu_seqlens, self.scale,
                                         self.n_heads, self.n_kv_heads,
                                         self.config.head_dim)
@@ -268,6 +270,8 @@ class LlamaForCausalLM:
         are prefill-chunk tokens attending over the paged cache
         (chunked prefill).  Returns logits for ``sample_idx`` rows."""
         def attn(i, qkv, k_cache, v_cache):
+            ops.rope_kv_append_qkv_(qkv, positions, self.cos_sin, k_cache,
+                                    v_cache, slot_mapping, self.n_heads)
             out = ops.prefill_paged_attn(
                 qkv, k_cache, v_cache, chunk_row0, chunk_pos0, chunk_nrows,
                 chunk_btrow, block_tables_pre, self.scale, self.n_heads)
@@ -286,8 +290,9 @@ class LlamaForCausalLM:
                        seq_lens: torch.Tensor) -> torch.Tensor:
         """One token per sequence; returns [batch, vocab] logits."""
         def attn(i, qkv, k_cache, v_cache):
-            return ops.decode_attn_qkv(qkv, k_cache, v_cache, block_tables,
-                                       seq_lens, self.scale, self.n_heads)
+            return ops.decode_attn_rope_qkv(
+                qkv, positions, self.cos_sin, k_cache, v_cache, slot_mapping,
+                block_tables, seq_lens, self.scale, self.n_heads)
         return self._forward(input_ids, positions, kv_cache, slot_mapping,
                              attn, last_idx=None)
 

@@ -121,6 +121,22 @@ def decode_attn_qkv(qkv: torch.Tensor, k_cache: torch.Tensor,
     return out.reshape(B, n_q * D)
 
 
+def decode_attn_rope_qkv(qkv, positions, cos_sin, k_cache, v_cache,
+                         slot_mapping, block_table, seq_lens, scale: float,
+                         n_q: int) -> torch.Tensor:
+    """Fused decode step: rope(q,k) + cache append + paged attention in
+    one kernel on GPU; composed from the unfused reference ops on CPU."""
+    if qkv.is_cuda:
+        _gpu()
+        return torch.ops.rlli.decode_attn_rope_qkv(
+            qkv, positions, cos_sin, k_cache, v_cache, slot_mapping,
+            block_table, seq_lens, scale, n_q)
+    rope_kv_append_qkv_(qkv, positions, cos_sin, k_cache, v_cache,
+                        slot_mapping, n_q)
+    return decode_attn_qkv(qkv, k_cache, v_cache, block_table, seq_lens,
+                           scale, n_q)
+
+
 def prefill_attn_qkv(qkv: torch.Tensor, cu_seqlens: torch.Tensor,
                      scale: float, n_q: int, n_kv: int,
                      head_dim: int) -> torch.Tensor:

@@ -24,7 +24,12 @@ namespace {
 
 constexpr float kNegInf = -1e30f;
 
-template <int GW, int GROUP>
+// FUSED variant: rope(q,k) + cache append of the new token + its
+// self-attention term live INSIDE this kernel (decode fuses 2 kernels
+// into 1; the self token is handled in registers by partial 0, so no
+// in-kernel cache write->read ordering exists — future steps read the
+// appended k/v after the kernel boundary).
+template <int GW, int GROUP, bool FUSED>
 __global__ __launch_bounds__(256)
 void decode_attn_kernel(const uint16_t* __restrict__ q,
                         const uint16_t* __restrict__ k_cache,
@@ -33,7 +38,14 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
                         const int32_t* __restrict__ seq_lens,
                         uint16_t* __restrict__ out,
                         int n_kv_heads, int block_size, int max_blocks,
-                        float scale, int q_stride) {
+                        float scale, int q_stride,
+                        const uint16_t* __restrict__ k_src,
+                        const uint16_t* __restrict__ v_src,
+                        const int32_t* __restrict__ positions,
+                        const float* __restrict__ cos_sin,
+                        const int32_t* __restrict__ slot_mapping,
+                        uint16_t* __restrict__ k_cache_w,
+                        uint16_t* __restrict__ v_cache_w) {
   constexpr int D = GW * 8;
   constexpr int GPW = 64 / GW;                 // lane-groups per wave
   const int seq = blockIdx.x / n_kv_heads;
@@ -63,6 +75,34 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
     for (int i = 0; i < 8; ++i) qv[h][i] = bf16_to_f32(qh.s[i]);
   }
 
+  // rope coefficients for this lane's 8 dims (pair index = d mod D/2);
+  // the rotation partner's raw value sits GW/2 lanes away in the group
+  float cs_c[FUSED ? 8 : 1], cs_s[FUSED ? 8 : 1];
+  bool first_half = true;
+  if constexpr (FUSED) {
+    constexpr int HALF = D / 2;
+    const int pos = positions[seq];
+    first_half = d0 < HALF;
+    const int pidx = d0 % HALF;
+    const float* cs = cos_sin + int64_t(pos) * D;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      cs_c[i] = cs[pidx + i];
+      cs_s[i] = cs[HALF + pidx + i];
+    }
+    // rotate every query head in place
+#pragma unroll
+    for (int h = 0; h < GROUP; ++h) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const float own = qv[h][i];
+        const float other = __shfl_xor(own, GW / 2, kWave);
+        qv[h][i] = first_half ? own * cs_c[i] - other * cs_s[i]
+                              : own * cs_c[i] + other * cs_s[i];
+      }
+    }
+  }
+
   float m[GROUP], l[GROUP], acc[GROUP][8];
 #pragma unroll
   for (int h = 0; h < GROUP; ++h) {
@@ -72,13 +112,55 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
     for (int i = 0; i < 8; ++i) acc[h][i] = 0.f;
   }
 
+  if constexpr (FUSED) {
+    // group 0 of wave 0: rope k, append k/v to the cache, and fold the
+    // NEW token's self-attention term into partial 0
+    if (wave == 0 && group == 0) {
+      bf16x8 kr, vr;
+      kr.u = *reinterpret_cast<const uint4*>(
+          k_src + int64_t(seq) * q_stride + kvh * D + d0);
+      vr.u = *reinterpret_cast<const uint4*>(
+          v_src + int64_t(seq) * q_stride + kvh * D + d0);
+      float kf[8], vf[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const float own = bf16_to_f32(kr.s[i]);
+        const float other = __shfl_xor(own, GW / 2, kWave);
+        kf[i] = first_half ? own * cs_c[i] - other * cs_s[i]
+                           : own * cs_c[i] + other * cs_s[i];
+        vf[i] = bf16_to_f32(vr.s[i]);
+      }
+      const int32_t slot = slot_mapping[seq];
+      bf16x8 kw;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) kw.s[i] = f32_to_bf16(kf[i]);
+      const int64_t cbase =
+          (int64_t(slot / block_size) * n_kv_heads + kvh) * block_size * D +
+          int64_t(slot % block_size) * D + d0;
+      *reinterpret_cast<uint4*>(k_cache_w + cbase) = kw.u;
+      *reinterpret_cast<uint4*>(v_cache_w + cbase) = vr.u;
+#pragma unroll
+      for (int h = 0; h < GROUP; ++h) {
+        float s = 0.f;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) s += qv[h][i] * kf[i];
+        s = group_sum<GW>(s) * scale;
+        m[h] = s;
+        l[h] = 1.f;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) acc[h][i] = vf[i];
+      }
+    }
+  }
+
   // Latency discipline: batch-issue ALL of a cache block's K/V vectors
   // for this lane-group (TPG k-loads + TPG v-loads back to back, plus
   // the NEXT block's K prefetched) before any arithmetic touches them —
   // one lane then has ~2*TPG+ loads in flight instead of 2, which is
   // what moves this kernel from latency-bound to bandwidth-bound
   // (guide Guideline 7/15: waits belong at the first consumer).
-  const int n_blocks = (len + block_size - 1) / block_size;
+  const int len_cache = FUSED ? len - 1 : len;   // FUSED: prefix only
+  const int n_blocks = (len_cache + block_size - 1) / block_size;
   constexpr int TPG = 16 / GPW >= 1 ? 16 / GPW : 1;   // tokens per group @bs16
   const int tok_per_grp = block_size / GPW;
 
@@ -103,7 +185,7 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
     for (int i = 0; i < TPG; ++i) {
       if (i >= tok_per_grp) break;
       const int tok = group + GPW * i;         // token within the block
-      if (b * block_size + tok >= len) continue;   // group-uniform tail
+      if (b * block_size + tok >= len_cache) continue;   // group-uniform tail
       bf16x8 kv, vv;
       kv.u = kraw[i];
       vv.u = vraw[i];
@@ -155,7 +237,7 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
           (int64_t(phys) * n_kv_heads + kvh) * block_size * D;
       for (int bi = 0; bi < tok_per_grp; ++bi) {
         const int tok = group + GPW * bi;
-        if (b * block_size + tok >= len) continue;
+        if (b * block_size + tok >= len_cache) continue;
         bf16x8 kv, vv;
         kv.u = *reinterpret_cast<const uint4*>(
             k_cache + kv_base + int64_t(tok) * D + d0);
@@ -217,36 +299,60 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
   }
 }
 
+struct FusedArgs {
+  const uint16_t* k_src = nullptr;
+  const uint16_t* v_src = nullptr;
+  const int32_t* positions = nullptr;
+  const float* cos_sin = nullptr;
+  const int32_t* slot_mapping = nullptr;
+  uint16_t* k_cache_w = nullptr;
+  uint16_t* v_cache_w = nullptr;
+};
+
 template <int GW, int GROUP>
 void dispatch_decode(const uint16_t* q, const uint16_t* k_cache,
                      const uint16_t* v_cache, const int32_t* block_table,
                      const int32_t* seq_lens, uint16_t* out, int batch,
                      int n_kv_heads, int block_size, int max_blocks,
-                     float scale, int q_stride, hipStream_t stream) {
+                     float scale, int q_stride, const FusedArgs* fa,
+                     hipStream_t stream) {
   constexpr int D = GW * 8;
   constexpr int GPW = 64 / GW;
   const int n_part = 4 * GPW;
   const size_t smem = size_t(n_part) * GROUP * (D + 2) * sizeof(float);
-  hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP>),
-                     dim3(batch * n_kv_heads), dim3(256), smem, stream,
-                     q, k_cache, v_cache, block_table, seq_lens, out,
-                     n_kv_heads, block_size, max_blocks, scale, q_stride);
+  if (fa != nullptr) {
+    hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP, true>),
+                       dim3(batch * n_kv_heads), dim3(256), smem, stream,
+                       q, k_cache, v_cache, block_table, seq_lens, out,
+                       n_kv_heads, block_size, max_blocks, scale, q_stride,
+                       fa->k_src, fa->v_src, fa->positions, fa->cos_sin,
+                       fa->slot_mapping, fa->k_cache_w, fa->v_cache_w);
+  } else {
+    hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP, false>),
+                       dim3(batch * n_kv_heads), dim3(256), smem, stream,
+                       q, k_cache, v_cache, block_table, seq_lens, out,
+                       n_kv_heads, block_size, max_blocks, scale, q_stride,
+                       nullptr, nullptr, nullptr, nullptr, nullptr,
+                       nullptr, nullptr);
+  }
 }
 
 }  // namespace
 
-void launch_decode_attn(const uint16_t* q, const uint16_t* k_cache,
-                        const uint16_t* v_cache, const int32_t* block_table,
-                        const int32_t* seq_lens, uint16_t* out, int batch,
-                        int n_q_heads, int n_kv_heads, int head_dim,
-                        int block_size, int max_blocks, float scale,
-                        int q_stride, hipStream_t stream) {
+void launch_decode_attn_impl(const uint16_t* q, const uint16_t* k_cache,
+                             const uint16_t* v_cache,
+                             const int32_t* block_table,
+                             const int32_t* seq_lens, uint16_t* out,
+                             int batch, int n_q_heads, int n_kv_heads,
+                             int head_dim, int block_size, int max_blocks,
+                             float scale, int q_stride, const FusedArgs* fa,
+                             hipStream_t stream) {
   if (batch == 0) return;
   const int group = n_q_heads / n_kv_heads;
   auto run = [&](auto gw_tag, auto group_tag) {
     dispatch_decode<decltype(gw_tag)::value, decltype(group_tag)::value>(
         q, k_cache, v_cache, block_table, seq_lens, out, batch, n_kv_heads,
-        block_size, max_blocks, scale, q_stride, stream);
+        block_size, max_blocks, scale, q_stride, fa, stream);
   };
   using I8 = std::integral_constant<int, 8>;
   using I16 = std::integral_constant<int, 16>;
@@ -267,6 +373,37 @@ void launch_decode_attn(const uint16_t* q, const uint16_t* k_cache,
     else if (group == 8) run(I8{}, G8{});
     else return;
   }
+}
+
+void launch_decode_attn(const uint16_t* q, const uint16_t* k_cache,
+                        const uint16_t* v_cache, const int32_t* block_table,
+                        const int32_t* seq_lens, uint16_t* out, int batch,
+                        int n_q_heads, int n_kv_heads, int head_dim,
+                        int block_size, int max_blocks, float scale,
+                        int q_stride, hipStream_t stream) {
+  launch_decode_attn_impl(q, k_cache, v_cache, block_table, seq_lens, out,
+                          batch, n_q_heads, n_kv_heads, head_dim, block_size,
+                          max_blocks, scale, q_stride, nullptr, stream);
+}
+
+void launch_decode_attn_fused(
+    const uint16_t* qkv, uint16_t* k_cache, uint16_t* v_cache,
+    const int32_t* block_table, const int32_t* seq_lens,
+    const int32_t* positions, const float* cos_sin,
+    const int32_t* slot_mapping, uint16_t* out, int batch, int n_q_heads,
+    int n_kv_heads, int head_dim, int block_size, int max_blocks,
+    float scale, int qkv_stride, hipStream_t stream) {
+  FusedArgs fa;
+  fa.k_src = qkv + n_q_heads * head_dim;
+  fa.v_src = qkv + (n_q_heads + n_kv_heads) * head_dim;
+  fa.positions = positions;
+  fa.cos_sin = cos_sin;
+  fa.slot_mapping = slot_mapping;
+  fa.k_cache_w = k_cache;
+  fa.v_cache_w = v_cache;
+  launch_decode_attn_impl(qkv, k_cache, v_cache, block_table, seq_lens, out,
+                          batch, n_q_heads, n_kv_heads, head_dim, block_size,
+                          max_blocks, scale, qkv_stride, &fa, stream);
 }
 
 }  // namespace rlli

@@ -212,6 +212,44 @@ Tensor decode_attn_qkv(const Tensor& qkv, const Tensor& k_cache,
   return out;
 }
 
+// Fused decode: rope + append + attention in one kernel.
+Tensor decode_attn_rope_qkv(const Tensor& qkv, const Tensor& positions,
+                            const Tensor& cos_sin, Tensor k_cache,
+                            Tensor v_cache, const Tensor& slot_mapping,
+                            const Tensor& block_table, const Tensor& seq_lens,
+                            double scale, int64_t n_q) {
+  check_bf16_contig(qkv, "qkv");
+  check_bf16_contig(k_cache, "k_cache");
+  check_bf16_contig(v_cache, "v_cache");
+  TORCH_CHECK(positions.scalar_type() == at::kInt && positions.is_contiguous());
+  TORCH_CHECK(slot_mapping.scalar_type() == at::kInt && slot_mapping.is_contiguous());
+  TORCH_CHECK(cos_sin.scalar_type() == at::kFloat && cos_sin.is_contiguous());
+  TORCH_CHECK(block_table.scalar_type() == at::kInt && block_table.is_contiguous());
+  TORCH_CHECK(seq_lens.scalar_type() == at::kInt && seq_lens.is_contiguous());
+  const int batch = int(qkv.size(0));
+  const int n_kv = int(k_cache.size(1));
+  const int block_size = int(k_cache.size(2));
+  const int D = int(k_cache.size(3));
+  const int stride = int(qkv.size(1));
+  TORCH_CHECK(stride == (n_q + 2 * n_kv) * D, "qkv width mismatch");
+  const int group = int(n_q) / n_kv;
+  TORCH_CHECK(group == 1 || group == 2 || group == 4 || group == 8);
+  TORCH_CHECK(D == 64 || D == 128);
+  TORCH_CHECK(cos_sin.size(1) == D);
+  const int max_blocks = int(block_table.size(1));
+  Tensor out = at::empty({batch, n_q * D}, qkv.options());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
+  rlli::launch_decode_attn_fused(
+      bf16_ptr(qkv), bf16_ptr(k_cache), bf16_ptr(v_cache),
+      block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
+      positions.data_ptr<int32_t>(), cos_sin.data_ptr<float>(),
+      slot_mapping.data_ptr<int32_t>(), bf16_ptr(out), batch, int(n_q),
+      n_kv, D, block_size, max_blocks, float(scale), stride,
+      current_stream(qkv));
+  check_launch("decode_attn_rope_qkv");
+  return out;
+}
+
 // --------------------------------------------------------- prefill_attn
 Tensor prefill_attn(const Tensor& q, const Tensor& k, const Tensor& v,
                     const Tensor& cu_seqlens, double scale) {
@@ -394,6 +432,9 @@ TORCH_LIBRARY(rlli, m) {
         "int n_q) -> ()");
   m.def("decode_attn_qkv(Tensor qkv, Tensor k_cache, Tensor v_cache, "
         "Tensor block_table, Tensor seq_lens, float scale, int n_q) -> Tensor");
+  m.def("decode_attn_rope_qkv(Tensor qkv, Tensor positions, Tensor cos_sin, "
+        "Tensor(a!) k_cache, Tensor(b!) v_cache, Tensor slot_mapping, "
+        "Tensor block_table, Tensor seq_lens, float scale, int n_q) -> Tensor");
   m.def("prefill_attn_qkv(Tensor qkv, Tensor cu_seqlens, float scale, "
         "int n_q, int n_kv, int head_dim) -> Tensor");
   m.def("prefill_attn(Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, "
@@ -415,6 +456,7 @@ TORCH_LIBRARY_IMPL(rlli, CUDA, m) {
   m.impl("decode_attn", &decode_attn);
   m.impl("rope_kv_append_qkv_", &rope_kv_append_qkv_);
   m.impl("decode_attn_qkv", &decode_attn_qkv);
+  m.impl("decode_attn_rope_qkv", &decode_attn_rope_qkv);
   m.impl("prefill_attn_qkv", &prefill_attn_qkv);
   m.impl("prefill_attn", &prefill_attn);
   m.impl("sample", &sample);

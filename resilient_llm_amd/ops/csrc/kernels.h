@@ -42,6 +42,16 @@ void launch_decode_attn(
     int block_size, int max_blocks, float scale, int q_stride,
     hipStream_t stream);
 
+// Fused decode: rope(q,k) + cache append + paged attention in ONE
+// kernel (decode path; qkv is the raw fused GEMM output).
+void launch_decode_attn_fused(
+    const uint16_t* qkv, uint16_t* k_cache, uint16_t* v_cache,
+    const int32_t* block_table, const int32_t* seq_lens,
+    const int32_t* positions, const float* cos_sin,
+    const int32_t* slot_mapping, uint16_t* out, int batch, int n_q_heads,
+    int n_kv_heads, int head_dim, int block_size, int max_blocks,
+    float scale, int qkv_stride, hipStream_t stream);
+
 // Varlen causal prefill attention over in-batch q/k/v.
 void launch_prefill_attn(
     const uint16_t* q, const uint16_t* k, const uint16_t* v,

@@ -273,3 +273,38 @@ def test_prefill_paged_vs_ref(n_q, n_kv, D):
         [0, 32, 40, 72], [0, 32, 40, 72], [32, 8, 32, 3], [0, 0, 0, 0],
         bt.cpu(), 0.088, n_q)
     assert_close_bf16(out.cpu(), out_ref)
+
+
+@pytest.mark.parametrize("n_q,n_kv,D", [(32, 8, 128), (8, 8, 128), (16, 2, 128),
+                                        (8, 2, 64)])
+def test_decode_attn_rope_fused_vs_unfused(n_q, n_kv, D):
+    """The fused rope+append+attention decode kernel matches the unfused
+    (rope_kv_append_qkv_ then decode_attn_qkv) sequence: identical cache
+    writes and attention output."""
+    torch.manual_seed(11)
+    B, bs = 5, 16
+    lens = [1, 17, 33, 64, 160]           # ATTENTION lengths incl. self
+    width = (n_q + 2 * n_kv) * D
+    nbt = max((l + bs - 1) // bs for l in lens)
+    nblocks = B * nbt + 2
+    kc = torch.randn(nblocks, n_kv, bs, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    bt = torch.arange(B * nbt, dtype=torch.int32, device=DEV).reshape(B, nbt).contiguous()
+    qkv = torch.randn(B, width, dtype=torch.bfloat16, device=DEV)
+    pos = torch.tensor([l - 1 for l in lens], dtype=torch.int32, device=DEV)
+    slots = torch.tensor([int(bt[i, (l - 1) // bs]) * bs + (l - 1) % bs
+                          for i, l in enumerate(lens)],
+                         dtype=torch.int32, device=DEV)
+    seq_lens = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    cs = ops.build_cos_sin(512, D, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+
+    qkv_u, kc_u, vc_u = qkv.clone(), kc.clone(), vc.clone()
+    ops.rope_kv_append_qkv_(qkv_u, pos, cs, kc_u, vc_u, slots, n_q)
+    out_u = ops.decode_attn_qkv(qkv_u, kc_u, vc_u, bt, seq_lens, scale, n_q)
+
+    out_f = torch.ops.rlli.decode_attn_rope_qkv(
+        qkv, pos, cs, kc, vc, slots, bt, seq_lens, scale, n_q)
+    assert_close_bf16(out_f.cpu(), out_u.cpu())
+    torch.testing.assert_close(kc.float(), kc_u.float())
+    torch.testing.assert_close(vc.float(), vc_u.float())
