@@ -92,40 +92,33 @@ def start_worker(loopth: LoopThread, device: str, label: str, sock: str,
     return loopth.run(_start(), timeout=600)
 
 
-def start_gateway(loopth: LoopThread, n_gpus: int, model: str, port: int,
-                  run_dir: str):
-    from resilient_llm_amd.config import load_config
-    from resilient_llm_amd.gateway.app import GatewayApp
-    from resilient_llm_amd.gateway.http import HttpServer
-    from resilient_llm_amd.workers.base import WorkerRegistry
-    from resilient_llm_amd.workers.rpc import RpcWorkerClient
-
-    data = {
-        "cluster": {"port": port},
-        "model_list": [
-            {"model_name": "bench-model",
-             "litellm_params": {"model": f"gpu/*/{model}"},
-             "model_info": {"id": "bench-spread"}},
-        ],
-        "router_settings": {"routing_strategy": "simple-shuffle",
-                            "enable_pre_call_checks": False},
-    }
-    config = load_config(data=data)
-
-    async def _start():
-        registry = WorkerRegistry()
-        for r in range(n_gpus):
-            client = RpcWorkerClient(f"gpu:{r}", {model},
-                                     os.path.join(run_dir, f"w{r}.sock"))
-            await client.connect(timeout=900)
-            registry.register("gpu", str(r), client)
-        app = GatewayApp(config, registry, health_interval_s=5.0)
-        server = HttpServer(app.handle, host="127.0.0.1", port=port)
-        await server.start()
-        await app.start_background()
-        return app
-
-    return loopth.run(_start(), timeout=1200)
+def spawn_gateways(n_gpus: int, model: str, port: int, run_dir: str,
+                   n_procs: int) -> list:
+    """Gateway PROCESSES sharing the port via SO_REUSEPORT (the same
+    scale-out LiteLLM's --num_workers provides); the bench config has no
+    rate limits, so per-process windows are irrelevant here."""
+    import subprocess
+    sockets = ",".join(os.path.join(run_dir, f"w{r}.sock")
+                       for r in range(n_gpus))
+    procs = []
+    ready_files = []
+    for g in range(n_procs):
+        rf = os.path.join(run_dir, f"gw{g}.ready")
+        ready_files.append(rf)
+        env = dict(os.environ)
+        env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
+        procs.append(subprocess.Popen(
+            [sys.executable, "-m", "resilient_llm_amd.gateway.bench_gateway",
+             "--port", str(port), "--model", model, "--sockets", sockets,
+             "--ready-file", rf], env=env))
+    deadline = time.time() + 900
+    while time.time() < deadline:
+        if all(os.path.exists(rf) for rf in ready_files):
+            return procs
+        for p in procs:
+            assert p.poll() is None, "gateway process died during startup"
+        time.sleep(0.25)
+    raise RuntimeError("gateways failed to come up")
 
 
 class LoadGen:
@@ -226,15 +219,17 @@ def main() -> None:
 
     port = None
     loadgen = None
+    gateways: list = []
     rounds_timed: list = []
     if rank == 0:
         port = free_port()
-        start_gateway(loopth, n_gpus if distributed else 1, args.model, port,
-                      run_dir)
+        n_gw = max(1, min(4, (n_gpus + 1) // 2))
+        gateways = spawn_gateways(n_gpus if distributed else 1, args.model,
+                                  port, run_dir, n_gw)
         loadgen = LoadGen(port, conc_total, args.prompt_tokens,
                           args.output_tokens)
-        log(f"gateway on :{port}, driving {conc_total} concurrent clients "
-            f"(loadgen subprocess)")
+        log(f"{n_gw} gateway process(es) on :{port}, driving {conc_total} "
+            f"concurrent clients (loadgen subprocess)")
 
     # ---- warmup ----
     for w in range(args.warmup):
@@ -266,6 +261,8 @@ def main() -> None:
 
     if rank == 0:
         loadgen.close()
+        for g in gateways:
+            g.terminate()
         n_req = sum(r["total"] for r in rounds_timed)
         n_ok = sum(r["ok"] for r in rounds_timed)
         lats = sorted(l for r in rounds_timed for l in r["latencies"])

@@ -87,10 +87,11 @@ class HttpServer:
         self.port = port
         self._server: Optional[asyncio.base_events.Server] = None
 
-    async def start(self) -> None:
+    async def start(self, reuse_port: bool = False) -> None:
         self._server = await asyncio.start_server(
             self._handle_conn, self.host, self.port,
-            limit=MAX_HEADER_BYTES, reuse_address=True)
+            limit=MAX_HEADER_BYTES, reuse_address=True,
+            reuse_port=reuse_port)
 
     async def stop(self) -> None:
         if self._server is not None:

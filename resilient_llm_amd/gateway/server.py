@@ -49,11 +49,12 @@ def build_registry(config: Config, stub_kwargs: Optional[dict] = None) -> Worker
 
 
 async def serve(config: Config, registry: Optional[WorkerRegistry] = None,
-                ready_event: Optional[asyncio.Event] = None) -> None:
+                ready_event: Optional[asyncio.Event] = None,
+                reuse_port: bool = False) -> None:
     registry = registry or build_registry(config)
     app = GatewayApp(config, registry)
     server = HttpServer(app.handle, host=config.cluster.host, port=config.cluster.port)
-    await server.start()
+    await server.start(reuse_port=reuse_port)
     for w in registry.all().values():
         if hasattr(w, "connect"):
             await w.connect(timeout=900)
@@ -84,6 +85,11 @@ def main() -> None:
     ap.add_argument("--config", default="config/config.yaml")
     ap.add_argument("--port", type=int, default=None, help="override cluster.port")
     ap.add_argument("--host", default=None)
+    ap.add_argument("--workers", type=int, default=1,
+                    help="gateway processes sharing the port via "
+                         "SO_REUSEPORT (NOTE: rate-limit windows and the "
+                         "ledger are per process — keep 1 unless the "
+                         "config is unlimited or limits are sharded)")
     args = ap.parse_args()
     config = load_config(args.config)
     if args.port is not None:
@@ -96,7 +102,21 @@ def main() -> None:
     if kinds & {"gpu", "pool"}:
         from ..workers.gpu import register_gpu_workers
         register_gpu_workers(config, registry)
-    asyncio.run(serve(config, registry))
+    if args.workers > 1:
+        log_with_timestamp(
+            f"SO_REUSEPORT scale-out: {args.workers} gateway processes; "
+            f"RPM/TPM windows are PER PROCESS", "yellow")
+        import multiprocessing
+        procs = []
+        for _ in range(args.workers - 1):
+            pr = multiprocessing.Process(
+                target=lambda: asyncio.run(serve(config, registry,
+                                                 reuse_port=True)))
+            pr.start()
+            procs.append(pr)
+        asyncio.run(serve(config, registry, reuse_port=True))
+    else:
+        asyncio.run(serve(config, registry))
 
 
 if __name__ == "__main__":
