@@ -18,6 +18,8 @@
 
 #include "common.h"
 
+#include <cstdlib>
+
 namespace rlli {
 
 namespace {
@@ -25,14 +27,13 @@ namespace {
 using bf16x8_vec = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-constexpr int KC = 256;         // k-chunk staged in LDS per iteration
 constexpr int BN = 64;          // column panel per workgroup
 
 DEV_INLINE int swz(int row, int byte_off) {
   return byte_off ^ ((row & 15) << 4);
 }
 
-template <int M_TILES>
+template <int M_TILES, int KC, int WDEPTH>
 __global__ __launch_bounds__(256)
 void skinny_gemm_kernel(const uint16_t* __restrict__ x,
                         const uint16_t* __restrict__ w,
@@ -85,33 +86,20 @@ void skinny_gemm_kernel(const uint16_t* __restrict__ x,
           swz(row, (row * KC + col8 * 8) * 2)) = piece;
     }
   };
-  // preload the whole chunk's B fragments: KSTEPS outstanding nt loads
-  u32x4 braw[KSTEPS];
-  auto load_b = [&](int c) {
+  u32x4 brawA[KSTEPS], brawB[KSTEPS];
+  auto load_b = [&](int c, u32x4 (&dst)[KSTEPS]) {
 #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks)
-      braw[ks] = __builtin_nontemporal_load(
+      dst[ks] = __builtin_nontemporal_load(
           reinterpret_cast<const u32x4*>(wrow + c * KC + ks * 32));
   };
-
-  stage_x(chunk0);
-  load_b(chunk0);
-  __syncthreads();
-
-  for (int c = chunk0; c < chunk1; ++c) {
+  auto compute = [&](int c, u32x4 (&bsrc)[KSTEPS]) {
     const char* xbase = reinterpret_cast<const char*>(xs) +
                         ((c & 1) ? MROWS * KC * 2 : 0);
-    u32x4 bcur[KSTEPS];
-#pragma unroll
-    for (int ks = 0; ks < KSTEPS; ++ks) bcur[ks] = braw[ks];
-    if (c + 1 < chunk1) {
-      stage_x(c + 1);        // issues into the other LDS buffer
-      load_b(c + 1);         // next chunk's B stream in flight
-    }
 #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
       const int k0 = ks * 32;
-      bf16x8_vec bfrag = *reinterpret_cast<bf16x8_vec*>(&bcur[ks]);
+      bf16x8_vec bfrag = *reinterpret_cast<bf16x8_vec*>(&bsrc[ks]);
 #pragma unroll
       for (int mt = 0; mt < M_TILES; ++mt) {
         const int row = mt * 16 + jcol;
@@ -120,6 +108,35 @@ void skinny_gemm_kernel(const uint16_t* __restrict__ x,
         bf16x8_vec afrag = *reinterpret_cast<bf16x8_vec*>(&araw);
         acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag, bfrag, acc[mt], 0, 0, 0);
+      }
+    }
+  };
+
+  stage_x(chunk0);
+  load_b(chunk0, brawA);
+  if (WDEPTH == 2 && chunk0 + 1 < chunk1) load_b(chunk0 + 1, brawB);
+  __syncthreads();
+
+  // A/B-alternating pipeline (statically indexed buffers, rule 20);
+  // WDEPTH=2 keeps TWO chunks of W-stream in flight per wave
+  for (int c = chunk0; c < chunk1; ++c) {
+    const bool useA = ((c - chunk0) & 1) == 0;
+    if (c + 1 < chunk1) stage_x(c + 1);
+    if (WDEPTH == 1) {
+      if (useA) {
+        if (c + 1 < chunk1) load_b(c + 1, brawB);
+        compute(c, brawA);
+      } else {
+        if (c + 1 < chunk1) load_b(c + 1, brawA);
+        compute(c, brawB);
+      }
+    } else {
+      if (useA) {
+        compute(c, brawA);
+        if (c + 2 < chunk1) load_b(c + 2, brawA);
+      } else {
+        compute(c, brawB);
+        if (c + 2 < chunk1) load_b(c + 2, brawB);
       }
     }
     __syncthreads();
@@ -160,13 +177,31 @@ void launch_skinny_gemm(const uint16_t* x, const uint16_t* w, float* ws,
                         uint16_t* out, int M, int N, int K, int splitk,
                         hipStream_t stream) {
   const int m_tiles = (M + 15) / 16;
+  const int KC = (std::getenv("RLLI_SKINNY_KC") &&
+                  atoi(std::getenv("RLLI_SKINNY_KC")) == 128) ? 128 : 256;
+  const int wdepth = (std::getenv("RLLI_SKINNY_WDEPTH") &&
+                      atoi(std::getenv("RLLI_SKINNY_WDEPTH")) == 1) ? 1 : 2;
   const int total_chunks = K / KC;
   const int chunks_per_slice = (total_chunks + splitk - 1) / splitk;
   const int blocks = (N / BN) * splitk;
   auto launch = [&](auto mt_tag) {
-    hipLaunchKernelGGL((skinny_gemm_kernel<decltype(mt_tag)::value>),
-                       dim3(blocks), dim3(256), 0, stream,
-                       x, w, ws, out, M, N, K, chunks_per_slice, splitk);
+    constexpr int MT = decltype(mt_tag)::value;
+    if (KC == 128 && wdepth == 2)
+      hipLaunchKernelGGL((skinny_gemm_kernel<MT, 128, 2>), dim3(blocks),
+                         dim3(256), 0, stream, x, w, ws, out, M, N, K,
+                         chunks_per_slice, splitk);
+    else if (KC == 128)
+      hipLaunchKernelGGL((skinny_gemm_kernel<MT, 128, 1>), dim3(blocks),
+                         dim3(256), 0, stream, x, w, ws, out, M, N, K,
+                         chunks_per_slice, splitk);
+    else if (wdepth == 2)
+      hipLaunchKernelGGL((skinny_gemm_kernel<MT, 256, 2>), dim3(blocks),
+                         dim3(256), 0, stream, x, w, ws, out, M, N, K,
+                         chunks_per_slice, splitk);
+    else
+      hipLaunchKernelGGL((skinny_gemm_kernel<MT, 256, 1>), dim3(blocks),
+                         dim3(256), 0, stream, x, w, ws, out, M, N, K,
+                         chunks_per_slice, splitk);
   };
   using T1 = std::integral_constant<int, 1>;
   using T2 = std::integral_constant<int, 2>;
