@@ -167,6 +167,15 @@ class GatewayApp:
                     ok = False
                 for model_id in self._deployments_on(worker):
                     self.router.set_healthy(model_id, ok)
+                if (not ok
+                        and getattr(worker, "proc_group", None)
+                        and worker.proc is not None
+                        and worker.proc.poll() is not None):
+                    # a dead TP-pool LEADER strands its followers in the
+                    # collective: reap the whole group so GPUs free up
+                    for pr in worker.proc_group[1:]:
+                        if pr.poll() is None:
+                            pr.terminate()
                 if (not ok and key not in respawning
                         and getattr(worker, "respawn", None) is not None
                         and getattr(worker, "proc", None) is not None

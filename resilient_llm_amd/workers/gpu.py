@@ -109,4 +109,5 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
                                   max_batch=max_batch)
         client = RpcWorkerClient(f"pool:{pool_name}", {model_name}, sock)
         client.proc = procs[0]
+        client.proc_group = procs
         registry.register("pool", pool_name, client)

@@ -147,6 +147,7 @@ class RpcWorkerClient(Worker):
         self._reader_task: Optional[asyncio.Task] = None
         self._in_flight = 0
         self.proc = None              # set by the spawner (subprocess handle)
+        self.proc_group = None        # TP pools: every rank's Popen
         self.respawn = None           # callable -> new Popen (elastic recovery)
         self.last_respawn = 0.0
 
@@ -291,9 +292,12 @@ class RpcWorkerClient(Worker):
                 self._writer.close()
             except Exception:
                 pass
-        if self.proc is not None and self.proc.poll() is None:
-            self.proc.terminate()
+        procs = self.proc_group or ([self.proc] if self.proc else [])
+        for p in procs:
+            if p.poll() is None:
+                p.terminate()
+        for p in procs:
             try:
-                self.proc.wait(timeout=10)
+                p.wait(timeout=10)
             except Exception:
-                self.proc.kill()
+                p.kill()
