@@ -23,7 +23,7 @@ from typing import Optional
 import torch
 
 from .. import ops
-from .kv_cache import PagedKVCache
+from .kv_cache import PagedKVCache, block_hash_chain
 
 # roctx phase markers for rocprofv3 --marker-trace:
 # torch.cuda.nvtx lowers to roctx on ROCm.  Opt-in: RLLI_ROCTX=1.
@@ -54,6 +54,7 @@ class SeqState:
     params: SamplingParams
     output_ids: list = dataclasses.field(default_factory=list)
     blocks: list = dataclasses.field(default_factory=list)
+    block_keys: Optional[list] = None      # prefix-cache content keys
     n_cached: int = 0
     arrived_at: float = dataclasses.field(default_factory=time.monotonic)
 
@@ -74,6 +75,7 @@ class LLMEngine:
                  max_model_len: Optional[int] = None,
                  chunk_size: int = 2048,
                  admission_window_s: float = 0.0,
+                 enable_prefix_caching: bool = True,
                  seed: int = 0) -> None:
         self.model = model
         self.kv = kv_cache
@@ -94,6 +96,7 @@ class LLMEngine:
         self.prefilling: list[SeqState] = []   # admitted, prompt not fully cached
         self.running: list[SeqState] = []
         self.chunk_size = chunk_size
+        self.enable_prefix_caching = enable_prefix_caching
         # coalesce request bursts into one prefill batch: hold admission
         # while arrivals are still landing (MUST be 0 under TP lockstep —
         # wall-clock decisions would diverge across ranks)
@@ -126,6 +129,8 @@ class LLMEngine:
         if len(self.waiting) >= self.max_queue:
             raise CapacityExceeded(f"queue full ({self.max_queue})")
         seq = SeqState(req_id=req_id, prompt_ids=list(prompt_ids), params=params)
+        if self.enable_prefix_caching:
+            seq.block_keys = block_hash_chain(seq.prompt_ids, self.block_size)
         if len(seq.prompt_ids) + params.max_tokens > self.max_model_len:
             raise CapacityExceeded(
                 f"prompt+max_tokens = "
@@ -166,10 +171,21 @@ class LLMEngine:
                 continue
             if tokens + len(seq.prompt_ids) > self.max_prefill_tokens and admitted:
                 break
-            if seq.reserved_blocks_needed > self.kv.free_blocks:
+            # prefix cache: acquire the longest cached chain of FULL
+            # prompt blocks (keep >= 1 suffix token so the first-output
+            # logits are always computed)
+            reused: list = []
+            if seq.block_keys:
+                max_reuse = (len(seq.prompt_ids) - 1) // self.block_size
+                reused = self.kv.lookup_prefix(seq.block_keys[:max_reuse])
+            needed = seq.reserved_blocks_needed - len(reused)
+            if needed > self.kv.free_blocks:
+                if reused:
+                    self.kv.free(reused)
                 break
-            seq.blocks = self.kv.allocate(seq.reserved_blocks_needed)
-            tokens += len(seq.prompt_ids)
+            seq.blocks = reused + self.kv.allocate(needed)
+            seq.n_cached = len(reused) * self.block_size
+            tokens += len(seq.prompt_ids) - seq.n_cached
             admitted.append(self.waiting.popleft())
         if admitted and len(self.stats["admit_events"]) < 1000:
             self.stats["admit_events"].append(
@@ -375,6 +391,14 @@ class LLMEngine:
         for seq, tok in zip(sampled_seqs, tokens):
             if id(seq) in decode_set:
                 seq.n_cached += 1
+            else:
+                # prompt fully prefilled: publish its full blocks to the
+                # prefix cache (idempotent for reused blocks)
+                if seq.block_keys:
+                    for i, key in enumerate(
+                            seq.block_keys[:len(seq.prompt_ids)
+                                           // self.block_size]):
+                        self.kv.register_block(seq.blocks[i], key)
             seq.output_ids.append(tok)
             if not self._finish(seq, outs, tok):
                 still_running.append(seq)
