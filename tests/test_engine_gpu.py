@@ -133,3 +133,20 @@ def test_chunked_prefill_gpu_matches_unchunked():
     e2.add_request("a", list(range(10, 150)), SamplingParams(max_tokens=6))
     got = drain(e2)["a"]
     assert got == want
+
+
+def test_prefix_cache_gpu_identical_outputs():
+    """GPU: repeated prompt reuses cached KV blocks with identical greedy
+    output (the reused blocks were written by the HIP kernels)."""
+    cfg = get_config("llama-3-8b")
+    model = LlamaForCausalLM(cfg, device="cuda:0", dtype=torch.bfloat16, seed=2)
+    kv = PagedKVCache.for_model(cfg, 128, device="cuda:0")
+    e = LLMEngine(model, kv, max_batch_size=2, enable_prefix_caching=True)
+    prompt = list(range(11, 140))
+    e.add_request("a", prompt, SamplingParams(max_tokens=5))
+    t1 = drain(e)["a"]
+    hits0 = kv.prefix_hits
+    e.add_request("b", prompt, SamplingParams(max_tokens=5))
+    t2 = drain(e)["b"]
+    assert t2 == t1
+    assert kv.prefix_hits > hits0
