@@ -26,9 +26,7 @@ import argparse
 import asyncio
 import json
 import os
-import random
 import socket
-import string
 import sys
 import tempfile
 import threading
@@ -51,12 +49,6 @@ def free_port() -> int:
     with socket.socket() as s:
         s.bind(("127.0.0.1", 0))
         return s.getsockname()[1]
-
-
-def make_prompt(rng: random.Random, n_tokens: int) -> str:
-    # ByteTokenizer: 1 token per byte (+bos +"user: ...\nassistant:" framing)
-    n = max(8, n_tokens - 18)
-    return "".join(rng.choice(string.ascii_lowercase + " ") for _ in range(n))
 
 
 class LoopThread:

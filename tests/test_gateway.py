@@ -288,3 +288,29 @@ def test_stream_client_disconnect_settles_ticket():
                     break
             time.sleep(0.25)
         assert settled, "ticket not settled after client disconnect"
+
+
+def test_consumer_limits_per_api_key():
+    """Optional per-consumer buckets on top of per-deployment ones
+    (SURVEY.md §3.4 MI355X equivalent)."""
+    from tests.gateway_harness import free_port, stub_config_dict
+    port = free_port()
+    cfg = stub_config_dict(port)
+    cfg["consumer_limits"] = {"keys": {"sk-limited": {"rpm": 2}}}
+    with run_gateway(cfg) as (client, *_):
+        from resilient_llm_amd.client import OpenAIClient
+        limited = OpenAIClient(f"http://127.0.0.1:{port}", api_key="sk-limited")
+        free = OpenAIClient(f"http://127.0.0.1:{port}", api_key="sk-free")
+        ok = limited_429 = 0
+        for _ in range(4):
+            try:
+                limited.chat.completions.create(model="llama-cris-demo",
+                                                messages=MSGS, max_tokens=2)
+                ok += 1
+            except RateLimitError:
+                limited_429 += 1
+        assert ok == 2 and limited_429 == 2
+        # other consumers unaffected
+        for _ in range(4):
+            free.chat.completions.create(model="llama-cris-demo",
+                                         messages=MSGS, max_tokens=2)
