@@ -34,6 +34,7 @@ _ROCTX = os.environ.get("RLLI_ROCTX") == "1"
 class SamplingParams:
     max_tokens: int = 128
     temperature: float = 0.0
+    top_p: float = 1.0
     seed: Optional[int] = None
     stop_on_eos: bool = True
 
@@ -55,6 +56,7 @@ class SeqState:
     output_ids: list = dataclasses.field(default_factory=list)
     blocks: list = dataclasses.field(default_factory=list)
     block_keys: Optional[list] = None      # prefix-cache content keys
+    default_seed: int = 0                  # deterministic per arrival order
     n_cached: int = 0
     arrived_at: float = dataclasses.field(default_factory=time.monotonic)
 
@@ -104,6 +106,7 @@ class LLMEngine:
         self.step_count = 0
         self.block_size = kv_cache.block_size
         self._aborted: set[str] = set()
+        self._arrival_counter = 0
         self._batch_dirty = True
         self._pending = None          # (seqs, event, host_tokens) 1-deep
         self._pinned = None
@@ -129,6 +132,9 @@ class LLMEngine:
         if len(self.waiting) >= self.max_queue:
             raise CapacityExceeded(f"queue full ({self.max_queue})")
         seq = SeqState(req_id=req_id, prompt_ids=list(prompt_ids), params=params)
+        self._arrival_counter += 1
+        seq.default_seed = (self.seed * 1000003
+                            + self._arrival_counter) & 0x7fffffffffffffff
         if self.enable_prefix_caching:
             seq.block_keys = block_hash_chain(seq.prompt_ids, self.block_size)
         if len(seq.prompt_ids) + params.max_tokens > self.max_model_len:
@@ -292,12 +298,35 @@ class LLMEngine:
                 self._batch_dirty = True
             setattr(self, attr, keep)
 
+    @staticmethod
+    def _seq_seed(s: SeqState) -> int:
+        base = s.params.seed if s.params.seed is not None else s.default_seed
+        return (base + 0x9e3779b97f4a7c15 * len(s.output_ids)) \
+            & 0x7fffffffffffffff
+
     def _sample(self, logits: torch.Tensor, seqs: list[SeqState]) -> list[int]:
         temps = torch.tensor([s.params.temperature for s in seqs],
                              dtype=torch.float32, device=self.device)
-        seed = (self.seed * 0x9e3779b9 + self.step_count) & 0x7fffffffffffffff
-        toks = ops.sample(logits, temps, seed)
+        seeds = torch.tensor([self._seq_seed(s) for s in seqs],
+                             dtype=torch.int64, device=self.device)
+        toks = ops.sample(logits, temps, seeds, 0)
+        topp = [i for i, s in enumerate(seqs)
+                if s.params.temperature > 0 and s.params.top_p < 1.0]
+        if topp:
+            toks = toks.clone()
+            for i in topp:
+                toks[i] = self._sample_top_p(logits[i], seqs[i])
         return toks.tolist()
+
+    def _sample_top_p(self, row_logits: torch.Tensor, s: SeqState) -> int:
+        """Nucleus sampling (torch path; rows that ask for top_p < 1)."""
+        probs = torch.softmax(row_logits.float() / s.params.temperature, -1)
+        sp, idx = probs.sort(descending=True)
+        keep = int((sp.cumsum(0) < s.params.top_p).sum()) + 1
+        sp = sp[:keep] / sp[:keep].sum()
+        gen = torch.Generator(device=row_logits.device)
+        gen.manual_seed(self._seq_seed(s))
+        return int(idx[int(torch.multinomial(sp, 1, generator=gen))])
 
     def _mixed_step(self) -> tuple[int, list[StepOutput]]:
         """One forward over [decode rows | prefill-chunk rows]: running
@@ -427,6 +456,11 @@ class LLMEngine:
         self._b_bt = bt.to(dev)
         self._b_temps = torch.tensor([s.params.temperature for s in seqs],
                                      dtype=torch.float32, device=dev)
+        self._b_seeds = torch.tensor([self._seq_seed(s) for s in seqs],
+                                     dtype=torch.int64, device=dev)
+        self._b_step_off = 0
+        self._b_has_topp = any(s.params.temperature > 0 and s.params.top_p < 1.0
+                               for s in seqs)
         self._batch_dirty = False
 
     def _decode_step(self) -> list[StepOutput]:
@@ -444,8 +478,14 @@ class LLMEngine:
         else:
             logits = self.model.forward_decode(self._b_ids, pos, self.kv,
                                                slots, bt, seq_lens)
-        seed = (self.seed * 0x9e3779b9 + self.step_count) & 0x7fffffffffffffff
-        tok_dev = ops.sample(logits, self._b_temps, seed)
+        if self._b_has_topp:
+            tokens = self._sample(logits, seqs)     # torch top-p path (sync)
+            tok_dev = torch.tensor(tokens, dtype=torch.int32,
+                                   device=self.device)
+        else:
+            tok_dev = ops.sample(logits, self._b_temps, self._b_seeds,
+                                 self._b_step_off)
+            self._b_step_off += 1
         for seq in seqs:
             seq.n_cached += 1          # KV of the fed token was appended
         # advance device state in place: the NEXT forward feeds tok_dev

@@ -171,11 +171,14 @@ def prefill_attn(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
 
 def sample(logits: torch.Tensor, temperatures: torch.Tensor,
-           seed: int = 0) -> torch.Tensor:
+           seeds: torch.Tensor, step: int = 0) -> torch.Tensor:
+    """Greedy argmax / Gumbel-max temperature sampling; ``seeds`` is a
+    per-row int64 tensor (per-request reproducibility), ``step`` a
+    uniform step offset mixed in-kernel."""
     if logits.is_cuda:
         _gpu()
-        return torch.ops.rlli.sample(logits, temperatures, seed)
-    return ref.sample(logits, temperatures, seed)
+        return torch.ops.rlli.sample(logits, temperatures, seeds, step)
+    return ref.sample(logits, temperatures, seeds, step)
 
 
 def prefill_paged_attn(qkv, k_cache, v_cache, chunk_row0, chunk_pos0,

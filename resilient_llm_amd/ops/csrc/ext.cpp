@@ -400,17 +400,21 @@ Tensor stream_probe(const Tensor& w, int64_t splitk) {
 }
 
 // --------------------------------------------------------------- sample
-Tensor sample(const Tensor& logits, const Tensor& temperatures, int64_t seed) {
+Tensor sample(const Tensor& logits, const Tensor& temperatures,
+              const Tensor& seeds, int64_t step) {
   check_bf16_contig(logits, "logits");
   TORCH_CHECK(temperatures.scalar_type() == at::kFloat &&
               temperatures.is_contiguous() && temperatures.is_cuda());
+  TORCH_CHECK(seeds.scalar_type() == at::kLong && seeds.is_contiguous() &&
+              seeds.is_cuda());
   const int batch = int(logits.size(0));
   const int vocab = int(logits.size(1));
-  TORCH_CHECK(temperatures.numel() == batch);
+  TORCH_CHECK(temperatures.numel() == batch && seeds.numel() == batch);
   Tensor out = at::empty({batch}, logits.options().dtype(at::kInt));
   c10::hip::HIPGuardMasqueradingAsCUDA guard(logits.device());
   rlli::launch_sample(bf16_ptr(logits), temperatures.data_ptr<float>(),
-                      uint64_t(seed), out.data_ptr<int32_t>(), batch, vocab,
+                      reinterpret_cast<const uint64_t*>(seeds.data_ptr<int64_t>()),
+                      uint64_t(step), out.data_ptr<int32_t>(), batch, vocab,
                       current_stream(logits));
   check_launch("sample");
   return out;
@@ -439,7 +443,7 @@ TORCH_LIBRARY(rlli, m) {
         "int n_q, int n_kv, int head_dim) -> Tensor");
   m.def("prefill_attn(Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, "
         "float scale) -> Tensor");
-  m.def("sample(Tensor logits, Tensor temperatures, int seed) -> Tensor");
+  m.def("sample(Tensor logits, Tensor temperatures, Tensor seeds, int step) -> Tensor");
   m.def("skinny_linear(Tensor x, Tensor w) -> Tensor");
   m.def("stream_probe(Tensor w, int splitk) -> Tensor");
   m.def("prefill_paged_attn(Tensor qkv, Tensor k_cache, Tensor v_cache, "

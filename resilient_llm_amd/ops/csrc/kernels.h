@@ -96,7 +96,8 @@ void launch_stream_probe(const uint16_t* w, float* sink, int N, int K,
 void launch_sample(
     const uint16_t* logits,            // [batch, vocab] bf16
     const float* temperatures,         // [batch]
-    uint64_t seed,
+    const uint64_t* seeds,             // [batch] per-request seeds
+    uint64_t step,                     // per-sequence step counter (mixed in)
     int32_t* out_tokens,               // [batch]
     int batch, int vocab, hipStream_t stream);
 

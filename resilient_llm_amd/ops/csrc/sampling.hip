@@ -41,13 +41,18 @@ DEV_INLINE BestPair better(BestPair a, BestPair b) {
 
 __global__ __launch_bounds__(256)
 void sample_kernel(const uint16_t* __restrict__ logits,
-                   const float* __restrict__ temperatures, uint64_t seed,
+                   const float* __restrict__ temperatures,
+                   const uint64_t* __restrict__ seeds, uint64_t step,
                    int32_t* __restrict__ out_tokens, int vocab) {
   const int row = blockIdx.x;
   const float temp = temperatures[row];
   const bool greedy = temp <= 0.f;
   const float inv_t = greedy ? 1.f : 1.f / temp;
-  const uint64_t row_ctr = uint64_t(row) << 32;
+  // per-request reproducibility: the row's seed (client-provided or
+  // engine default) mixed with the per-sequence step counter in-kernel,
+  // so the decode pipeline never uploads per-step randomness
+  const uint64_t seed = splitmix64(seeds[row] + 0x9e3779b97f4a7c15ull * step);
+  const uint64_t row_ctr = 0;
 
   BestPair best{-1e30f, 0};
   const int nvec = vocab / 8;
@@ -102,11 +107,11 @@ void sample_kernel(const uint16_t* __restrict__ logits,
 }  // namespace
 
 void launch_sample(const uint16_t* logits, const float* temperatures,
-                   uint64_t seed, int32_t* out_tokens, int batch, int vocab,
-                   hipStream_t stream) {
+                   const uint64_t* seeds, uint64_t step, int32_t* out_tokens,
+                   int batch, int vocab, hipStream_t stream) {
   if (batch == 0) return;
   hipLaunchKernelGGL(sample_kernel, dim3(batch), dim3(256), 0, stream,
-                     logits, temperatures, seed, out_tokens, vocab);
+                     logits, temperatures, seeds, step, out_tokens, vocab);
 }
 
 }  // namespace rlli

@@ -149,19 +149,19 @@ def prefill_paged_attn(qkv: torch.Tensor, k_cache: torch.Tensor,
 
 
 def sample(logits: torch.Tensor, temperatures: torch.Tensor,
-           seed: int = 0) -> torch.Tensor:
-    """Greedy rows match the kernel exactly; stochastic rows use torch's
-    RNG (the kernel's hash RNG is GPU-side — distribution-level tests)."""
+           seeds: torch.Tensor, step: int = 0) -> torch.Tensor:
+    """Greedy rows match the kernel exactly; stochastic rows use torch
+    RNG seeded per row from (seeds[i], step) — same reproducibility
+    contract as the kernel, not bitwise the same stream."""
     out = torch.empty(logits.shape[0], dtype=torch.int32, device=logits.device)
     lf = logits.float()
     greedy = temperatures <= 0
     if greedy.any():
         out[greedy] = lf[greedy].argmax(-1).to(torch.int32)
-    if (~greedy).any():
+    C = 0x9e3779b97f4a7c15
+    for i in (~greedy).nonzero(as_tuple=True)[0].tolist():
         gen = torch.Generator(device=logits.device)
-        gen.manual_seed(seed)
-        idx = (~greedy).nonzero(as_tuple=True)[0]
-        scaled = lf[idx] / temperatures[idx].unsqueeze(-1)
-        probs = torch.softmax(scaled, dim=-1)
-        out[idx] = torch.multinomial(probs, 1, generator=gen).squeeze(-1).to(torch.int32)
+        gen.manual_seed((int(seeds[i]) + C * step) & 0x7fffffffffffffff)
+        probs = torch.softmax(lf[i] / float(temperatures[i]), dim=-1)
+        out[i] = int(torch.multinomial(probs, 1, generator=gen))
     return out

@@ -165,6 +165,7 @@ class EngineWorker(Worker):
         prompt_ids = self.tokenizer.encode(render_prompt(req.messages))
         params = SamplingParams(max_tokens=req.max_tokens,
                                 temperature=req.temperature,
+                                top_p=req.top_p,
                                 seed=req.seed)
         self._req_counter += 1
         rid = f"{req.request_id}-{self._req_counter}"
@@ -185,6 +186,7 @@ class EngineWorker(Worker):
                     ("add", rid, prompt_ids,
                      {"max_tokens": params.max_tokens,
                       "temperature": params.temperature,
+                      "top_p": params.top_p,
                       "seed": params.seed}))
             else:
                 self.engine.add_request(rid, prompt_ids, params)

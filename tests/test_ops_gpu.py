@@ -117,11 +117,15 @@ def test_prefill_attn(n_q, n_kv, D, lens):
     assert_close_bf16(out.cpu(), out_ref)
 
 
+def _seeds(vals):
+    return torch.tensor(vals, dtype=torch.int64, device=DEV)
+
+
 def test_sample_greedy_matches_argmax():
     torch.manual_seed(7)
     logits = torch.randn(16, 128256, dtype=torch.bfloat16, device=DEV)
     temps = torch.zeros(16, device=DEV)
-    toks = ops.sample(logits, temps, seed=1)
+    toks = ops.sample(logits, temps, _seeds([1] * 16), 0)
     ref_toks = logits.float().argmax(-1).int()
     assert torch.equal(toks.cpu(), ref_toks.cpu())
 
@@ -136,7 +140,7 @@ def test_sample_temperature_distribution():
     counts = torch.zeros(vocab)
     n = 2000
     for s in range(n):
-        t = ops.sample(logits, temps, seed=s)
+        t = ops.sample(logits, temps, _seeds([s]), 0)
         counts[t.item()] += 1
     probs = torch.softmax(logits[0, :4].float(), -1)
     emp = counts[:4] / n
@@ -144,14 +148,16 @@ def test_sample_temperature_distribution():
     assert torch.allclose(emp, probs.cpu(), atol=0.05), (emp, probs)
 
 
-def test_sample_deterministic_given_seed():
+def test_sample_deterministic_given_seed_and_step():
     logits = torch.randn(4, 1000, dtype=torch.bfloat16, device=DEV)
     temps = torch.full((4,), 0.8, device=DEV)
-    a = ops.sample(logits, temps, seed=42)
-    b = ops.sample(logits, temps, seed=42)
-    c = ops.sample(logits, temps, seed=43)
+    a = ops.sample(logits, temps, _seeds([42, 42, 7, 7]), 3)
+    b = ops.sample(logits, temps, _seeds([42, 42, 7, 7]), 3)
+    c = ops.sample(logits, temps, _seeds([43, 42, 7, 7]), 3)
+    d = ops.sample(logits, temps, _seeds([42, 42, 7, 7]), 4)
     assert torch.equal(a, b)
-    assert not torch.equal(a, c)   # overwhelmingly likely
+    assert a[0] == a[1] and a[2] == a[3]   # same seed+step -> same draw
+    assert not torch.equal(a, c) or not torch.equal(a, d)
 
 
 def test_fused_qkv_ops_match_unfused():

@@ -97,5 +97,6 @@ def test_rope_kv_roundtrip():
 
 def test_sample_greedy():
     logits = torch.randn(5, 100)
-    out = ref.sample(logits.bfloat16(), torch.zeros(5))
+    out = ref.sample(logits.bfloat16(), torch.zeros(5),
+                     torch.zeros(5, dtype=torch.int64))
     assert torch.equal(out.long(), logits.bfloat16().float().argmax(-1))
