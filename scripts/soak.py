@@ -1,0 +1,111 @@
+#!/usr/bin/env python3
+"""Mixed-load soak: streaming + non-streaming + temperatures + mid-run
+worker kill/respawn against a 2-replica single-GPU gateway.  Reports
+success rates and verifies no in-flight leaks at the end."""
+import os
+import random
+import sys
+import threading
+import time
+from types import SimpleNamespace
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+sys.path.insert(0, os.path.join(REPO, "demos"))
+import importlib.util
+spec = importlib.util.spec_from_file_location(
+    "_common", os.path.join(REPO, "demos", "_common.py"))
+common = importlib.util.module_from_spec(spec)
+spec.loader.exec_module(common)
+
+from resilient_llm_amd.client import APIError, OpenAIClient  # noqa: E402
+
+CONFIG = {
+    "cluster": {"port": 4999},
+    "model_list": [
+        {"model_name": "soak", "litellm_params": {"model": "gpu/0/llama-3-8b"},
+         "model_info": {"id": "gpu0/soak"}},
+        {"model_name": "soak", "litellm_params": {"model": "gpu/0.1/llama-3-8b"},
+         "model_info": {"id": "gpu0.1/soak"}},
+    ],
+    "router_settings": {"routing_strategy": "simple-shuffle",
+                        "enable_pre_call_checks": False},
+}
+
+
+def main(duration_s=420, n_threads=24):
+    import yaml, tempfile
+    cfgfile = os.path.join(tempfile.mkdtemp(), "soak.yaml")
+    with open(cfgfile, "w") as f:
+        yaml.safe_dump(CONFIG, f)
+    args = SimpleNamespace(base_url=None, config=cfgfile, gpu=True)
+    stats = {"ok": 0, "err": 0, "stream_ok": 0, "stream_err": 0}
+    lock = threading.Lock()
+    stop = threading.Event()
+
+    with common.gateway_session(args) as (client, config):
+        base = f"http://{client.host}:{client.port}"
+
+        def worker(tid):
+            rng = random.Random(tid)
+            c = OpenAIClient(base, api_key=f"sk-{tid}")
+            while not stop.is_set():
+                msgs = [{"role": "user",
+                         "content": "x" * rng.randint(16, 400)}]
+                stream = rng.random() < 0.3
+                try:
+                    if stream:
+                        s = c.chat.completions.create(
+                            model="soak", messages=msgs,
+                            max_tokens=rng.randint(4, 48), stream=True,
+                            temperature=rng.choice([0.0, 0.8]), timeout=120)
+                        text, _ = s.collect_text()
+                        with lock:
+                            stats["stream_ok"] += 1
+                    else:
+                        c.chat.completions.create(
+                            model="soak", messages=msgs,
+                            max_tokens=rng.randint(4, 48),
+                            temperature=rng.choice([0.0, 0.8]),
+                            top_p=rng.choice([1.0, 0.9]), timeout=120)
+                        with lock:
+                            stats["ok"] += 1
+                except Exception:
+                    with lock:
+                        stats["stream_err" if stream else "err"] += 1
+
+        threads = [threading.Thread(target=worker, args=(i,), daemon=True)
+                   for i in range(n_threads)]
+        for t in threads:
+            t.start()
+        t0 = time.time()
+        killed = 0
+        while time.time() - t0 < duration_s:
+            time.sleep(5)
+            el = time.time() - t0
+            if killed < 2 and el > 120 * (killed + 1):
+                dev = "gpu:0" if killed == 0 else "gpu:0.1"
+                print(f"[soak {el:.0f}s] killing {dev}", flush=True)
+                try:
+                    client.inject_fault(dev, "kill")
+                except APIError:
+                    pass
+                killed += 1
+            with lock:
+                print(f"[soak {el:.0f}s] {dict(stats)}", flush=True)
+        stop.set()
+        for t in threads:
+            t.join(timeout=130)
+        time.sleep(2)
+        rows = {r["model_id"]: r for r in client.router_state()["deployments"]}
+        inflight = {k: r["in_flight"] for k, r in rows.items()}
+        healthy = {k: r["healthy"] for k, r in rows.items()}
+        total = sum(stats.values())
+        ok = stats["ok"] + stats["stream_ok"]
+        print(f"SOAK DONE: {ok}/{total} ok ({100*ok/max(total,1):.1f}%), "
+              f"2 worker kills+respawns; in_flight={inflight}; "
+              f"healthy={healthy}", flush=True)
+
+
+if __name__ == "__main__":
+    main()
