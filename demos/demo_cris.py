@@ -24,7 +24,7 @@ from _common import (
     add_common_args, gateway_session, log_with_timestamp, print_table,
 )
 
-from resilient_llm_amd.client import APIError, OpenAIClient, RateLimitError
+from resilient_llm_amd.client import OpenAIClient, RateLimitError
 
 QUESTIONS = [
     "Summarize the benefits of cross-device inference in one sentence.",

@@ -26,7 +26,7 @@ from _common import (
     print_table,
 )
 
-from resilient_llm_amd.client import OpenAIClient, RateLimitError
+from resilient_llm_amd.client import RateLimitError
 from resilient_llm_amd.utils.logging import sanitize_error
 
 ALIAS = "llama-fallback-demo"

@@ -60,9 +60,8 @@ class SeqState:
     n_cached: int = 0
     arrived_at: float = dataclasses.field(default_factory=time.monotonic)
 
-    @property
-    def reserved_blocks_needed(self) -> int:
-        return -(-(len(self.prompt_ids) + self.params.max_tokens) // 16)
+    def blocks_needed(self, block_size: int) -> int:
+        return -(-(len(self.prompt_ids) + self.params.max_tokens) // block_size)
 
 
 class CapacityExceeded(RuntimeError):
@@ -142,10 +141,10 @@ class LLMEngine:
                 f"prompt+max_tokens = "
                 f"{len(seq.prompt_ids) + params.max_tokens} exceeds "
                 f"max_model_len {self.max_model_len}")
-        if seq.reserved_blocks_needed > self.kv.num_blocks:
+        if seq.blocks_needed(self.block_size) > self.kv.num_blocks:
             raise CapacityExceeded(
-                f"request needs {seq.reserved_blocks_needed} KV blocks, "
-                f"cache has {self.kv.num_blocks}")
+                f"request needs {seq.blocks_needed(self.block_size)} KV "
+                f"blocks, cache has {self.kv.num_blocks}")
         with self._queue_lock:
             self.waiting.append(seq)
 
@@ -184,7 +183,7 @@ class LLMEngine:
             if seq.block_keys:
                 max_reuse = (len(seq.prompt_ids) - 1) // self.block_size
                 reused = self.kv.lookup_prefix(seq.block_keys[:max_reuse])
-            needed = seq.reserved_blocks_needed - len(reused)
+            needed = seq.blocks_needed(self.block_size) - len(reused)
             if needed > self.kv.free_blocks:
                 if reused:
                     self.kv.free(reused)

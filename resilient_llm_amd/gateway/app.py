@@ -35,7 +35,7 @@ from ..router.core import (
 )
 from ..router.token_bucket import MinuteWindowLimiter
 from ..workers.base import (
-    GenerationRequest, GenerationResult, Worker, WorkerDead, WorkerError,
+    GenerationRequest, GenerationResult, Worker, WorkerError,
     WorkerRegistry, WorkerThrottled,
 )
 from ..workers.stub import estimate_tokens

@@ -16,7 +16,6 @@ from __future__ import annotations
 
 import argparse
 import asyncio
-import os
 
 
 def main() -> None:

@@ -11,8 +11,6 @@ what the no-GPU plumbing tests and the gloo multi-process tests use.
 from __future__ import annotations
 
 import os
-from typing import Optional
-
 import torch
 
 from . import ref  # noqa: F401

@@ -175,7 +175,8 @@ class EngineWorker(Worker):
         try:
             if self.tp_control is not None:
                 # leader: validate locally, then lockstep-broadcast the op
-                blocks_needed = -(-(len(prompt_ids) + params.max_tokens) // 16)
+                blocks_needed = -(-(len(prompt_ids) + params.max_tokens)
+                                  // self.engine.block_size)
                 if blocks_needed > self.engine.kv.num_blocks:
                     raise CapacityExceeded(
                         f"request needs {blocks_needed} KV blocks, cache has "

@@ -10,7 +10,7 @@ import threading
 
 from resilient_llm_amd.client import OpenAIClient
 from resilient_llm_amd.config import load_config
-from resilient_llm_amd.gateway.server import build_registry, serve
+from resilient_llm_amd.gateway.server import build_registry
 
 
 def free_port() -> int:

@@ -6,8 +6,6 @@ import os
 import subprocess
 import sys
 
-import pytest
-
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 DEMOS = os.path.join(REPO, "demos")
 

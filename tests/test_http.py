@@ -4,8 +4,6 @@ errors, chunked SSE framing)."""
 import asyncio
 import json
 
-import pytest
-
 from resilient_llm_amd.gateway.http import HttpServer, Request, Response
 from tests.gateway_harness import free_port
 

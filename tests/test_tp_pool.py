@@ -10,7 +10,6 @@ import os
 import tempfile
 
 import pytest
-import torch
 
 from resilient_llm_amd.config import PoolDef
 from resilient_llm_amd.workers.base import GenerationRequest

@@ -4,7 +4,6 @@ end-to-end slice: gateway -> engine worker -> tiny model."""
 import asyncio
 import json
 import os
-import tempfile
 
 import pytest
 import torch
