@@ -70,7 +70,8 @@ void prefill_mfma_kernel(const uint16_t* __restrict__ qkv,
   // LDS: V^T [D][PPAD] shared + per-wave P [QBLK][PPAD]
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   uint16_t* vt = reinterpret_cast<uint16_t*>(smem_raw);          // D*PPAD
-  uint16_t* p_lds = vt + D * PPAD + wave * QBLK * PPAD;          // per wave
+  uint16_t* ks = vt + D * PPAD;                                  // KVBLK*D, swizzled
+  uint16_t* p_lds = ks + KVBLK * D + wave * QBLK * PPAD;         // per wave
 
   // ---- load Q fragments: [2 q-tiles][DC] x 4 VGPR ----
   bf16x8_vec qf[2][DC];
@@ -117,14 +118,25 @@ void prefill_mfma_kernel(const uint16_t* __restrict__ qkv,
         const int d8 = (idx % (D / 8)) * 8;
         const int row = kv0 + kv;
         uint4 raw = {0, 0, 0, 0};
-        if (row < seq_end)
+        uint4 kraw = {0, 0, 0, 0};
+        if (row < seq_end) {
           raw = *reinterpret_cast<const uint4*>(
               qkv + int64_t(row) * qkv_stride + v_off + kvh * D + d8);
+          kraw = *reinterpret_cast<const uint4*>(
+              qkv + int64_t(row) * qkv_stride + kv_off + kvh * D + d8);
+        }
         bf16x8 piece;
         piece.u = raw;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           vt[(d8 + j) * PPAD + kv] = piece.s[j];
+        // K stays row-major but XOR-swizzled so the B-fragment read
+        // (16 lanes, 16 different rows, same column range) is
+        // bank-conflict-free (guide T2 / G4); mask by row length so the
+        // XOR never escapes the row (D=64 rows are only 128 B)
+        *reinterpret_cast<uint4*>(
+            reinterpret_cast<char*>(ks) +
+            (kv * D * 2 + ((d8 * 2) ^ ((kv & (D / 8 - 1)) << 4)))) = kraw;
       }
     }
     __syncthreads();
@@ -138,16 +150,16 @@ void prefill_mfma_kernel(const uint16_t* __restrict__ qkv,
     if (active) {
 #pragma unroll
       for (int kt = 0; kt < 2; ++kt) {
-        // B fragments for this 16-kv column tile, all DC chunks
+        // B fragments from the swizzled LDS K tile (staged once per
+        // workgroup instead of 4x-redundant scattered L2 reads)
         bf16x8_vec kf[DC];
+        const int krow = kt * 16 + jcol;
 #pragma unroll
         for (int dc = 0; dc < DC; ++dc) {
-          const int krow = kv0 + kt * 16 + jcol;
-          uint4 raw = {0, 0, 0, 0};
-          if (krow < seq_end)
-            raw = *reinterpret_cast<const uint4*>(
-                qkv + int64_t(krow) * qkv_stride + kv_off + kvh * D +
-                dc * 32 + koct * 8);
+          uint4 raw = *reinterpret_cast<const uint4*>(
+              reinterpret_cast<const char*>(ks) +
+              (krow * D * 2 +
+               (((dc * 32 + koct * 8) * 2) ^ ((krow & (D / 8 - 1)) << 4))));
           kf[dc] = *reinterpret_cast<bf16x8_vec*>(&raw);
         }
 #pragma unroll
@@ -245,7 +257,9 @@ void launch_prefill_mfma(const uint16_t* qkv, const int32_t* chunk_t0,
                          hipStream_t stream) {
   const int n_hw = group > 4 ? group / 4 : 1;
   const int blocks = n_chunks * n_kv_heads * n_hw;
-  const size_t smem = size_t(head_dim) * PPAD * 2 + size_t(4) * QBLK * PPAD * 2;
+  const size_t smem = size_t(head_dim) * PPAD * 2 +
+                      size_t(KVBLK) * head_dim * 2 +
+                      size_t(4) * QBLK * PPAD * 2;
   if (head_dim == 128) {
     hipLaunchKernelGGL(prefill_mfma_kernel<128>, dim3(blocks), dim3(256),
                        smem, stream, qkv, chunk_t0, chunk_seq_start,

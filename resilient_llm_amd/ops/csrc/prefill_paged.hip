@@ -65,7 +65,8 @@ void prefill_paged_kernel(const uint16_t* __restrict__ qkv,
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   uint16_t* vt = reinterpret_cast<uint16_t*>(smem_raw);
-  uint16_t* p_lds = vt + D * PPAD + wave * QBLK * PPAD;
+  uint16_t* ks = vt + D * PPAD;                                  // KVBLK*D, swizzled
+  uint16_t* p_lds = ks + KVBLK * D + wave * QBLK * PPAD;
 
   bf16x8_vec qf[2][DC];
 #pragma unroll
@@ -116,13 +117,20 @@ void prefill_paged_kernel(const uint16_t* __restrict__ qkv,
         const int d8 = (idx % (D / 8)) * 8;
         const int p = kv0 + kv;
         uint4 raw = {0, 0, 0, 0};
-        if (p < kv_hi)
-          raw = *reinterpret_cast<const uint4*>(v_cache + kv_addr(p) + d8);
+        uint4 kraw = {0, 0, 0, 0};
+        if (p < kv_hi) {
+          const int64_t a = kv_addr(p);
+          raw = *reinterpret_cast<const uint4*>(v_cache + a + d8);
+          kraw = *reinterpret_cast<const uint4*>(k_cache + a + d8);
+        }
         bf16x8 piece;
         piece.u = raw;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           vt[(d8 + j) * PPAD + kv] = piece.s[j];
+        *reinterpret_cast<uint4*>(
+            reinterpret_cast<char*>(ks) +
+            (kv * D * 2 + ((d8 * 2) ^ ((kv & (D / 8 - 1)) << 4)))) = kraw;
       }
     }
     __syncthreads();
@@ -136,14 +144,13 @@ void prefill_paged_kernel(const uint16_t* __restrict__ qkv,
 #pragma unroll
       for (int kt = 0; kt < 2; ++kt) {
         bf16x8_vec kf[DC];
-        const int p = kv0 + kt * 16 + jcol;
-        const int64_t kaddr = p < kv_hi ? kv_addr(p) : 0;
+        const int krow = kt * 16 + jcol;
 #pragma unroll
         for (int dc = 0; dc < DC; ++dc) {
-          uint4 raw = {0, 0, 0, 0};
-          if (p < kv_hi)
-            raw = *reinterpret_cast<const uint4*>(
-                k_cache + kaddr + dc * 32 + koct * 8);
+          uint4 raw = *reinterpret_cast<const uint4*>(
+              reinterpret_cast<const char*>(ks) +
+              (krow * D * 2 +
+               (((dc * 32 + koct * 8) * 2) ^ ((krow & (D / 8 - 1)) << 4))));
           kf[dc] = *reinterpret_cast<bf16x8_vec*>(&raw);
         }
 #pragma unroll
@@ -237,7 +244,9 @@ void launch_prefill_paged(const uint16_t* qkv, const uint16_t* k_cache,
                           int block_size, float scale, hipStream_t stream) {
   const int n_hw = group > 4 ? group / 4 : 1;
   const int blocks = n_chunks * n_kv_heads * n_hw;
-  const size_t smem = size_t(head_dim) * PPAD * 2 + size_t(4) * QBLK * PPAD * 2;
+  const size_t smem = size_t(head_dim) * PPAD * 2 +
+                      size_t(KVBLK) * head_dim * 2 +
+                      size_t(4) * QBLK * PPAD * 2;
   if (head_dim == 128) {
     hipLaunchKernelGGL(prefill_paged_kernel<128>, dim3(blocks), dim3(256),
                        smem, stream, qkv, k_cache, v_cache, chunk_row0,
