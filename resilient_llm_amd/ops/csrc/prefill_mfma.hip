@@ -34,8 +34,8 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 
 constexpr float kNegInf = -1e30f;
 constexpr int QBLK = 32;
-constexpr int KVBLK = 32;
-constexpr int PPAD = 40;      // padded row stride (elements) for P / V^T
+constexpr int KVBLK = 64;
+constexpr int PPAD = KVBLK + 8;   // padded row stride (elements) for P / V^T
 
 template <int D>
 __global__ __launch_bounds__(256)
@@ -141,15 +141,16 @@ void prefill_mfma_kernel(const uint16_t* __restrict__ qkv,
     }
     __syncthreads();
 
-    // ---- S = Q K^T over this tile (2x2 16x16 tiles) ----
-    f32x4 s_acc[2][2];
+    // ---- S = Q K^T over this tile (2 x KVBLK/16 16x16 tiles) ----
+    constexpr int KT = KVBLK / 16;
+    f32x4 s_acc[2][KT];
 #pragma unroll
     for (int qt = 0; qt < 2; ++qt)
 #pragma unroll
-      for (int kt = 0; kt < 2; ++kt) s_acc[qt][kt] = f32x4{0.f, 0.f, 0.f, 0.f};
+      for (int kt = 0; kt < KT; ++kt) s_acc[qt][kt] = f32x4{0.f, 0.f, 0.f, 0.f};
     if (active) {
 #pragma unroll
-      for (int kt = 0; kt < 2; ++kt) {
+      for (int kt = 0; kt < KT; ++kt) {
         // B fragments from the swizzled LDS K tile (staged once per
         // workgroup instead of 4x-redundant scattered L2 reads)
         bf16x8_vec kf[DC];
@@ -177,31 +178,36 @@ void prefill_mfma_kernel(const uint16_t* __restrict__ qkv,
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row_g = t0 + qt * 16 + koct * 4 + r;
-          float s0 = s_acc[qt][0][r] * scale;
-          float s1 = s_acc[qt][1][r] * scale;
-          const int c0 = kv0 + jcol;
-          const int c1 = kv0 + 16 + jcol;
-          if (c0 > row_g || c0 >= seq_end) s0 = kNegInf;
-          if (c1 > row_g || c1 >= seq_end) s1 = kNegInf;
+          float sv[KT];
+          float mx = kNegInf;
+#pragma unroll
+          for (int kt = 0; kt < KT; ++kt) {
+            float s = s_acc[qt][kt][r] * scale;
+            const int c = kv0 + kt * 16 + jcol;
+            if (c > row_g || c >= seq_end) s = kNegInf;
+            sv[kt] = s;
+            mx = fmaxf(mx, s);
+          }
           // row max across the 16-lane group
-          float mx = fmaxf(s0, s1);
 #pragma unroll
           for (int off = 8; off > 0; off >>= 1)
             mx = fmaxf(mx, __shfl_xor(mx, off, kWave));
           const float m_new = fmaxf(m_st[qt][r], mx);
           corr[r] = __expf(m_st[qt][r] - m_new);
           m_st[qt][r] = m_new;
-          const float p0 = __expf(s0 - m_new);
-          const float p1 = __expf(s1 - m_new);
-          float rs = p0 + p1;
+          float rs = 0.f;
+          // store P row (bf16) to the per-wave LDS buffer
+          const int prow = qt * 16 + koct * 4 + r;
+#pragma unroll
+          for (int kt = 0; kt < KT; ++kt) {
+            const float p = __expf(sv[kt] - m_new);
+            rs += p;
+            p_lds[prow * PPAD + kt * 16 + jcol] = f32_to_bf16(p);
+          }
 #pragma unroll
           for (int off = 8; off > 0; off >>= 1)
             rs += __shfl_xor(rs, off, kWave);
           l_st[qt][r] = l_st[qt][r] * corr[r] + rs;
-          // store P row (bf16) to the per-wave LDS buffer
-          const int prow = qt * 16 + koct * 4 + r;
-          p_lds[prow * PPAD + jcol] = f32_to_bf16(p0);
-          p_lds[prow * PPAD + 16 + jcol] = f32_to_bf16(p1);
         }
         // rescale O accumulators for this q-tile
 #pragma unroll
@@ -213,16 +219,19 @@ void prefill_mfma_kernel(const uint16_t* __restrict__ qkv,
       // ---- PV: A = P (from LDS, A-frag layout), B = V^T rows ----
 #pragma unroll
       for (int qt = 0; qt < 2; ++qt) {
-        uint4 praw = *reinterpret_cast<const uint4*>(
-            p_lds + (qt * 16 + jcol) * PPAD + koct * 8);
-        bf16x8_vec pfrag = *reinterpret_cast<bf16x8_vec*>(&praw);
 #pragma unroll
-        for (int ct = 0; ct < CT; ++ct) {
-          uint4 vraw = *reinterpret_cast<const uint4*>(
-              vt + (ct * 16 + jcol) * PPAD + koct * 8);
-          bf16x8_vec vfrag = *reinterpret_cast<bf16x8_vec*>(&vraw);
-          o_acc[qt][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              pfrag, vfrag, o_acc[qt][ct], 0, 0, 0);
+        for (int kb = 0; kb < KVBLK / 32; ++kb) {
+          uint4 praw = *reinterpret_cast<const uint4*>(
+              p_lds + (qt * 16 + jcol) * PPAD + kb * 32 + koct * 8);
+          bf16x8_vec pfrag = *reinterpret_cast<bf16x8_vec*>(&praw);
+#pragma unroll
+          for (int ct = 0; ct < CT; ++ct) {
+            uint4 vraw = *reinterpret_cast<const uint4*>(
+                vt + (ct * 16 + jcol) * PPAD + kb * 32 + koct * 8);
+            bf16x8_vec vfrag = *reinterpret_cast<bf16x8_vec*>(&vraw);
+            o_acc[qt][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                pfrag, vfrag, o_acc[qt][ct], 0, 0, 0);
+          }
         }
       }
     }

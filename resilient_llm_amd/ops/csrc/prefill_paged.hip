@@ -12,8 +12,7 @@
 //
 // Cache block row [phys, kvh, tok, 0:D] is contiguous, so the QK^T
 // B fragment is still a single 16-B load per lane, one block-table
-// lookup per 16 kv positions (hoisted per 32-token tile: a tile spans
-// exactly two cache blocks).
+// lookup per 16 kv positions (a KVBLK=64 tile spans four cache blocks).
 
 #include "common.h"
 
@@ -26,11 +25,11 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 
 constexpr float kNegInf = -1e30f;
 constexpr int QBLK = 32;
-constexpr int KVBLK = 32;
-constexpr int PPAD = 40;
+constexpr int KVBLK = 64;
+constexpr int PPAD = KVBLK + 8;
 
 template <int D>
-__global__ __launch_bounds__(256)
+__global__ __launch_bounds__(256, 2)
 void prefill_paged_kernel(const uint16_t* __restrict__ qkv,
                           const uint16_t* __restrict__ k_cache,
                           const uint16_t* __restrict__ v_cache,
@@ -135,14 +134,15 @@ void prefill_paged_kernel(const uint16_t* __restrict__ qkv,
     }
     __syncthreads();
 
-    f32x4 s_acc[2][2];
+    constexpr int KT = KVBLK / 16;
+    f32x4 s_acc[2][KT];
 #pragma unroll
     for (int qt = 0; qt < 2; ++qt)
 #pragma unroll
-      for (int kt = 0; kt < 2; ++kt) s_acc[qt][kt] = f32x4{0.f, 0.f, 0.f, 0.f};
+      for (int kt = 0; kt < KT; ++kt) s_acc[qt][kt] = f32x4{0.f, 0.f, 0.f, 0.f};
     if (active) {
 #pragma unroll
-      for (int kt = 0; kt < 2; ++kt) {
+      for (int kt = 0; kt < KT; ++kt) {
         bf16x8_vec kf[DC];
         const int krow = kt * 16 + jcol;
 #pragma unroll
@@ -167,29 +167,34 @@ void prefill_paged_kernel(const uint16_t* __restrict__ qkv,
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int qpos = pos0 + qt * 16 + koct * 4 + r;   // seq position
-          float s0 = s_acc[qt][0][r] * scale;
-          float s1 = s_acc[qt][1][r] * scale;
-          const int c0 = kv0 + jcol;
-          const int c1 = kv0 + 16 + jcol;
-          if (c0 > qpos || c0 >= kv_hi) s0 = kNegInf;
-          if (c1 > qpos || c1 >= kv_hi) s1 = kNegInf;
-          float mx = fmaxf(s0, s1);
+          float sv[KT];
+          float mx = kNegInf;
+#pragma unroll
+          for (int kt = 0; kt < KT; ++kt) {
+            float s = s_acc[qt][kt][r] * scale;
+            const int c = kv0 + kt * 16 + jcol;
+            if (c > qpos || c >= kv_hi) s = kNegInf;
+            sv[kt] = s;
+            mx = fmaxf(mx, s);
+          }
 #pragma unroll
           for (int off = 8; off > 0; off >>= 1)
             mx = fmaxf(mx, __shfl_xor(mx, off, kWave));
           const float m_new = fmaxf(m_st[qt][r], mx);
           corr[r] = __expf(m_st[qt][r] - m_new);
           m_st[qt][r] = m_new;
-          const float p0 = __expf(s0 - m_new);
-          const float p1 = __expf(s1 - m_new);
-          float rs = p0 + p1;
+          float rs = 0.f;
+          const int prow = qt * 16 + koct * 4 + r;
+#pragma unroll
+          for (int kt = 0; kt < KT; ++kt) {
+            const float p = __expf(sv[kt] - m_new);
+            rs += p;
+            p_lds[prow * PPAD + kt * 16 + jcol] = f32_to_bf16(p);
+          }
 #pragma unroll
           for (int off = 8; off > 0; off >>= 1)
             rs += __shfl_xor(rs, off, kWave);
           l_st[qt][r] = l_st[qt][r] * corr[r] + rs;
-          const int prow = qt * 16 + koct * 4 + r;
-          p_lds[prow * PPAD + jcol] = f32_to_bf16(p0);
-          p_lds[prow * PPAD + 16 + jcol] = f32_to_bf16(p1);
         }
 #pragma unroll
         for (int ct = 0; ct < CT; ++ct)
@@ -199,16 +204,19 @@ void prefill_paged_kernel(const uint16_t* __restrict__ qkv,
 
 #pragma unroll
       for (int qt = 0; qt < 2; ++qt) {
-        uint4 praw = *reinterpret_cast<const uint4*>(
-            p_lds + (qt * 16 + jcol) * PPAD + koct * 8);
-        bf16x8_vec pfrag = *reinterpret_cast<bf16x8_vec*>(&praw);
 #pragma unroll
-        for (int ct = 0; ct < CT; ++ct) {
-          uint4 vraw = *reinterpret_cast<const uint4*>(
-              vt + (ct * 16 + jcol) * PPAD + koct * 8);
-          bf16x8_vec vfrag = *reinterpret_cast<bf16x8_vec*>(&vraw);
-          o_acc[qt][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              pfrag, vfrag, o_acc[qt][ct], 0, 0, 0);
+        for (int kb = 0; kb < KVBLK / 32; ++kb) {
+          uint4 praw = *reinterpret_cast<const uint4*>(
+              p_lds + (qt * 16 + jcol) * PPAD + kb * 32 + koct * 8);
+          bf16x8_vec pfrag = *reinterpret_cast<bf16x8_vec*>(&praw);
+#pragma unroll
+          for (int ct = 0; ct < CT; ++ct) {
+            uint4 vraw = *reinterpret_cast<const uint4*>(
+                vt + (ct * 16 + jcol) * PPAD + kb * 32 + koct * 8);
+            bf16x8_vec vfrag = *reinterpret_cast<bf16x8_vec*>(&vraw);
+            o_acc[qt][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                pfrag, vfrag, o_acc[qt][ct], 0, 0, 0);
+          }
         }
       }
     }
