@@ -108,4 +108,9 @@ def main(duration_s=420, n_threads=24):
 
 
 if __name__ == "__main__":
-    main()
+    import argparse
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--duration", type=int, default=420)
+    ap.add_argument("--threads", type=int, default=24)
+    a = ap.parse_args()
+    main(duration_s=a.duration, n_threads=a.threads)
