@@ -110,4 +110,7 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
         client = RpcWorkerClient(f"pool:{pool_name}", {model_name}, sock)
         client.proc = procs[0]
         client.proc_group = procs
+        client.respawn = (lambda pd=pool_def, m=model_name, s=sock:
+                          spawn_pool_worker(pd, m, s, kv_gb=kv_gb,
+                                            max_batch=max_batch))
         registry.register("pool", pool_name, client)

@@ -264,13 +264,27 @@ class RpcWorkerClient(Worker):
 
     async def respawn_now(self) -> bool:
         """Elastic recovery: restart a dead worker
-        process and reconnect.  Returns True once serving again."""
+        process and reconnect.  Returns True once serving again.
+
+        TP pools restart as a WHOLE group (every rank must rejoin a
+        fresh RCCL rendezvous): surviving followers are reaped first,
+        and a ``respawn`` hook that returns a list re-populates
+        ``proc_group`` with the new ranks (leader first)."""
         import time as _time
         if self.respawn is None:
             return False
         if self.proc is not None and self.proc.poll() is None:
             return True
         self.last_respawn = _time.monotonic()
+        followers = (self.proc_group or [])[1:]
+        for p in followers:
+            if p.poll() is None:
+                p.terminate()
+        for p in followers:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                p.kill()
         if self._reader_task:
             self._reader_task.cancel()
             self._reader_task = None
@@ -280,7 +294,12 @@ class RpcWorkerClient(Worker):
             _os.unlink(self.socket_path)
         except OSError:
             pass
-        self.proc = self.respawn()
+        spawned = self.respawn()
+        if isinstance(spawned, list):
+            self.proc_group = spawned
+            self.proc = spawned[0]
+        else:
+            self.proc = spawned
         await self.connect(timeout=900)
         return True
 

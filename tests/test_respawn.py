@@ -103,3 +103,59 @@ def test_worker_respawn_after_kill():
     finally:
         loop.call_soon_threadsafe(holder["stop"].set)
         th.join(timeout=15)
+
+
+def test_pool_group_respawn():
+    """A dead TP pool restarts as a WHOLE group: surviving followers are
+    reaped, a fresh rank group is spawned (new RCCL/gloo rendezvous),
+    and the pool serves again (ROADMAP item: pool elastic recovery)."""
+    import os
+    import tempfile
+
+    from resilient_llm_amd.config import PoolDef
+    from resilient_llm_amd.workers.base import GenerationRequest
+    from resilient_llm_amd.workers.pool import spawn_pool_worker
+    from resilient_llm_amd.workers.rpc import RpcWorkerClient
+
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    sock = os.path.join(tempfile.mkdtemp(prefix="rlli-pr-"), "pool.sock")
+    pool = PoolDef(name="pr", gpus=[0, 1], tensor_parallel=2)
+
+    def spawn():
+        return spawn_pool_worker(pool, "tiny", sock, device_override="cpu",
+                                 tp_backend="gloo", max_batch=4)
+
+    procs = spawn()
+
+    def greq(rid):
+        return GenerationRequest(request_id=rid, model="tiny",
+                                 messages=[{"role": "user", "content": "hi"}],
+                                 max_tokens=4)
+
+    async def run():
+        client = RpcWorkerClient("pool:pr", {"tiny"}, sock)
+        client.proc = procs[0]
+        client.proc_group = procs
+        client.respawn = spawn
+        await client.connect(timeout=180)
+        res = await client.generate(greq("a"))
+        assert res.completion_tokens == 4
+
+        # kill the LEADER hard; follower becomes an orphan
+        procs[0].kill()
+        procs[0].wait(timeout=10)
+
+        assert await client.respawn_now()
+        # group was replaced wholesale
+        assert client.proc_group is not procs
+        assert all(p.poll() is not None for p in procs)
+        res = await client.generate(greq("b"))
+        assert res.completion_tokens == 4
+        await client.close()
+
+    try:
+        asyncio.run(run())
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                p.kill()
