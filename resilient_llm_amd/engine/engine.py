@@ -77,6 +77,7 @@ class LLMEngine:
                  chunk_size: int = 2048,
                  admission_window_s: float = 0.0,
                  enable_prefix_caching: bool = True,
+                 target_step_ms: Optional[float] = None,
                  seed: int = 0) -> None:
         self.model = model
         self.kv = kv_cache
@@ -102,6 +103,14 @@ class LLMEngine:
         # while arrivals are still landing (MUST be 0 under TP lockstep —
         # wall-clock decisions would diverge across ranks)
         self.admission_window_s = admission_window_s
+        # latency-targeted scheduler: when set, the per-step prefill
+        # token budget is AIMD-tuned so mixed steps stay near this bound
+        # (decode cadence / inter-token-latency SLO).  MUST be None
+        # under TP lockstep — wall-clock decisions diverge across ranks
+        # (same rule as admission_window_s).
+        self.target_step_ms = target_step_ms
+        self._prefill_budget = max_prefill_tokens
+        self._budget_floor = min(max_prefill_tokens, 256)
         self.step_count = 0
         self.block_size = kv_cache.block_size
         self._aborted: set[str] = set()
@@ -242,9 +251,11 @@ class LLMEngine:
             outs += new_outs
             if _ROCTX:
                 torch.cuda.nvtx.range_pop()
+            elapsed = time.monotonic() - t0
             self.stats["prefill_steps"] += 1
-            self.stats["prefill_time"] += time.monotonic() - t0
+            self.stats["prefill_time"] += elapsed
             self.stats["prefill_tokens"] += n_chunk_tokens
+            self._tune_budget(elapsed)
             return outs
         if self.running:
             t0 = time.monotonic()
@@ -327,6 +338,23 @@ class LLMEngine:
         gen.manual_seed(self._seq_seed(s))
         return int(idx[int(torch.multinomial(sp, 1, generator=gen))])
 
+    def _tune_budget(self, elapsed_s: float) -> None:
+        """AIMD latency-targeted scheduling: multiplicative decrease of
+        the per-step prefill token budget proportional to overshoot of
+        target_step_ms, additive increase while comfortably under it —
+        long prompt bursts cannot stall the decode cadence for longer
+        than (roughly) the SLO."""
+        if self.target_step_ms is None:
+            return
+        ms = elapsed_s * 1e3
+        if ms > self.target_step_ms:
+            scaled = int(self._prefill_budget * self.target_step_ms / ms * 0.9)
+            self._prefill_budget = max(self._budget_floor, scaled)
+        elif ms < 0.8 * self.target_step_ms:
+            self._prefill_budget = min(
+                self.max_prefill_tokens,
+                self._prefill_budget + max(64, self.chunk_size // 4))
+
     def _mixed_step(self) -> tuple[int, list[StepOutput]]:
         """One forward over [decode rows | prefill-chunk rows]: running
         sequences decode while prefilling sequences advance by up to
@@ -342,7 +370,7 @@ class LLMEngine:
         slots = [self._slot(s, s.n_cached) for s in seqs_d]
         seq_lens_d = [s.n_cached + 1 for s in seqs_d]
 
-        budget = self.max_prefill_tokens
+        budget = self._prefill_budget
         chunk_plan: list = []
         for seq in self.prefilling:
             if budget <= 0:

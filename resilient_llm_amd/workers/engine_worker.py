@@ -62,6 +62,7 @@ class EngineWorker(Worker):
                  use_graphs: bool = False,
                  tp_rank: int = 0, tp_world: int = 1, tp_group=None,
                  tp_control=None,
+                 target_step_ms: Optional[float] = None,
                  seed: int = 0) -> None:
         super().__init__(device=device_label or f"gpu:{device}",
                          models={model_name})
@@ -86,7 +87,11 @@ class EngineWorker(Worker):
                                 max_prefill_tokens=max_prefill_tokens,
                                 chunk_size=chunk_size,
                                 admission_window_s=0.0 if tp_control is not None
-                                else 0.006)
+                                else 0.006,
+                                # wall-clock tuning diverges under TP
+                                # lockstep, like admission_window_s
+                                target_step_ms=None if tp_control is not None
+                                else target_step_ms)
         if use_graphs and torch_device != "cpu":
             from ..engine.graph import install_graph_runner
             install_graph_runner(self.engine)

@@ -18,6 +18,9 @@ def main() -> None:
     ap.add_argument("--max-batch", type=int, default=64)
     ap.add_argument("--graphs", action="store_true")
     ap.add_argument("--num-blocks", type=int, default=None)
+    ap.add_argument("--target-step-ms", type=float, default=None,
+                    help="AIMD prefill budget: keep mixed steps near "
+                         "this bound (decode-cadence SLO)")
     ap.add_argument("--device", default=None,
                     help="torch device override (tests: cpu)")
     args = ap.parse_args()
@@ -45,7 +48,8 @@ def main() -> None:
                               kv_gb=args.kv_gb,
                               max_batch_size=args.max_batch,
                               num_blocks=args.num_blocks,
-                              use_graphs=args.graphs)
+                              use_graphs=args.graphs,
+                              target_step_ms=args.target_step_ms)
         log_with_timestamp(
             f"worker {args.device_label} ready: {args.model} on {device}, "
             f"{worker.engine.kv.num_blocks} KV blocks "

@@ -77,3 +77,44 @@ def test_many_seqs_chunked_concurrently():
                       SamplingParams(max_tokens=5))
     outs = drain(e)
     assert len(outs) == 6 and all(len(v) == 5 for v in outs.values())
+
+
+def test_latency_targeted_budget_shrinks_and_regrows():
+    """AIMD scheduler (target_step_ms): an impossible SLO collapses the
+    per-step prefill budget to the floor; a generous SLO grows it back
+    to max_prefill_tokens.  Output tokens must be unaffected (the budget
+    only re-slices chunks across steps)."""
+    prompt = list(range(10, 93))
+    ref = make_engine(chunk_size=2048)
+    ref.add_request("a", prompt, SamplingParams(max_tokens=7))
+    want = drain(ref)["a"]
+
+    # CPU steps take milliseconds >> 1e-6 ms -> every mixed step shrinks
+    e = make_engine(chunk_size=8, max_prefill_tokens=1024,
+                    target_step_ms=1e-6)
+    e.add_request("a", prompt, SamplingParams(max_tokens=7))
+    got = drain(e)["a"]
+    assert got == want
+    assert e._prefill_budget == e._budget_floor
+
+    # generous SLO: budget climbs back (additive increase per step)
+    e.target_step_ms = 1e9
+    for _ in range(600):
+        if e._prefill_budget >= e.max_prefill_tokens:
+            break
+        e.add_request(f"r{_}", prompt[:16], SamplingParams(max_tokens=1))
+        drain(e)
+    assert e._prefill_budget == e.max_prefill_tokens
+
+
+def test_budget_never_starves_prefill():
+    """Even at the floor, at least one chunk advances per step — a
+    too-tight SLO degrades throughput, never liveness."""
+    e = make_engine(chunk_size=8, max_prefill_tokens=1024,
+                    target_step_ms=1e-6)
+    for i in range(6):
+        e.add_request(f"s{i}", list(range(3, 70)),
+                      SamplingParams(max_tokens=3))
+    outs = drain(e)
+    assert len(outs) == 6
+    assert all(len(v) == 3 for v in outs.values())
