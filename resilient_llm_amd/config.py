@@ -104,6 +104,10 @@ class ClusterConfig:
     gpus: Optional[int] = None            # None -> autodetect
     pools: dict[str, PoolDef] = dataclasses.field(default_factory=dict)
     ledger_path: Optional[str] = None     # JSONL invocation log (X12)
+    # decode-cadence SLO for single-GPU workers: AIMD-bounds the tokens
+    # a mixed prefill step may take (engine target_step_ms); ignored by
+    # TP pools (wall-clock tuning diverges under lockstep)
+    target_step_ms: Optional[float] = None
 
 
 @dataclasses.dataclass
@@ -209,6 +213,7 @@ def load_config(path: str | os.PathLike[str] | None = None,
         gpus=cluster_raw.get("gpus"),
         pools=_parse_pools(cluster_raw.get("pools")),
         ledger_path=cluster_raw.get("ledger_path"),
+        target_step_ms=cluster_raw.get("target_step_ms"),
     )
 
     model_list = data.get("model_list") or []

@@ -30,6 +30,7 @@ REPO_ROOT = os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__fi
 def spawn_gpu_worker(device_index: str, model_name: str,
                      socket_path: str, kv_gb: float = 24.0,
                      max_batch: int = 64, use_graphs: bool = True,
+                     target_step_ms: float | None = None,
                      extra_env: dict | None = None) -> subprocess.Popen:
     """``device_index`` may be a sub-device replica like ``0.1``: several
     worker processes co-located on physical GPU 0 — 288 GB of HBM3E holds
@@ -48,6 +49,8 @@ def spawn_gpu_worker(device_index: str, model_name: str,
            "--max-batch", str(max_batch)]
     if use_graphs:
         cmd.append("--graphs")
+    if target_step_ms is not None:
+        cmd += ["--target-step-ms", str(target_step_ms)]
     return subprocess.Popen(cmd, env=env)
 
 
@@ -88,14 +91,17 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
         (model_name,) = models
         sock = os.path.join(run_dir, f"gpu{target}.sock")
         log_with_timestamp(f"spawning worker gpu:{target} ({model_name})", "grey")
+        slo = config.cluster.target_step_ms
         proc = spawn_gpu_worker(target, model_name, sock, kv_gb=kv_gb,
-                                max_batch=max_batch, use_graphs=use_graphs)
+                                max_batch=max_batch, use_graphs=use_graphs,
+                                target_step_ms=slo)
         client = RpcWorkerClient(f"gpu:{target}", {model_name}, sock)
         client.proc = proc
         client.respawn = (lambda t=target, m=model_name, s=sock:
                           spawn_gpu_worker(t, m, s, kv_gb=kv_gb,
                                            max_batch=max_batch,
-                                           use_graphs=use_graphs))
+                                           use_graphs=use_graphs,
+                                           target_step_ms=slo))
         registry.register("gpu", target, client)
 
     for pool_name, model_name in sorted(pools.items()):
