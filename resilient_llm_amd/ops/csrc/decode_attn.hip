@@ -45,11 +45,21 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
                         const float* __restrict__ cos_sin,
                         const int32_t* __restrict__ slot_mapping,
                         uint16_t* __restrict__ k_cache_w,
-                        uint16_t* __restrict__ v_cache_w) {
+                        uint16_t* __restrict__ v_cache_w,
+                        int n_split,
+                        float* __restrict__ part_out,
+                        float* __restrict__ part_ml) {
   constexpr int D = GW * 8;
   constexpr int GPW = 64 / GW;                 // lane-groups per wave
-  const int seq = blockIdx.x / n_kv_heads;
-  const int kvh = blockIdx.x % n_kv_heads;
+  // split-K (flash-decode): when batch*n_kv_heads can't fill the chip,
+  // n_split workgroups share one (sequence, kv-head) — each walks an
+  // interleaved subset of the cache blocks and emits an f32 partial
+  // (unnormalized acc + running max/denominator); a tiny combine
+  // kernel reduces them.  n_split==1 is the classic single-WG path.
+  const int split = blockIdx.x % n_split;
+  const int sk = blockIdx.x / n_split;
+  const int seq = sk / n_kv_heads;
+  const int kvh = sk % n_kv_heads;
   const int len = seq_lens[seq];
   const int lane = threadIdx.x & (kWave - 1);
   const int wave = threadIdx.x >> 6;
@@ -115,7 +125,7 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
   if constexpr (FUSED) {
     // group 0 of wave 0: rope k, append k/v to the cache, and fold the
     // NEW token's self-attention term into partial 0
-    if (wave == 0 && group == 0) {
+    if (wave == 0 && group == 0 && split == 0) {
       bf16x8 kr, vr;
       kr.u = *reinterpret_cast<const uint4*>(
           k_src + int64_t(seq) * q_stride + kvh * D + d0);
@@ -213,25 +223,26 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
     }
   };
 
+  const int bstride = n_waves * n_split;     // global block-chain count
   if (tok_per_grp <= TPG) {
     // depth-2 software pipeline over cache blocks: the NEXT block's
     // K/V loads are in flight while this block computes (named A/B
     // buffers, statically indexed — guide §5.4 rule 20)
     uint4 kA[TPG], vA[TPG], kB[TPG], vB[TPG];
-    int b = wave;
+    int b = split * n_waves + wave;
     if (b < n_blocks) issue_block(b, kA, vA);
     while (b < n_blocks) {
-      if (b + n_waves < n_blocks) issue_block(b + n_waves, kB, vB);
+      if (b + bstride < n_blocks) issue_block(b + bstride, kB, vB);
       consume_block(b, kA, vA);
-      b += n_waves;
+      b += bstride;
       if (b >= n_blocks) break;
-      if (b + n_waves < n_blocks) issue_block(b + n_waves, kA, vA);
+      if (b + bstride < n_blocks) issue_block(b + bstride, kA, vA);
       consume_block(b, kB, vB);
-      b += n_waves;
+      b += bstride;
     }
   } else {
     // oversized cache blocks: plain per-sub-batch loop
-    for (int b = wave; b < n_blocks; b += n_waves) {
+    for (int b = split * n_waves + wave; b < n_blocks; b += bstride) {
       const int phys = block_table[int64_t(seq) * max_blocks + b];
       const int64_t kv_base =
           (int64_t(phys) * n_kv_heads + kvh) * block_size * D;
@@ -294,8 +305,46 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
       num += w * accs[(int64_t(p) * GROUP + h) * D + d];
       den += w * ml[(p * GROUP + h) * 2 + 1];
     }
-    out[(int64_t(seq) * n_kv_heads * GROUP + kvh * GROUP + h) * D + d] =
-        f32_to_bf16(den > 0.f ? num / den : 0.f);
+    if (n_split == 1) {
+      out[(int64_t(seq) * n_kv_heads * GROUP + kvh * GROUP + h) * D + d] =
+          f32_to_bf16(den > 0.f ? num / den : 0.f);
+    } else {
+      // unnormalized partial in this split's max-frame; a split whose
+      // block range was empty publishes M=-inf/den=0 and combines to 0
+      const int64_t pb = (int64_t(sk) * n_split + split) * GROUP + h;
+      part_out[pb * D + d] = num;
+      if (d == 0) {
+        part_ml[pb * 2 + 0] = M;
+        part_ml[pb * 2 + 1] = den;
+      }
+    }
+  }
+}
+
+// Reduce n_split partials per (sequence, kv-head): one WG per (seq,
+// kv-head), threads stride over GROUP*D output elements.
+__global__ __launch_bounds__(256)
+void decode_combine_kernel(const float* __restrict__ part,
+                           const float* __restrict__ ml,
+                           uint16_t* __restrict__ out,
+                           int n_split, int group, int D) {
+  const int sk = blockIdx.x;
+  const int ghd = group * D;
+  for (int idx = threadIdx.x; idx < ghd; idx += blockDim.x) {
+    const int h = idx / D;
+    float M = kNegInf;
+    for (int s = 0; s < n_split; ++s)
+      M = fmaxf(M, ml[((int64_t(sk) * n_split + s) * group + h) * 2]);
+    float num = 0.f, den = 0.f;
+    if (M > kNegInf) {
+      for (int s = 0; s < n_split; ++s) {
+        const int64_t b = (int64_t(sk) * n_split + s) * group + h;
+        const float w = __expf(ml[b * 2] - M);
+        num += w * part[(int64_t(sk) * n_split + s) * ghd + idx];
+        den += w * ml[b * 2 + 1];
+      }
+    }
+    out[int64_t(sk) * ghd + idx] = f32_to_bf16(den > 0.f ? num / den : 0.f);
   }
 }
 
@@ -315,29 +364,47 @@ void dispatch_decode(const uint16_t* q, const uint16_t* k_cache,
                      const int32_t* seq_lens, uint16_t* out, int batch,
                      int n_kv_heads, int block_size, int max_blocks,
                      float scale, int q_stride, const FusedArgs* fa,
+                     int n_split, float* part_out, float* part_ml,
                      hipStream_t stream) {
   constexpr int D = GW * 8;
   constexpr int GPW = 64 / GW;
   const int n_part = 4 * GPW;
   const size_t smem = size_t(n_part) * GROUP * (D + 2) * sizeof(float);
+  const dim3 grid(batch * n_kv_heads * n_split);
   if (fa != nullptr) {
     hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP, true>),
-                       dim3(batch * n_kv_heads), dim3(256), smem, stream,
+                       grid, dim3(256), smem, stream,
                        q, k_cache, v_cache, block_table, seq_lens, out,
                        n_kv_heads, block_size, max_blocks, scale, q_stride,
                        fa->k_src, fa->v_src, fa->positions, fa->cos_sin,
-                       fa->slot_mapping, fa->k_cache_w, fa->v_cache_w);
+                       fa->slot_mapping, fa->k_cache_w, fa->v_cache_w,
+                       n_split, part_out, part_ml);
   } else {
     hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP, false>),
-                       dim3(batch * n_kv_heads), dim3(256), smem, stream,
+                       grid, dim3(256), smem, stream,
                        q, k_cache, v_cache, block_table, seq_lens, out,
                        n_kv_heads, block_size, max_blocks, scale, q_stride,
                        nullptr, nullptr, nullptr, nullptr, nullptr,
-                       nullptr, nullptr);
+                       nullptr, nullptr, n_split, part_out, part_ml);
+  }
+  if (n_split > 1) {
+    hipLaunchKernelGGL(decode_combine_kernel, dim3(batch * n_kv_heads),
+                       dim3(256), 0, stream, part_out, part_ml, out,
+                       n_split, GROUP, D);
   }
 }
 
 }  // namespace
+
+// Split factor so decode fills the chip: 256 CUs want >=512 workgroups
+// in flight; small batches (interactive / long-context) get their
+// sequences split across block-chains instead of idling CUs.
+int decode_attn_n_split(int batch, int n_kv_heads) {
+  const int wgs = batch * n_kv_heads;
+  if (wgs >= 384) return 1;
+  int n = (512 + wgs - 1) / wgs;
+  return n > 16 ? 16 : n;
+}
 
 void launch_decode_attn_impl(const uint16_t* q, const uint16_t* k_cache,
                              const uint16_t* v_cache,
@@ -346,13 +413,16 @@ void launch_decode_attn_impl(const uint16_t* q, const uint16_t* k_cache,
                              int batch, int n_q_heads, int n_kv_heads,
                              int head_dim, int block_size, int max_blocks,
                              float scale, int q_stride, const FusedArgs* fa,
+                             int n_split, float* part_out, float* part_ml,
                              hipStream_t stream) {
   if (batch == 0) return;
+  if (part_out == nullptr) n_split = 1;
   const int group = n_q_heads / n_kv_heads;
   auto run = [&](auto gw_tag, auto group_tag) {
     dispatch_decode<decltype(gw_tag)::value, decltype(group_tag)::value>(
         q, k_cache, v_cache, block_table, seq_lens, out, batch, n_kv_heads,
-        block_size, max_blocks, scale, q_stride, fa, stream);
+        block_size, max_blocks, scale, q_stride, fa, n_split, part_out,
+        part_ml, stream);
   };
   using I8 = std::integral_constant<int, 8>;
   using I16 = std::integral_constant<int, 16>;
@@ -380,10 +450,12 @@ void launch_decode_attn(const uint16_t* q, const uint16_t* k_cache,
                         const int32_t* seq_lens, uint16_t* out, int batch,
                         int n_q_heads, int n_kv_heads, int head_dim,
                         int block_size, int max_blocks, float scale,
-                        int q_stride, hipStream_t stream) {
+                        int q_stride, int n_split, float* part_out,
+                        float* part_ml, hipStream_t stream) {
   launch_decode_attn_impl(q, k_cache, v_cache, block_table, seq_lens, out,
                           batch, n_q_heads, n_kv_heads, head_dim, block_size,
-                          max_blocks, scale, q_stride, nullptr, stream);
+                          max_blocks, scale, q_stride, nullptr,
+                          n_split, part_out, part_ml, stream);
 }
 
 void launch_decode_attn_fused(
@@ -392,7 +464,8 @@ void launch_decode_attn_fused(
     const int32_t* positions, const float* cos_sin,
     const int32_t* slot_mapping, uint16_t* out, int batch, int n_q_heads,
     int n_kv_heads, int head_dim, int block_size, int max_blocks,
-    float scale, int qkv_stride, hipStream_t stream) {
+    float scale, int qkv_stride, int n_split, float* part_out,
+    float* part_ml, hipStream_t stream) {
   FusedArgs fa;
   fa.k_src = qkv + n_q_heads * head_dim;
   fa.v_src = qkv + (n_q_heads + n_kv_heads) * head_dim;
@@ -403,7 +476,8 @@ void launch_decode_attn_fused(
   fa.v_cache_w = v_cache;
   launch_decode_attn_impl(qkv, k_cache, v_cache, block_table, seq_lens, out,
                           batch, n_q_heads, n_kv_heads, head_dim, block_size,
-                          max_blocks, scale, qkv_stride, &fa, stream);
+                          max_blocks, scale, qkv_stride, &fa,
+                          n_split, part_out, part_ml, stream);
 }
 
 }  // namespace rlli

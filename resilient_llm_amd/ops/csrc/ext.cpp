@@ -152,6 +152,29 @@ void rope_kv_append_qkv_(Tensor qkv, const Tensor& positions,
   check_launch("rope_kv_append_qkv_");
 }
 
+
+// split-K scratch for decode attention (empty tensors when n_split==1);
+// allocation participates in hipGraph capture pools, so graphed decode
+// buckets capture their scratch once and reuse it on replay.
+struct SplitScratch {
+  int n_split = 1;
+  Tensor part, ml;
+  float* part_ptr() { return n_split > 1 ? part.data_ptr<float>() : nullptr; }
+  float* ml_ptr() { return n_split > 1 ? ml.data_ptr<float>() : nullptr; }
+};
+
+static SplitScratch make_split_scratch(const Tensor& ref, int batch,
+                                       int n_kv, int group, int D) {
+  SplitScratch s;
+  s.n_split = rlli::decode_attn_n_split(batch, n_kv);
+  if (s.n_split > 1) {
+    auto opt = ref.options().dtype(at::kFloat);
+    s.part = at::empty({int64_t(batch) * n_kv * s.n_split * group * D}, opt);
+    s.ml = at::empty({int64_t(batch) * n_kv * s.n_split * group * 2}, opt);
+  }
+  return s;
+}
+
 // ---------------------------------------------------------- decode_attn
 Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
                    const Tensor& v_cache, const Tensor& block_table,
@@ -177,11 +200,13 @@ Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
   TORCH_CHECK(block_table.size(0) == batch && seq_lens.numel() == batch);
   Tensor out = at::empty_like(q);
   c10::hip::HIPGuardMasqueradingAsCUDA guard(q.device());
+  SplitScratch ss = make_split_scratch(q, batch, n_kv, group, D);
   rlli::launch_decode_attn(
       bf16_ptr(q), bf16_ptr(k_cache), bf16_ptr(v_cache),
       block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       bf16_ptr(out), batch, n_q, n_kv, D, block_size, max_blocks,
-      float(scale), n_q * D, current_stream(q));
+      float(scale), n_q * D, ss.n_split, ss.part_ptr(), ss.ml_ptr(),
+      current_stream(q));
   check_launch("decode_attn");
   return out;
 }
@@ -203,11 +228,13 @@ Tensor decode_attn_qkv(const Tensor& qkv, const Tensor& k_cache,
   TORCH_CHECK(D == 64 || D == 128);
   Tensor out = at::empty({batch, n_q * D}, qkv.options());
   c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
+  SplitScratch ss = make_split_scratch(qkv, batch, n_kv, group, D);
   rlli::launch_decode_attn(
       bf16_ptr(qkv), bf16_ptr(k_cache), bf16_ptr(v_cache),
       block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       bf16_ptr(out), batch, int(n_q), n_kv, D, block_size, max_blocks,
-      float(scale), stride, current_stream(qkv));
+      float(scale), stride, ss.n_split, ss.part_ptr(), ss.ml_ptr(),
+      current_stream(qkv));
   check_launch("decode_attn_qkv");
   return out;
 }
@@ -239,13 +266,14 @@ Tensor decode_attn_rope_qkv(const Tensor& qkv, const Tensor& positions,
   const int max_blocks = int(block_table.size(1));
   Tensor out = at::empty({batch, n_q * D}, qkv.options());
   c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
+  SplitScratch ss = make_split_scratch(qkv, batch, n_kv, group, D);
   rlli::launch_decode_attn_fused(
       bf16_ptr(qkv), bf16_ptr(k_cache), bf16_ptr(v_cache),
       block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       positions.data_ptr<int32_t>(), cos_sin.data_ptr<float>(),
       slot_mapping.data_ptr<int32_t>(), bf16_ptr(out), batch, int(n_q),
       n_kv, D, block_size, max_blocks, float(scale), stride,
-      current_stream(qkv));
+      ss.n_split, ss.part_ptr(), ss.ml_ptr(), current_stream(qkv));
   check_launch("decode_attn_rope_qkv");
   return out;
 }

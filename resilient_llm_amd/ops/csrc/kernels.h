@@ -30,6 +30,13 @@ void launch_rope_kv_append(
     int tokens, int n_q_heads, int n_kv_heads, int head_dim,
     int block_size, int q_stride, int kv_stride, hipStream_t stream);
 
+// Split-K factor for decode attention (flash-decode): >1 when
+// batch*n_kv_heads workgroups cannot fill 256 CUs.  The caller sizes
+// the f32 scratch as [batch*n_kv*n_split*group*head_dim] (part_out)
+// and [batch*n_kv*n_split*group*2] (part_ml) and passes both; n_split=1
+// (or null scratch) is the classic single-workgroup-per-(seq,head) path.
+int decode_attn_n_split(int batch, int n_kv_heads);
+
 // Paged GQA decode attention: one new q token per sequence.
 void launch_decode_attn(
     const uint16_t* q,                 // [batch, n_q_heads, head_dim]
@@ -40,7 +47,7 @@ void launch_decode_attn(
     uint16_t* out,                     // [batch, n_q_heads, head_dim]
     int batch, int n_q_heads, int n_kv_heads, int head_dim,
     int block_size, int max_blocks, float scale, int q_stride,
-    hipStream_t stream);
+    int n_split, float* part_out, float* part_ml, hipStream_t stream);
 
 // Fused decode: rope(q,k) + cache append + paged attention in ONE
 // kernel (decode path; qkv is the raw fused GEMM output).
@@ -50,7 +57,8 @@ void launch_decode_attn_fused(
     const int32_t* positions, const float* cos_sin,
     const int32_t* slot_mapping, uint16_t* out, int batch, int n_q_heads,
     int n_kv_heads, int head_dim, int block_size, int max_blocks,
-    float scale, int qkv_stride, hipStream_t stream);
+    float scale, int qkv_stride, int n_split, float* part_out,
+    float* part_ml, hipStream_t stream);
 
 // Varlen causal prefill attention over in-batch q/k/v.
 void launch_prefill_attn(
