@@ -11,6 +11,7 @@
 
 #include "kernels.h"
 
+#include <cstdlib>
 #include <vector>
 
 namespace {
@@ -166,6 +167,8 @@ struct SplitScratch {
 static SplitScratch make_split_scratch(const Tensor& ref, int batch,
                                        int n_kv, int group, int D) {
   SplitScratch s;
+  static const bool disabled = std::getenv("RLLI_NO_SPLITK") != nullptr;
+  if (disabled) return s;
   s.n_split = rlli::decode_attn_n_split(batch, n_kv);
   if (s.n_split > 1) {
     auto opt = ref.options().dtype(at::kFloat);
