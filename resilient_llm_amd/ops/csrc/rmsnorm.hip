@@ -137,6 +137,93 @@ __global__ void rmsnorm_two_pass(const uint16_t* __restrict__ x,
   }
 }
 
+// One WAVE per row, every load batch-issued.  The one-WG-per-row
+// geometry above is kept for dim > 4096; at decode batch sizes it was
+// latency-bound (batch 64 -> 64 WGs, one outstanding 16-B load per
+// lane, 0.39 TB/s measured): a single wave with 2*VPL loads in flight
+// streams its row at HBM rate and reduces wave-locally (no LDS, no
+// barrier), and the small WGs pack many-per-CU.
+template <int VPL, bool FUSE_RESIDUAL>
+__global__ __launch_bounds__(64)
+void rmsnorm_wave(const uint16_t* __restrict__ x,
+                  uint16_t* __restrict__ residual,
+                  const uint16_t* __restrict__ w,
+                  uint16_t* __restrict__ y, int dim, float eps) {
+  const int row = blockIdx.x;
+  const int64_t base = int64_t(row) * dim;
+  const int lane = threadIdx.x;
+  const int nvec = dim / 8;
+
+  uint4 xr[VPL], rr[VPL];
+#pragma unroll
+  for (int p = 0; p < VPL; ++p) {
+    const int v8 = p * kWave + lane;
+    xr[p] = uint4{0, 0, 0, 0};
+    if (v8 < nvec) {
+      xr[p] = *reinterpret_cast<const uint4*>(x + base + v8 * 8);
+      if constexpr (FUSE_RESIDUAL)
+        rr[p] = *reinterpret_cast<const uint4*>(residual + base + v8 * 8);
+    }
+  }
+
+  float vals[VPL][8];
+  float ss = 0.f;
+#pragma unroll
+  for (int p = 0; p < VPL; ++p) {
+    const int v8 = p * kWave + lane;
+    bf16x8 xv;
+    xv.u = xr[p];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) vals[p][i] = bf16_to_f32(xv.s[i]);
+    if constexpr (FUSE_RESIDUAL) {
+      if (v8 < nvec) {
+        bf16x8 rv, outv;
+        rv.u = rr[p];
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          vals[p][i] += bf16_to_f32(rv.s[i]);
+          outv.s[i] = f32_to_bf16(vals[p][i]);
+          // y must equal rmsnorm(STORED residual): round-trip via bf16
+          vals[p][i] = bf16_to_f32(outv.s[i]);
+        }
+        *reinterpret_cast<uint4*>(residual + base + v8 * 8) = outv.u;
+      }
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) ss += vals[p][i] * vals[p][i];
+  }
+  ss = wave_sum(ss);
+  const float inv = rsqrtf(ss / dim + eps);
+
+#pragma unroll
+  for (int p = 0; p < VPL; ++p) {
+    const int v8 = p * kWave + lane;
+    if (v8 >= nvec) continue;
+    bf16x8 wv, outv;
+    wv.u = *reinterpret_cast<const uint4*>(w + v8 * 8);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      outv.s[i] = f32_to_bf16(vals[p][i] * inv * bf16_to_f32(wv.s[i]));
+    *reinterpret_cast<uint4*>(y + base + v8 * 8) = outv.u;
+  }
+}
+
+template <bool FR>
+void launch_wave(const uint16_t* x, uint16_t* residual, const uint16_t* w,
+                 uint16_t* y, int rows, int dim, float eps,
+                 hipStream_t stream) {
+  const int nvec = dim / 8;
+  const int vpl = (nvec + kWave - 1) / kWave;
+  auto go = [&](auto tag) {
+    hipLaunchKernelGGL((rmsnorm_wave<decltype(tag)::value, FR>), dim3(rows),
+                       dim3(kWave), 0, stream, x, residual, w, y, dim, eps);
+  };
+  if (vpl <= 1) go(std::integral_constant<int, 1>{});
+  else if (vpl <= 2) go(std::integral_constant<int, 2>{});
+  else if (vpl <= 4) go(std::integral_constant<int, 4>{});
+  else go(std::integral_constant<int, 8>{});
+}
+
 }  // namespace
 
 void launch_rmsnorm(const uint16_t* x, uint16_t* residual, const uint16_t* w,
@@ -144,7 +231,12 @@ void launch_rmsnorm(const uint16_t* x, uint16_t* residual, const uint16_t* w,
                     hipStream_t stream) {
   if (rows == 0) return;
   const int smem = 16 * sizeof(float);
-  if (dim <= 8192) {
+  if (dim <= 4096) {
+    if (residual)
+      launch_wave<true>(x, residual, w, y, rows, dim, eps, stream);
+    else
+      launch_wave<false>(x, nullptr, w, y, rows, dim, eps, stream);
+  } else if (dim <= 8192) {
     int threads = ((dim / 8 + kWave - 1) / kWave) * kWave;
     if (residual)
       hipLaunchKernelGGL((rmsnorm_one_vec<true>), dim3(rows), dim3(threads),
