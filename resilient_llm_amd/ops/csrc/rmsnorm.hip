@@ -11,6 +11,8 @@
 
 #include "common.h"
 
+#include <cstdlib>
+
 namespace rlli {
 
 namespace {
@@ -231,7 +233,8 @@ void launch_rmsnorm(const uint16_t* x, uint16_t* residual, const uint16_t* w,
                     hipStream_t stream) {
   if (rows == 0) return;
   const int smem = 16 * sizeof(float);
-  if (dim <= 4096) {
+  static const bool no_wave = std::getenv("RLLI_NO_WAVE_NORM") != nullptr;
+  if (dim <= 4096 && !no_wave) {
     if (residual)
       launch_wave<true>(x, residual, w, y, rows, dim, eps, stream);
     else
