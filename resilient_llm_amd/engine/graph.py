@@ -55,6 +55,11 @@ class DecodeGraphRunner:
         with torch.cuda.stream(s):
             for _ in range(2):
                 self._run_eager(self.buckets[-1])
+            # visit every bucket eagerly so the measured-GEMM autotuner
+            # (ops/autotune.py) pins per-shape algos BEFORE capture —
+            # tuning needs syncs and cannot run mid-capture
+            for b in self.buckets:
+                self._run_eager(b)
         torch.cuda.current_stream(self.device).wait_stream(s)
         torch.cuda.synchronize(self.device)
 

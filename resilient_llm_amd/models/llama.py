@@ -27,15 +27,25 @@ import torch
 import torch.nn.functional as F
 
 from .. import ops
+from ..ops import autotune as lt_autotune
 
 # Opt-in: route decode-shaped GEMMs (M<=64) through the custom skinny-M
 # MFMA kernel instead of hipBLASLt (A/B lever; see ops/csrc/skinny_gemm.hip)
 _USE_SKINNY = os.environ.get("RLLI_SKINNY") == "1"
 
 
+# Row counts that the measured-GEMM autotuner may tune: the decode graph
+# buckets (+ the small row counts prefill's lm_head sees).  Anything else
+# (prefill/mixed token counts — unbounded variety) goes straight to
+# torch so the tuning cache stays finite.
+_LT_ROWS = frozenset({1, 2, 4, 8, 12, 16, 24, 32, 48, 64, 96, 128})
+
+
 def _linear(x, w):
     if _USE_SKINNY:
         return ops.linear_auto(x, w)
+    if x.is_cuda and x.dim() == 2 and x.shape[0] in _LT_ROWS:
+        return lt_autotune.tuned_linear(x, w)
     return F.linear(x, w)
 
 

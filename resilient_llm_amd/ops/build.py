@@ -28,6 +28,7 @@ SAN_FLAGS = (["-fsanitize=address"] if os.environ.get("RLLI_ASAN") == "1" else
 
 SOURCES = [
     "ext.cpp",
+    "lt_gemm.cpp",
     "rmsnorm.hip",
     "silu_mul.hip",
     "rope_kv.hip",
@@ -48,7 +49,7 @@ def _torch_flags() -> tuple[list[str], list[str]]:
     includes = [f"-I{p}" for p in ce.include_paths()]
     libs = [f"-L{p}" for p in ce.library_paths()]
     libs += ["-ltorch", "-ltorch_cpu", "-lc10", "-ltorch_hip", "-lc10_hip",
-             "-lamdhip64"]
+             "-lamdhip64", "-lhipblaslt"]
     abi = int(torch._C._GLIBCXX_USE_CXX11_ABI)
     includes.append(f"-D_GLIBCXX_USE_CXX11_ABI={abi}")
     return includes, libs

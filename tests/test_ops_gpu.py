@@ -314,3 +314,34 @@ def test_decode_attn_rope_fused_vs_unfused(n_q, n_kv, D):
     assert_close_bf16(out_f.cpu(), out_u.cpu())
     torch.testing.assert_close(kc.float(), kc_u.float())
     torch.testing.assert_close(vc.float(), vc_u.float())
+
+
+@pytest.mark.gpu
+def test_lt_linear_matches_torch():
+    """Direct hipBLASLt linear with explicit algo index == F.linear
+    (bf16 in, f32 accumulate — summation-order tolerance only)."""
+    import resilient_llm_amd.ops as ops  # noqa: F401  (loads the library)
+    for M, N, K in [(8, 4096, 4096), (64, 6144, 4096), (64, 4096, 14336)]:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+        w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.02
+        want = torch.nn.functional.linear(x, w).float()
+        idxs = torch.ops.rlli.lt_heuristics(x, w, 4).tolist()
+        assert idxs, "no hipblaslt heuristics returned"
+        for idx in idxs[:3]:
+            got = torch.ops.rlli.lt_linear(x, w, idx).float()
+            torch.testing.assert_close(got, want, rtol=3e-2, atol=3e-1)
+
+
+@pytest.mark.gpu
+def test_tuned_linear_caches_and_matches():
+    from resilient_llm_amd.ops import autotune
+    x = torch.randn(16, 2048, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(1024, 2048, dtype=torch.bfloat16, device="cuda") * 0.02
+    want = torch.nn.functional.linear(x, w).float()
+    got = autotune.tuned_linear(x, w).float()
+    torch.testing.assert_close(got, want, rtol=3e-2, atol=3e-1)
+    key = (16, 1024, 2048)
+    assert key in autotune.tuned_shapes()
+    # second call hits the cache (either pinned algo or torch fallback)
+    got2 = autotune.tuned_linear(x, w).float()
+    torch.testing.assert_close(got2, want, rtol=3e-2, atol=3e-1)
