@@ -25,7 +25,13 @@ import torch.nn.functional as F
 _cache: dict[tuple[int, int, int], int | None] = {}
 _MAX_CANDIDATES = 24
 _TIME_ITERS = 20
-_DISABLED = os.environ.get("RLLI_NO_LT") == "1"
+# Default OFF: the tuner's per-GEMM wins (13-40% in isolation, see
+# scripts/probe_hipblaslt.cpp) did NOT survive in context — the full
+# serving bench measured 129.4 reqs/s tuned vs 139.1 with torch's picks
+# on the same box.  The fast isolated algos are split-K-heavy and their
+# workspace traffic competes with the rest of the decode stream.  Kept
+# as an A/B lever (RLLI_LT=1) and for other model geometries.
+_DISABLED = os.environ.get("RLLI_LT") != "1"
 
 
 def _time_fn(fn, iters: int = _TIME_ITERS) -> float:
