@@ -333,8 +333,9 @@ def test_lt_linear_matches_torch():
 
 
 @pytest.mark.gpu
-def test_tuned_linear_caches_and_matches():
+def test_tuned_linear_caches_and_matches(monkeypatch):
     from resilient_llm_amd.ops import autotune
+    monkeypatch.setattr(autotune, "_DISABLED", False)   # opt-in lever
     x = torch.randn(16, 2048, dtype=torch.bfloat16, device="cuda")
     w = torch.randn(1024, 2048, dtype=torch.bfloat16, device="cuda") * 0.02
     want = torch.nn.functional.linear(x, w).float()
