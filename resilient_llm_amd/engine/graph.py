@@ -13,6 +13,8 @@ from __future__ import annotations
 
 import torch
 
+from ..ops import autotune as lt_autotune
+
 BUCKETS = [1, 2, 4, 8, 16, 24, 32, 40, 48, 56, 64, 96, 128]
 
 
@@ -60,6 +62,15 @@ class DecodeGraphRunner:
             # tuning needs syncs and cannot run mid-capture
             for b in self.buckets:
                 self._run_eager(b)
+            if lt_autotune.enabled():
+                # re-decide per shape against the REAL objective: a
+                # full decode step at the largest bucket, then copy the
+                # per-weight decisions to the other buckets
+                bmax = self.buckets[-1]
+                keys = [k for k in lt_autotune.tuned_shapes() if k[0] == bmax]
+                lt_autotune.tune_in_context(lambda: self._run_eager(bmax),
+                                            keys=keys)
+                lt_autotune.propagate(keys)
         torch.cuda.current_stream(self.device).wait_stream(s)
         torch.cuda.synchronize(self.device)
 

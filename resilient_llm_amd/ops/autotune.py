@@ -23,6 +23,7 @@ import torch
 import torch.nn.functional as F
 
 _cache: dict[tuple[int, int, int], int | None] = {}
+_candidates: dict[tuple[int, int, int], list[int]] = {}
 _MAX_CANDIDATES = 24
 _TIME_ITERS = 20
 # Default OFF: the tuner's per-GEMM wins (13-40% in isolation, see
@@ -47,9 +48,12 @@ def _time_fn(fn, iters: int = _TIME_ITERS) -> float:
 
 def _tune(x: torch.Tensor, w: torch.Tensor) -> int | None:
     """Race torch's pick against the library's heuristic candidates;
-    returns the winning algo index, or None when torch wins."""
+    returns the winning algo index, or None when torch wins.  The top
+    few candidates are remembered for the in-context pass."""
+    key = (x.shape[0], w.shape[0], w.shape[1])
     torch_t = _time_fn(lambda: F.linear(x, w))
     best_algo, best_t = None, torch_t
+    timed: list[tuple[float, int]] = []
     try:
         idxs = torch.ops.rlli.lt_heuristics(x, w, _MAX_CANDIDATES).tolist()
     except Exception:
@@ -59,8 +63,10 @@ def _tune(x: torch.Tensor, w: torch.Tensor) -> int | None:
             t = _time_fn(lambda: torch.ops.rlli.lt_linear(x, w, idx))
         except Exception:
             continue
+        timed.append((t, idx))
         if t < best_t:
             best_algo, best_t = idx, t
+    _candidates[key] = [i for _, i in sorted(timed)[:6]]
     # require a real margin over torch before pinning an algo: the
     # tuning sample is small and torch's pick is the safe default
     if best_algo is not None and best_t < 0.97 * torch_t:
@@ -83,7 +89,51 @@ def tuned_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
         _cache[key] = algo
     if algo is None:
         return F.linear(x, w)
-    return torch.ops.rlli.lt_linear(x, w, algo)
+    try:
+        return torch.ops.rlli.lt_linear(x, w, algo)
+    except Exception:
+        # an algo propagated to a row count it does not support
+        _cache[key] = None
+        return F.linear(x, w)
+
+
+def enabled() -> bool:
+    return not _DISABLED
+
+
+def tune_in_context(run_fn, keys=None, iters: int = 6) -> dict:
+    """Second tuning pass with the CORRECT objective: isolated per-GEMM
+    wins did not survive the real decode stream (129.4 vs 139.1 reqs/s
+    e2e — split-K workspace traffic competes with neighbouring
+    kernels), so re-decide each shape by timing ``run_fn`` — a full
+    decode step — with the cache pinned to torch vs each remembered
+    candidate (greedy coordinate descent, one pass).  Call before graph
+    capture; needs device syncs."""
+    if _DISABLED:
+        return {}
+    report: dict = {}
+    for key in (keys if keys is not None else list(_candidates)):
+        if key not in _candidates:
+            continue
+        best, best_t = None, None
+        for opt in [None] + _candidates[key]:
+            _cache[key] = opt
+            t = _time_fn(run_fn, iters)
+            if best_t is None or t < best_t:
+                best, best_t = opt, t
+        _cache[key] = best
+        report[key] = (best, best_t)
+    return report
+
+
+def propagate(tuned_keys) -> None:
+    """Copy each in-context (N, K) decision to every other cached row
+    count of the same weight (decode graph buckets share weights)."""
+    by_nk = {(k[1], k[2]): _cache[k] for k in tuned_keys if k in _cache}
+    for key in list(_cache):
+        nk = (key[1], key[2])
+        if nk in by_nk:
+            _cache[key] = by_nk[nk]
 
 
 def tuned_shapes() -> dict:
