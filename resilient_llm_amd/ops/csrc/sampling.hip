@@ -56,7 +56,35 @@ void sample_kernel(const uint16_t* __restrict__ logits,
 
   BestPair best{-1e30f, 0};
   const int nvec = vocab / 8;
-  for (int v8 = threadIdx.x; v8 < nvec; v8 += blockDim.x) {
+  const int stride = blockDim.x;
+  int v8 = threadIdx.x;
+  // 4 row-vectors in flight per lane before any arithmetic (one
+  // outstanding load per lane left this walk latency-bound); `better`
+  // is associative with a lowest-index tie-break, so the reordering
+  // cannot change the sampled token
+  for (; v8 + 3 * stride < nvec; v8 += 4 * stride) {
+    uint4 raw[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      raw[j] = *reinterpret_cast<const uint4*>(
+          logits + int64_t(row) * vocab + (v8 + j * stride) * 8);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      bf16x8 lv;
+      lv.u = raw[j];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const int idx = (v8 + j * stride) * 8 + i;
+        float val = bf16_to_f32(lv.s[i]) * inv_t;
+        if (!greedy) {
+          const float u = uniform01(seed, row_ctr | uint64_t(idx));
+          val += -__logf(-__logf(u));
+        }
+        best = better(best, BestPair{val, idx});
+      }
+    }
+  }
+  for (; v8 < nvec; v8 += stride) {
     bf16x8 lv;
     lv.u = *reinterpret_cast<const uint4*>(logits + int64_t(row) * vocab + v8 * 8);
 #pragma unroll
