@@ -50,8 +50,11 @@ def main(duration_s=420, n_threads=24):
             rng = random.Random(tid)
             c = OpenAIClient(base, api_key=f"sk-{tid}")
             while not stop.is_set():
-                msgs = [{"role": "user",
-                         "content": "x" * rng.randint(16, 400)}]
+                # 10% long prompts: chunked prefill + split-K decode
+                # under churn, not just short-prompt steady state
+                n_chars = (rng.randint(2000, 8000) if rng.random() < 0.1
+                           else rng.randint(16, 400))
+                msgs = [{"role": "user", "content": "x" * n_chars}]
                 stream = rng.random() < 0.3
                 try:
                     if stream:
