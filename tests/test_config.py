@@ -113,3 +113,18 @@ def test_shipped_config_loads():
     cfg = load_config(path)
     assert cfg.deployments
     assert cfg.router.routing_strategy == "simple-shuffle"
+
+
+def test_cluster_target_step_ms_parsed():
+    cfg = load_config(data={
+        "cluster": {"port": 4101, "target_step_ms": 25.0},
+        "model_list": [{"model_name": "m",
+                        "litellm_params": {"model": "stub/0/llama-3-8b"}}],
+    })
+    assert cfg.cluster.target_step_ms == 25.0
+    cfg2 = load_config(data={
+        "cluster": {"port": 4101},
+        "model_list": [{"model_name": "m",
+                        "litellm_params": {"model": "stub/0/llama-3-8b"}}],
+    })
+    assert cfg2.cluster.target_step_ms is None
