@@ -86,7 +86,7 @@ Tensor lt_linear(const Tensor& x, const Tensor& w, int64_t algo_index) {
   c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   Tensor y = at::empty({x.size(0), w.size(0)}, x.options());
   static const float alpha = 1.f, beta = 0.f;
-  auto gemm = make_gemm(x, const_cast<Tensor&>(w), y, &alpha, &beta);
+  auto gemm = make_gemm(x, w, y, &alpha, &beta);
 
   std::vector<int> want{int(algo_index)};
   std::vector<hipblasLtMatmulHeuristicResult_t> algos;
@@ -111,7 +111,7 @@ Tensor lt_heuristics(const Tensor& x, const Tensor& w, int64_t max_n) {
   c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   Tensor y = at::empty({x.size(0), w.size(0)}, x.options());
   static const float alpha = 1.f, beta = 0.f;
-  auto gemm = make_gemm(x, const_cast<Tensor&>(w), y, &alpha, &beta);
+  auto gemm = make_gemm(x, w, y, &alpha, &beta);
   hipblaslt_ext::GemmPreference pref;
   pref.setMaxWorkspaceBytes(kWorkspaceBytes);
   std::vector<hipblasLtMatmulHeuristicResult_t> algos;
