@@ -24,7 +24,6 @@
 #include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
 #include <hipblaslt/hipblaslt-ext.hpp>
 
-#include <mutex>
 #include <vector>
 
 namespace {
