@@ -128,3 +128,15 @@ def test_cluster_target_step_ms_parsed():
                         "litellm_params": {"model": "stub/0/llama-3-8b"}}],
     })
     assert cfg2.cluster.target_step_ms is None
+
+
+def test_70b_pool_config_loads():
+    cfg = load_config("config/config.70b.yaml")
+    assert set(cfg.cluster.pools) == {"pool-a", "pool-b"}
+    assert cfg.cluster.pools["pool-a"].tensor_parallel == 4
+    assert cfg.cluster.pools["pool-a"].gpus == [0, 1, 2, 3]
+    pool_deps = [d for d in cfg.deployments if d.backend_kind == "pool"]
+    assert len(pool_deps) == 4
+    assert all(d.backend_model == "llama-3-70b" for d in pool_deps)
+    assert len(cfg.deployments_for("llama-70b")) == 2
+    assert cfg.router.fallbacks["llama-70b-pool-a"] == ["llama-70b-pool-b"]
