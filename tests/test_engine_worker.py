@@ -120,3 +120,24 @@ def test_rpc_worker_process_end_to_end(rpc_worker):
         await client.close()
 
     asyncio.run(run())
+
+
+def test_worker_target_step_ms_plumbed():
+    """The decode-cadence SLO reaches the engine and the AIMD controller
+    converges to the floor under an impossible target (CPU steps are
+    milliseconds); generation output is unaffected."""
+    async def run():
+        w = EngineWorker(device="cpu", model_name="tiny", device_label="slo",
+                         num_blocks=64, seed=0, target_step_ms=1e-6,
+                         chunk_size=8)
+        try:
+            assert w.engine.target_step_ms == 1e-6
+            res = await w.generate(GenerationRequest(
+                request_id="s1", model="tiny",
+                messages=[{"role": "user", "content": "hello " * 30}],
+                max_tokens=4))
+            assert res.completion_tokens == 4
+            assert w.engine._prefill_budget == w.engine._budget_floor
+        finally:
+            await w.close()
+    asyncio.run(run())
