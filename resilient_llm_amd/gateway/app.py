@@ -165,6 +165,15 @@ class GatewayApp:
                     ok = True
                 except Exception:
                     ok = False
+                if ok and getattr(worker, "proc_group", None):
+                    # a dead FOLLOWER strands the collective while the
+                    # leader still answers health RPCs: detect it, take
+                    # the leader down too, and let whole-group respawn
+                    # rebuild the pool on a fresh rendezvous
+                    if any(p.poll() is not None for p in worker.proc_group):
+                        ok = False
+                        if worker.proc is not None and worker.proc.poll() is None:
+                            worker.proc.terminate()
                 for model_id in self._deployments_on(worker):
                     self.router.set_healthy(model_id, ok)
                 if (not ok

@@ -159,3 +159,71 @@ def test_pool_group_respawn():
         for p in procs:
             if p.poll() is None:
                 p.kill()
+
+
+def test_health_loop_detects_dead_pool_follower():
+    """A dead pool FOLLOWER strands the RCCL collective while the leader
+    still answers health RPCs.  The health loop must notice (proc_group
+    poll), take the leader down, and whole-group respawn must rebuild —
+    end to end on a real tiny TP=2 gloo pool."""
+    import tempfile as _tf
+
+    from resilient_llm_amd.config import PoolDef, load_config
+    from resilient_llm_amd.workers.base import GenerationRequest
+    from resilient_llm_amd.workers.pool import spawn_pool_worker
+
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    sock = os.path.join(_tf.mkdtemp(prefix="rlli-fd-"), "pool.sock")
+    pool = PoolDef(name="fd", gpus=[0, 1], tensor_parallel=2)
+
+    def spawn():
+        return spawn_pool_worker(pool, "tiny", sock, device_override="cpu",
+                                 tp_backend="gloo", max_batch=4)
+
+    procs = spawn()
+    client = RpcWorkerClient("pool:fd", {"tiny"}, sock)
+    client.proc = procs[0]
+    client.proc_group = procs
+    client.respawn = spawn
+    client.last_respawn = 0.0
+
+    config = load_config(data={
+        "cluster": {"port": free_port()},
+        "model_list": [{"model_name": "fd-model",
+                        "litellm_params": {"model": "stub/0/llama-3-8b"}}],
+    })
+    registry = WorkerRegistry()
+    registry.register("pool", "fd", client)
+
+    async def run():
+        await client.connect(timeout=180)
+        app = GatewayApp(config, registry, health_interval_s=0.3)
+        app.respawn_cooldown_s = 0.0
+        await app.start_background()
+        try:
+            # kill the FOLLOWER only; leader keeps serving health RPCs
+            procs[1].kill()
+            procs[1].wait(timeout=10)
+            deadline = asyncio.get_event_loop().time() + 120
+            while asyncio.get_event_loop().time() < deadline:
+                await asyncio.sleep(1.0)
+                if (client.proc_group is not procs
+                        and client.proc.poll() is None):
+                    break
+            assert client.proc_group is not procs, \
+                "health loop never rebuilt the pool after follower death"
+            res = await client.generate(GenerationRequest(
+                request_id="fd1", model="tiny",
+                messages=[{"role": "user", "content": "after respawn"}],
+                max_tokens=3))
+            assert res.completion_tokens == 3
+        finally:
+            await app.stop_background()
+            await client.close()
+
+    try:
+        asyncio.run(run())
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                p.kill()
