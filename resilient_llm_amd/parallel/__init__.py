@@ -29,8 +29,11 @@ def init_pool_groups(rank: int, world: int, master_port: int,
     share a rendezvous."""
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ["MASTER_PORT"] = str(master_port)
+    # collective timeout: must exceed the leader's idle-heartbeat period
+    # (EngineWorker.TP_HEARTBEAT_S) or idle followers die in broadcast
+    timeout_s = float(os.environ.get("RLLI_TP_TIMEOUT_S", "600"))
     dist.init_process_group(backend="gloo", rank=rank, world_size=world,
-                            timeout=datetime.timedelta(seconds=600))
+                            timeout=datetime.timedelta(seconds=timeout_s))
     control_group = dist.group.WORLD
     if tp_backend == "gloo":
         tp_group = dist.new_group(backend="gloo")

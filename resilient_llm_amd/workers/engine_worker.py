@@ -14,6 +14,7 @@ killing the actual GPU process.
 from __future__ import annotations
 
 import asyncio
+import os
 import threading
 import time
 from typing import AsyncIterator, Optional
@@ -117,7 +118,14 @@ class EngineWorker(Worker):
             self._thread.start()
 
     # --------------------------------------------------------- engine loop
+    # TP lockstep heartbeat: followers BLOCK inside the control-group
+    # broadcast; without periodic empty syncs an IDLE pool would hit the
+    # gloo collective timeout (parallel.init_pool_groups) and the
+    # followers would crash.  Env-tunable for tests.
+    TP_HEARTBEAT_S = float(os.environ.get("RLLI_TP_HEARTBEAT_S", "30"))
+
     def _engine_loop(self) -> None:
+        last_sync = time.monotonic()
         while not self._stop:
             if self.fault_mode == "hang":
                 time.sleep(0.05)
@@ -125,11 +133,16 @@ class EngineWorker(Worker):
             has_work = self.engine.has_work() or (
                 self.tp_control is not None and self.tp_control.pending())
             if not has_work:
+                if (self.tp_control is not None and self.tp_control.is_leader
+                        and time.monotonic() - last_sync > self.TP_HEARTBEAT_S):
+                    self.tp_control.sync(self.engine)
+                    last_sync = time.monotonic()
                 self._work_event.wait(timeout=0.01)
                 self._work_event.clear()
                 continue
             if self.tp_control is not None:
                 self.tp_control.sync(self.engine)
+                last_sync = time.monotonic()
                 if not self.engine.has_work():
                     continue
             try:

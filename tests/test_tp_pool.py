@@ -79,3 +79,38 @@ def test_tp2_matches_tp1_and_serves_concurrently(tp2_pool):
         await client.close()
 
     asyncio.run(run())
+
+
+def test_idle_pool_survives_short_collective_timeout():
+    """Idle-pool liveness: followers block in the control-group
+    broadcast, so with a 3 s collective timeout an idle pool would die
+    without the leader's heartbeat (RLLI_TP_HEARTBEAT_S) — run idle for
+    3x the timeout, then serve."""
+    env = {"RLLI_TP_TIMEOUT_S": "3", "RLLI_TP_HEARTBEAT_S": "0.5"}
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    old = {k: os.environ.get(k) for k in env}
+    os.environ.update(env)
+    sock = os.path.join(tempfile.mkdtemp(prefix="rlli-hb-"), "pool.sock")
+    pool = PoolDef(name="hb", gpus=[0, 1], tensor_parallel=2)
+    procs = spawn_pool_worker(pool, "tiny", sock, device_override="cpu",
+                              tp_backend="gloo", max_batch=4)
+    try:
+        async def run():
+            client = RpcWorkerClient("pool:hb", {"tiny"}, sock)
+            client.proc = procs[0]
+            await client.connect(timeout=180)
+            res = await client.generate(greq("w", "warm", 2))
+            assert res.completion_tokens == 2
+            await asyncio.sleep(9.0)          # idle >> collective timeout
+            assert all(p.poll() is None for p in procs), \
+                "a pool rank died while idle"
+            res = await client.generate(greq("a", "after idle", 3))
+            assert res.completion_tokens == 3
+            await client.close()
+        asyncio.run(run())
+    finally:
+        for k, v in old.items():
+            os.environ.pop(k, None) if v is None else os.environ.__setitem__(k, v)
+        for p in procs:
+            if p.poll() is None:
+                p.kill()
