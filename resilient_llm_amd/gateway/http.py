@@ -81,6 +81,11 @@ Handler = Callable[[Request], Awaitable[Response]]
 
 
 class HttpServer:
+    # a client that sends headers but never the body must not pin the
+    # connection forever (idle KEEP-ALIVE waits between requests are
+    # fine and unbounded; only the body read is deadline-bounded)
+    body_timeout_s: float = 120.0
+
     def __init__(self, handler: Handler, host: str = "127.0.0.1", port: int = 4000) -> None:
         self.handler = handler
         self.host = host
@@ -121,7 +126,16 @@ class HttpServer:
         length = int(headers.get("content-length", "0") or "0")
         if length > MAX_BODY_BYTES:
             raise ValueError("body too large")
-        body = await reader.readexactly(length) if length else b""
+        if length:
+            try:
+                body = await asyncio.wait_for(reader.readexactly(length),
+                                              timeout=self.body_timeout_s)
+            except asyncio.TimeoutError:
+                raise ValueError("body read timeout")
+            except asyncio.IncompleteReadError:
+                return None
+        else:
+            body = b""
         return Request(method, target, headers, body, version)
 
     # ----------------------------------------------------------- writing

@@ -118,3 +118,31 @@ def test_chunked_streaming_response():
         writer.close()
         await server.stop()
     run(main())
+
+
+def test_body_read_timeout():
+    """Headers promising a body that never arrives: the server answers
+    400 within the body deadline instead of pinning the connection."""
+    import socket as _socket
+
+    from resilient_llm_amd.gateway.http import HttpServer, Response
+
+    async def handler(req):
+        return Response.json_response({"ok": True})
+
+    async def run():
+        srv = HttpServer(handler, port=free_port())
+        srv.body_timeout_s = 0.5
+        await srv.start()
+        try:
+            reader, writer = await asyncio.open_connection("127.0.0.1", srv.port)
+            writer.write(b"POST /chat/completions HTTP/1.1\r\n"
+                         b"content-length: 100\r\n\r\n")   # body never sent
+            await writer.drain()
+            data = await asyncio.wait_for(reader.read(4096), timeout=5.0)
+            assert b"400" in data and b"body read timeout" in data
+            writer.close()
+        finally:
+            await srv.stop()
+
+    asyncio.run(run())
