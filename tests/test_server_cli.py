@@ -1,0 +1,47 @@
+"""The real gateway server process (what bin/start-gateway.sh execs):
+spawn `python -m resilient_llm_amd.gateway.server` on the stub config
+and serve one completion through it."""
+
+import os
+import subprocess
+import sys
+import time
+
+from resilient_llm_amd.client import OpenAIClient
+from tests.gateway_harness import free_port
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_server_cli_serves_stub_config():
+    port = free_port()
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "resilient_llm_amd.gateway.server",
+         "--config", os.path.join(REPO, "config", "config.stub.yaml"),
+         "--port", str(port)],
+        env={**os.environ, "PYTHONPATH": REPO},
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    try:
+        client = OpenAIClient(f"http://127.0.0.1:{port}")
+        deadline = time.time() + 30
+        last = None
+        while time.time() < deadline:
+            try:
+                h = client.health()
+                if h.get("status") == "ok":
+                    break
+            except Exception as e:
+                last = e
+                time.sleep(0.3)
+        else:
+            raise AssertionError(f"server never became healthy: {last}")
+        r = client.chat.completions.create(
+            model="llama-fallback-demo",
+            messages=[{"role": "user", "content": "hi"}], max_tokens=4)
+        assert r.usage.completion_tokens == 4
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=10)
+        except Exception:
+            proc.kill()
