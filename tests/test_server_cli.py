@@ -45,3 +45,38 @@ def test_server_cli_serves_stub_config():
             proc.wait(timeout=10)
         except Exception:
             proc.kill()
+
+
+def test_server_cli_multiple_workers_reuse_port():
+    """--workers N forks N SO_REUSEPORT server processes on one port
+    (the LiteLLM --num_workers analogue); all requests succeed."""
+    port = free_port()
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "resilient_llm_amd.gateway.server",
+         "--config", os.path.join(REPO, "config", "config.stub.yaml"),
+         "--port", str(port), "--workers", "2"],
+        env={**os.environ, "PYTHONPATH": REPO},
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    try:
+        client = OpenAIClient(f"http://127.0.0.1:{port}")
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            try:
+                if client.health().get("status") == "ok":
+                    break
+            except Exception:
+                time.sleep(0.3)
+        else:
+            raise AssertionError("server never became healthy")
+        for i in range(6):
+            r = OpenAIClient(f"http://127.0.0.1:{port}").chat.completions.create(
+                model="llama-fallback-loadbalance",
+                messages=[{"role": "user", "content": f"req {i}"}],
+                max_tokens=3)
+            assert r.usage.completion_tokens == 3
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=10)
+        except Exception:
+            proc.kill()
