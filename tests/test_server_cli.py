@@ -80,3 +80,36 @@ def test_server_cli_multiple_workers_reuse_port():
             proc.wait(timeout=10)
         except Exception:
             proc.kill()
+
+
+def test_start_gateway_script():
+    """bin/start-gateway.sh (C7): validate config then exec the server."""
+    port = free_port()
+    proc = subprocess.Popen(
+        ["bash", os.path.join(REPO, "bin", "start-gateway.sh"),
+         os.path.join(REPO, "config", "config.stub.yaml"), str(port)],
+        env={**os.environ, "PYTHONPATH": REPO},
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+        start_new_session=True)
+    try:
+        client = OpenAIClient(f"http://127.0.0.1:{port}")
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            try:
+                if client.health().get("status") == "ok":
+                    break
+            except Exception:
+                time.sleep(0.3)
+        else:
+            raise AssertionError("script-launched gateway never healthy")
+        r = client.chat.completions.create(
+            model="llama-fallback-demo",
+            messages=[{"role": "user", "content": "hi"}], max_tokens=2)
+        assert r.usage.completion_tokens == 2
+    finally:
+        import signal
+        os.killpg(proc.pid, signal.SIGTERM)
+        try:
+            proc.wait(timeout=10)
+        except Exception:
+            os.killpg(proc.pid, signal.SIGKILL)
