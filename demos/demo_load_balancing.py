@@ -131,7 +131,7 @@ def main() -> int:
     ap = argparse.ArgumentParser(description="Load-balancing demo")
     ap.add_argument("--requests", type=int, default=10)
     ap.add_argument("--loop", action="store_true")
-    ap.add_argument("--interval", type=int, default=10,
+    ap.add_argument("--interval", type=int, default=30,
                     help="seconds between loop rounds (>=5)")
     add_common_args(ap)
     args = ap.parse_args()

@@ -145,7 +145,7 @@ def run_loop_mode(args, client: OpenAIClient, base_url: str) -> dict:
 def main() -> int:
     ap = argparse.ArgumentParser(description="Quota isolation demo")
     ap.add_argument("--loop", action="store_true")
-    ap.add_argument("--interval", type=int, default=65,
+    ap.add_argument("--interval", type=int, default=30,
                     help="seconds between rounds (>=5; quotas reset per minute)")
     add_common_args(ap)
     args = ap.parse_args()
