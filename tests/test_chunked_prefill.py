@@ -118,3 +118,28 @@ def test_budget_never_starves_prefill():
     outs = drain(e)
     assert len(outs) == 6
     assert all(len(v) == 3 for v in outs.values())
+
+
+def test_abort_mid_chunked_prefill_frees_blocks():
+    """Aborting while a long prompt is mid-chunk returns ALL its blocks
+    (reserved at admission) and other requests are unaffected."""
+    e = make_engine(chunk_size=8, num_blocks=128)
+    free0 = e.kv.free_blocks
+    e.add_request("long", list(range(3, 120)), SamplingParams(max_tokens=4))
+    e.add_request("other", list(range(7, 30)), SamplingParams(max_tokens=3))
+    e.step()                               # admit + first chunks
+    assert e.kv.free_blocks < free0
+    e.abort("long")
+    outs = drain(e)
+    assert "long" not in outs
+    assert len(outs["other"]) == 3
+    assert e.kv.free_blocks == free0       # everything returned
+
+
+def test_abort_waiting_request_never_allocates():
+    e = make_engine(chunk_size=8, num_blocks=128)
+    free0 = e.kv.free_blocks
+    e.add_request("w", list(range(3, 40)), SamplingParams(max_tokens=2))
+    e.abort("w")                           # aborted while still queued
+    assert drain(e) == {}
+    assert e.kv.free_blocks == free0
