@@ -82,6 +82,18 @@ MODEL_PRESETS: dict[str, LlamaConfig] = {
         name="llama-3-70b", hidden_size=8192, n_layers=80, n_heads=64,
         n_kv_heads=8, head_dim=128, intermediate_size=28672,
         vocab_size=128256),
+    "llama-3-1b": LlamaConfig(
+        # Llama-3.2-1B geometry: small-footprint replica for dense
+        # multi-replica serving on one GPU (head_dim 64 path)
+        name="llama-3-1b", hidden_size=2048, n_layers=16, n_heads=32,
+        n_kv_heads=8, head_dim=64, intermediate_size=8192,
+        vocab_size=128256),
+    "mistral-7b": LlamaConfig(
+        # Mistral-7B-v0.3 geometry (no sliding window: full attention,
+        # which is exact for contexts <= max_position)
+        name="mistral-7b", hidden_size=4096, n_layers=32, n_heads=32,
+        n_kv_heads=8, head_dim=128, intermediate_size=14336,
+        vocab_size=32768, rope_theta=1000000.0),
     # CPU-testable models
     "tiny": LlamaConfig(
         name="tiny", hidden_size=256, n_layers=2, n_heads=4, n_kv_heads=2,

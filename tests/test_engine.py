@@ -1,5 +1,7 @@
 """Engine + model tests on CPU (tiny model, fp32 reference ops)."""
 
+import dataclasses
+
 import pytest
 import torch
 
@@ -178,3 +180,30 @@ def test_top_p_sampling_path():
                                   seed=5))
     assert [o.token_id for o in drain(e2)["p"]] == \
         [o.token_id for o in outs["p"]]
+
+
+def test_additional_model_presets_generate():
+    """Every non-flagship preset steps through the engine on CPU (the
+    same LlamaForCausalLM covers the whole family — dims only)."""
+    from resilient_llm_amd.models import get_config
+
+    for name in ("llama-3-1b", "mistral-7b"):
+        cfg = get_config(name)
+        assert cfg.n_heads * cfg.head_dim in (cfg.hidden_size, cfg.hidden_size * 2)
+        # shrink to CPU-testable size but keep the preset's SHAPE rules
+        small = dataclasses.replace(cfg, n_layers=2, vocab_size=256,
+                                    hidden_size=cfg.hidden_size // 16,
+                                    intermediate_size=cfg.intermediate_size // 16,
+                                    n_heads=max(2, cfg.n_heads // 16),
+                                    n_kv_heads=max(1, cfg.n_kv_heads // 8),
+                                    head_dim=64, max_position=512)
+        model = LlamaForCausalLM(small, device="cpu", dtype=torch.float32,
+                                 seed=1)
+        kv = PagedKVCache.for_model(small, 64, device="cpu")
+        kv.k = kv.k.float(); kv.v = kv.v.float()
+        eng = LLMEngine(model, kv, max_batch_size=2)
+        eng.add_request("a", list(range(5, 25)), SamplingParams(max_tokens=3))
+        toks = []
+        while eng.has_work():
+            toks += [o.token_id for o in eng.step()]
+        assert len(toks) == 3
