@@ -98,6 +98,12 @@ class HttpServer:
             limit=MAX_HEADER_BYTES, reuse_address=True,
             reuse_port=reuse_port)
 
+    def close_listener(self) -> None:
+        """Stop ACCEPTING; existing connections keep being served
+        (graceful-drain phase 1)."""
+        if self._server is not None:
+            self._server.close()
+
     async def stop(self) -> None:
         if self._server is not None:
             self._server.close()

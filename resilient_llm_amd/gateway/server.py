@@ -50,7 +50,7 @@ def build_registry(config: Config, stub_kwargs: Optional[dict] = None) -> Worker
 
 async def serve(config: Config, registry: Optional[WorkerRegistry] = None,
                 ready_event: Optional[asyncio.Event] = None,
-                reuse_port: bool = False) -> None:
+                reuse_port: bool = False, drain_s: float = 30.0) -> None:
     registry = registry or build_registry(config)
     app = GatewayApp(config, registry)
     server = HttpServer(app.handle, host=config.cluster.host, port=config.cluster.port)
@@ -74,6 +74,19 @@ async def serve(config: Config, registry: Optional[WorkerRegistry] = None,
             pass
     try:
         await stop.wait()
+        # graceful drain: stop accepting, let in-flight requests finish
+        server.close_listener()
+        deadline = loop.time() + drain_s
+        while loop.time() < deadline:
+            busy = sum(s.in_flight for s in app.router.states)
+            if busy == 0:
+                break
+            await asyncio.sleep(0.1)
+        else:
+            log_with_timestamp(
+                f"drain timeout after {drain_s}s; "
+                f"{sum(s.in_flight for s in app.router.states)} still "
+                f"in flight", "yellow")
     finally:
         await app.stop_background()
         await server.stop()
@@ -85,6 +98,9 @@ def main() -> None:
     ap.add_argument("--config", default="config/config.yaml")
     ap.add_argument("--port", type=int, default=None, help="override cluster.port")
     ap.add_argument("--host", default=None)
+    ap.add_argument("--drain-s", type=float, default=30.0,
+                    help="graceful-shutdown drain window for in-flight "
+                         "requests on SIGTERM/SIGINT")
     ap.add_argument("--workers", type=int, default=1,
                     help="gateway processes sharing the port via "
                          "SO_REUSEPORT (NOTE: rate-limit windows and the "
@@ -111,12 +127,14 @@ def main() -> None:
         for _ in range(args.workers - 1):
             pr = multiprocessing.Process(
                 target=lambda: asyncio.run(serve(config, registry,
-                                                 reuse_port=True)))
+                                                 reuse_port=True,
+                                                 drain_s=args.drain_s)))
             pr.start()
             procs.append(pr)
-        asyncio.run(serve(config, registry, reuse_port=True))
+        asyncio.run(serve(config, registry, reuse_port=True,
+                          drain_s=args.drain_s))
     else:
-        asyncio.run(serve(config, registry))
+        asyncio.run(serve(config, registry, drain_s=args.drain_s))
 
 
 if __name__ == "__main__":

@@ -113,3 +113,62 @@ def test_start_gateway_script():
             proc.wait(timeout=10)
         except Exception:
             os.killpg(proc.pid, signal.SIGKILL)
+
+
+def test_graceful_drain_on_sigterm():
+    """SIGTERM mid-request: the gateway stops accepting, the in-flight
+    request COMPLETES, and the process exits 0 — no dropped work."""
+    import json as _json
+    import signal as _signal
+    import threading
+    import urllib.request
+
+    port = free_port()
+    # slow stub (~1.2 s per request) so SIGTERM lands mid-request
+    cfg = os.path.join(REPO, "config", "config.stub.yaml")
+    proc = subprocess.Popen(
+        [sys.executable, "-c",
+         "import sys; sys.argv=['x','--config',%r,'--port','%d']; "
+         "from resilient_llm_amd.gateway import server as S; "
+         "from resilient_llm_amd.config import load_config; "
+         "import asyncio; "
+         "cfg=load_config(%r); cfg.cluster.port=%d; "
+         "reg=S.build_registry(cfg, stub_kwargs={'first_token_ms':1200}); "
+         "asyncio.run(S.serve(cfg, reg))" % (cfg, port, cfg, port)],
+        env={**os.environ, "PYTHONPATH": REPO},
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    try:
+        client = OpenAIClient(f"http://127.0.0.1:{port}")
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            try:
+                if client.health().get("status") == "ok":
+                    break
+            except Exception:
+                time.sleep(0.2)
+        else:
+            raise AssertionError("gateway never healthy")
+
+        result = {}
+
+        def slow_request():
+            body = _json.dumps({
+                "model": "llama-fallback-loadbalance",
+                "messages": [{"role": "user", "content": "slow"}],
+                "max_tokens": 3}).encode()
+            req = urllib.request.Request(
+                f"http://127.0.0.1:{port}/chat/completions", data=body,
+                headers={"content-type": "application/json"})
+            with urllib.request.urlopen(req, timeout=30) as r:
+                result["data"] = _json.loads(r.read())
+
+        th = threading.Thread(target=slow_request)
+        th.start()
+        time.sleep(0.4)                  # request is in flight
+        proc.send_signal(_signal.SIGTERM)
+        th.join(timeout=30)
+        assert result["data"]["usage"]["completion_tokens"] == 3
+        assert proc.wait(timeout=20) == 0
+    finally:
+        if proc.poll() is None:
+            proc.kill()
