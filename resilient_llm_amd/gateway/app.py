@@ -272,9 +272,9 @@ class GatewayApp:
         return prompt_est, prompt_est + max_tokens
 
     def _gen_request(self, body: dict, consumer: str, ticket: Ticket,
-                     stream: bool) -> GenerationRequest:
+                     stream: bool, rid: Optional[str] = None) -> GenerationRequest:
         return GenerationRequest(
-            request_id=f"req-{uuid.uuid4().hex[:16]}",
+            request_id=rid or f"req-{uuid.uuid4().hex[:16]}",
             model=ticket.deployment.backend_model,
             messages=body["messages"],
             max_tokens=int(body.get("max_tokens") or DEFAULT_MAX_TOKENS),
@@ -319,9 +319,13 @@ class GatewayApp:
         if not self.consumers.try_acquire(consumer, total_est):
             return Response.error(429, f"consumer {consumer!r} rate limit exceeded",
                                   err_type="rate_limit_error", code="rate_limit_exceeded")
+        # request-id propagation: honor the client's x-request-id (or
+        # mint one); it tags the ledger rows and every response
+        rid = req.headers.get("x-request-id") or f"req-{uuid.uuid4().hex[:16]}"
         if bool(body.get("stream")):
-            return await self._chat_stream(body, alias, consumer, total_est)
-        return await self._chat_once(body, alias, consumer, total_est)
+            return await self._chat_stream(body, alias, consumer, total_est,
+                                           rid)
+        return await self._chat_once(body, alias, consumer, total_est, rid)
 
     # max attempts across replicas + fallback chain before giving up
     def _max_attempts(self, alias: str) -> int:
@@ -334,7 +338,7 @@ class GatewayApp:
         return max(2, min(n + 1, 8))
 
     async def _chat_once(self, body: dict, alias: str, consumer: str,
-                         total_est: int) -> Response:
+                         total_est: int, rid: Optional[str] = None) -> Response:
         t0 = time.monotonic()
         exclude: set = set()
         last_err: Optional[Exception] = None
@@ -349,7 +353,8 @@ class GatewayApp:
             except RouterRateLimit as e:
                 return Response.error(429, str(e), err_type="rate_limit_error",
                                       code="rate_limit_exceeded")
-            greq = self._gen_request(body, consumer, ticket, stream=False)
+            greq = self._gen_request(body, consumer, ticket, stream=False,
+                                     rid=rid)
             try:
                 worker = self._worker_for(ticket)
                 result = await asyncio.wait_for(worker.generate(greq),
@@ -395,12 +400,13 @@ class GatewayApp:
             "x-gateway-model-id": ticket.deployment.model_id,
             "x-gateway-device": worker.device,
             "x-gateway-fallback": "true" if ticket.is_fallback else "false",
+            "x-request-id": greq.request_id,
         }
         return Response.json_response(payload, headers=headers)
 
     # -------------------------------------------------------- streaming
     async def _chat_stream(self, body: dict, alias: str, consumer: str,
-                           total_est: int) -> Response:
+                           total_est: int, rid: Optional[str] = None) -> Response:
         """SSE streaming (X14) with hot failover: a mid-stream worker
         death/hang re-routes to the next deployment, which regenerates;
         tokens the client already received are discarded (stateless
@@ -423,7 +429,7 @@ class GatewayApp:
             ticket: Optional[Ticket] = first_ticket
             last_err: Optional[Exception] = None
             attempts = app._max_attempts(alias)
-            comp_id = f"chatcmpl-{uuid.uuid4().hex[:16]}"
+            comp_id = f"chatcmpl-{rid or uuid.uuid4().hex[:16]}"
             try:
                 for attempt in range(attempts):
                     if ticket is None:
@@ -432,7 +438,8 @@ class GatewayApp:
                         except NoDeploymentAvailable as e:
                             last_err = e
                             break
-                    greq = app._gen_request(body, consumer, ticket, stream=True)
+                    greq = app._gen_request(body, consumer, ticket,
+                                            stream=True, rid=rid)
                     try:
                         worker = app._worker_for(ticket)
                     except WorkerError as e:
@@ -520,6 +527,7 @@ class GatewayApp:
             "x-gateway-model-id": first_ticket.deployment.model_id,
             "x-gateway-fallback": "true" if first_ticket.is_fallback else "false",
             "cache-control": "no-cache",
+            "x-request-id": rid or "",
         }
         return Response(status=200, headers=headers,
                         content_type="text/event-stream", body_iter=body_iter())

@@ -2,6 +2,7 @@
 BASELINE.json config 1 (plumbing, no GPU)."""
 
 import concurrent.futures as cf
+import json
 import time
 
 import pytest
@@ -314,3 +315,38 @@ def test_consumer_limits_per_api_key():
         for _ in range(4):
             free.chat.completions.create(model="llama-cris-demo",
                                          messages=MSGS, max_tokens=2)
+
+
+def test_request_id_propagation():
+    """x-request-id: client-supplied ids echo back and tag the ledger;
+    absent ids are minted."""
+    import http.client
+
+    with run_gateway() as (client, registry, config):
+        conn = http.client.HTTPConnection(client.host, client.port, timeout=15)
+        body = json.dumps({"model": "llama-fallback-loadbalance",
+                           "messages": [{"role": "user", "content": "id"}],
+                           "max_tokens": 2})
+        conn.request("POST", "/chat/completions", body=body,
+                     headers={"content-type": "application/json",
+                              "x-request-id": "req-my-trace-42"})
+        resp = conn.getresponse()
+        assert resp.status == 200
+        assert resp.getheader("x-request-id") == "req-my-trace-42"
+        resp.read()
+
+        # minted when absent
+        conn.request("POST", "/chat/completions", body=body,
+                     headers={"content-type": "application/json"})
+        resp2 = conn.getresponse()
+        minted = resp2.getheader("x-request-id")
+        assert minted and minted.startswith("req-") and minted != "req-my-trace-42"
+        resp2.read()
+        conn.close()
+
+        import urllib.request
+        with urllib.request.urlopen(
+                f"http://{client.host}:{client.port}/admin/requests?n=10",
+                timeout=10) as r:
+            rows = json.loads(r.read())["requests"]
+        assert "req-my-trace-42" in [row["request_id"] for row in rows]
