@@ -422,6 +422,9 @@ class GatewayApp:
 
         app = self
 
+        include_usage = bool((body.get("stream_options") or {})
+                             .get("include_usage"))
+
         async def body_iter() -> AsyncIterator[bytes]:
             t0 = time.monotonic()
             sent_tokens = 0
@@ -484,6 +487,22 @@ class GatewayApp:
                             finish_reason=finish or "stop")
                         app.router.complete(ticket, actual_tokens=None)
                         app._record(ticket, greq, consumer, "ok", t0, result, worker.device)
+                        if include_usage:
+                            # OpenAI stream_options.include_usage: one
+                            # final chunk with empty choices + usage
+                            usage_evt = {
+                                "id": comp_id,
+                                "object": "chat.completion.chunk",
+                                "created": int(time.time()),
+                                "model": model_id, "choices": [],
+                                "usage": {
+                                    "prompt_tokens": result.prompt_tokens,
+                                    "completion_tokens": sent_tokens,
+                                    "total_tokens":
+                                        result.prompt_tokens + sent_tokens,
+                                },
+                            }
+                            yield f"data: {json.dumps(usage_evt)}\n\n".encode()
                         yield b"data: [DONE]\n\n"
                         return
                     except (WorkerError, asyncio.TimeoutError) as e:

@@ -350,3 +350,30 @@ def test_request_id_propagation():
                 timeout=10) as r:
             rows = json.loads(r.read())["requests"]
         assert "req-my-trace-42" in [row["request_id"] for row in rows]
+
+
+def test_stream_include_usage():
+    """stream_options.include_usage emits a final usage chunk with
+    empty choices before [DONE] (OpenAI semantics)."""
+    import http.client
+
+    with run_gateway() as (client, registry, config):
+        conn = http.client.HTTPConnection(client.host, client.port, timeout=15)
+        body = json.dumps({"model": "llama-fallback-loadbalance",
+                           "messages": [{"role": "user", "content": "u"}],
+                           "max_tokens": 3, "stream": True,
+                           "stream_options": {"include_usage": True}})
+        conn.request("POST", "/chat/completions", body=body,
+                     headers={"content-type": "application/json"})
+        resp = conn.getresponse()
+        raw = resp.read().decode()
+        conn.close()
+        datas = [json.loads(l[6:]) for l in raw.splitlines()
+                 if l.startswith("data: ") and l != "data: [DONE]"]
+        assert raw.rstrip().endswith("data: [DONE]")
+        usage_chunks = [d for d in datas if d.get("usage")]
+        assert len(usage_chunks) == 1
+        assert usage_chunks[0]["choices"] == []
+        assert usage_chunks[0]["usage"]["completion_tokens"] == 3
+        # content chunks don't carry usage
+        assert all("usage" not in d for d in datas[:-1])
