@@ -77,6 +77,16 @@ class ConsumerLimiter:
             lim.reconcile(estimated, actual)
 
 
+def _normalize_stop(stop) -> Optional[list]:
+    """OpenAI `stop`: a string or up to 4 strings."""
+    if stop is None:
+        return None
+    if isinstance(stop, str):
+        stop = [stop]
+    stop = [s for s in stop if isinstance(s, str) and s]
+    return stop[:4] or None
+
+
 class AccessPolicy:
     """API-key -> alias ACL — the least-privilege IAM policy analogue
     (reference iam/policy.json, C8 in SURVEY.md §2.1): which consumers
@@ -283,6 +293,7 @@ class GatewayApp:
             seed=body.get("seed"),
             stream=stream,
             consumer=consumer,
+            stop=_normalize_stop(body.get("stop")),
         )
 
     def _record(self, ticket: Ticket, greq: Optional[GenerationRequest],

@@ -43,6 +43,7 @@ class GenerationRequest:
     seed: Optional[int] = None
     stream: bool = False
     consumer: str = "anonymous"
+    stop: Optional[list] = None      # OpenAI stop sequences (<= 4 strings)
 
 
 @dataclasses.dataclass

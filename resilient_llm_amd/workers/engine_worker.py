@@ -50,6 +50,16 @@ def render_prompt(messages: list) -> str:
     return "\n".join(parts)
 
 
+def _earliest_stop(text: str, stops) -> int:
+    """Index of the earliest stop-sequence occurrence, or -1."""
+    cut = -1
+    for s in stops or []:
+        i = text.find(s)
+        if i >= 0 and (cut < 0 or i < cut):
+            cut = i
+    return cut
+
+
 class EngineWorker(Worker):
     def __init__(self, device: str, model_name: str = "llama-3-8b",
                  device_label: Optional[str] = None,
@@ -235,9 +245,14 @@ class EngineWorker(Worker):
             outs, first_t = got
             token_ids = [o.token_id for o in outs]
             finish = outs[-1].finish_reason or "stop"
+            text = self.tokenizer.decode(token_ids)
+            cut = _earliest_stop(text, req.stop)
+            if cut >= 0:
+                text = text[:cut]
+                finish = "stop"
             self.total_served += 1
             return GenerationResult(
-                text=self.tokenizer.decode(token_ids),
+                text=text,
                 prompt_tokens=n_prompt,
                 completion_tokens=len(token_ids),
                 finish_reason=finish,
@@ -257,6 +272,10 @@ class EngineWorker(Worker):
         try:
             emitted = ""
             token_ids: list[int] = []
+            stops = req.stop or []
+            # hold back enough text that a stop sequence spanning chunk
+            # boundaries is never partially emitted
+            hold = max((len(s) - 1 for s in stops), default=0)
             while True:
                 out = await q.get()
                 self._check_fault()
@@ -264,8 +283,17 @@ class EngineWorker(Worker):
                     raise WorkerError(f"engine error: {out}") from out
                 token_ids.append(out.token_id)
                 full = self.tokenizer.decode(token_ids)
-                delta = full[len(emitted):]
-                emitted = full
+                cut = _earliest_stop(full, stops)
+                if cut >= 0:
+                    delta = full[len(emitted):cut]
+                    emitted = full[:cut]
+                    yield GenerationChunk(text=delta, token_id=out.token_id,
+                                          finish_reason="stop")
+                    break
+                visible = len(full) if out.finished else max(
+                    len(emitted), len(full) - hold)
+                delta = full[len(emitted):visible]
+                emitted = full[:visible]
                 yield GenerationChunk(
                     text=delta, token_id=out.token_id,
                     finish_reason=out.finish_reason if out.finished else None)

@@ -76,11 +76,17 @@ class StubWorker(Worker):
                 await asyncio.sleep(self.token_delay_ms * n / 1000.0)
             self._check_fault()
             self.total_served += 1
+            text = " ".join(words)
+            finish = "length" if n == req.max_tokens else "stop"
+            from .engine_worker import _earliest_stop
+            cut = _earliest_stop(text, req.stop)
+            if cut >= 0:
+                text, finish = text[:cut], "stop"
             return GenerationResult(
-                text=" ".join(words),
+                text=text,
                 prompt_tokens=self._prompt_tokens(req),
                 completion_tokens=n,
-                finish_reason="length" if n == req.max_tokens else "stop",
+                finish_reason=finish,
                 ttft_ms=(time.monotonic() - t0) * 1000.0,
             )
         finally:

@@ -141,3 +141,38 @@ def test_worker_target_step_ms_plumbed():
         finally:
             await w.close()
     asyncio.run(run())
+
+
+def test_stop_sequences_final_and_stream():
+    """OpenAI `stop`: generation truncates at the earliest stop string
+    (never emitted), finish_reason becomes "stop" — both modes."""
+    async def run():
+        w = EngineWorker(device="cpu", model_name="tiny", device_label="st",
+                         num_blocks=64, seed=0)
+        try:
+            base = await w.generate(GenerationRequest(
+                request_id="b", model="tiny",
+                messages=[{"role": "user", "content": "stop test"}],
+                max_tokens=8))
+            assert len(base.text) > 2
+            stop_s = base.text[2:4]     # guaranteed to occur
+            res = await w.generate(GenerationRequest(
+                request_id="s", model="tiny",
+                messages=[{"role": "user", "content": "stop test"}],
+                max_tokens=8, stop=[stop_s]))
+            assert stop_s not in res.text
+            assert res.finish_reason == "stop"
+            assert base.text.startswith(res.text)
+
+            chunks = []
+            async for c in w.generate_stream(GenerationRequest(
+                    request_id="t", model="tiny",
+                    messages=[{"role": "user", "content": "stop test"}],
+                    max_tokens=8, stop=[stop_s], stream=True)):
+                chunks.append(c)
+            text = "".join(c.text for c in chunks)
+            assert text == res.text
+            assert chunks[-1].finish_reason == "stop"
+        finally:
+            await w.close()
+    asyncio.run(run())
