@@ -436,9 +436,12 @@ class GatewayApp:
         include_usage = bool((body.get("stream_options") or {})
                              .get("include_usage"))
 
+        prompt_est, _ = self._estimate(body)
+
         async def body_iter() -> AsyncIterator[bytes]:
             t0 = time.monotonic()
             sent_tokens = 0
+            prompt_toks = prompt_est       # refined by the first chunk
             exclude: set = set()
             ticket: Optional[Ticket] = first_ticket
             last_err: Optional[Exception] = None
@@ -477,6 +480,8 @@ class GatewayApp:
                             except StopAsyncIteration:
                                 break
                             completion_tokens += 1
+                            if chunk.prompt_tokens is not None:
+                                prompt_toks = chunk.prompt_tokens
                             if completion_tokens <= sent_tokens:
                                 finish = chunk.finish_reason or finish
                                 continue   # replay skip after failover
@@ -494,7 +499,8 @@ class GatewayApp:
                             yield f"data: {json.dumps(evt)}\n\n".encode()
                         sent_tokens += n_emitted
                         result = GenerationResult(
-                            text="", prompt_tokens=total_est, completion_tokens=sent_tokens,
+                            text="", prompt_tokens=prompt_toks,
+                            completion_tokens=sent_tokens,
                             finish_reason=finish or "stop")
                         app.router.complete(ticket, actual_tokens=None)
                         app._record(ticket, greq, consumer, "ok", t0, result, worker.device)

@@ -51,6 +51,7 @@ class GenerationChunk:
     text: str
     token_id: Optional[int] = None
     finish_reason: Optional[str] = None   # set on the last chunk
+    prompt_tokens: Optional[int] = None   # set on the FIRST chunk
 
 
 @dataclasses.dataclass

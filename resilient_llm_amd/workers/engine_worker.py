@@ -288,7 +288,9 @@ class EngineWorker(Worker):
                     delta = full[len(emitted):cut]
                     emitted = full[:cut]
                     yield GenerationChunk(text=delta, token_id=out.token_id,
-                                          finish_reason="stop")
+                                          finish_reason="stop",
+                                          prompt_tokens=n_prompt
+                                          if len(token_ids) == 1 else None)
                     break
                 visible = len(full) if out.finished else max(
                     len(emitted), len(full) - hold)
@@ -296,7 +298,8 @@ class EngineWorker(Worker):
                 emitted = full[:visible]
                 yield GenerationChunk(
                     text=delta, token_id=out.token_id,
-                    finish_reason=out.finish_reason if out.finished else None)
+                    finish_reason=out.finish_reason if out.finished else None,
+                    prompt_tokens=n_prompt if len(token_ids) == 1 else None)
                 if out.finished:
                     break
             self.total_served += 1

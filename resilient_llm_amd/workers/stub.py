@@ -109,6 +109,7 @@ class StubWorker(Worker):
                 yield GenerationChunk(
                     text=(w if i == 0 else " " + w),
                     finish_reason=("length" if n == req.max_tokens else "stop") if last else None,
+                    prompt_tokens=self._prompt_tokens(req) if i == 0 else None,
                 )
             self.total_served += 1
         finally:

@@ -375,5 +375,8 @@ def test_stream_include_usage():
         assert len(usage_chunks) == 1
         assert usage_chunks[0]["choices"] == []
         assert usage_chunks[0]["usage"]["completion_tokens"] == 3
+        # prompt_tokens is the worker's ACTUAL count (1-word prompt),
+        # not the admission estimate that includes max_tokens
+        assert usage_chunks[0]["usage"]["prompt_tokens"] < 10
         # content chunks don't carry usage
         assert all("usage" not in d for d in datas[:-1])
