@@ -502,7 +502,9 @@ class GatewayApp:
                             text="", prompt_tokens=prompt_toks,
                             completion_tokens=sent_tokens,
                             finish_reason=finish or "stop")
-                        app.router.complete(ticket, actual_tokens=None)
+                        actual = prompt_toks + sent_tokens
+                        app.router.complete(ticket, actual_tokens=actual)
+                        app.consumers.reconcile(consumer, total_est, actual)
                         app._record(ticket, greq, consumer, "ok", t0, result, worker.device)
                         if include_usage:
                             # OpenAI stream_options.include_usage: one
