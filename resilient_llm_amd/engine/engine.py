@@ -35,8 +35,19 @@ class SamplingParams:
     max_tokens: int = 128
     temperature: float = 0.0
     top_p: float = 1.0
+    presence_penalty: float = 0.0     # subtract once per seen output token
+    frequency_penalty: float = 0.0    # subtract per occurrence
     seed: Optional[int] = None
     stop_on_eos: bool = True
+
+    @property
+    def needs_torch_sampling(self) -> bool:
+        """Rows the fused gumbel-max kernel cannot serve: nucleus
+        filtering or repetition penalties (penalties need the row's
+        output-token histogram, kept host-side)."""
+        return ((self.temperature > 0 and self.top_p < 1.0)
+                or self.presence_penalty != 0.0
+                or self.frequency_penalty != 0.0)
 
 
 @dataclasses.dataclass
@@ -320,20 +331,38 @@ class LLMEngine:
         seeds = torch.tensor([self._seq_seed(s) for s in seqs],
                              dtype=torch.int64, device=self.device)
         toks = ops.sample(logits, temps, seeds, 0)
-        topp = [i for i, s in enumerate(seqs)
-                if s.params.temperature > 0 and s.params.top_p < 1.0]
-        if topp:
+        special = [i for i, s in enumerate(seqs)
+                   if s.params.needs_torch_sampling]
+        if special:
             toks = toks.clone()
-            for i in topp:
-                toks[i] = self._sample_top_p(logits[i], seqs[i])
+            for i in special:
+                toks[i] = self._sample_torch(logits[i], seqs[i])
         return toks.tolist()
 
-    def _sample_top_p(self, row_logits: torch.Tensor, s: SeqState) -> int:
-        """Nucleus sampling (torch path; rows that ask for top_p < 1)."""
-        probs = torch.softmax(row_logits.float() / s.params.temperature, -1)
-        sp, idx = probs.sort(descending=True)
-        keep = int((sp.cumsum(0) < s.params.top_p).sum()) + 1
-        sp = sp[:keep] / sp[:keep].sum()
+    def _sample_torch(self, row_logits: torch.Tensor, s: SeqState) -> int:
+        """Torch sampling path for rows the fused kernel cannot serve:
+        repetition penalties (OpenAI presence/frequency semantics) and/or
+        nucleus (top_p) filtering.  Greedy rows with penalties argmax
+        the penalized logits."""
+        p = s.params
+        lg = row_logits.float()
+        if p.presence_penalty or p.frequency_penalty:
+            lg = lg.clone()
+            counts: dict[int, int] = {}
+            for t in s.output_ids:
+                counts[t] = counts.get(t, 0) + 1
+            for t, c in counts.items():
+                lg[t] -= p.presence_penalty + p.frequency_penalty * c
+        if p.temperature <= 0:
+            return int(lg.argmax())
+        probs = torch.softmax(lg / p.temperature, -1)
+        if p.top_p < 1.0:
+            sp, idx = probs.sort(descending=True)
+            keep = int((sp.cumsum(0) < p.top_p).sum()) + 1
+            sp = sp[:keep] / sp[:keep].sum()
+        else:
+            sp, idx = probs, torch.arange(probs.shape[0],
+                                          device=probs.device)
         gen = torch.Generator(device=row_logits.device)
         gen.manual_seed(self._seq_seed(s))
         return int(idx[int(torch.multinomial(sp, 1, generator=gen))])
@@ -486,8 +515,7 @@ class LLMEngine:
         self._b_seeds = torch.tensor([self._seq_seed(s) for s in seqs],
                                      dtype=torch.int64, device=dev)
         self._b_step_off = 0
-        self._b_has_topp = any(s.params.temperature > 0 and s.params.top_p < 1.0
-                               for s in seqs)
+        self._b_has_topp = any(s.params.needs_torch_sampling for s in seqs)
         self._batch_dirty = False
 
     def _decode_step(self) -> list[StepOutput]:

@@ -290,6 +290,8 @@ class GatewayApp:
             max_tokens=int(body.get("max_tokens") or DEFAULT_MAX_TOKENS),
             temperature=float(body.get("temperature", 0.0)),
             top_p=float(body.get("top_p", 1.0)),
+            presence_penalty=float(body.get("presence_penalty", 0.0)),
+            frequency_penalty=float(body.get("frequency_penalty", 0.0)),
             seed=body.get("seed"),
             stream=stream,
             consumer=consumer,

@@ -40,6 +40,8 @@ class GenerationRequest:
     max_tokens: int = 128
     temperature: float = 0.0
     top_p: float = 1.0
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
     seed: Optional[int] = None
     stream: bool = False
     consumer: str = "anonymous"

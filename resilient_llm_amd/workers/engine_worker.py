@@ -194,6 +194,8 @@ class EngineWorker(Worker):
         params = SamplingParams(max_tokens=req.max_tokens,
                                 temperature=req.temperature,
                                 top_p=req.top_p,
+                                presence_penalty=req.presence_penalty,
+                                frequency_penalty=req.frequency_penalty,
                                 seed=req.seed)
         self._req_counter += 1
         rid = f"{req.request_id}-{self._req_counter}"
