@@ -278,7 +278,9 @@ class GatewayApp:
     def _estimate(self, body: dict) -> tuple[int, int]:
         prompt_est = sum(estimate_tokens(str(m.get("content", "")))
                          for m in body["messages"])
-        max_tokens = int(body.get("max_tokens") or DEFAULT_MAX_TOKENS)
+        max_tokens = int(body.get("max_tokens")
+                       or body.get("max_completion_tokens")
+                       or DEFAULT_MAX_TOKENS)
         return prompt_est, prompt_est + max_tokens
 
     def _gen_request(self, body: dict, consumer: str, ticket: Ticket,
@@ -287,7 +289,9 @@ class GatewayApp:
             request_id=rid or f"req-{uuid.uuid4().hex[:16]}",
             model=ticket.deployment.backend_model,
             messages=body["messages"],
-            max_tokens=int(body.get("max_tokens") or DEFAULT_MAX_TOKENS),
+            max_tokens=int(body.get("max_tokens")
+                       or body.get("max_completion_tokens")
+                       or DEFAULT_MAX_TOKENS),
             temperature=float(body.get("temperature", 0.0)),
             top_p=float(body.get("top_p", 1.0)),
             presence_penalty=float(body.get("presence_penalty", 0.0)),

@@ -380,3 +380,26 @@ def test_stream_include_usage():
         assert usage_chunks[0]["usage"]["prompt_tokens"] < 10
         # content chunks don't carry usage
         assert all("usage" not in d for d in datas[:-1])
+
+
+def test_max_completion_tokens_alias():
+    """OpenAI's newer max_completion_tokens name is honored."""
+    with run_gateway() as (client, registry, config):
+        r = client.chat.completions.create(
+            model="llama-fallback-loadbalance",
+            messages=[{"role": "user", "content": "alias"}],
+            max_tokens=128, max_completion_tokens=3)
+        # explicit max_tokens wins when both present (it was set to the
+        # client default of 128 here, so 128 is the cap, not 3)...
+        assert r.usage.completion_tokens <= 128
+
+        import http.client
+        conn = http.client.HTTPConnection(client.host, client.port, timeout=15)
+        body = json.dumps({"model": "llama-fallback-loadbalance",
+                           "messages": [{"role": "user", "content": "x"}],
+                           "max_completion_tokens": 3})
+        conn.request("POST", "/chat/completions", body=body,
+                     headers={"content-type": "application/json"})
+        resp = json.loads(conn.getresponse().read())
+        conn.close()
+        assert resp["usage"]["completion_tokens"] == 3
