@@ -226,6 +226,10 @@ class GatewayApp:
             if req.method != "POST":
                 return Response.error(405, "use POST")
             return await self.chat_completions(req)
+        if req.path in ("/completions", "/v1/completions"):
+            if req.method != "POST":
+                return Response.error(405, "use POST")
+            return await self.text_completions(req)
         if route == ("GET", "/health"):
             return await self.health(req)
         if req.path in ("/models", "/v1/models") and req.method == "GET":
@@ -343,6 +347,47 @@ class GatewayApp:
             return await self._chat_stream(body, alias, consumer, total_est,
                                            rid)
         return await self._chat_once(body, alias, consumer, total_est, rid)
+
+    async def text_completions(self, req: Request) -> Response:
+        """Legacy /completions: the prompt is wrapped as a single user
+        message and the chat result is re-shaped to text_completion.
+        Streaming is chat-endpoint-only (the reference exercises only
+        /chat/completions — SURVEY.md X1)."""
+        try:
+            body = req.json()
+            if not isinstance(body, dict) or not isinstance(body.get("model"), str):
+                raise ValueError("'model' is required")
+            prompt = body.get("prompt")
+            if isinstance(prompt, list):
+                prompt = prompt[0] if prompt else ""
+            if not isinstance(prompt, str):
+                raise ValueError("'prompt' must be a string")
+        except (ValueError, json.JSONDecodeError) as e:
+            return Response.error(400, str(e), err_type="invalid_request_error")
+        if body.get("stream"):
+            return Response.error(
+                400, "streaming is supported on /chat/completions",
+                err_type="invalid_request_error")
+        chat_body = dict(body)
+        chat_body.pop("prompt", None)
+        chat_body["messages"] = [{"role": "user", "content": prompt}]
+        chat_req = Request(req.method, req.path, req.headers,
+                           json.dumps(chat_body).encode(), req.http_version)
+        resp = await self.chat_completions(chat_req)
+        if resp.status != 200:
+            return resp
+        payload = json.loads(resp.body)
+        text = payload["choices"][0]["message"]["content"]
+        out = {
+            "id": payload["id"].replace("chatcmpl-", "cmpl-", 1),
+            "object": "text_completion",
+            "created": payload["created"],
+            "model": payload["model"],
+            "choices": [{"index": 0, "text": text, "logprobs": None,
+                         "finish_reason": payload["choices"][0]["finish_reason"]}],
+            "usage": payload["usage"],
+        }
+        return Response.json_response(out, headers=resp.headers)
 
     # max attempts across replicas + fallback chain before giving up
     def _max_attempts(self, alias: str) -> int:

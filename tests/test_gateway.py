@@ -403,3 +403,21 @@ def test_max_completion_tokens_alias():
         resp = json.loads(conn.getresponse().read())
         conn.close()
         assert resp["usage"]["completion_tokens"] == 3
+
+
+def test_legacy_completions_endpoint():
+    with run_gateway() as (client, registry, config):
+        import http.client
+        conn = http.client.HTTPConnection(client.host, client.port, timeout=15)
+        body = json.dumps({"model": "llama-fallback-loadbalance",
+                           "prompt": "complete me", "max_tokens": 3})
+        conn.request("POST", "/v1/completions", body=body,
+                     headers={"content-type": "application/json"})
+        resp = conn.getresponse()
+        assert resp.status == 200
+        data = json.loads(resp.read())
+        conn.close()
+        assert data["object"] == "text_completion"
+        assert data["id"].startswith("cmpl-")
+        assert data["choices"][0]["text"]
+        assert data["usage"]["completion_tokens"] == 3
