@@ -277,6 +277,23 @@ class GatewayApp:
         msgs = body.get("messages")
         if not isinstance(msgs, list) or not msgs:
             raise ValueError("'messages' must be a non-empty list")
+        # numeric fields must coerce NOW so a bad type is a 400, not a
+        # 500 from deep inside the dispatch path (found by the API fuzz
+        # test)
+        for field, kind in (("max_tokens", int),
+                            ("max_completion_tokens", int),
+                            ("seed", int),
+                            ("temperature", float), ("top_p", float),
+                            ("presence_penalty", float),
+                            ("frequency_penalty", float)):
+            v = body.get(field)
+            if v is None:
+                continue
+            if isinstance(v, bool) or not isinstance(v, (int, float)):
+                raise ValueError(f"'{field}' must be a number")
+            body[field] = kind(v)
+        if body.get("max_tokens") is not None and body["max_tokens"] < 1:
+            raise ValueError("'max_tokens' must be >= 1")
         return body
 
     def _estimate(self, body: dict) -> tuple[int, int]:

@@ -421,3 +421,59 @@ def test_legacy_completions_endpoint():
         assert data["id"].startswith("cmpl-")
         assert data["choices"][0]["text"]
         assert data["usage"]["completion_tokens"] == 3
+
+
+def test_api_fuzz_never_5xx():
+    """Deterministic mini-fuzz over the request surface: malformed and
+    edge-case bodies must map to 2xx/4xx — never a 5xx (handler bugs)."""
+    import http.client
+    import random
+
+    rng = random.Random(1234)
+    bodies = [
+        {},                                        # missing everything
+        {"model": "nope"},                         # unknown alias, no messages
+        {"model": "llama-fallback-loadbalance"},   # no messages
+        {"model": "llama-fallback-loadbalance", "messages": []},
+        {"model": "llama-fallback-loadbalance",
+         "messages": [{"role": "user", "content": "x"}],
+         "max_tokens": "three"},                   # bad type
+        {"model": 5, "messages": [{"role": "user", "content": "x"}]},
+        {"model": "llama-fallback-loadbalance",
+         "messages": [{"role": "user", "content": "x"}],
+         "temperature": -1.0, "top_p": 0.0},
+        {"model": "llama-fallback-loadbalance",
+         "messages": [{"role": "user", "content": "x"}],
+         "stop": [""] * 10},
+        {"model": "llama-fallback-loadbalance",
+         "messages": [{"role": "user", "content": "x"}],
+         "stop": 42},
+        {"model": "llama-fallback-loadbalance",
+         "messages": [{"role": "user", "content": "x"}],
+         "presence_penalty": "NaNish"},
+        {"model": "unknown-model",
+         "messages": [{"role": "user", "content": "x"}]},
+    ]
+    # plus a few random-garbage structures
+    for _ in range(8):
+        bodies.append({"model": rng.choice(["llama-fallback-loadbalance", 7]),
+                       "messages": rng.choice([None, "hi", [{"role": "user",
+                                                            "content": "y"}]]),
+                       rng.choice(["seed", "stream", "max_tokens"]):
+                           rng.choice(["x", -5, 1.5, {}, []])})
+
+    with run_gateway() as (client, registry, config):
+        conn = http.client.HTTPConnection(client.host, client.port, timeout=15)
+        for body in bodies:
+            conn.request("POST", "/chat/completions", body=json.dumps(body),
+                         headers={"content-type": "application/json"})
+            resp = conn.getresponse()
+            resp.read()
+            assert resp.status < 500, (resp.status, body)
+        # raw non-JSON body
+        conn.request("POST", "/chat/completions", body=b"\x00\x01notjson",
+                     headers={"content-type": "application/json"})
+        resp = conn.getresponse()
+        resp.read()
+        conn.close()
+        assert resp.status == 400
