@@ -294,6 +294,11 @@ class GatewayApp:
             body[field] = kind(v)
         if body.get("max_tokens") is not None and body["max_tokens"] < 1:
             raise ValueError("'max_tokens' must be >= 1")
+        stop = body.get("stop")
+        if stop is not None and not isinstance(stop, str) and not (
+                isinstance(stop, list)
+                and all(isinstance(s, str) for s in stop)):
+            raise ValueError("'stop' must be a string or list of strings")
         return body
 
     def _estimate(self, body: dict) -> tuple[int, int]:
