@@ -78,6 +78,12 @@ class WorkerRpcServer:
                     msg = await recv_frame(reader)
                 except (asyncio.IncompleteReadError, ConnectionResetError):
                     return
+                except Exception:
+                    # malformed frame (bad length, not msgpack): drop the
+                    # CONNECTION, never the worker
+                    return
+                if not isinstance(msg, dict):
+                    continue
                 asyncio.ensure_future(self._dispatch(msg, reply))
         finally:
             writer.close()

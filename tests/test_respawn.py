@@ -227,3 +227,47 @@ def test_health_loop_detects_dead_pool_follower():
         for p in procs:
             if p.poll() is None:
                 p.kill()
+
+
+def test_rpc_server_survives_malformed_frames():
+    """Garbage on the worker socket (bad lengths, non-msgpack, wrong
+    types) drops that connection but never the worker: a well-formed
+    client keeps working."""
+    import struct
+    import tempfile as _tf
+
+    import msgpack
+
+    from resilient_llm_amd.workers.engine_worker import EngineWorker
+    from resilient_llm_amd.workers.rpc import WorkerRpcServer
+
+    sock = os.path.join(_tf.mkdtemp(prefix="rlli-fz-"), "w.sock")
+
+    async def run():
+        w = EngineWorker(device="cpu", model_name="tiny", device_label="fz",
+                         num_blocks=64, seed=0)
+        srv = WorkerRpcServer(w, sock)
+        await srv.start()
+        try:
+            for payload in (
+                    b"\xff\xff\xff\xff",                     # 4 GiB length
+                    b"\x05\x00\x00\x00hello",                # not msgpack
+                    struct.pack("<I", 1) + msgpack.packb(7), # non-dict msg
+                    b"\x02\x00",                             # truncated
+            ):
+                r, wtr = await asyncio.open_unix_connection(sock)
+                wtr.write(payload)
+                await wtr.drain()
+                wtr.close()
+            await asyncio.sleep(0.2)
+
+            client = RpcWorkerClient("fz", {"tiny"}, sock)
+            await client.connect(timeout=30)
+            h = await client.health()
+            assert h["status"] == "ok"
+            await client.close()
+        finally:
+            await srv.stop()
+            await w.close()
+
+    asyncio.run(run())
