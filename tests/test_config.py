@@ -140,3 +140,47 @@ def test_70b_pool_config_loads():
     assert all(d.backend_model == "llama-3-70b" for d in pool_deps)
     assert len(cfg.deployments_for("llama-70b")) == 2
     assert cfg.router.fallbacks["llama-70b-pool-a"] == ["llama-70b-pool-b"]
+
+
+def test_config_fuzz_always_config_error():
+    """Malformed configs raise ConfigError (a ValueError) with a clear
+    message — never KeyError/TypeError/AttributeError from deep inside."""
+    from resilient_llm_amd.config import ConfigError
+
+    bad = [
+        None,
+        [],
+        {"model_list": None},
+        {"model_list": []},
+        {"model_list": "nope"},
+        {"model_list": [None]},
+        {"model_list": [{"litellm_params": {"model": "gpu/0/m"}}]},   # no name
+        {"model_list": [{"model_name": 7,
+                         "litellm_params": {"model": "gpu/0/m"}}]},
+        {"model_list": [{"model_name": "m", "litellm_params": None}]},
+        {"model_list": [{"model_name": "m",
+                         "litellm_params": {"model": "nodelimiter"}}]},
+        {"model_list": [{"model_name": "m",
+                         "litellm_params": {"model": "weird/0/m"}}]},
+        {"model_list": [{"model_name": "m",
+                         "litellm_params": {"model": "gpu/0/x"},
+                         "rpm": -3}]},
+        {"cluster": {"port": 80},          # privileged port rejected
+         "model_list": [{"model_name": "m",
+                         "litellm_params": {"model": "stub/0/x"}}]},
+        {"cluster": {"pools": {"p": {"gpus": "zero"}}},
+         "model_list": [{"model_name": "m",
+                         "litellm_params": {"model": "stub/0/x"}}]},
+        {"cluster": {"pools": {"p": {"gpus": [0, 1], "tensor_parallel": 3}}},
+         "model_list": [{"model_name": "m",
+                         "litellm_params": {"model": "stub/0/x"}}]},
+        {"model_list": [{"model_name": "m",
+                         "litellm_params": {"model": "stub/0/x"}}],
+         "router_settings": {"fallbacks": "broken"}},
+        {"model_list": [{"model_name": "m",
+                         "litellm_params": {"model": "stub/0/x"}}],
+         "router_settings": {"fallbacks": [{"a": "not-a-list"}]}},
+    ]
+    for data in bad:
+        with pytest.raises(ConfigError):
+            load_config(data=data)
