@@ -515,7 +515,7 @@ class LLMEngine:
         self._b_seeds = torch.tensor([self._seq_seed(s) for s in seqs],
                                      dtype=torch.int64, device=dev)
         self._b_step_off = 0
-        self._b_has_topp = any(s.params.needs_torch_sampling for s in seqs)
+        self._b_torch_sampling = any(s.params.needs_torch_sampling for s in seqs)
         self._batch_dirty = False
 
     def _decode_step(self) -> list[StepOutput]:
@@ -533,7 +533,7 @@ class LLMEngine:
         else:
             logits = self.model.forward_decode(self._b_ids, pos, self.kv,
                                                slots, bt, seq_lens)
-        if self._b_has_topp:
+        if self._b_torch_sampling:
             tokens = self._sample(logits, seqs)     # torch top-p path (sync)
             tok_dev = torch.tensor(tokens, dtype=torch.int32,
                                    device=self.device)
