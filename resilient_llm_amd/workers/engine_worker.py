@@ -327,6 +327,10 @@ class EngineWorker(Worker):
                 "queued": len(self.engine.waiting),
                 "running": len(self.engine.running),
                 "kv_free_blocks": self.engine.kv.free_blocks,
+                "prefix_cache": {
+                    "hits": self.engine.kv.prefix_hits,
+                    "lookups": self.engine.kv.prefix_lookups,
+                },
                 "total_served": self.total_served,
                 "engine_stats": {k: v for k, v in self.engine.stats.items()
                                  if k != "admit_events"}}

@@ -107,3 +107,29 @@ def test_full_block_prompt_keeps_one_suffix_token():
     e.add_request("b", prompt, SamplingParams(max_tokens=4))
     t2 = drain(e)["b"]
     assert t1 == t2
+
+
+def test_prefix_stats_in_worker_health():
+    import asyncio
+
+    from resilient_llm_amd.workers.engine_worker import EngineWorker
+    from resilient_llm_amd.workers.base import GenerationRequest
+
+    async def run():
+        w = EngineWorker(device="cpu", model_name="tiny", device_label="pc",
+                         num_blocks=64, seed=0)
+        try:
+            req = GenerationRequest(
+                request_id="a", model="tiny",
+                messages=[{"role": "user", "content": "same prompt " * 8}],
+                max_tokens=2)
+            await w.generate(req)
+            await w.generate(GenerationRequest(
+                request_id="b", model="tiny", messages=req.messages,
+                max_tokens=2))
+            h = await w.health()
+            pc = h["prefix_cache"]
+            assert pc["lookups"] >= 2 and pc["hits"] >= 1
+        finally:
+            await w.close()
+    asyncio.run(run())
