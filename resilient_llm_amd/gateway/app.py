@@ -129,6 +129,7 @@ class GatewayApp:
         self.policy = AccessPolicy(config.raw.get("auth"))
         self.health_interval_s = health_interval_s
         self.respawn_cooldown_s = 10.0
+        self.last_health: dict = {}     # worker key -> last health dict
         self._health_task: Optional[asyncio.Task] = None
         self.started_at = time.time()
 
@@ -170,8 +171,9 @@ class GatewayApp:
             await asyncio.sleep(self.health_interval_s)
             for key, worker in self.registry.all().items():
                 try:
-                    await asyncio.wait_for(worker.health(),
-                                           timeout=self.health_interval_s)
+                    h = await asyncio.wait_for(worker.health(),
+                                               timeout=self.health_interval_s)
+                    self.last_health[key] = h
                     ok = True
                 except Exception:
                     ok = False
@@ -716,5 +718,15 @@ class GatewayApp:
         st = self.ledger.stats()
         for k in ("total", "ok", "errors", "throttled", "fallbacks"):
             lines.append(f'gateway_ledger_{k} {st[k]}')
+        for key, h in self.last_health.items():
+            wl = f'worker="{key}"'
+            for field in ("in_flight", "queued", "running",
+                          "kv_free_blocks", "total_served"):
+                if field in h:
+                    lines.append(f'worker_{field}{{{wl}}} {h[field]}')
+            pc = h.get("prefix_cache")
+            if pc:
+                lines.append(f'worker_prefix_hits{{{wl}}} {pc["hits"]}')
+                lines.append(f'worker_prefix_lookups{{{wl}}} {pc["lookups"]}')
         return Response(body="\n".join(lines) + "\n",
                         content_type="text/plain; version=0.0.4")

@@ -201,6 +201,17 @@ def test_metrics_endpoint():
         conn.close()
         assert "gateway_requests_total" in body
         assert "gateway_ledger_ok 1" in body
+        # worker gauges appear once the health loop has swept
+        import time as _t
+        deadline = _t.time() + 5
+        while _t.time() < deadline and "worker_in_flight" not in body:
+            _t.sleep(0.3)
+            conn = client._connect(None)
+            conn.request("GET", "/metrics")
+            body = conn.getresponse().read().decode()
+            conn.close()
+        assert "worker_in_flight" in body
+        assert "worker_total_served" in body
 
 
 def test_throttled_worker_typed_status():
