@@ -10,8 +10,9 @@ synchronous read of per-GPU counters with no propagation delay (SURVEY.md
 pool ("account"), deployment id, consumer — plus serving-quality fields
 (latency, time-to-first-token, token counts) CloudWatch never had.
 
-Optionally appends JSONL for post-hoc analysis, and feeds the Prometheus
-exporter in :mod:`resilient_llm_amd.obs.metrics`.
+Optionally appends JSONL for post-hoc analysis, and feeds the gateway's
+Prometheus exposition (``GET /metrics`` — gateway/app.py
+prometheus_metrics).
 """
 
 from __future__ import annotations
