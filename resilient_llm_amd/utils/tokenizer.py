@@ -32,3 +32,26 @@ class ByteTokenizer:
         data = bytes(i - BYTE_OFFSET for i in ids
                      if BYTE_OFFSET <= i < BYTE_OFFSET + 256)
         return data.decode("utf-8", errors="replace")
+
+    def decode_stream(self, ids: list[int], final: bool) -> str:
+        """Streaming-safe decode: while not ``final``, a trailing
+        INCOMPLETE multi-byte UTF-8 sequence is held back instead of
+        decoding to U+FFFD — otherwise an already-emitted delta would
+        retroactively change once the remaining continuation bytes
+        arrive, and the concatenated stream would diverge from the
+        final text."""
+        data = bytes(i - BYTE_OFFSET for i in ids
+                     if BYTE_OFFSET <= i < BYTE_OFFSET + 256)
+        if not final and data:
+            # find a lead byte within the last 3 positions; if its
+            # sequence is unfinished, strip it
+            for k in range(1, min(4, len(data) + 1)):
+                b = data[-k]
+                if (b & 0xC0) == 0x80:       # continuation, keep looking
+                    continue
+                if b >= 0xC0:                # lead byte of 2-4 byte seq
+                    need = 2 if b < 0xE0 else 3 if b < 0xF0 else 4
+                    if k < need:
+                        data = data[:-k]
+                break                        # ASCII or complete: stop
+        return data.decode("utf-8", errors="replace")

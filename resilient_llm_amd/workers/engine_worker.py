@@ -284,7 +284,7 @@ class EngineWorker(Worker):
                 if isinstance(out, Exception):
                     raise WorkerError(f"engine error: {out}") from out
                 token_ids.append(out.token_id)
-                full = self.tokenizer.decode(token_ids)
+                full = self.tokenizer.decode_stream(token_ids, out.finished)
                 cut = _earliest_stop(full, stops)
                 if cut >= 0:
                     delta = full[len(emitted):cut]

@@ -176,3 +176,22 @@ def test_stop_sequences_final_and_stream():
         finally:
             await w.close()
     asyncio.run(run())
+
+
+def test_stream_decode_holds_back_partial_utf8():
+    from resilient_llm_amd.utils.tokenizer import BYTE_OFFSET, ByteTokenizer
+
+    tok = ByteTokenizer(512)
+    ids = tok.encode("é", add_bos=False)        # 2 bytes
+    assert len(ids) == 2
+    assert tok.decode_stream(ids[:1], final=False) == ""     # held back
+    assert tok.decode_stream(ids, final=False) == "é"
+    assert tok.decode_stream(ids[:1], final=True) == "�"
+    # 4-byte emoji split at every point
+    e = tok.encode("\N{ROCKET}", add_bos=False)
+    for cut in range(1, 4):
+        assert tok.decode_stream(e[:cut], final=False) == ""
+    assert tok.decode_stream(e, final=False) == "\N{ROCKET}"
+    # ascii passes through untouched
+    a = tok.encode("ok", add_bos=False)
+    assert tok.decode_stream(a, final=False) == "ok"
