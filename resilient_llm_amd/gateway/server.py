@@ -129,10 +129,20 @@ def main() -> None:
                 target=lambda: asyncio.run(serve(config, registry,
                                                  reuse_port=True,
                                                  drain_s=args.drain_s)))
+            # die with the parent: SIGTERM is delivered to the parent
+            # only, and an orphaned reuse-port sibling would keep the
+            # port serving forever
+            pr.daemon = True
             pr.start()
             procs.append(pr)
-        asyncio.run(serve(config, registry, reuse_port=True,
-                          drain_s=args.drain_s))
+        try:
+            asyncio.run(serve(config, registry, reuse_port=True,
+                              drain_s=args.drain_s))
+        finally:
+            for pr in procs:
+                pr.terminate()
+            for pr in procs:
+                pr.join(timeout=10)
     else:
         asyncio.run(serve(config, registry, drain_s=args.drain_s))
 

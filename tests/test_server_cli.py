@@ -80,6 +80,14 @@ def test_server_cli_multiple_workers_reuse_port():
             proc.wait(timeout=10)
         except Exception:
             proc.kill()
+    # the forked reuse-port sibling must die with the parent (it leaked
+    # an orphan per run before the daemon+terminate fix): the port must
+    # stop answering entirely
+    time.sleep(1.0)
+    import socket as _s
+    with _s.socket() as s:
+        assert s.connect_ex(("127.0.0.1", port)) != 0, \
+            "orphaned reuse-port worker still serving"
 
 
 def test_start_gateway_script():
