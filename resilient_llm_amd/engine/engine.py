@@ -89,6 +89,7 @@ class LLMEngine:
                  admission_window_s: float = 0.0,
                  enable_prefix_caching: bool = True,
                  target_step_ms: Optional[float] = None,
+                 eos_ids=None,
                  seed: int = 0) -> None:
         self.model = model
         self.kv = kv_cache
@@ -99,7 +100,13 @@ class LLMEngine:
         self.max_model_len = max_model_len or getattr(
             model.config, "max_position", 8192)
         self.seed = seed
-        self.eos_id = 2
+        # per-model EOS: taken from the model config
+        # (Llama-3 presets carry 128001/128009) unless the caller
+        # overrides — the worker passes the tokenizer's/deployment's ids
+        if eos_ids is not None:
+            self.eos_ids = frozenset(int(e) for e in eos_ids)
+        else:
+            self.eos_ids = frozenset(getattr(model.config, "eos_ids", (2,)))
         # waiting/abort are the ONLY cross-thread surfaces: guarded by a
         # small queue lock so enqueue NEVER waits on a running step (a
         # coarse lock here serialized arrivals behind 100ms prefills and
@@ -125,6 +132,7 @@ class LLMEngine:
         self.step_count = 0
         self.block_size = kv_cache.block_size
         self._aborted: set[str] = set()
+        self._live: set[str] = set()   # rids currently owned by the engine
         self._arrival_counter = 0
         self._batch_dirty = True
         self._pending = None          # (seqs, event, host_tokens) 1-deep
@@ -167,10 +175,15 @@ class LLMEngine:
                 f"blocks, cache has {self.kv.num_blocks}")
         with self._queue_lock:
             self.waiting.append(seq)
+            self._live.add(req_id)
 
     def abort(self, req_id: str) -> None:
+        """Abort is a no-op for rids the engine no longer owns (finished /
+        never added), so post-completion aborts cannot grow ``_aborted``
+        without bound (one leaked entry per request)."""
         with self._queue_lock:
-            self._aborted.add(req_id)
+            if req_id in self._live:
+                self._aborted.add(req_id)
 
     # --------------------------------------------------------- scheduling
     def _admit(self) -> list[SeqState]:
@@ -193,6 +206,7 @@ class LLMEngine:
             if seq.req_id in self._aborted:
                 self.waiting.popleft()
                 self._aborted.discard(seq.req_id)
+                self._live.discard(seq.req_id)
                 continue
             if tokens + len(seq.prompt_ids) > self.max_prefill_tokens and admitted:
                 break
@@ -224,7 +238,7 @@ class LLMEngine:
     def _finish(self, seq: SeqState, outs: list[StepOutput], token: int) -> bool:
         p = seq.params
         reason = None
-        if p.stop_on_eos and token == self.eos_id:
+        if p.stop_on_eos and token in self.eos_ids:
             reason = "stop"
         elif len(seq.output_ids) >= p.max_tokens:
             reason = "length"
@@ -235,6 +249,7 @@ class LLMEngine:
         if reason is not None:
             self.kv.free(seq.blocks)
             seq.blocks = []
+            self._live.discard(seq.req_id)
             return True
         return False
 
@@ -313,17 +328,29 @@ class LLMEngine:
                     self.kv.free(seq.blocks)
                     seq.blocks = []
                     self._aborted.discard(seq.req_id)
+                    self._live.discard(seq.req_id)
                 else:
                     keep.append(seq)
             if len(keep) != len(getattr(self, attr)):
                 self._batch_dirty = True
             setattr(self, attr, keep)
+        # an abort that raced a normal finish matches no live sequence:
+        # clear it here so _aborted cannot accumulate stale rids
+        with self._queue_lock:
+            self._aborted &= self._live
 
     @staticmethod
     def _seq_seed(s: SeqState) -> int:
+        """Per-(request, absolute output index) seed, wrapped to SIGNED
+        int64 (full 64-bit value — a 63-bit mask here would discard the
+        bit-63 carry at whatever step _rebuild_batch happened to run,
+        while the sampling kernel's later ``+ GOLDEN*step`` additions wrap
+        mod 2^64: tokens would depend on rebuild timing).
+        Both the kernel (uint64 reinterpret) and the torch fallback see
+        the same premix regardless of where the step offset was added."""
         base = s.params.seed if s.params.seed is not None else s.default_seed
-        return (base + 0x9e3779b97f4a7c15 * len(s.output_ids)) \
-            & 0x7fffffffffffffff
+        v = (base + 0x9e3779b97f4a7c15 * len(s.output_ids)) & 0xffffffffffffffff
+        return v - 0x10000000000000000 if v >= 0x8000000000000000 else v
 
     def _sample(self, logits: torch.Tensor, seqs: list[SeqState]) -> list[int]:
         temps = torch.tensor([s.params.temperature for s in seqs],

@@ -55,3 +55,66 @@ class ByteTokenizer:
                         data = data[:-k]
                 break                        # ASCII or complete: stop
         return data.decode("utf-8", errors="replace")
+
+
+class HFTokenizer:
+    """Real tokenizer loaded from a weights directory (tokenizer.json /
+    tokenizer.model next to the safetensors shards) — VERDICT r01 #6:
+    real-checkpoint serving needs real token boundaries and the model's
+    actual EOS ids, not the byte codec's hardcoded 2."""
+
+    def __init__(self, path: str) -> None:
+        from transformers import AutoTokenizer
+        self._tok = AutoTokenizer.from_pretrained(path)
+        self.vocab_size = len(self._tok)
+        eos = self._tok.eos_token_id
+        ids = set()
+        if eos is not None:
+            ids.add(int(eos))
+        # Llama-3 instruct checkpoints end turns with <|eot_id|> while
+        # eos_token is <|end_of_text|>: both must stop generation
+        for special in ("<|eot_id|>", "<|end_of_text|>", "</s>", "<|im_end|>"):
+            try:
+                i = self._tok.convert_tokens_to_ids(special)
+            except Exception:
+                i = None
+            if i is not None and i >= 0 and i != getattr(
+                    self._tok, "unk_token_id", None):
+                ids.add(int(i))
+        self.eos_ids = frozenset(ids) or frozenset({EOS_ID})
+        self.eos_id = min(self.eos_ids)
+        self.bos_id = self._tok.bos_token_id
+
+    def encode(self, text: str, add_bos: bool = True) -> list:
+        return self._tok.encode(text, add_special_tokens=add_bos)
+
+    def decode(self, ids: list) -> str:
+        return self._tok.decode(ids, skip_special_tokens=True)
+
+    def decode_stream(self, ids: list, final: bool) -> str:
+        text = self._tok.decode(ids, skip_special_tokens=True)
+        # hold back a trailing replacement char: a multi-byte character
+        # split across BPE tokens must not emit U+FFFD mid-stream
+        if not final:
+            text = text.rstrip("�")
+        return text
+
+
+def _has_tokenizer_files(path: str) -> bool:
+    import os
+    return any(os.path.exists(os.path.join(path, f))
+               for f in ("tokenizer.json", "tokenizer.model",
+                         "tokenizer_config.json"))
+
+
+def load_tokenizer(weights_dir, vocab_size: int):
+    """HF tokenizer when the weights dir ships one (and transformers is
+    importable), else the byte codec — the synthetic-workload default."""
+    if weights_dir and _has_tokenizer_files(weights_dir):
+        try:
+            return HFTokenizer(weights_dir)
+        except Exception as e:          # missing transformers / bad files
+            from .logging import log_with_timestamp
+            log_with_timestamp(f"tokenizer load from {weights_dir} failed "
+                               f"({e}); using byte codec", "yellow")
+    return ByteTokenizer(vocab_size)

@@ -14,6 +14,7 @@ killing the actual GPU process.
 from __future__ import annotations
 
 import asyncio
+import dataclasses
 import os
 import threading
 import time
@@ -24,7 +25,7 @@ import torch
 from ..engine import LLMEngine, PagedKVCache, SamplingParams
 from ..engine.engine import CapacityExceeded
 from ..models import LlamaForCausalLM, get_config
-from ..utils.tokenizer import ByteTokenizer
+from ..utils.tokenizer import load_tokenizer
 from .base import (
     GenerationChunk, GenerationRequest, GenerationResult, Worker,
     WorkerDead, WorkerError, WorkerThrottled,
@@ -74,6 +75,8 @@ class EngineWorker(Worker):
                  tp_rank: int = 0, tp_world: int = 1, tp_group=None,
                  tp_control=None,
                  target_step_ms: Optional[float] = None,
+                 weights: Optional[str] = None,
+                 eos_id=None,
                  seed: int = 0) -> None:
         super().__init__(device=device_label or f"gpu:{device}",
                          models={model_name})
@@ -87,6 +90,21 @@ class EngineWorker(Worker):
         self.model = LlamaForCausalLM(config, device=torch_device, dtype=dtype,
                                       tp_rank=tp_rank, tp_world=tp_world,
                                       tp_group=tp_group, seed=seed)
+        if weights:
+            n = self.model.load_safetensors(weights)
+            from ..utils.logging import log_with_timestamp
+            log_with_timestamp(f"{self.device}: loaded {n} tensors from "
+                               f"{weights} (tp {tp_rank}/{tp_world})", "grey")
+        self.tokenizer = load_tokenizer(weights, config.vocab_size)
+        # EOS precedence: explicit deployment override > tokenizer-declared
+        # ids (real HF tokenizer) > model-preset ids
+        if eos_id is not None:
+            eos_ids = [eos_id] if isinstance(eos_id, int) else list(eos_id)
+        elif getattr(self.tokenizer, "eos_ids", None):
+            eos_ids = list(self.tokenizer.eos_ids)
+        else:
+            eos_ids = list(getattr(config, "eos_ids", (2,)))
+        self.eos_ids = frozenset(eos_ids)
         nb = num_blocks or default_num_blocks(config, kv_gb, tp_world=tp_world)
         kv = PagedKVCache.for_model(config, nb, device=torch_device,
                                     tp_world=tp_world)
@@ -97,6 +115,7 @@ class EngineWorker(Worker):
                                 max_queue=max_queue, seed=seed,
                                 max_prefill_tokens=max_prefill_tokens,
                                 chunk_size=chunk_size,
+                                eos_ids=self.eos_ids,
                                 admission_window_s=0.0 if tp_control is not None
                                 else 0.006,
                                 # wall-clock tuning diverges under TP
@@ -107,7 +126,6 @@ class EngineWorker(Worker):
             from ..engine.graph import install_graph_runner
             install_graph_runner(self.engine)
         self.tp_control = tp_control
-        self.tokenizer = ByteTokenizer(config.vocab_size)
         self.fault_mode = "none"
         self.total_served = 0
         self._req_counter = 0
@@ -213,12 +231,11 @@ class EngineWorker(Worker):
                         f"{self.engine.kv.num_blocks}")
                 if len(self.engine.waiting) >= self.engine.max_queue:
                     raise CapacityExceeded(f"queue full ({self.engine.max_queue})")
+                # the FULL sampling params ride the broadcast: a dropped
+                # field (r01 dropped the penalties) silently no-ops that
+                # feature on TP pools and can diverge rank sampling
                 self.tp_control.submit(
-                    ("add", rid, prompt_ids,
-                     {"max_tokens": params.max_tokens,
-                      "temperature": params.temperature,
-                      "top_p": params.top_p,
-                      "seed": params.seed}))
+                    ("add", rid, prompt_ids, dataclasses.asdict(params)))
             else:
                 self.engine.add_request(rid, prompt_ids, params)
         except CapacityExceeded as e:
@@ -235,16 +252,25 @@ class EngineWorker(Worker):
         return self._in_flight
 
     # --------------------------------------------------------------- API
+    def _abort(self, rid: str) -> None:
+        if self.tp_control is not None:
+            self.tp_control.submit(("abort", rid))
+            self._work_event.set()
+        else:
+            self.engine.abort(rid)
+
     async def generate(self, req: GenerationRequest) -> GenerationResult:
         t0 = time.monotonic()
         rid, q, n_prompt = self._enqueue(req, mode="final")
         self._in_flight += 1
+        finished = False
         try:
             got = await q.get()
             self._check_fault()
             if isinstance(got, Exception):
                 raise WorkerError(f"engine error: {got}") from got
             outs, first_t = got
+            finished = True    # engine freed the sequence itself: no abort
             token_ids = [o.token_id for o in outs]
             finish = outs[-1].finish_reason or "stop"
             text = self.tokenizer.decode(token_ids)
@@ -262,15 +288,13 @@ class EngineWorker(Worker):
         finally:
             self._in_flight -= 1
             self._cleanup(rid)
-            if self.tp_control is not None:
-                self.tp_control.submit(("abort", rid))
-                self._work_event.set()
-            else:
-                self.engine.abort(rid)
+            if not finished:   # abort only requests the engine still owns
+                self._abort(rid)
 
     async def _stream_impl(self, req: GenerationRequest) -> AsyncIterator[GenerationChunk]:
         rid, q, n_prompt = self._enqueue(req)
         self._in_flight += 1
+        finished = False
         try:
             emitted = ""
             token_ids: list[int] = []
@@ -303,16 +327,16 @@ class EngineWorker(Worker):
                     finish_reason=out.finish_reason if out.finished else None,
                     prompt_tokens=n_prompt if len(token_ids) == 1 else None)
                 if out.finished:
+                    finished = True   # engine-side finish: no abort needed
                     break
             self.total_served += 1
         finally:
             self._in_flight -= 1
             self._cleanup(rid)
-            if self.tp_control is not None:
-                self.tp_control.submit(("abort", rid))
-                self._work_event.set()
-            else:
-                self.engine.abort(rid)
+            # a stop-sequence cut or client disconnect leaves the engine
+            # still generating: abort it; a normal finish needs nothing
+            if not finished:
+                self._abort(rid)
 
     def generate_stream(self, req: GenerationRequest) -> AsyncIterator[GenerationChunk]:
         return self._stream_impl(req)

@@ -31,6 +31,8 @@ def spawn_gpu_worker(device_index: str, model_name: str,
                      socket_path: str, kv_gb: float = 24.0,
                      max_batch: int = 64, use_graphs: bool = True,
                      target_step_ms: float | None = None,
+                     weights: str | None = None,
+                     eos_ids: list | None = None,
                      extra_env: dict | None = None) -> subprocess.Popen:
     """``device_index`` may be a sub-device replica like ``0.1``: several
     worker processes co-located on physical GPU 0 — 288 GB of HBM3E holds
@@ -51,6 +53,10 @@ def spawn_gpu_worker(device_index: str, model_name: str,
         cmd.append("--graphs")
     if target_step_ms is not None:
         cmd += ["--target-step-ms", str(target_step_ms)]
+    if weights:
+        cmd += ["--weights", str(weights)]
+    for e in eos_ids or []:
+        cmd += ["--eos-id", str(e)]
     return subprocess.Popen(cmd, env=env)
 
 
@@ -67,6 +73,7 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
     has_star = False
     star_models: set[str] = set()
     pools: dict[str, str] = {}
+    opts: dict[str, dict] = {}   # target -> extra litellm_params (weights, eos_id)
     for d in config.deployments:
         if d.backend_kind == "gpu":
             if d.backend_target == "*":
@@ -74,8 +81,12 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
                 star_models.add(d.backend_model)
             else:
                 gpu_models.setdefault(d.backend_target, set()).add(d.backend_model)
+                if d.params:
+                    opts.setdefault(d.backend_target, d.params)
         elif d.backend_kind == "pool":
             pools[d.backend_target] = d.backend_model
+            if d.params:
+                opts.setdefault(f"pool:{d.backend_target}", d.params)
     if has_star and not gpu_models:
         n = torch.cuda.device_count()
         for i in range(n):
@@ -92,8 +103,14 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
         sock = os.path.join(run_dir, f"gpu{target}.sock")
         log_with_timestamp(f"spawning worker gpu:{target} ({model_name})", "grey")
         slo = config.cluster.target_step_ms
+        extra = opts.get(target, {})
+        weights = extra.get("weights")
+        eos = extra.get("eos_id")
+        eos_list = None if eos is None else (
+            [int(eos)] if isinstance(eos, int) else [int(e) for e in eos])
         proc = spawn_gpu_worker(target, model_name, sock, kv_gb=kv_gb,
                                 max_batch=max_batch, use_graphs=use_graphs,
+                                weights=weights, eos_ids=eos_list,
                                 target_step_ms=slo)
         client = RpcWorkerClient(f"gpu:{target}", {model_name}, sock)
         client.proc = proc
@@ -101,6 +118,7 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
                           spawn_gpu_worker(t, m, s, kv_gb=kv_gb,
                                            max_batch=max_batch,
                                            use_graphs=use_graphs,
+                                           weights=weights, eos_ids=eos_list,
                                            target_step_ms=slo))
         registry.register("gpu", target, client)
 
@@ -111,12 +129,14 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
         log_with_timestamp(
             f"spawning TP={pool_def.tensor_parallel} pool worker "
             f"pool:{pool_name} on GPUs {pool_def.gpus} ({model_name})", "grey")
+        weights = opts.get(f"pool:{pool_name}", {}).get("weights")
         procs = spawn_pool_worker(pool_def, model_name, sock, kv_gb=kv_gb,
-                                  max_batch=max_batch)
+                                  max_batch=max_batch, weights=weights)
         client = RpcWorkerClient(f"pool:{pool_name}", {model_name}, sock)
         client.proc = procs[0]
         client.proc_group = procs
         client.respawn = (lambda pd=pool_def, m=model_name, s=sock:
                           spawn_pool_worker(pd, m, s, kv_gb=kv_gb,
-                                            max_batch=max_batch))
+                                            max_batch=max_batch,
+                                            weights=weights))
         registry.register("pool", pool_name, client)

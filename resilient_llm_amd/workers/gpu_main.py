@@ -21,6 +21,11 @@ def main() -> None:
     ap.add_argument("--target-step-ms", type=float, default=None,
                     help="AIMD prefill budget: keep mixed steps near "
                          "this bound (decode-cadence SLO)")
+    ap.add_argument("--weights", default=None,
+                    help="safetensors dir (HF Llama names); random init "
+                         "when absent")
+    ap.add_argument("--eos-id", type=int, action="append", default=None,
+                    help="override EOS token id(s); repeatable")
     ap.add_argument("--device", default=None,
                     help="torch device override (tests: cpu)")
     args = ap.parse_args()
@@ -49,6 +54,8 @@ def main() -> None:
                               max_batch_size=args.max_batch,
                               num_blocks=args.num_blocks,
                               use_graphs=args.graphs,
+                              weights=args.weights,
+                              eos_id=args.eos_id,
                               target_step_ms=args.target_step_ms)
         log_with_timestamp(
             f"worker {args.device_label} ready: {args.model} on {device}, "

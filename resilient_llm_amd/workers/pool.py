@@ -85,6 +85,7 @@ def _free_port() -> int:
 def spawn_pool_worker(pool: PoolDef, model_name: str, socket_path: str,
                       kv_gb: float = 24.0, max_batch: int = 64,
                       device_override: Optional[str] = None,
+                      weights: Optional[str] = None,
                       tp_backend: str = "nccl") -> list:
     """Spawn pool.tensor_parallel processes; returns the Popen list
     (leader first)."""
@@ -109,5 +110,7 @@ def spawn_pool_worker(pool: PoolDef, model_name: str, socket_path: str,
                "--tp-backend", tp_backend]
         if device_override is not None:
             cmd += ["--device", device_override]
+        if weights:
+            cmd += ["--weights", str(weights)]
         procs.append(subprocess.Popen(cmd, env=env))
     return procs

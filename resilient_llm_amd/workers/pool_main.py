@@ -20,6 +20,8 @@ def main() -> None:
     ap.add_argument("--max-batch", type=int, default=64)
     ap.add_argument("--num-blocks", type=int, default=None)
     ap.add_argument("--tp-backend", default="nccl")
+    ap.add_argument("--weights", default=None,
+                    help="safetensors dir — each rank loads its TP shard")
     ap.add_argument("--device", default=None, help="override (tests: cpu)")
     args = ap.parse_args()
 
@@ -48,7 +50,7 @@ def main() -> None:
         device=device, model_name=args.model,
         device_label=f"pool:{args.pool_name}",
         kv_gb=args.kv_gb, max_batch_size=args.max_batch,
-        num_blocks=args.num_blocks,
+        num_blocks=args.num_blocks, weights=args.weights,
         tp_rank=args.rank, tp_world=args.world, tp_group=tp_group,
         tp_control=control)
     log_with_timestamp(

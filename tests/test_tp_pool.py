@@ -20,10 +20,10 @@ from resilient_llm_amd.workers.rpc import RpcWorkerClient
 pytestmark = pytest.mark.timeout(240)
 
 
-def greq(rid, text, n):
+def greq(rid, text, n, **kw):
     return GenerationRequest(request_id=rid, model="tiny",
                              messages=[{"role": "user", "content": text}],
-                             max_tokens=n)
+                             max_tokens=n, **kw)
 
 
 @pytest.fixture()
@@ -44,12 +44,12 @@ def tp2_pool():
             p.kill()
 
 
-def tp1_reference_tokens(text, n):
+def tp1_reference_tokens(text, n, **kw):
     async def run():
         w = EngineWorker(device="cpu", model_name="tiny", device_label="ref",
                          num_blocks=64, seed=0)
         try:
-            res = await w.generate(greq("ref", text, n))
+            res = await w.generate(greq("ref", text, n, **kw))
             return res.text
         finally:
             await w.close()
@@ -59,6 +59,10 @@ def tp1_reference_tokens(text, n):
 def test_tp2_matches_tp1_and_serves_concurrently(tp2_pool):
     sock, procs = tp2_pool
     expected = tp1_reference_tokens("hello tensor parallel", 8)
+    expected_pen = tp1_reference_tokens(
+        "penalty parity", 6, presence_penalty=0.7, frequency_penalty=0.4)
+    expected_sampled = tp1_reference_tokens(
+        "sampled parity", 6, temperature=0.9, seed=424242)
 
     async def run():
         client = RpcWorkerClient("pool:t", {"tiny"}, sock)
@@ -73,6 +77,16 @@ def test_tp2_matches_tp1_and_serves_concurrently(tp2_pool):
             *[client.generate(greq(f"c{i}", f"prompt number {i}", 5))
               for i in range(4)])
         assert all(r.completion_tokens == 5 for r in results)
+
+        # the lockstep 'add' broadcast must carry the FULL
+        # sampling params — penalties and seeded sampling must match TP=1
+        pen = await client.generate(greq("p", "penalty parity", 6,
+                                         presence_penalty=0.7,
+                                         frequency_penalty=0.4))
+        assert pen.text == expected_pen
+        sampled = await client.generate(greq("s", "sampled parity", 6,
+                                             temperature=0.9, seed=424242))
+        assert sampled.text == expected_sampled
 
         h = await client.health()
         assert h["status"] == "ok"
