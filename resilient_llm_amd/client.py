@@ -90,15 +90,18 @@ class _Completions:
     def create(self, *, model: str, messages: list, max_tokens: int = 128,
                temperature: float = 0.0, top_p: float = 1.0,
                stream: bool = False, timeout: Optional[float] = None,
-               seed: Optional[int] = None, **kw):
+               seed: Optional[int] = None,
+               extra_headers: Optional[dict] = None, **kw):
         payload = {"model": model, "messages": messages, "max_tokens": max_tokens,
                    "temperature": temperature, "top_p": top_p, "stream": stream}
         if seed is not None:
             payload["seed"] = seed
         payload.update(kw)
         if stream:
-            return self._client._post_stream("/chat/completions", payload, timeout)
-        return self._client._post_json("/chat/completions", payload, timeout)
+            return self._client._post_stream("/chat/completions", payload,
+                                             timeout, extra_headers)
+        return self._client._post_json("/chat/completions", payload, timeout,
+                                       extra_headers)
 
 
 class _Chat:
@@ -132,12 +135,13 @@ class OpenAIClient:
         raise APIError(status, msg or "request failed", data)
 
     def _post_json(self, path: str, payload: dict,
-                   timeout: Optional[float]) -> ChatCompletionResponse:
+                   timeout: Optional[float],
+                   extra_headers: Optional[dict] = None) -> ChatCompletionResponse:
         conn = self._connect(timeout)
         try:
             try:
                 conn.request("POST", path, body=json.dumps(payload),
-                             headers=self._headers())
+                             headers={**self._headers(), **(extra_headers or {})})
                 resp = conn.getresponse()
                 body = resp.read()
             except (ConnectionError, OSError) as e:
@@ -150,10 +154,12 @@ class OpenAIClient:
             conn.close()
 
     def _post_stream(self, path: str, payload: dict,
-                     timeout: Optional[float]) -> "StreamResponse":
+                     timeout: Optional[float],
+                     extra_headers: Optional[dict] = None) -> "StreamResponse":
         conn = self._connect(timeout)
         try:
-            conn.request("POST", path, body=json.dumps(payload), headers=self._headers())
+            conn.request("POST", path, body=json.dumps(payload),
+                         headers={**self._headers(), **(extra_headers or {})})
             resp = conn.getresponse()
         except (ConnectionError, OSError) as e:
             conn.close()

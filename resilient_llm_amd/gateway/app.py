@@ -38,7 +38,7 @@ from ..workers.base import (
     GenerationRequest, GenerationResult, Worker, WorkerError,
     WorkerRegistry, WorkerThrottled,
 )
-from ..workers.stub import estimate_tokens
+
 from .http import Request, Response
 
 DEFAULT_MAX_TOKENS = 128
@@ -153,12 +153,18 @@ class GatewayApp:
 
     def _deployments_on(self, worker: Worker) -> list[str]:
         """model_ids of deployments that resolve to this worker (for
-        health marking).  ``*`` targets are skipped — they re-resolve."""
+        health marking).  ``*`` targets are skipped — they re-resolve.
+        The MODEL must match too: two
+        different-model deployments co-located on one target must not
+        share health flags."""
         kind, _, target = worker.device.partition(":")
         out = []
         for d in self.config.deployments:
-            if d.backend_kind == kind and d.backend_target == target:
-                out.append(d.model_id)
+            if d.backend_kind != kind or d.backend_target != target:
+                continue
+            if (d.backend_model or d.model_id) not in worker.models:
+                continue
+            out.append(d.model_id)
         return out
 
     async def _health_loop(self) -> None:
@@ -304,7 +310,8 @@ class GatewayApp:
         return body
 
     def _estimate(self, body: dict) -> tuple[int, int]:
-        prompt_est = sum(estimate_tokens(str(m.get("content", "")))
+        cpt = self.config.router.estimate_chars_per_token
+        prompt_est = sum(max(1, len(str(m.get("content", ""))) // cpt)
                          for m in body["messages"])
         max_tokens = int(body.get("max_tokens")
                        or body.get("max_completion_tokens")
@@ -367,6 +374,16 @@ class GatewayApp:
         # request-id propagation: honor the client's x-request-id (or
         # mint one); it tags the ledger rows and every response
         rid = req.headers.get("x-request-id") or f"req-{uuid.uuid4().hex[:16]}"
+        # sampled-request seed is minted HERE, once per request (VERDICT
+        # r01 #5): a mid-stream failover replays with the SAME seed, so
+        # the replacement worker regenerates the identical prefix and the
+        # discard-by-count skip concatenates coherently.  (Engine-side
+        # default seeds are arrival-order-derived and differ per worker.)
+        if body.get("seed") is None and float(body.get("temperature") or 0.0) > 0:
+            import hashlib
+            body["seed"] = int.from_bytes(
+                hashlib.blake2b(rid.encode(), digest_size=8).digest(),
+                "big") >> 1
         if bool(body.get("stream")):
             return await self._chat_stream(body, alias, consumer, total_est,
                                            rid)
@@ -431,12 +448,19 @@ class GatewayApp:
         try:
             attempts = self._max_attempts(alias)
         except UnknownAlias:
+            self.consumers.reconcile(consumer, total_est, 0)
             return Response.error(404, f"unknown model {alias!r}",
                                   err_type="invalid_request_error", code="model_not_found")
         for _ in range(attempts):
             try:
                 ticket = self.router.acquire(alias, total_est, exclude=exclude)
             except RouterRateLimit as e:
+                # once a worker has FAILED (non-throttle), an
+                # exhausted router is a consequence of the failure, not a
+                # rate limit — fall through to the 502 path below
+                if last_err is not None and not isinstance(last_err, WorkerThrottled):
+                    break
+                self.consumers.reconcile(consumer, total_est, 0)
                 return Response.error(429, str(e), err_type="rate_limit_error",
                                       code="rate_limit_exceeded")
             greq = self._gen_request(body, consumer, ticket, stream=False,
@@ -459,6 +483,10 @@ class GatewayApp:
             self._record(ticket, greq, consumer, "ok", t0, result, worker.device)
             return self._completion_response(greq, ticket, result, worker)
         from ..utils.logging import sanitize_error
+        # no backend work happened: release the consumer's token charge
+        # (failed attempts over-throttled clients for a full
+        # minute window during backend outages)
+        self.consumers.reconcile(consumer, total_est, 0)
         msg = sanitize_error(last_err) if last_err else "no deployment available"
         status = 429 if isinstance(last_err, WorkerThrottled) else 502
         return Response.error(status, f"all deployments failed: {msg}",
@@ -500,9 +528,11 @@ class GatewayApp:
         try:
             first_ticket = self.router.acquire(alias, total_est)
         except UnknownAlias:
+            self.consumers.reconcile(consumer, total_est, 0)
             return Response.error(404, f"unknown model {alias!r}",
                                   err_type="invalid_request_error", code="model_not_found")
         except RouterRateLimit as e:
+            self.consumers.reconcile(consumer, total_est, 0)
             return Response.error(429, str(e), err_type="rate_limit_error",
                                   code="rate_limit_exceeded")
 
@@ -617,6 +647,10 @@ class GatewayApp:
                             pass
                         continue
                 from ..utils.logging import sanitize_error
+                # settle the consumer charge to what was actually served
+                app.consumers.reconcile(
+                    consumer, total_est,
+                    prompt_toks + sent_tokens if sent_tokens else 0)
                 err_evt = {"error": {"message": f"stream failed: "
                                                 f"{sanitize_error(last_err) if last_err else 'exhausted'}",
                                      "type": "api_error"}}
@@ -628,6 +662,9 @@ class GatewayApp:
                 # ledger stay truthful (X12)
                 if ticket is not None and not ticket.done:
                     app.router.complete(ticket, actual_tokens=None)
+                    app.consumers.reconcile(
+                        consumer, total_est,
+                        prompt_toks + sent_tokens if sent_tokens else 0)
                     app.ledger.record(InvocationRecord(
                         ts=time.time(), request_id=comp_id, alias=alias,
                         model_id=ticket.deployment.model_id, device="",

@@ -73,7 +73,8 @@ def stub_config_dict(port: int, **overrides) -> dict:
 
 
 @contextlib.contextmanager
-def run_gateway(config_dict=None, stub_kwargs=None, health_interval_s=0.2):
+def run_gateway(config_dict=None, stub_kwargs=None, health_interval_s=0.2,
+                registry=None):
     port = None
     if config_dict is None:
         port = free_port()
@@ -81,7 +82,8 @@ def run_gateway(config_dict=None, stub_kwargs=None, health_interval_s=0.2):
     else:
         port = config_dict["cluster"]["port"]
     config = load_config(data=config_dict)
-    registry = build_registry(config, stub_kwargs=stub_kwargs)
+    if registry is None:
+        registry = build_registry(config, stub_kwargs=stub_kwargs)
 
     loop = asyncio.new_event_loop()
     ready = threading.Event()
