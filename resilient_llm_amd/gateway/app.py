@@ -119,10 +119,12 @@ class GatewayApp:
     def __init__(self, config: Config, registry: WorkerRegistry,
                  router: Optional[Router] = None,
                  ledger: Optional[InvocationLedger] = None,
-                 health_interval_s: float = 2.0) -> None:
+                 health_interval_s: float = 2.0,
+                 shared_limits_path: Optional[str] = None) -> None:
         self.config = config
         self.registry = registry
-        self.router = router or Router(config.deployments, config.router)
+        self.router = router or Router(config.deployments, config.router,
+                                       shared_limits_path=shared_limits_path)
         self.ledger = ledger or InvocationLedger(
             jsonl_path=config.cluster.ledger_path)
         self.consumers = ConsumerLimiter(config.raw.get("consumer_limits"))

@@ -10,6 +10,7 @@ from __future__ import annotations
 
 import argparse
 import asyncio
+import os
 import signal
 from typing import Optional
 
@@ -50,9 +51,10 @@ def build_registry(config: Config, stub_kwargs: Optional[dict] = None) -> Worker
 
 async def serve(config: Config, registry: Optional[WorkerRegistry] = None,
                 ready_event: Optional[asyncio.Event] = None,
-                reuse_port: bool = False, drain_s: float = 30.0) -> None:
+                reuse_port: bool = False, drain_s: float = 30.0,
+                shared_limits_path: Optional[str] = None) -> None:
     registry = registry or build_registry(config)
-    app = GatewayApp(config, registry)
+    app = GatewayApp(config, registry, shared_limits_path=shared_limits_path)
     server = HttpServer(app.handle, host=config.cluster.host, port=config.cluster.port)
     await server.start(reuse_port=reuse_port)
     for w in registry.all().values():
@@ -103,9 +105,11 @@ def main() -> None:
                          "requests on SIGTERM/SIGINT")
     ap.add_argument("--workers", type=int, default=1,
                     help="gateway processes sharing the port via "
-                         "SO_REUSEPORT (NOTE: rate-limit windows and the "
-                         "ledger are per process — keep 1 unless the "
-                         "config is unlimited or limits are sharded)")
+                         "SO_REUSEPORT; deployment RPM/TPM windows are "
+                         "SHARED across processes (mmap'd counter file), "
+                         "so rpm=3 admits exactly 3 cluster-wide. "
+                         "Consumer-key limits and the in-memory ledger "
+                         "view stay per process (JSONL ledger is shared).")
     args = ap.parse_args()
     config = load_config(args.config)
     if args.port is not None:
@@ -119,16 +123,23 @@ def main() -> None:
         from ..workers.gpu import register_gpu_workers
         register_gpu_workers(config, registry)
     if args.workers > 1:
+        import multiprocessing
+        import tempfile
+        limited = any(d.rpm or d.tpm for d in config.deployments)
+        shared_path = None
+        if limited:
+            shared_path = os.path.join(
+                tempfile.mkdtemp(prefix="rlli-gw-"), "windows.bin")
         log_with_timestamp(
             f"SO_REUSEPORT scale-out: {args.workers} gateway processes; "
-            f"RPM/TPM windows are PER PROCESS", "yellow")
-        import multiprocessing
+            + ("RPM/TPM windows SHARED via " + shared_path if limited
+               else "no rate limits configured"), "green")
         procs = []
         for _ in range(args.workers - 1):
             pr = multiprocessing.Process(
-                target=lambda: asyncio.run(serve(config, registry,
-                                                 reuse_port=True,
-                                                 drain_s=args.drain_s)))
+                target=lambda: asyncio.run(serve(
+                    config, registry, reuse_port=True, drain_s=args.drain_s,
+                    shared_limits_path=shared_path)))
             # die with the parent: SIGTERM is delivered to the parent
             # only, and an orphaned reuse-port sibling would keep the
             # port serving forever
@@ -137,7 +148,8 @@ def main() -> None:
             procs.append(pr)
         try:
             asyncio.run(serve(config, registry, reuse_port=True,
-                              drain_s=args.drain_s))
+                              drain_s=args.drain_s,
+                              shared_limits_path=shared_path))
         finally:
             for pr in procs:
                 pr.terminate()

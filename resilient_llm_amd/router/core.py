@@ -80,16 +80,31 @@ class Ticket:
 class Router:
     def __init__(self, deployments: Iterable[Deployment], settings: RouterSettings,
                  clock: Callable[[], float] = time.monotonic,
-                 rng: Optional[random.Random] = None) -> None:
+                 rng: Optional[random.Random] = None,
+                 shared_limits_path: Optional[str] = None) -> None:
         self.settings = settings
         self._clock = clock
         self._rng = rng or random.Random()
         self._lock = threading.Lock()
-        self.states: list[DeploymentState] = [
-            DeploymentState(dep=d,
-                            limiter=MinuteWindowLimiter(d.rpm, d.tpm, clock=clock))
-            for d in deployments
-        ]
+        deployments = list(deployments)
+        if shared_limits_path:
+            # multi-process gateways (SO_REUSEPORT): ONE budget across all
+            # processes via an mmap'd counter file; slots
+            # are assigned by config order, identical in every process
+            from .shared_window import SharedMinuteWindowLimiter, SharedWindowFile
+            shared = SharedWindowFile(shared_limits_path, len(deployments))
+            self.states = [
+                DeploymentState(dep=d, limiter=SharedMinuteWindowLimiter(
+                    d.rpm, d.tpm, shared, i))
+                for i, d in enumerate(deployments)
+            ]
+        else:
+            self.states = [
+                DeploymentState(dep=d,
+                                limiter=MinuteWindowLimiter(d.rpm, d.tpm,
+                                                            clock=clock))
+                for d in deployments
+            ]
         self._by_alias: dict[str, list[DeploymentState]] = {}
         for s in self.states:
             self._by_alias.setdefault(s.dep.model_name, []).append(s)
