@@ -40,9 +40,32 @@ PROMPT_TOKENS = 128
 OUTPUT_TOKENS = 64
 CONCURRENCY_PER_GPU = 64
 
+# distributed hardening (the first 8-GPU run happens
+# blind — a dead rank must produce a diagnosis, not a silent hang):
+# rendezvous + collective timeout, and a faulthandler watchdog that dumps
+# all thread stacks and exits non-zero if a barrier never completes.
+RENDEZVOUS_TIMEOUT_S = float(os.environ.get("RLLI_BENCH_RDZV_TIMEOUT_S", "300"))
+WATCHDOG_S = float(os.environ.get("RLLI_BENCH_WATCHDOG_S", "1200"))
+
+_RANK = os.environ.get("RANK", "0")
+
 
 def log(msg):
-    print(f"[bench] {msg}", file=sys.stderr, flush=True)
+    print(f"[bench r{_RANK}] {msg}", file=sys.stderr, flush=True)
+
+
+def arm_watchdog(tag: str, seconds: float = WATCHDOG_S):
+    """If this rank is still stuck here after ``seconds``, dump every
+    thread's stack to stderr and hard-exit so torchrun tears the job
+    down with a readable cause instead of hanging the driver."""
+    import faulthandler
+    log(f"watchdog armed: {tag} ({seconds:.0f}s)")
+    faulthandler.dump_traceback_later(seconds, exit=True)
+
+
+def disarm_watchdog():
+    import faulthandler
+    faulthandler.cancel_dump_traceback_later()
 
 
 def free_port() -> int:
@@ -172,19 +195,32 @@ def main() -> None:
         f"WORLD_SIZE {world} must equal --gpus {n_gpus}"
     distributed = world > 1
     if distributed:
+        import datetime
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group(backend="gloo", rank=rank, world_size=world)
+        arm_watchdog("rendezvous", RENDEZVOUS_TIMEOUT_S + 60)
+        dist.init_process_group(
+            backend="gloo", rank=rank, world_size=world,
+            timeout=datetime.timedelta(seconds=RENDEZVOUS_TIMEOUT_S))
+        disarm_watchdog()
+        log(f"rendezvous complete: rank {rank}/{world}")
 
     on_gpu = args.device is None and torch.cuda.is_available()
-    device = args.device or (f"cuda:{local_rank}" if on_gpu else "cpu")
+    # co-located ranks (rehearsal: 2 ranks on a 1-GPU box) wrap onto the
+    # devices that exist; on a full node this is the identity mapping
+    dev_index = local_rank % max(1, torch.cuda.device_count()) if on_gpu else 0
+    device = args.device or (f"cuda:{dev_index}" if on_gpu else "cpu")
     if on_gpu:
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(dev_index)
+        if dev_index != local_rank:
+            log(f"co-located: local_rank {local_rank} -> cuda:{dev_index}")
 
-    def sync():
+    def sync(tag="sync"):
+        arm_watchdog(tag)
         if on_gpu:
             torch.cuda.synchronize()
         if distributed:
             dist.barrier()
+        disarm_watchdog()
 
     # shared run dir for worker sockets (rank 0 creates, broadcasts)
     if distributed:
@@ -207,7 +243,7 @@ def main() -> None:
                           use_graphs=on_gpu and not args.no_graphs)
     log(f"rank {rank}: worker up on {device} in {time.time() - t_init:.1f}s "
         f"({num_blocks} KV blocks)")
-    sync()
+    sync("post-worker-startup barrier")
 
     port = None
     loadgen = None
@@ -228,15 +264,17 @@ def main() -> None:
         if rank == 0:
             res = loadgen.round()
             log(f"warmup {w}: {res['ok']}/{res['total']} ok")
-        sync()
+        sync(f"warmup {w} barrier")
 
     # ---- timed ----
-    sync()
+    sync("pre-timed barrier")
     t0 = time.monotonic()
+    arm_watchdog("timed rounds", WATCHDOG_S)
     for k in range(args.steps):
         if rank == 0:
             rounds_timed.append(loadgen.round())
-    sync()
+    disarm_watchdog()
+    sync("post-timed barrier")
     elapsed = time.monotonic() - t0
     try:
         h = loopth.run(worker.health(), timeout=30)
