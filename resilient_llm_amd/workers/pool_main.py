@@ -43,7 +43,8 @@ def main() -> None:
     from .rpc import WorkerRpcServer
 
     control_group, tp_group = init_pool_groups(
-        args.rank, args.world, args.master_port, tp_backend=args.tp_backend)
+        args.rank, args.world, args.master_port, tp_backend=args.tp_backend,
+        device=device if device.startswith("cuda") else None)
     control = TPControl(control_group, is_leader=args.rank == 0)
 
     worker = EngineWorker(

@@ -41,13 +41,13 @@ def test_tp2_pool_colocated_gpu():
     lockstep request broadcast, sharded bf16 kernels and the per-layer
     all-reduces execute on the GPU.  RCCL refuses two ranks on one device
     ('Duplicate GPU detected', measured r02 — profiles/r02_rccl_rehearsal.md),
-    so the co-located collectives run over gloo; the REAL-RCCL variant
-    below runs whenever >= 2 logical devices exist (8-GPU node or CPX
-    partitioning)."""
+    so this ALSO exercises the startup probe + unanimous gloo fallback in
+    parallel.init_pool_groups; the REAL-RCCL variant below runs whenever
+    >= 2 logical devices exist (8-GPU node)."""
     sock = os.path.join(tempfile.mkdtemp(prefix="rlli-tp2gpu-"), "p.sock")
     pool = PoolDef(name="tp2", gpus=[0, 0], tensor_parallel=2)
     procs = spawn_pool_worker(pool, "tiny", sock, max_batch=4,
-                              device_override="cuda:0", tp_backend="gloo")
+                              device_override="cuda:0", tp_backend="nccl")
     try:
         async def run():
             client = RpcWorkerClient("pool:tp2", {"tiny"}, sock)
