@@ -38,6 +38,7 @@ SOURCES = [
     "prefill_paged.hip",
     "sampling.hip",
     "skinny_gemm.hip",
+    "skinny2.hip",
     "streamprobe.hip",
 ]
 HEADERS = ["common.h", "kernels.h"]

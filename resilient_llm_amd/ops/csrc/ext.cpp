@@ -419,6 +419,71 @@ Tensor skinny_linear(const Tensor& x, const Tensor& w) {
   return out;
 }
 
+// ------------------------------------------------------------- skinny2
+// v2 zero-LDS skinny GEMM; splitk <= 0 picks the grid-fill heuristic
+// (>= ~512 workgroups).  Requirements: M <= 64, K % 256 == 0, N % 16 == 0.
+static int skinny2_auto_splitk(int N, int K) {
+  const int groups = N / 16;
+  int sk = std::max(1, (512 + groups - 1) / groups);
+  return std::min<int>(sk, std::max(1, K / 256));
+}
+
+Tensor skinny2_linear(const Tensor& x, const Tensor& w, int64_t splitk) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(w, "w");
+  const int M = int(x.size(0));
+  const int K = int(x.size(1));
+  const int N = int(w.size(0));
+  TORCH_CHECK(w.size(1) == K, "x/W K mismatch");
+  TORCH_CHECK(M >= 1 && M <= 64, "skinny2_linear needs 1 <= M <= 64, got ", M);
+  TORCH_CHECK(K % 256 == 0 && N % 16 == 0,
+              "skinny2_linear needs K%256==0 and N%16==0, got K=", K, " N=", N);
+  int sk = int(splitk) > 0 ? int(splitk) : skinny2_auto_splitk(N, K);
+  sk = std::min(sk, K / 256);
+  Tensor out = at::empty({M, N}, x.options());
+  Tensor ws;
+  float* ws_ptr = nullptr;
+  if (sk > 1) {
+    ws = at::empty({sk, M, N}, x.options().dtype(at::kFloat));
+    ws_ptr = ws.data_ptr<float>();
+  }
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  rlli::launch_skinny2(bf16_ptr(x), bf16_ptr(w), ws_ptr, bf16_ptr(out),
+                       M, N, K, sk, current_stream(x));
+  check_launch("skinny2_linear");
+  return out;
+}
+
+// down-proj with the SwiGLU fused into the A-fragment path:
+// out = (silu(gu[:, :K]) * gu[:, K:]) @ w^T,  gu = [M, 2K].
+Tensor skinny2_silu_linear(const Tensor& gu, const Tensor& w, int64_t splitk) {
+  check_bf16_contig(gu, "gu");
+  check_bf16_contig(w, "w");
+  const int M = int(gu.size(0));
+  const int K2 = int(gu.size(1));
+  TORCH_CHECK(K2 % 2 == 0, "gu must be [M, 2K]");
+  const int K = K2 / 2;
+  const int N = int(w.size(0));
+  TORCH_CHECK(w.size(1) == K, "gu/W K mismatch");
+  TORCH_CHECK(M >= 1 && M <= 64, "skinny2_silu needs 1 <= M <= 64, got ", M);
+  TORCH_CHECK(K % 256 == 0 && N % 16 == 0,
+              "skinny2_silu needs K%256==0 and N%16==0, got K=", K, " N=", N);
+  int sk = int(splitk) > 0 ? int(splitk) : skinny2_auto_splitk(N, K);
+  sk = std::min(sk, K / 256);
+  Tensor out = at::empty({M, N}, gu.options());
+  Tensor ws;
+  float* ws_ptr = nullptr;
+  if (sk > 1) {
+    ws = at::empty({sk, M, N}, gu.options().dtype(at::kFloat));
+    ws_ptr = ws.data_ptr<float>();
+  }
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(gu.device());
+  rlli::launch_skinny2_silu(bf16_ptr(gu), bf16_ptr(w), ws_ptr, bf16_ptr(out),
+                            M, N, K, sk, current_stream(gu));
+  check_launch("skinny2_silu_linear");
+  return out;
+}
+
 Tensor stream_probe(const Tensor& w, int64_t splitk) {
   check_bf16_contig(w, "w");
   Tensor sink = at::zeros({256}, w.options().dtype(at::kFloat));
@@ -476,6 +541,8 @@ TORCH_LIBRARY(rlli, m) {
         "float scale) -> Tensor");
   m.def("sample(Tensor logits, Tensor temperatures, Tensor seeds, int step) -> Tensor");
   m.def("skinny_linear(Tensor x, Tensor w) -> Tensor");
+  m.def("skinny2_linear(Tensor x, Tensor w, int splitk) -> Tensor");
+  m.def("skinny2_silu_linear(Tensor gu, Tensor w, int splitk) -> Tensor");
   m.def("stream_probe(Tensor w, int splitk) -> Tensor");
   m.def("prefill_paged_attn(Tensor qkv, Tensor k_cache, Tensor v_cache, "
         "Tensor chunk_row0, Tensor chunk_pos0, Tensor chunk_nrows, "
@@ -496,6 +563,8 @@ TORCH_LIBRARY_IMPL(rlli, CUDA, m) {
   m.impl("prefill_attn", &prefill_attn);
   m.impl("sample", &sample);
   m.impl("skinny_linear", &skinny_linear);
+  m.impl("skinny2_linear", &skinny2_linear);
+  m.impl("skinny2_silu_linear", &skinny2_silu_linear);
   m.impl("stream_probe", &stream_probe);
   m.impl("prefill_paged_attn", &prefill_paged_attn);
 }

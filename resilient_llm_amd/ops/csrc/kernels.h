@@ -90,6 +90,16 @@ void launch_prefill_paged(const uint16_t* qkv, const uint16_t* k_cache,
 
 // Skinny-M GEMM (decode projections): out[M,N] = x[M,K] @ W[N,K]^T.
 // ws is a [splitk, M, N] f32 workspace (unused when splitk == 1).
+// v2: zero-LDS one-wave-per-16-columns register-dataflow skinny GEMM
+// (skinny2.hip); ws is the [splitk, M, N] f32 slab when splitk > 1.
+void launch_skinny2(const uint16_t* x, const uint16_t* w, float* ws,
+                    uint16_t* out, int M, int N, int K, int splitk,
+                    hipStream_t stream);
+// silu(g)*u fused into the A-fragment path: gu is [M, 2K] (gate|up).
+void launch_skinny2_silu(const uint16_t* gu, const uint16_t* w, float* ws,
+                         uint16_t* out, int M, int N, int K, int splitk,
+                         hipStream_t stream);
+
 void launch_skinny_gemm(const uint16_t* x, const uint16_t* w, float* ws,
                         uint16_t* out, int M, int N, int K, int splitk,
                         hipStream_t stream);

@@ -1,0 +1,109 @@
+#!/usr/bin/env python3
+"""A/B the v2 skinny GEMM against hipBLASLt (F.linear) and the r01 kernel
+on the Llama-3-8B decode shapes, plus the fused-SiLU down-proj against
+silu_mul + F.linear.  Correctness-checks every config against fp32
+matmul first; timing is 200 reps between cuda events, L2-cold by
+construction (weights >> 32 MB L2 stream every reps).
+
+Usage (GPU box):  python scripts/bench_skinny2.py [--m 64] [--sk 1 2 4 8]
+"""
+
+import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from resilient_llm_amd import ops
+
+SHAPES = [
+    ("qkv", 6144, 4096),
+    ("o", 4096, 4096),
+    ("gate_up", 28672, 4096),
+    ("down", 4096, 14336),
+]
+
+
+def timeit(fn, reps=200, warmup=20):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    s = torch.cuda.Event(enable_timing=True)
+    e = torch.cuda.Event(enable_timing=True)
+    s.record()
+    for _ in range(reps):
+        fn()
+    e.record()
+    torch.cuda.synchronize()
+    return s.elapsed_time(e) / reps * 1000.0  # us
+
+
+def check(a, b, tag, tol=2e-2):
+    ref = b.float()
+    err = (a.float() - ref).abs().max().item()
+    scale = ref.abs().max().item() + 1e-6
+    assert err / scale < tol, f"{tag}: rel err {err/scale:.4f}"
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--m", type=int, nargs="*", default=[64])
+    ap.add_argument("--sk", type=int, nargs="*", default=[1, 2, 4, 8])
+    ap.add_argument("--reps", type=int, default=200)
+    args = ap.parse_args()
+    ops.load_extension(required=True)
+    dev = "cuda:0"
+    torch.manual_seed(0)
+
+    for M in args.m:
+        print(f"==== M={M} ====", flush=True)
+        for name, N, K in SHAPES:
+            x = torch.randn(M, K, device=dev, dtype=torch.bfloat16) * 0.3
+            w = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.02
+            ref = (x.float() @ w.float().t())
+            check(torch.nn.functional.linear(x, w), ref, f"{name} blaslt")
+            t_lt = timeit(lambda: torch.nn.functional.linear(x, w), args.reps)
+            row = [f"{name:8s} lt={t_lt:7.1f}us"]
+            best = (t_lt, "lt")
+            for sk in args.sk:
+                if sk > K // 256:
+                    continue
+                out = ops.skinny2_linear(x, w, sk)
+                check(out, ref, f"{name} sk{sk}")
+                t = timeit(lambda: ops.skinny2_linear(x, w, sk), args.reps)
+                row.append(f"sk{sk}={t:7.1f}")
+                if t < best[0]:
+                    best = (t, f"sk{sk}")
+            print(" ".join(row) + f"   best={best[1]} "
+                  f"({t_lt / best[0]:.2f}x vs lt)", flush=True)
+
+        # fused SwiGLU down-proj: vs silu_mul + F.linear
+        I = 14336
+        gu = torch.randn(M, 2 * I, device=dev, dtype=torch.bfloat16) * 0.3
+        wd = torch.randn(4096, I, device=dev, dtype=torch.bfloat16) * 0.02
+        gf, uf = gu.float().chunk(2, -1)
+        ref = (torch.nn.functional.silu(gf) * uf) @ wd.float().t()
+
+        def eager():
+            return torch.nn.functional.linear(ops.silu_mul(gu), wd)
+        check(eager(), ref, "down eager", tol=3e-2)
+        t_e = timeit(eager, args.reps)
+        row = [f"down+silu eager={t_e:7.1f}us"]
+        best = (t_e, "eager")
+        for sk in args.sk:
+            if sk > I // 256:
+                continue
+            out = ops.skinny2_silu_linear(gu, wd, sk)
+            check(out, ref, f"down fused sk{sk}", tol=3e-2)
+            t = timeit(lambda: ops.skinny2_silu_linear(gu, wd, sk), args.reps)
+            row.append(f"fused-sk{sk}={t:7.1f}")
+            if t < best[0]:
+                best = (t, f"fused-sk{sk}")
+        print(" ".join(row) + f"   best={best[1]} "
+              f"({t_e / best[0]:.2f}x vs eager)", flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -220,6 +220,38 @@ def test_skinny_linear_matches_blaslt(M, N, K):
     torch.testing.assert_close(y.float(), y_ref, atol=0.05, rtol=0.05)
 
 
+@pytest.mark.parametrize("M,N,K", [
+    (64, 6144, 4096),    # 8B qkv
+    (64, 4096, 4096),    # 8B o-proj
+    (64, 4096, 14336),   # 8B down
+    (1, 4096, 4096), (17, 4096, 4096), (33, 1024, 512), (48, 64, 256),
+])
+@pytest.mark.parametrize("splitk", [1, 2, 4])
+def test_skinny2_linear_matches_fp32(M, N, K, splitk):
+    if splitk > K // 256:
+        pytest.skip("splitk > K/256")
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.1
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.1
+    y = ops.skinny2_linear(x, w, splitk)
+    y_ref = torch.nn.functional.linear(x.float(), w.float())
+    torch.testing.assert_close(y.float(), y_ref, atol=0.05, rtol=0.05)
+
+
+@pytest.mark.parametrize("M,I,H", [
+    (64, 14336, 4096),   # 8B down-proj geometry
+    (13, 1024, 512),
+    (64, 3584, 1024),    # 70B TP=8-ish shard geometry
+])
+@pytest.mark.parametrize("splitk", [1, 4])
+def test_skinny2_silu_linear_matches_fp32(M, I, H, splitk):
+    gu = torch.randn(M, 2 * I, dtype=torch.bfloat16, device=DEV) * 0.3
+    w = torch.randn(H, I, dtype=torch.bfloat16, device=DEV) * 0.05
+    y = ops.skinny2_silu_linear(gu, w, splitk)
+    gf, uf = gu.float().chunk(2, -1)
+    y_ref = (torch.nn.functional.silu(gf) * uf) @ w.float().t()
+    torch.testing.assert_close(y.float(), y_ref, atol=0.08, rtol=0.08)
+
+
 @pytest.mark.parametrize("n_q,n_kv,D,lens", [
     (32, 8, 128, [5, 128, 63, 200]),     # 8B GQA shape, ragged
     (8, 8, 128, [33]),                   # group 1
