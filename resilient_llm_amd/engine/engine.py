@@ -140,6 +140,11 @@ class LLMEngine:
         self.stats = {"prefill_steps": 0, "decode_steps": 0,
                       "prefill_time": 0.0, "decode_time": 0.0,
                       "prefill_tokens": 0, "decode_tokens": 0,
+                      # host-side phase split of the mixed/prefill step
+                      # (prep = python+tensor build, fwd = launch, sample
+                      # = kernel + tolist sync): finds host-bound stalls
+                      "prefill_prep_time": 0.0, "prefill_fwd_time": 0.0,
+                      "prefill_sample_time": 0.0,
                       "admit_events": []}
         # decode graph runner installed by engine/graph.py (GPU only)
         self.graph_runner = None
@@ -468,6 +473,7 @@ class LLMEngine:
                 t[i, :len(r)] = torch.tensor(r, dtype=torch.int32)
             return t.to(dev)
 
+        t_prep = time.monotonic()
         # fast path: no decode rows and every chunk is a whole prompt ->
         # classic batched varlen prefill over in-batch K/V (no paged
         # reads, no row-copy) — this is the bench/burst-arrival shape
@@ -476,23 +482,33 @@ class LLMEngine:
             cu = [0]
             for s, _, n in chunk_plan:
                 cu.append(cu[-1] + n)
-            logits = self.model.forward_prefill(
-                i32(input_ids), i32(positions), self.kv, i32(slots), i32(cu))
+            ids_t, pos_t, slots_t, cu_t = (i32(input_ids), i32(positions),
+                                           i32(slots), i32(cu))
+            self.stats["prefill_prep_time"] += time.monotonic() - t_prep
+            t_fwd = time.monotonic()
+            logits = self.model.forward_prefill(ids_t, pos_t, self.kv,
+                                                slots_t, cu_t)
+            self.stats["prefill_fwd_time"] += time.monotonic() - t_fwd
             return self._mixed_finish(chunk_plan, sampled_seqs, logits)
 
-        logits = self.model.forward_mixed(
-            i32(input_ids), i32(positions), self.kv, i32(slots), B_d,
-            bt_tensor([s.blocks for s in seqs_d]) if B_d else None,
-            i32(seq_lens_d) if B_d else None,
-            i32(c_row0), i32(c_pos0), i32(c_nrows), i32(c_btrow),
-            bt_tensor(bt_rows),
-            torch.tensor(sample_idx, dtype=torch.long, device=dev))
+        args = (i32(input_ids), i32(positions), self.kv, i32(slots), B_d,
+                bt_tensor([s.blocks for s in seqs_d]) if B_d else None,
+                i32(seq_lens_d) if B_d else None,
+                i32(c_row0), i32(c_pos0), i32(c_nrows), i32(c_btrow),
+                bt_tensor(bt_rows),
+                torch.tensor(sample_idx, dtype=torch.long, device=dev))
+        self.stats["prefill_prep_time"] += time.monotonic() - t_prep
+        t_fwd = time.monotonic()
+        logits = self.model.forward_mixed(*args)
+        self.stats["prefill_fwd_time"] += time.monotonic() - t_fwd
         return self._mixed_finish(chunk_plan, sampled_seqs, logits,
                                   decode_seqs=seqs_d)
 
     def _mixed_finish(self, chunk_plan, sampled_seqs, logits,
                       decode_seqs=()) -> tuple[int, list[StepOutput]]:
+        t_s = time.monotonic()
         tokens = self._sample(logits, sampled_seqs)
+        self.stats["prefill_sample_time"] += time.monotonic() - t_s
         n_chunk_tokens = 0
         for seq, start, n in chunk_plan:
             seq.n_cached = start + n

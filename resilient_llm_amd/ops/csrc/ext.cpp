@@ -423,8 +423,8 @@ Tensor skinny_linear(const Tensor& x, const Tensor& w) {
 // v2 zero-LDS skinny GEMM; splitk <= 0 picks the grid-fill heuristic
 // (>= ~512 workgroups).  Requirements: M <= 64, K % 256 == 0, N % 16 == 0.
 static int skinny2_auto_splitk(int N, int K) {
-  const int groups = N / 16;
-  int sk = std::max(1, (512 + groups - 1) / groups);
+  const int panels = N / 64;   // one 64-column panel per wave
+  int sk = std::max(1, (512 + panels - 1) / panels);
   return std::min<int>(sk, std::max(1, K / 256));
 }
 
@@ -436,8 +436,8 @@ Tensor skinny2_linear(const Tensor& x, const Tensor& w, int64_t splitk) {
   const int N = int(w.size(0));
   TORCH_CHECK(w.size(1) == K, "x/W K mismatch");
   TORCH_CHECK(M >= 1 && M <= 64, "skinny2_linear needs 1 <= M <= 64, got ", M);
-  TORCH_CHECK(K % 256 == 0 && N % 16 == 0,
-              "skinny2_linear needs K%256==0 and N%16==0, got K=", K, " N=", N);
+  TORCH_CHECK(K % 256 == 0 && N % 64 == 0,
+              "skinny2_linear needs K%256==0 and N%64==0, got K=", K, " N=", N);
   int sk = int(splitk) > 0 ? int(splitk) : skinny2_auto_splitk(N, K);
   sk = std::min(sk, K / 256);
   Tensor out = at::empty({M, N}, x.options());
@@ -466,8 +466,8 @@ Tensor skinny2_silu_linear(const Tensor& gu, const Tensor& w, int64_t splitk) {
   const int N = int(w.size(0));
   TORCH_CHECK(w.size(1) == K, "gu/W K mismatch");
   TORCH_CHECK(M >= 1 && M <= 64, "skinny2_silu needs 1 <= M <= 64, got ", M);
-  TORCH_CHECK(K % 256 == 0 && N % 16 == 0,
-              "skinny2_silu needs K%256==0 and N%16==0, got K=", K, " N=", N);
+  TORCH_CHECK(K % 256 == 0 && N % 64 == 0,
+              "skinny2_silu needs K%256==0 and N%64==0, got K=", K, " N=", N);
   int sk = int(splitk) > 0 ? int(splitk) : skinny2_auto_splitk(N, K);
   sk = std::min(sk, K / 256);
   Tensor out = at::empty({M, N}, gu.options());

@@ -3,26 +3,26 @@
 // LDS-staged design, which measured 46-62% of the streaming ceiling
 // (profiles/r01_gemm_streaming_analysis.md).
 //
-// Why v2 is shaped this way (the r01 postmortem):
-//  - r01 staged x in LDS with a __syncthreads per K-chunk: four waves in
-//    lockstep, W-stream stalls at every barrier.  v2 has ZERO LDS and
-//    ZERO barriers — each workgroup is ONE wave owning 16 output columns.
-//  - Both MFMA operands load straight from memory in fragment order
-//    (mfma_f32_16x16x32_bf16: lane l holds 8 CONTIGUOUS k of row/col
-//    l&15 at k-octet l>>4, guide §3 Fragment layout):
-//      B-frag = 16 B of one W row (HBM, nontemporal: read-once stream),
-//      A-frag = 16 B of one x row (L2-hot: all N/16 groups re-read the
-//               same tiny x, so HBM sees it ~once per XCD).
-//    No transpose, no staging, no cross-lane traffic: the only
-//    instructions in the hot loop are loads and MFMAs, so the W stream
-//    can stay as deep in flight as the probe's (streamprobe.hip).
-//  - Split-K for grid fill on small N (one wave per 16 columns gives
-//    N=4096 only 256 groups): slices write f32 slabs, a second kernel
-//    reduces.  Slab traffic is M*N*4*splitk — tiny next to W.
+// Shape of the design (r01 + first-v2 postmortems):
+//  - ZERO LDS, ZERO barriers: one wave per workgroup, owning a 64-column
+//    output panel and a K-slice.  Both MFMA operands load straight from
+//    memory in fragment order (mfma_f32_16x16x32_bf16: lane l holds 8
+//    CONTIGUOUS k of row/col l&15 at k-octet l>>4 — guide §3):
+//      B-frag = 16 B of one W row (HBM, nontemporal read-once stream),
+//      A-frag = 16 B of one x row (tiny, L2-resident).
+//  - 64 columns per wave, not 16: each A-fragment feeds FOUR MFMAs, so
+//    the redundant x re-reads across panels stay at 1x the W stream
+//    (the 16-col variant measured L2-bound: A-traffic was 4x W).
+//  - Double-buffered chunk pipeline in STATICALLY indexed register
+//    arrays: chunk c+1's loads issue while chunk c computes, so the
+//    s_waitcnt ahead of each MFMA is a partial vmcnt (the naive loop
+//    compiled to vmcnt(0) before every MFMA — zero overlap).
+//  - Split-K for grid fill (N=4096 is only 64 panels): slices write f32
+//    slabs, a second kernel reduces.  Slab traffic is M*N*4*splitk —
+//    tiny next to W.
 //
 // PROLOGUE variants fuse the producer elementwise op into the A-frag
-// load path (the activation is tiny; the fusion deletes a whole kernel
-// launch + an intermediate tensor round-trip):
+// consume path (deletes a kernel launch + an intermediate round-trip):
 //   PLAIN:    a = x[m][k]
 //   SILU_MUL: a = silu(g[m][k]) * u[m][k],  x = gu[M, 2K] (down-proj)
 
@@ -39,6 +39,9 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 using u32x4 = __attribute__((ext_vector_type(4))) unsigned int;
 
 enum { PRO_PLAIN = 0, PRO_SILU = 1 };
+
+constexpr int BN = 64;          // columns per wave (4 MFMA column tiles)
+constexpr int CT = BN / 16;     // column tiles
 
 DEV_INLINE bf16x8_vec as_bf16x8(u32x4 v) {
   return *reinterpret_cast<bf16x8_vec*>(&v);
@@ -60,12 +63,9 @@ DEV_INLINE u32x4 silu_mul_frag(u32x4 graw, u32x4 uraw) {
   return out;
 }
 
-// One wave = one 16-column output group x one K-slice.
-//  blockIdx.x = group + n_groups * slice
-// KSTEP canonical chunk: 8 mfma k-steps = 256 k per iteration, fully
-// unrolled so LLVM hoists the whole chunk's loads ahead of its MFMAs
-// (the deep-prefetch ring, without hand-rolled buffers).
-template <int M_TILES, int PROLOGUE>
+// One wave = one 64-column panel x one K-slice.
+//  blockIdx.x = panel + n_panels * slice
+template <int M_TILES, int PROLOGUE, int KSTEPS>
 __global__ __launch_bounds__(64)
 void skinny2_kernel(const uint16_t* __restrict__ x,
                     const uint16_t* __restrict__ w,
@@ -73,82 +73,91 @@ void skinny2_kernel(const uint16_t* __restrict__ x,
                     uint16_t* __restrict__ out_bf16,  // [M][N] when splitk==1
                     int M, int N, int K,
                     int k_per_slice, int splitk) {
-  const int n_groups = N / 16;
-  const int group = blockIdx.x % n_groups;
-  const int slice = blockIdx.x / n_groups;
+  const int n_panels = N / BN;
+  const int panel = blockIdx.x % n_panels;
+  const int slice = blockIdx.x / n_panels;
   const int kbeg = slice * k_per_slice;
   const int kend = min(kbeg + k_per_slice, K);
   if (kbeg >= kend) return;
 
   const int lane = threadIdx.x;
-  const int jcol = lane & 15;   // B column / A row (within tile)
+  const int jcol = lane & 15;   // B column / A row (within a 16-tile)
   const int koct = lane >> 4;   // k-octet: 8 contiguous bf16
 
   // x row stride in bf16 elements (SILU reads gu[M, 2K])
   const int xstride = PROLOGUE == PRO_SILU ? 2 * K : K;
-  // hardware-bounds-checked x loads: rows >= M return 0 (padded M tiles
-  // contribute nothing) — guide T8/T20: descriptor from wave-uniform ptr
+  // Both operands load through buffer descriptors (T8/T20): the k part
+  // of every address is WAVE-UNIFORM and rides the SGPR soffset, so a
+  // load costs zero VALU and zero address VGPRs — the per-lane voffset
+  // of each fragment stream is a loop-invariant register.  x rows >= M
+  // are hardware-bounds-checked to 0 (padded M tiles add nothing).
   const auto xrsrc = __builtin_amdgcn_make_buffer_rsrc(
       const_cast<uint16_t*>(x), /*stride*/ (short)0,
       /*bytes*/ M * xstride * 2, /*flags*/ 0x00020000);
+  const auto wrsrc = __builtin_amdgcn_make_buffer_rsrc(
+      const_cast<uint16_t*>(w), /*stride*/ (short)0,
+      /*bytes*/ int(int64_t(N) * K * 2), /*flags*/ 0x00020000);
 
-  const uint16_t* wrow = w + int64_t(group * 16 + jcol) * K + koct * 8;
-
-  f32x4 acc[M_TILES];
+  int voff_b[CT];
 #pragma unroll
-  for (int mt = 0; mt < M_TILES; ++mt) acc[mt] = f32x4{0.f, 0.f, 0.f, 0.f};
+  for (int ct = 0; ct < CT; ++ct)
+    voff_b[ct] = ((panel * BN + ct * 16 + jcol) * K + koct * 8) * 2;
+  int voff_a[M_TILES];
+#pragma unroll
+  for (int mt = 0; mt < M_TILES; ++mt)
+    voff_a[mt] = ((mt * 16 + jcol) * xstride + koct * 8) * 2;
 
-  // byte offset of this lane's A fragment for row tile mt at k
-  auto a_off = [&](int mt, int k) {
-    return ((mt * 16 + jcol) * xstride + k + koct * 8) * 2;
-  };
+  f32x4 acc[M_TILES][CT];
+#pragma unroll
+  for (int mt = 0; mt < M_TILES; ++mt)
+#pragma unroll
+    for (int ct = 0; ct < CT; ++ct) acc[mt][ct] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  // Double-buffered chunk pipeline with STATICALLY indexed register
-  // arrays (they must stay in VGPRs): load chunk c+1 while computing
-  // chunk c, so every MFMA's operands were issued a full chunk earlier
-  // and the s_waitcnt before it is a PARTIAL vmcnt (the naive loop
-  // compiled to vmcnt(0) before every MFMA — zero overlap).
-  constexpr int KSTEPS = PROLOGUE == PRO_SILU ? 2 : 4;  // x 32 k per chunk
   constexpr int CHUNK = 32 * KSTEPS;
-  constexpr int NA = PROLOGUE == PRO_SILU ? 2 : 1;      // regs per A frag
+  constexpr int NA = PROLOGUE == PRO_SILU ? 2 : 1;   // regs per A frag
+  constexpr int AUX_NT = 2;                          // nontemporal policy
 
-  u32x4 bbuf[2][KSTEPS];
+  u32x4 bbuf[2][KSTEPS][CT];
   u32x4 abuf[2][KSTEPS][M_TILES][NA];
 
   auto load_chunk = [&](int buf, int k) {
 #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
-      const int kk = k + ks * 32;
-      bbuf[buf][ks] = __builtin_nontemporal_load(
-          reinterpret_cast<const u32x4*>(wrow + kk));
+      const int soff = (k + ks * 32) * 2;    // SGPR: k is wave-uniform
+#pragma unroll
+      for (int ct = 0; ct < CT; ++ct)
+        bbuf[buf][ks][ct] = __builtin_amdgcn_raw_buffer_load_b128(
+            wrsrc, voff_b[ct], soff, AUX_NT);
 #pragma unroll
       for (int mt = 0; mt < M_TILES; ++mt) {
         abuf[buf][ks][mt][0] = __builtin_amdgcn_raw_buffer_load_b128(
-            xrsrc, a_off(mt, kk), 0, 0);
+            xrsrc, voff_a[mt], soff, 0);
         if (PROLOGUE == PRO_SILU)
           abuf[buf][ks][mt][1] = __builtin_amdgcn_raw_buffer_load_b128(
-              xrsrc, a_off(mt, kk) + K * 2, 0, 0);
+              xrsrc, voff_a[mt], soff + K * 2, 0);
       }
     }
   };
   auto compute_chunk = [&](int buf) {
 #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
-      const bf16x8_vec bfrag = as_bf16x8(bbuf[buf][ks]);
 #pragma unroll
       for (int mt = 0; mt < M_TILES; ++mt) {
         u32x4 araw = abuf[buf][ks][mt][0];
         if (PROLOGUE == PRO_SILU)
           araw = silu_mul_frag(araw, abuf[buf][ks][mt][1]);
-        acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            as_bf16x8(araw), bfrag, acc[mt], 0, 0, 0);
+        const bf16x8_vec afrag = as_bf16x8(araw);
+#pragma unroll
+        for (int ct = 0; ct < CT; ++ct)
+          acc[mt][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag, as_bf16x8(bbuf[buf][ks][ct]), acc[mt][ct], 0, 0, 0);
       }
     }
   };
 
   // k_per_slice and K are multiples of 256, so every slice has an even
-  // number of 128-k chunks (and of 64-k chunks for SILU) — the 2x
-  // unrolled steady-state below never needs a tail.
+  // number of chunks for KSTEPS <= 4 — the 2x unrolled steady state
+  // never needs a tail.
   load_chunk(0, kbeg);
   if (kbeg + CHUNK < kend) load_chunk(1, kbeg + CHUNK);
   int k = kbeg;
@@ -162,17 +171,22 @@ void skinny2_kernel(const uint16_t* __restrict__ x,
   if (k + CHUNK < kend) compute_chunk(1);
 
   // C layout (guide §3): col = lane&15, row = (lane>>4)*4 + reg
-  const int out_col = group * 16 + jcol;
 #pragma unroll
   for (int mt = 0; mt < M_TILES; ++mt) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int out_row = mt * 16 + koct * 4 + r;
       if (out_row >= M) continue;
-      if (splitk == 1) {
-        out_bf16[int64_t(out_row) * N + out_col] = f32_to_bf16(acc[mt][r]);
-      } else {
-        out_ws[(int64_t(slice) * M + out_row) * N + out_col] = acc[mt][r];
+#pragma unroll
+      for (int ct = 0; ct < CT; ++ct) {
+        const int out_col = panel * BN + ct * 16 + jcol;
+        if (splitk == 1) {
+          out_bf16[int64_t(out_row) * N + out_col] =
+              f32_to_bf16(acc[mt][ct][r]);
+        } else {
+          out_ws[(int64_t(slice) * M + out_row) * N + out_col] =
+              acc[mt][ct][r];
+        }
       }
     }
   }
@@ -196,11 +210,29 @@ void launch_any(const uint16_t* x, const uint16_t* w, float* ws,
                 hipStream_t stream) {
   const int m_tiles = (M + 15) / 16;
   const int k_per_slice = ((K / 256 + splitk - 1) / splitk) * 256;
-  const int blocks = (N / 16) * splitk;
+  // rounding can leave trailing slices with no K range: drop them, or
+  // the reduce would sum their UNINITIALIZED slabs
+  splitk = (K + k_per_slice - 1) / k_per_slice;
+  const int blocks = (N / BN) * splitk;
+  // KSTEPS: prefetch window per chunk.  Large fragment sets (M_TILES>2,
+  // or SILU's 2x A regs) use the narrow window so the kernel stays under
+  // ~256 VGPRs and multi-wave occupancy hides the latency instead
+  // (372-VGPR variants measured slower: AGPR shuffling + drained vmcnt).
+  // RLLI_SK2_KS=2|4 overrides for tuning.
+  static const char* ks_env = std::getenv("RLLI_SK2_KS");
+  const int ks_req = ks_env ? atoi(ks_env) : 0;
   auto launch = [&](auto mt_tag) {
     constexpr int MT = decltype(mt_tag)::value;
-    hipLaunchKernelGGL((skinny2_kernel<MT, PROLOGUE>), dim3(blocks), dim3(64),
-                       0, stream, x, w, ws, out, M, N, K, k_per_slice, splitk);
+    const int ks = ks_req ? ks_req
+                          : ((PROLOGUE == PRO_SILU || MT > 2) ? 2 : 4);
+    if (ks == 4)
+      hipLaunchKernelGGL((skinny2_kernel<MT, PROLOGUE, 4>), dim3(blocks),
+                         dim3(64), 0, stream, x, w, ws, out, M, N, K,
+                         k_per_slice, splitk);
+    else
+      hipLaunchKernelGGL((skinny2_kernel<MT, PROLOGUE, 2>), dim3(blocks),
+                         dim3(64), 0, stream, x, w, ws, out, M, N, K,
+                         k_per_slice, splitk);
   };
   using T1 = std::integral_constant<int, 1>;
   using T2 = std::integral_constant<int, 2>;
