@@ -38,10 +38,16 @@ def main():
         from resilient_llm_amd.engine.graph import install_graph_runner
         install_graph_runner(engine)
 
+    import random
+    rng = random.Random(7)
     for rep in range(args.repeat):
         for i in range(args.batch):
-            engine.add_request(f"p{rep}-{i}",
-                               list(range(10, 10 + args.prompt_tokens)),
+            # DISTINCT prompts: identical prompts hit the prefix cache and
+            # skip almost all prefill compute (r01 profiled that by
+            # mistake and under-read prefill cost ~20x)
+            prompt = [rng.randrange(10, 28000)
+                      for _ in range(args.prompt_tokens)]
+            engine.add_request(f"p{rep}-{i}", prompt,
                                SamplingParams(max_tokens=args.decode_steps))
         torch.cuda.synchronize()
         t0 = time.monotonic()
