@@ -282,6 +282,15 @@ def main() -> None:
         log(f"rank {rank} engine stats: {stats}")
         admits = worker.engine.stats.get("admit_events", [])
         log(f"rank {rank} admits (t, n, still_waiting): {admits[:40]}")
+        if worker.engine._trace:
+            tr = worker.engine._trace[-220:]
+            lines = []
+            prev = None
+            for t0, kind, ms in tr:
+                gap = 0.0 if prev is None else (t0 - prev) * 1e3
+                lines.append(f"{t0:.4f} +{gap:6.2f}ms {kind} host={ms}ms")
+                prev = t0
+            log("step trace (last 220):\n" + "\n".join(lines))
     except Exception:
         pass
     if distributed:

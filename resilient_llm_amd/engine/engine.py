@@ -148,6 +148,10 @@ class LLMEngine:
                       "admit_events": []}
         # decode graph runner installed by engine/graph.py (GPU only)
         self.graph_runner = None
+        # RLLI_STEP_TRACE=1: per-step wall-clock ring (t_start, kind,
+        # batch, ms) — finds serving-context stalls the aggregate
+        # counters average away
+        self._trace = [] if os.environ.get("RLLI_STEP_TRACE") == "1" else None
 
     # ------------------------------------------------------------- admin
     @property
@@ -278,6 +282,7 @@ class LLMEngine:
             if _ROCTX:
                 torch.cuda.nvtx.range_push(
                     f"mixed d{len(self.running)}+p{len(self.prefilling)}")
+            nd, np_ = len(self.running), len(self.prefilling)
             n_chunk_tokens, new_outs = self._mixed_step()
             outs += new_outs
             if _ROCTX:
@@ -286,17 +291,24 @@ class LLMEngine:
             self.stats["prefill_steps"] += 1
             self.stats["prefill_time"] += elapsed
             self.stats["prefill_tokens"] += n_chunk_tokens
+            if self._trace is not None:
+                self._trace.append((t0, f"mix d{nd}+p{np_}({n_chunk_tokens}t)",
+                                    round(elapsed * 1e3, 2)))
             self._tune_budget(elapsed)
             return outs
         if self.running:
             t0 = time.monotonic()
             if _ROCTX:
                 torch.cuda.nvtx.range_push(f"decode b{len(self.running)}")
+            nb = len(self.running)
             outs += self._decode_step()
             if _ROCTX:
                 torch.cuda.nvtx.range_pop()
             self.stats["decode_steps"] += 1
-            self.stats["decode_time"] += time.monotonic() - t0
+            elapsed = time.monotonic() - t0
+            self.stats["decode_time"] += elapsed
+            if self._trace is not None:
+                self._trace.append((t0, f"dec b{nb}", round(elapsed * 1e3, 2)))
             return outs
         return outs
 
