@@ -18,13 +18,17 @@ import sys
 
 HERE = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(HERE, "csrc")
-BUILD = os.path.join(HERE, "build")
-OUT_SO = os.path.join(HERE, "_rlli_hip.so")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 # Host-side sanitizers for the extension: RLLI_ASAN=1 /
 # RLLI_TSAN=1 add -fsanitize to host code (device code unaffected).
-SAN_FLAGS = (["-fsanitize=address"] if os.environ.get("RLLI_ASAN") == "1" else
-             ["-fsanitize=thread"] if os.environ.get("RLLI_TSAN") == "1" else [])
+# Sanitized builds go to their OWN objects + .so so they never clobber
+# the production artifact that ships to GPU boxes.
+_SAN = ("asan" if os.environ.get("RLLI_ASAN") == "1" else
+        "tsan" if os.environ.get("RLLI_TSAN") == "1" else "")
+SAN_FLAGS = {"asan": ["-fsanitize=address"],
+             "tsan": ["-fsanitize=thread"], "": []}[_SAN]
+BUILD = os.path.join(HERE, "build" + (f"-{_SAN}" if _SAN else ""))
+OUT_SO = os.path.join(HERE, f"_rlli_hip{'_' + _SAN if _SAN else ''}.so")
 
 SOURCES = [
     "ext.cpp",
