@@ -15,7 +15,7 @@ import torch
 
 from ..ops import autotune as lt_autotune
 
-BUCKETS = [1, 2, 4, 8, 16, 24, 32, 40, 48, 56, 64, 96, 128]
+BUCKETS = [1, 2, 4, 8, 16, 24, 32, 40, 48, 56, 64, 96, 128, 160, 192, 224, 256]
 
 
 class DecodeGraphRunner:
