@@ -61,3 +61,27 @@ DEV_INLINE float group_sum(float v) {
   return v;
 }
 
+// Same reduction on the VALU pipe via DPP row ops — no ds_swizzle, no
+// lgkmcnt stall chain (__shfl_xor issues on the DS pipe: the decode-attn
+// hot loop measured 120 ds_swizzles + 94 lgkmcnt waits per iteration
+// group before this).  Only valid for REDUCTIONS (W <= 16): after stage
+// k every lane of a 2^k sub-group holds that sub-group's total, so a
+// mirror — any representative of the complementary sub-group — replaces
+// the exact xor partner that DPP cannot express.
+template <int CTRL>
+DEV_INLINE float dpp_mov_f32(float v) {
+  return __builtin_bit_cast(
+      float, __builtin_amdgcn_update_dpp(
+                 0, __builtin_bit_cast(int, v), CTRL, 0xf, 0xf, true));
+}
+
+template <int W>
+DEV_INLINE float group_sum_dpp(float v) {
+  static_assert(W == 2 || W == 4 || W == 8 || W == 16);
+  if constexpr (W >= 2) v += dpp_mov_f32<0xB1>(v);    // quad_perm ^1
+  if constexpr (W >= 4) v += dpp_mov_f32<0x4E>(v);    // quad_perm ^2
+  if constexpr (W >= 8) v += dpp_mov_f32<0x141>(v);   // row_half_mirror
+  if constexpr (W >= 16) v += dpp_mov_f32<0x140>(v);  // row_mirror
+  return v;
+}
+

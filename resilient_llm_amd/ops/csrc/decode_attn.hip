@@ -36,7 +36,16 @@ constexpr float kNegInf = -1e30f;
 // into 1; the self token is handled in registers by partial 0, so no
 // in-kernel cache write->read ordering exists — future steps read the
 // appended k/v after the kernel boundary).
-template <int GW, int GROUP, bool FUSED>
+// BS: compile-time cache block size (0 = runtime).  The engine's paged
+// cache uses 16; baking it makes tok_per_grp a constant, so the K/V
+// issue loop fully unrolls with no per-load `i < tok_per_grp` branch
+// (the runtime form emitted s_cbranch between every load pair,
+// fragmenting the batch and forcing early vmcnt waits).
+// ABL: perf-ablation variants (RLLI_ATTN_ABLATE, results intentionally
+// WRONG, never used for real inference): 1 = skip LDS publish/merge,
+// 2 = skip online softmax, 3 = loads only.  Isolates which phase owns
+// the wall time (guide pitfall 8: ablate before optimizing).
+template <int GW, int GROUP, bool FUSED, int BS = 0, int ABL = 0>
 __global__ __launch_bounds__(256)
 void decode_attn_kernel(const uint16_t* __restrict__ q,
                         const uint16_t* __restrict__ k_cache,
@@ -44,7 +53,7 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
                         const int32_t* __restrict__ block_table,
                         const int32_t* __restrict__ seq_lens,
                         uint16_t* __restrict__ out,
-                        int n_kv_heads, int block_size, int max_blocks,
+                        int n_kv_heads, int block_size_rt, int max_blocks,
                         float scale, int q_stride,
                         const uint16_t* __restrict__ k_src,
                         const uint16_t* __restrict__ v_src,
@@ -58,6 +67,7 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
                         float* __restrict__ part_ml) {
   constexpr int D = GW * 8;
   constexpr int GPW = 64 / GW;                 // lane-groups per wave
+  const int block_size = BS ? BS : block_size_rt;
   // split-K (flash-decode): when batch*n_kv_heads can't fill the chip,
   // n_split workgroups share one (sequence, kv-head) — each walks an
   // interleaved subset of the cache blocks and emits an f32 partial
@@ -74,8 +84,7 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
   const int group = lane / GW;
   const int gl = lane % GW;
   const int d0 = gl * 8;
-  const int n_part = n_waves * GPW;
-  const int pid = wave * GPW + group;
+  const int n_part = n_waves;    // one partial per wave (in-wave merged)
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   float* accs = reinterpret_cast<float*>(smem_raw);       // [n_part][GROUP][D]
@@ -161,7 +170,7 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
         float s = 0.f;
 #pragma unroll
         for (int i = 0; i < 8; ++i) s += qv[h][i] * kf[i];
-        s = group_sum<GW>(s) * scale;
+        s = group_sum_dpp<GW>(s) * scale;
         m[h] = s;
         l[h] = 1.f;
 #pragma unroll
@@ -181,8 +190,15 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
   constexpr int TPG = 16 / GPW >= 1 ? 16 / GPW : 1;   // tokens per group @bs16
   const int tok_per_grp = block_size / GPW;
 
-  auto issue_block = [&](int b, uint4 (&kraw)[TPG], uint4 (&vraw)[TPG]) {
-    const int phys = block_table[int64_t(seq) * max_blocks + b];
+  // phys index fetched one pipeline stage AHEAD of its block's K/V
+  // issue: with the load inside issue_block the address of every K/V
+  // load depended on a just-issued block_table load and the compiler
+  // had to drain vmcnt before computing it, serializing the double
+  // buffer.
+  auto fetch_phys = [&](int b) -> int {
+    return block_table[int64_t(seq) * max_blocks + b];
+  };
+  auto issue_block = [&](int phys, uint4 (&kraw)[TPG], uint4 (&vraw)[TPG]) {
     const int64_t kv_base =
         (int64_t(phys) * n_kv_heads + kvh) * block_size * D;
 #pragma unroll
@@ -212,13 +228,24 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
 #pragma unroll
       for (int j = 0; j < 8; ++j) vf[j] = bf16_to_f32(vv.s[j]);
 
+      if constexpr (ABL == 3) {              // loads only
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[0][j] += kf[j] + vf[j];
+        continue;
+      }
 #pragma unroll
       for (int h = 0; h < GROUP; ++h) {
         float s = 0.f;
 #pragma unroll
         for (int j = 0; j < 8; ++j) s += qv[h][j] * kf[j];
-        s = group_sum<GW>(s);
+        s = group_sum_dpp<GW>(s);
         s *= scale;
+        if constexpr (ABL == 2) {            // no online softmax
+          l[h] += s;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) acc[h][j] += s * vf[j];
+          continue;
+        }
         const float m_new = fmaxf(m[h], s);
         const float corr = __expf(m[h] - m_new);
         const float p = __expf(s - m_new);
@@ -230,6 +257,71 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
     }
   };
 
+  // Block-batched softmax consume: all TPG scores per head are computed
+  // before ONE running-max update for the whole block — the per-token
+  // form's serial corr/exp chain was 4.6 us of the 20.8 us kernel at
+  // batch 64 (RLLI_ATTN_ABLATE=2 vs 1); batching makes the TPG exps
+  // independent and pays one corr per (block, head).  Invalid tail
+  // tokens contribute exp(kNegInf - m_new) == 0 (kNegInf is a finite
+  // -1e30, so no -inf minus -inf NaN is reachable; every walked block
+  // holds at least one valid token, so m_new is always finite).
+  auto consume_block_batched = [&](int bb, const uint4 (&kraw)[TPG],
+                                   const uint4 (&vraw)[TPG]) {
+    float vfs[TPG][8];
+    float svals[GROUP][TPG];
+#pragma unroll
+    for (int i = 0; i < TPG; ++i) {
+      if (i >= tok_per_grp) break;
+      const int tok = group + GPW * i;
+      const bool valid = bb * block_size + tok < len_cache;
+      bf16x8 kv, vv;
+      kv.u = kraw[i];
+      vv.u = vraw[i];
+      float kf[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) kf[j] = bf16_to_f32(kv.s[j]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vfs[i][j] = bf16_to_f32(vv.s[j]);
+#pragma unroll
+      for (int h = 0; h < GROUP; ++h) {
+        float s = 0.f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) s += qv[h][j] * kf[j];
+        s = group_sum_dpp<GW>(s) * scale;
+        svals[h][i] = valid ? s : kNegInf;
+      }
+    }
+#pragma unroll
+    for (int h = 0; h < GROUP; ++h) {
+      float m_new = m[h];
+#pragma unroll
+      for (int i = 0; i < TPG; ++i) {
+        if (i >= tok_per_grp) break;
+        m_new = fmaxf(m_new, svals[h][i]);
+      }
+      const float corr = __expf(m[h] - m_new);
+      float p[TPG], psum = 0.f;
+#pragma unroll
+      for (int i = 0; i < TPG; ++i) {
+        if (i >= tok_per_grp) break;
+        p[i] = __expf(svals[h][i] - m_new);
+        psum += p[i];
+      }
+      m[h] = m_new;
+      l[h] = l[h] * corr + psum;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float t = acc[h][j] * corr;
+#pragma unroll
+        for (int i = 0; i < TPG; ++i) {
+          if (i >= tok_per_grp) break;
+          t += p[i] * vfs[i][j];
+        }
+        acc[h][j] = t;
+      }
+    }
+  };
+
   const int bstride = n_waves * n_split;     // global block-chain count
   if (tok_per_grp <= TPG) {
     // depth-2 software pipeline over cache blocks: the NEXT block's
@@ -237,14 +329,32 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
     // buffers, statically indexed — guide §5.4 rule 20)
     uint4 kA[TPG], vA[TPG], kB[TPG], vB[TPG];
     int b = split * n_waves + wave;
-    if (b < n_blocks) issue_block(b, kA, vA);
+    int physA = 0, physB = 0;
+    if (b < n_blocks) {
+      physA = fetch_phys(b);
+      if (b + bstride < n_blocks) physB = fetch_phys(b + bstride);
+      issue_block(physA, kA, vA);
+    }
+    auto consume = [&](int bb, const uint4 (&kraw)[TPG],
+                       const uint4 (&vraw)[TPG]) {
+      if constexpr (ABL == 0)
+        consume_block_batched(bb, kraw, vraw);
+      else
+        consume_block(bb, kraw, vraw);
+    };
     while (b < n_blocks) {
-      if (b + bstride < n_blocks) issue_block(b + bstride, kB, vB);
-      consume_block(b, kA, vA);
+      if (b + bstride < n_blocks) {
+        issue_block(physB, kB, vB);
+        if (b + 2 * bstride < n_blocks) physA = fetch_phys(b + 2 * bstride);
+      }
+      consume(b, kA, vA);
       b += bstride;
       if (b >= n_blocks) break;
-      if (b + bstride < n_blocks) issue_block(b + bstride, kA, vA);
-      consume_block(b, kB, vB);
+      if (b + bstride < n_blocks) {
+        issue_block(physA, kA, vA);
+        if (b + 2 * bstride < n_blocks) physB = fetch_phys(b + 2 * bstride);
+      }
+      consume(b, kB, vB);
       b += bstride;
     }
   } else {
@@ -271,7 +381,7 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
           float s = 0.f;
 #pragma unroll
           for (int j = 0; j < 8; ++j) s += qv[h][j] * kf[j];
-          s = group_sum<GW>(s);
+          s = group_sum_dpp<GW>(s);
           s *= scale;
           const float m_new = fmaxf(m[h], s);
           const float corr = __expf(m[h] - m_new);
@@ -286,28 +396,74 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
     }
   }
 
-  // ---- publish per-lane-group partials ----
+  if constexpr (ABL != 0) {
+    // ablation: no LDS publish/merge — wave 0's lane-groups dump their
+    // raw partials straight to out (wrong values, comparable traffic)
+    if (wave == 0) {
 #pragma unroll
-  for (int h = 0; h < GROUP; ++h) {
-    float* dst = accs + ((int64_t(pid) * GROUP + h) * D) + d0;
+      for (int h = 0; h < GROUP; ++h) {
+        bf16x8 o;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) dst[j] = acc[h][j];
-    if (gl == 0) {
-      ml[(pid * GROUP + h) * 2 + 0] = m[h];
-      ml[(pid * GROUP + h) * 2 + 1] = l[h];
+        for (int j = 0; j < 8; ++j) o.s[j] = f32_to_bf16(acc[h][j] + l[h]);
+        if (group == 0)
+          *reinterpret_cast<uint4*>(
+              out + (int64_t(seq) * n_kv_heads * GROUP + kvh * GROUP + h) * D +
+              d0) = o.u;
+      }
+    }
+    return;
+  }
+
+  // ---- in-wave cross-group merge (log-sum-exp): lanes off apart hold
+  // the same d0 slice for sibling lane-groups, so GPW partials combine
+  // with shuffles instead of LDS round-trips — the LDS image shrinks
+  // GPWx (16 -> 4 partials at GW=16) and the final scan with it (the
+  // full publish+merge was 4.1 us of the 20.8 us kernel at batch 64,
+  // RLLI_ATTN_ABLATE=1 vs 0).  kNegInf is finite, so idle lane-groups
+  // (m = kNegInf, l = 0) combine as weight-0 with no -inf-inf NaN.
+#pragma unroll
+  for (int off = GW; off < 64; off <<= 1) {
+#pragma unroll
+    for (int h = 0; h < GROUP; ++h) {
+      const float m_o = __shfl_xor(m[h], off, kWave);
+      const float l_o = __shfl_xor(l[h], off, kWave);
+      const float M = fmaxf(m[h], m_o);
+      const float wa = __expf(m[h] - M);
+      const float wb = __expf(m_o - M);
+      l[h] = l[h] * wa + l_o * wb;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        acc[h][j] = acc[h][j] * wa + __shfl_xor(acc[h][j], off, kWave) * wb;
+      m[h] = M;
+    }
+  }
+
+  // ---- publish one partial per WAVE (group 0 lanes carry the merged
+  // value for their d0 slice) ----
+  if (group == 0) {
+#pragma unroll
+    for (int h = 0; h < GROUP; ++h) {
+      float* dst = accs + ((int64_t(wave) * GROUP + h) * D) + d0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dst[j] = acc[h][j];
+      if (gl == 0) {
+        ml[(wave * GROUP + h) * 2 + 0] = m[h];
+        ml[(wave * GROUP + h) * 2 + 1] = l[h];
+      }
     }
   }
   __syncthreads();
 
   // ---- merge: each thread owns (head, dim) output elements ----
+  const int n_wpart = n_waves;
   for (int idx = threadIdx.x; idx < GROUP * D; idx += blockDim.x) {
     const int h = idx / D;
     const int d = idx % D;
     float M = kNegInf;
-    for (int p = 0; p < n_part; ++p)
+    for (int p = 0; p < n_wpart; ++p)
       M = fmaxf(M, ml[(p * GROUP + h) * 2 + 0]);
     float num = 0.f, den = 0.f;
-    for (int p = 0; p < n_part; ++p) {
+    for (int p = 0; p < n_wpart; ++p) {
       const float w = __expf(ml[(p * GROUP + h) * 2 + 0] - M);
       num += w * accs[(int64_t(p) * GROUP + h) * D + d];
       den += w * ml[(p * GROUP + h) * 2 + 1];
@@ -374,25 +530,45 @@ void dispatch_decode(const uint16_t* q, const uint16_t* k_cache,
                      int n_split, float* part_out, float* part_ml,
                      hipStream_t stream) {
   constexpr int D = GW * 8;
-  constexpr int GPW = 64 / GW;
-  const int n_part = 4 * GPW;
+  const int n_part = 4;          // one in-wave-merged partial per wave
   const size_t smem = size_t(n_part) * GROUP * (D + 2) * sizeof(float);
   const dim3 grid(batch * n_kv_heads * n_split);
-  if (fa != nullptr) {
-    hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP, true>),
-                       grid, dim3(256), smem, stream,
-                       q, k_cache, v_cache, block_table, seq_lens, out,
-                       n_kv_heads, block_size, max_blocks, scale, q_stride,
-                       fa->k_src, fa->v_src, fa->positions, fa->cos_sin,
-                       fa->slot_mapping, fa->k_cache_w, fa->v_cache_w,
-                       n_split, part_out, part_ml);
+  auto launch = [&](auto bs_tag, auto abl_tag) {
+    constexpr int BS = decltype(bs_tag)::value;
+    constexpr int ABL = decltype(abl_tag)::value;
+    if (fa != nullptr) {
+      hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP, true, BS, ABL>),
+                         grid, dim3(256), smem, stream,
+                         q, k_cache, v_cache, block_table, seq_lens, out,
+                         n_kv_heads, block_size, max_blocks, scale, q_stride,
+                         fa->k_src, fa->v_src, fa->positions, fa->cos_sin,
+                         fa->slot_mapping, fa->k_cache_w, fa->v_cache_w,
+                         n_split, part_out, part_ml);
+    } else {
+      hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP, false, BS, ABL>),
+                         grid, dim3(256), smem, stream,
+                         q, k_cache, v_cache, block_table, seq_lens, out,
+                         n_kv_heads, block_size, max_blocks, scale, q_stride,
+                         nullptr, nullptr, nullptr, nullptr, nullptr,
+                         nullptr, nullptr, n_split, part_out, part_ml);
+    }
+  };
+  using I0 = std::integral_constant<int, 0>;
+  // RLLI_ATTN_ABLATE: perf-ablation kernels (WRONG results — bench only)
+  static const int abl = [] {
+    const char* e = std::getenv("RLLI_ATTN_ABLATE");
+    return e ? atoi(e) : 0;
+  }();
+  if (block_size == 16) {
+    using B16 = std::integral_constant<int, 16>;
+    switch (abl) {
+      case 1: launch(B16{}, std::integral_constant<int, 1>{}); break;
+      case 2: launch(B16{}, std::integral_constant<int, 2>{}); break;
+      case 3: launch(B16{}, std::integral_constant<int, 3>{}); break;
+      default: launch(B16{}, I0{});
+    }
   } else {
-    hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP, false>),
-                       grid, dim3(256), smem, stream,
-                       q, k_cache, v_cache, block_table, seq_lens, out,
-                       n_kv_heads, block_size, max_blocks, scale, q_stride,
-                       nullptr, nullptr, nullptr, nullptr, nullptr,
-                       nullptr, nullptr, n_split, part_out, part_ml);
+    launch(I0{}, I0{});
   }
   if (n_split > 1) {
     hipLaunchKernelGGL(decode_combine_kernel, dim3(batch * n_kv_heads),
@@ -407,6 +583,12 @@ void dispatch_decode(const uint16_t* q, const uint16_t* k_cache,
 // in flight; small batches (interactive / long-context) get their
 // sequences split across block-chains instead of idling CUs.
 int decode_attn_n_split(int batch, int n_kv_heads) {
+  // RLLI_ATTN_SPLIT forces the factor (A/B tuning); 0/unset = heuristic
+  static const int forced = [] {
+    const char* e = std::getenv("RLLI_ATTN_SPLIT");
+    return e ? atoi(e) : 0;
+  }();
+  if (forced > 0) return forced > 16 ? 16 : forced;
   const int wgs = batch * n_kv_heads;
   if (wgs >= 384) return 1;
   int n = (512 + wgs - 1) / wgs;

@@ -93,7 +93,7 @@ void prefill_attn_kernel(const uint16_t* __restrict__ q,
       float s = 0.f;
 #pragma unroll
       for (int i = 0; i < 8; ++i) s += qv[h][i] * kf[i];
-      s = group_sum<GW>(s);
+      s = group_sum_dpp<GW>(s);
       s *= scale;
       const float m_new = fmaxf(m[h], s);
       const float corr = __expf(m[h] - m_new);
