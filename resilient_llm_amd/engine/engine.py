@@ -135,6 +135,7 @@ class LLMEngine:
         self._live: set[str] = set()   # rids currently owned by the engine
         self._arrival_counter = 0
         self._batch_dirty = True
+        self._graph_loaded = False
         self._pending = None          # (seqs, event, host_tokens) 1-deep
         self._pinned = None
         self.stats = {"prefill_steps": 0, "decode_steps": 0,
@@ -572,22 +573,44 @@ class LLMEngine:
         self._b_step_off = 0
         self._b_torch_sampling = any(s.params.needs_torch_sampling for s in seqs)
         self._batch_dirty = False
+        self._graph_loaded = False
 
     def _decode_step(self) -> list[StepOutput]:
         seqs = self.running
         if getattr(self, "_batch_dirty", True) or self._b_ids.shape[0] != len(seqs):
             self._rebuild_batch()
+        gr = self.graph_runner
+        if gr is not None and gr.steady_ok(len(seqs), self._b_bt.shape[1]):
+            # steady path: the WHOLE step (slots, forward, sample, state
+            # advance) is one graph replay; the static buffers are loaded
+            # only when the batch composition changed (graph.py)
+            if not self._graph_loaded:
+                gr.load(self._b_ids, self._b_pos, self._b_bt, self._b_temps,
+                        self._b_seeds)
+                self._graph_loaded = True
+            tok_dev, logits = gr.step()
+            if self._b_torch_sampling:
+                # top-p / penalty rows: recompute on the host from the
+                # captured logits and correct the fed tokens before the
+                # next replay (_sample reproduces the fused kernel's
+                # choice for plain rows — same seeds, same premix)
+                tokens = self._sample(logits, seqs)
+                tok_dev = torch.tensor(tokens, dtype=torch.int32,
+                                       device=self.device)
+                gr.override_tokens(tok_dev)
+            for seq in seqs:
+                seq.n_cached += 1      # KV of the fed token was appended
+            return self._queue_pending(seqs, tok_dev)
+        # eager path (no graphs, torch-side sampling, or out-of-envelope
+        # block tables)
         pos = self._b_pos
         bt = self._b_bt
         block_idx = (pos // self.block_size).long()
         slots = bt.gather(1, block_idx.unsqueeze(1)).squeeze(1) \
             * self.block_size + (pos - block_idx.int() * self.block_size)
         seq_lens = pos + 1
-        if self.graph_runner is not None:
-            logits = self.graph_runner.run(self._b_ids, pos, slots, bt, seq_lens)
-        else:
-            logits = self.model.forward_decode(self._b_ids, pos, self.kv,
-                                               slots, bt, seq_lens)
+        logits = self.model.forward_decode(self._b_ids, pos, self.kv,
+                                           slots, bt, seq_lens)
         if self._b_torch_sampling:
             tokens = self._sample(logits, seqs)     # torch top-p path (sync)
             tok_dev = torch.tensor(tokens, dtype=torch.int32,
@@ -603,6 +626,13 @@ class LLMEngine:
         # batch composition changes when the pending tokens land)
         self._b_ids.copy_(tok_dev)
         self._b_pos += 1
+        return self._queue_pending(seqs, tok_dev)
+
+    def _queue_pending(self, seqs: list, tok_dev: torch.Tensor) \
+            -> list[StepOutput]:
+        """Hand the sampled tokens to the async landing pipeline: D2H
+        into the pinned ring, event-fenced; processed at the NEXT step
+        so steady decode never synchronizes."""
         B = len(seqs)
         if self.device.type == "cuda":
             if self._pinned is None or self._pinned.shape[0] < self.max_batch_size:

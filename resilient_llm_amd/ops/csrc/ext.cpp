@@ -508,10 +508,19 @@ Tensor sample(const Tensor& logits, const Tensor& temperatures,
   TORCH_CHECK(temperatures.numel() == batch && seeds.numel() == batch);
   Tensor out = at::empty({batch}, logits.options().dtype(at::kInt));
   c10::hip::HIPGuardMasqueradingAsCUDA guard(logits.device());
+  const int n_split = rlli::sample_n_split(batch);
+  uint64_t* pptr = nullptr;
+  Tensor partials;
+  if (n_split > 1) {
+    // caching-allocator tensor: participates in hipGraph capture pools
+    partials = at::empty({int64_t(batch) * n_split},
+                         logits.options().dtype(at::kLong));
+    pptr = reinterpret_cast<uint64_t*>(partials.data_ptr<int64_t>());
+  }
   rlli::launch_sample(bf16_ptr(logits), temperatures.data_ptr<float>(),
                       reinterpret_cast<const uint64_t*>(seeds.data_ptr<int64_t>()),
                       uint64_t(step), out.data_ptr<int32_t>(), batch, vocab,
-                      current_stream(logits));
+                      pptr, n_split, current_stream(logits));
   check_launch("sample");
   return out;
 }

@@ -111,12 +111,15 @@ void launch_stream_probe(const uint16_t* w, float* sink, int N, int K,
 // Fused sampling: greedy argmax when temperature[i] == 0, else Gumbel-max
 // sampling of softmax(logits / temperature[i]) with an in-kernel counter
 // hash RNG keyed on (seed, row, column) — no 32 MB random tensor per step.
+int sample_n_split(int batch);       // phase-A split factor (1 = single)
 void launch_sample(
     const uint16_t* logits,            // [batch, vocab] bf16
     const float* temperatures,         // [batch]
     const uint64_t* seeds,             // [batch] per-request seeds
     uint64_t step,                     // per-sequence step counter (mixed in)
     int32_t* out_tokens,               // [batch]
-    int batch, int vocab, hipStream_t stream);
+    int batch, int vocab,
+    uint64_t* partials, int n_split,   // [batch, n_split] workspace
+    hipStream_t stream);
 
 }  // namespace rlli

@@ -108,6 +108,52 @@ def test_graph_runner_matches_eager():
     assert eager_out == graph_out
 
 
+def test_graph_runner_matches_eager_sampled():
+    """Temperature sampling through the steady graph (seeds advanced
+    in-graph by GOLDEN per replay) matches the eager path's step-offset
+    premix token for token."""
+    from resilient_llm_amd.engine.graph import install_graph_runner
+    prompts = {f"s{i}": list(range(7 + i, 52 + 3 * i)) for i in range(4)}
+    params = lambda: SamplingParams(max_tokens=10, temperature=0.8, seed=123)  # noqa: E731
+
+    eager = make_engine("tiny-128", num_blocks=256, max_batch_size=8, seed=9)
+    for rid, p in prompts.items():
+        eager.add_request(rid, p, params())
+    eager_out = drain(eager)
+
+    graphed = make_engine("tiny-128", num_blocks=256, max_batch_size=8, seed=9)
+    install_graph_runner(graphed)
+    for rid, p in prompts.items():
+        graphed.add_request(rid, p, params())
+    graph_out = drain(graphed)
+    assert eager_out == graph_out
+
+
+def test_graph_runner_top_p_override_matches_eager():
+    """A batch mixing top-p rows (torch-side sampling) with greedy rows
+    still replays the steady graph; the override path must match the
+    eager engine token for token."""
+    from resilient_llm_amd.engine.graph import install_graph_runner
+
+    def load(e):
+        e.add_request("p0", list(range(5, 40)),
+                      SamplingParams(max_tokens=8, temperature=0.9,
+                                     top_p=0.8, seed=7))
+        e.add_request("p1", list(range(9, 52)), SamplingParams(max_tokens=8))
+        e.add_request("p2", list(range(3, 61)),
+                      SamplingParams(max_tokens=8, temperature=0.7,
+                                     presence_penalty=0.5, seed=11))
+
+    eager = make_engine("tiny-128", num_blocks=256, max_batch_size=8, seed=3)
+    load(eager)
+    eager_out = drain(eager)
+
+    graphed = make_engine("tiny-128", num_blocks=256, max_batch_size=8, seed=3)
+    install_graph_runner(graphed)
+    load(graphed)
+    assert drain(graphed) == eager_out
+
+
 def test_long_context_beyond_graph_envelope():
     """A sequence longer than the graph runner's block-table width falls
     back to the eager decode path and stays correct."""
