@@ -65,7 +65,16 @@ DEV_INLINE u32x4 silu_mul_frag(u32x4 graw, u32x4 uraw) {
 
 // One wave = one 64-column panel x one K-slice.
 //  blockIdx.x = panel + n_panels * slice
-template <int M_TILES, int PROLOGUE, int KSTEPS>
+// G = pipeline depth in 32-k GRANULES (one granule = CT B-frags +
+// M_TILES A-frags = the operand set of 16 MFMAs).  The r02 chunk
+// pipeline (2 buffers of KSTEPS granules, compute-all-then-load-all)
+// measured ~6 GB/s of stream per wave: the compiler placed every
+// next-chunk load after the whole MFMA block, so only one chunk was
+// ever in flight.  The granule ring consumes granule p and immediately
+// refills its registers with granule p+G — (G-1) granules stay in
+// flight at all times — and sched_group_barrier pins the
+// [16 MFMA][8 VMEM] alternation so the scheduler cannot re-bunch them.
+template <int M_TILES, int PROLOGUE, int G>
 __global__ __launch_bounds__(64)
 void skinny2_kernel(const uint16_t* __restrict__ x,
                     const uint16_t* __restrict__ w,
@@ -113,62 +122,66 @@ void skinny2_kernel(const uint16_t* __restrict__ x,
 #pragma unroll
     for (int ct = 0; ct < CT; ++ct) acc[mt][ct] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  constexpr int CHUNK = 32 * KSTEPS;
   constexpr int NA = PROLOGUE == PRO_SILU ? 2 : 1;   // regs per A frag
   constexpr int AUX_NT = 2;                          // nontemporal policy
 
-  u32x4 bbuf[2][KSTEPS][CT];
-  u32x4 abuf[2][KSTEPS][M_TILES][NA];
+  u32x4 bbuf[G][CT];
+  u32x4 abuf[G][M_TILES][NA];
 
-  auto load_chunk = [&](int buf, int k) {
+  auto load_gran = [&](int p, int kk) {
+    const int soff = kk * 2;                 // SGPR: kk is wave-uniform
 #pragma unroll
-    for (int ks = 0; ks < KSTEPS; ++ks) {
-      const int soff = (k + ks * 32) * 2;    // SGPR: k is wave-uniform
+    for (int ct = 0; ct < CT; ++ct)
+      bbuf[p][ct] = __builtin_amdgcn_raw_buffer_load_b128(
+          wrsrc, voff_b[ct], soff, AUX_NT);
+#pragma unroll
+    for (int mt = 0; mt < M_TILES; ++mt) {
+      abuf[p][mt][0] = __builtin_amdgcn_raw_buffer_load_b128(
+          xrsrc, voff_a[mt], soff, 0);
+      if (PROLOGUE == PRO_SILU)
+        abuf[p][mt][1] = __builtin_amdgcn_raw_buffer_load_b128(
+            xrsrc, voff_a[mt], soff + K * 2, 0);
+    }
+  };
+  auto mfma_gran = [&](int p) {
+#pragma unroll
+    for (int mt = 0; mt < M_TILES; ++mt) {
+      u32x4 araw = abuf[p][mt][0];
+      if (PROLOGUE == PRO_SILU)
+        araw = silu_mul_frag(araw, abuf[p][mt][1]);
+      const bf16x8_vec afrag = as_bf16x8(araw);
 #pragma unroll
       for (int ct = 0; ct < CT; ++ct)
-        bbuf[buf][ks][ct] = __builtin_amdgcn_raw_buffer_load_b128(
-            wrsrc, voff_b[ct], soff, AUX_NT);
-#pragma unroll
-      for (int mt = 0; mt < M_TILES; ++mt) {
-        abuf[buf][ks][mt][0] = __builtin_amdgcn_raw_buffer_load_b128(
-            xrsrc, voff_a[mt], soff, 0);
-        if (PROLOGUE == PRO_SILU)
-          abuf[buf][ks][mt][1] = __builtin_amdgcn_raw_buffer_load_b128(
-              xrsrc, voff_a[mt], soff + K * 2, 0);
-      }
-    }
-  };
-  auto compute_chunk = [&](int buf) {
-#pragma unroll
-    for (int ks = 0; ks < KSTEPS; ++ks) {
-#pragma unroll
-      for (int mt = 0; mt < M_TILES; ++mt) {
-        u32x4 araw = abuf[buf][ks][mt][0];
-        if (PROLOGUE == PRO_SILU)
-          araw = silu_mul_frag(araw, abuf[buf][ks][mt][1]);
-        const bf16x8_vec afrag = as_bf16x8(araw);
-#pragma unroll
-        for (int ct = 0; ct < CT; ++ct)
-          acc[mt][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag, as_bf16x8(bbuf[buf][ks][ct]), acc[mt][ct], 0, 0, 0);
-      }
+        acc[mt][ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, as_bf16x8(bbuf[p][ct]), acc[mt][ct], 0, 0, 0);
     }
   };
 
-  // k_per_slice and K are multiples of 256, so every slice has an even
-  // number of chunks for KSTEPS <= 4 — the 2x unrolled steady state
-  // never needs a tail.
-  load_chunk(0, kbeg);
-  if (kbeg + CHUNK < kend) load_chunk(1, kbeg + CHUNK);
-  int k = kbeg;
-  for (; k + 2 * CHUNK < kend; k += 2 * CHUNK) {
-    compute_chunk(0);
-    load_chunk(0, k + 2 * CHUNK);
-    compute_chunk(1);
-    if (k + 3 * CHUNK < kend) load_chunk(1, k + 3 * CHUNK);
+  const int n_g = (kend - kbeg) / 32;        // granules in this K-slice
+  // prologue: fill the ring
+#pragma unroll
+  for (int p = 0; p < G; ++p)
+    if (p < n_g) load_gran(p, kbeg + p * 32);
+  // steady: consume granule g (ring slot p), refill slot p with g+G.
+  // The refill MUST follow the MFMAs (same registers), but only ONE
+  // granule's MFMAs — the other G-1 granules' loads stay in flight, so
+  // the wait ahead of each MFMA group is a counted vmcnt, not a drain.
+  int g = 0;
+  for (; g + G < n_g; g += G) {
+#pragma unroll
+    for (int p = 0; p < G; ++p) {
+      mfma_gran(p);
+      if (g + p + G < n_g) load_gran(p, kbeg + (g + p + G) * 32);
+      // pin the alternation: one granule of MFMAs, then its refill
+      // loads — the scheduler may not re-bunch loads behind compute
+      __builtin_amdgcn_sched_group_barrier(0x008, M_TILES * CT, 0);
+      __builtin_amdgcn_sched_group_barrier(0x020, CT + M_TILES * NA, 0);
+    }
   }
-  compute_chunk(0);
-  if (k + CHUNK < kend) compute_chunk(1);
+  // epilogue: drain the ring
+#pragma unroll
+  for (int p = 0; p < G; ++p)
+    if (g + p < n_g) mfma_gran(p);
 
   // C layout (guide §3): col = lane&15, row = (lane>>4)*4 + reg
 #pragma unroll
@@ -214,25 +227,28 @@ void launch_any(const uint16_t* x, const uint16_t* w, float* ws,
   // the reduce would sum their UNINITIALIZED slabs
   splitk = (K + k_per_slice - 1) / k_per_slice;
   const int blocks = (N / BN) * splitk;
-  // KSTEPS: prefetch window per chunk.  Large fragment sets (M_TILES>2,
-  // or SILU's 2x A regs) use the narrow window so the kernel stays under
-  // ~256 VGPRs and multi-wave occupancy hides the latency instead
-  // (372-VGPR variants measured slower: AGPR shuffling + drained vmcnt).
-  // RLLI_SK2_KS=2|4 overrides for tuning.
-  static const char* ks_env = std::getenv("RLLI_SK2_KS");
-  const int ks_req = ks_env ? atoi(ks_env) : 0;
+  // G: granule-ring depth = granules kept in flight.  Deep rings buy
+  // per-wave streaming rate (in-flight bytes/latency) until the VGPR
+  // budget forces spills or 1-wave occupancy; big fragment sets
+  // (M_TILES > 2, SILU's 2x A regs) take a shallower ring.
+  // RLLI_SK2_G=2..8 overrides for tuning.
+  static const char* g_env = std::getenv("RLLI_SK2_G");
+  const int g_req = g_env ? atoi(g_env) : 0;
   auto launch = [&](auto mt_tag) {
     constexpr int MT = decltype(mt_tag)::value;
-    const int ks = ks_req ? ks_req
-                          : ((PROLOGUE == PRO_SILU || MT > 2) ? 2 : 4);
-    if (ks == 4)
-      hipLaunchKernelGGL((skinny2_kernel<MT, PROLOGUE, 4>), dim3(blocks),
-                         dim3(64), 0, stream, x, w, ws, out, M, N, K,
-                         k_per_slice, splitk);
-    else
-      hipLaunchKernelGGL((skinny2_kernel<MT, PROLOGUE, 2>), dim3(blocks),
-                         dim3(64), 0, stream, x, w, ws, out, M, N, K,
-                         k_per_slice, splitk);
+    const int g = g_req ? g_req
+                        : (PROLOGUE == PRO_SILU ? 3 : (MT > 2 ? 4 : 6));
+    auto go = [&](auto g_tag) {
+      hipLaunchKernelGGL((skinny2_kernel<MT, PROLOGUE,
+                                         decltype(g_tag)::value>),
+                         dim3(blocks), dim3(64), 0, stream, x, w, ws, out,
+                         M, N, K, k_per_slice, splitk);
+    };
+    if (g <= 2) go(std::integral_constant<int, 2>{});
+    else if (g == 3) go(std::integral_constant<int, 3>{});
+    else if (g == 4) go(std::integral_constant<int, 4>{});
+    else if (g <= 6) go(std::integral_constant<int, 6>{});
+    else go(std::integral_constant<int, 8>{});
   };
   using T1 = std::integral_constant<int, 1>;
   using T2 = std::integral_constant<int, 2>;
