@@ -237,7 +237,7 @@ def load_config(path: str | os.PathLike[str] | None = None,
     )
     _require(router.estimate_chars_per_token >= 1,
              "router_settings.estimate_chars_per_token must be >= 1")
-    _require(router.routing_strategy in ("simple-shuffle", "least-busy", "round-robin"),
+    _require(router.routing_strategy in ("simple-shuffle", "least-busy", "round-robin", "prefix-affinity"),
              f"unknown routing_strategy {router.routing_strategy!r}")
 
     cris_model = (data.get("cris") or {}).get("model_id")

@@ -442,6 +442,24 @@ class GatewayApp:
                 pass
         return max(2, min(n + 1, 8))
 
+    def _affinity_key(self, body: dict) -> Optional[str]:
+        """Prompt-prefix key for ``prefix-affinity`` routing: the first
+        512 chars of the serialized messages.  Requests sharing a system
+        prompt / conversation head map to the same replica, whose engine
+        prefix cache then serves the shared blocks (None for any other
+        strategy — zero cost on the default paths)."""
+        if self.router.settings.routing_strategy != "prefix-affinity":
+            return None
+        parts: list[str] = []
+        n = 0
+        for m in body.get("messages") or []:
+            frag = f"{m.get('role', '')}\x1f{m.get('content', '')}\x1e"
+            parts.append(frag)
+            n += len(frag)
+            if n >= 512:
+                break
+        return "".join(parts)[:512] or None
+
     async def _chat_once(self, body: dict, alias: str, consumer: str,
                          total_est: int, rid: Optional[str] = None) -> Response:
         t0 = time.monotonic()
@@ -455,7 +473,8 @@ class GatewayApp:
                                   err_type="invalid_request_error", code="model_not_found")
         for _ in range(attempts):
             try:
-                ticket = self.router.acquire(alias, total_est, exclude=exclude)
+                ticket = self.router.acquire(alias, total_est, exclude=exclude,
+                                             affinity_key=self._affinity_key(body))
             except RouterRateLimit as e:
                 # once a worker has FAILED (non-throttle), an
                 # exhausted router is a consequence of the failure, not a
@@ -528,7 +547,8 @@ class GatewayApp:
         tokens the client already received are discarded (stateless
         replay)."""
         try:
-            first_ticket = self.router.acquire(alias, total_est)
+            first_ticket = self.router.acquire(
+                alias, total_est, affinity_key=self._affinity_key(body))
         except UnknownAlias:
             self.consumers.reconcile(consumer, total_est, 0)
             return Response.error(404, f"unknown model {alias!r}",
@@ -558,7 +578,9 @@ class GatewayApp:
                 for attempt in range(attempts):
                     if ticket is None:
                         try:
-                            ticket = app.router.acquire(alias, total_est, exclude=exclude)
+                            ticket = app.router.acquire(
+                                alias, total_est, exclude=exclude,
+                                affinity_key=app._affinity_key(body))
                         except NoDeploymentAvailable as e:
                             last_err = e
                             break

@@ -23,6 +23,7 @@ gateway executes them against workers and reports success/failure back.
 
 from __future__ import annotations
 
+import hashlib
 import dataclasses
 import random
 import threading
@@ -133,11 +134,24 @@ class Router:
             return False
         return True
 
-    def _shuffle_pick(self, alias: str,
-                      candidates: list[DeploymentState]) -> DeploymentState:
+    def _shuffle_pick(self, alias: str, candidates: list[DeploymentState],
+                      affinity_key: Optional[str] = None) -> DeploymentState:
         if len(candidates) == 1:
             return candidates[0]
         strategy = self.settings.routing_strategy
+        if strategy == "prefix-affinity" and affinity_key:
+            # Rendezvous (highest-random-weight) hash of (prompt prefix,
+            # deployment): identical prompt prefixes land on the same
+            # replica, so its engine prefix cache (engine/kv_cache.py
+            # content-addressed blocks) serves the shared context from
+            # cache instead of re-prefilling it on a random replica.
+            # Stable across gateway processes (blake2b, not PYTHONHASH-
+            # seeded hash()) and self-healing: when the preferred
+            # deployment is unhealthy/full the next-highest score takes
+            # over deterministically, and affinity restores on recovery.
+            return max(candidates, key=lambda s: hashlib.blake2b(
+                (affinity_key + "\0" + s.dep.model_id).encode(),
+                digest_size=8).digest())
         if strategy == "round-robin":
             # positional round-robin: a per-alias cursor cycles the
             # available candidates in order, independent of request
@@ -154,11 +168,12 @@ class Router:
 
     # ------------------------------------------------------------ acquire
     def _try_alias(self, alias: str, tokens: int, exclude: set,
-                   attempted: list) -> Optional[DeploymentState]:
+                   attempted: list,
+                   affinity_key: Optional[str] = None) -> Optional[DeploymentState]:
         states = self.alias_states(alias)
         candidates = [s for s in states if self._available(s, tokens, exclude)]
         while candidates:
-            s = self._shuffle_pick(alias, candidates)
+            s = self._shuffle_pick(alias, candidates, affinity_key)
             if s.limiter.try_acquire(tokens):
                 return s
             attempted.append(s.dep.model_id)
@@ -166,7 +181,8 @@ class Router:
         return None
 
     def acquire(self, alias: str, tokens_estimate: int = 0,
-                exclude: Optional[set] = None) -> Ticket:
+                exclude: Optional[set] = None,
+                affinity_key: Optional[str] = None) -> Ticket:
         """Admit a request for ``alias``: pick a deployment, charge its
         window, or walk the fallback chain.  Raises
         :class:`RouterRateLimit` when everything is exhausted and
@@ -174,12 +190,14 @@ class Router:
         exclude = exclude or set()
         attempted: list = []
         with self._lock:
-            s = self._try_alias(alias, tokens_estimate, exclude, attempted)
+            s = self._try_alias(alias, tokens_estimate, exclude, attempted,
+                                affinity_key)
             if s is not None:
                 return self._issue(s, alias, tokens_estimate, False, attempted)
             for fb_alias in self.settings.fallbacks.get(alias, []):
                 try:
-                    s = self._try_alias(fb_alias, tokens_estimate, exclude, attempted)
+                    s = self._try_alias(fb_alias, tokens_estimate, exclude,
+                                        attempted, affinity_key)
                 except UnknownAlias:
                     continue
                 if s is not None:

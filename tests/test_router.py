@@ -141,3 +141,62 @@ def test_least_busy_strategy():
     t1 = r.acquire("lb", 1)
     t2 = r.acquire("lb", 1)
     assert {t1.deployment.model_id, t2.deployment.model_id} == {"gpu0/r0", "gpu1/r1"}
+
+
+# ------------------------------------------------------- prefix affinity
+def _affinity_router(n=3, rpm=1000):
+    deps = [Deployment("aff", f"gpu/{i}/llama-3-8b", f"gpu{i}/r{i}",
+                       rpm=rpm, tpm=10**7, weight=1) for i in range(n)]
+    settings = RouterSettings(routing_strategy="prefix-affinity", fallbacks={})
+    return Router(deps, settings, clock=FakeClock(0.0),
+                  rng=random.Random(7))
+
+
+def test_prefix_affinity_sticky_and_spread():
+    """Identical prompt prefixes always land on the same deployment
+    (prefix-cache locality); distinct prefixes spread across replicas."""
+    r = _affinity_router()
+    picks = set()
+    for _ in range(10):
+        t = r.acquire("aff", 10, affinity_key="system: you are a bot\x1eQ1")
+        picks.add(t.state.dep.model_id)
+        r.complete(t, 10)
+    assert len(picks) == 1
+    spread = set()
+    for i in range(40):
+        t = r.acquire("aff", 10, affinity_key=f"prefix-{i}")
+        spread.add(t.state.dep.model_id)
+        r.complete(t, 10)
+    assert len(spread) == 3   # rendezvous hashing uses every replica
+
+
+def test_prefix_affinity_failover_and_restore():
+    """When the preferred replica is down, the SAME alternate is chosen
+    every time (deterministic next-highest score); affinity returns to
+    the original replica on recovery."""
+    r = _affinity_router()
+    key = "shared-context"
+    t = r.acquire("aff", 10, affinity_key=key)
+    preferred = t.state
+    r.complete(t, 10)
+    preferred.healthy = False
+    alts = set()
+    for _ in range(8):
+        t = r.acquire("aff", 10, affinity_key=key)
+        alts.add(t.state.dep.model_id)
+        r.complete(t, 10)
+    assert len(alts) == 1 and preferred.dep.model_id not in alts
+    preferred.healthy = True
+    t = r.acquire("aff", 10, affinity_key=key)
+    assert t.state is preferred
+    r.complete(t, 10)
+
+
+def test_prefix_affinity_without_key_falls_back_to_shuffle():
+    r = _affinity_router()
+    picks = set()
+    for _ in range(40):
+        t = r.acquire("aff", 10)
+        picks.add(t.state.dep.model_id)
+        r.complete(t, 10)
+    assert len(picks) == 3
