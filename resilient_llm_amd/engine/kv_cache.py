@@ -61,10 +61,11 @@ class PagedKVCache:
 
     @classmethod
     def for_model(cls, config, num_blocks: int, device="cpu",
-                  block_size: int = 16, tp_world: int = 1) -> "PagedKVCache":
+                  block_size: int = 16, tp_world: int = 1,
+                  kv_dtype: torch.dtype = torch.bfloat16) -> "PagedKVCache":
         return cls(config.n_layers, num_blocks,
                    config.n_kv_heads // tp_world, block_size,
-                   config.head_dim, device=device)
+                   config.head_dim, device=device, dtype=kv_dtype)
 
     def layer(self, i: int) -> tuple[torch.Tensor, torch.Tensor]:
         return self.k[i], self.v[i]

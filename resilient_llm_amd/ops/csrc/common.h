@@ -39,6 +39,53 @@ DEV_INLINE uint16_t f32_to_bf16(float x) {
   return uint16_t(u >> 16);
 }
 
+// ---- KV-cache element codecs --------------------------------------
+// The paged cache stores bf16 (default) or OCP fp8-e4m3 (opt-in,
+// `kv_dtype: fp8`): half the HBM traffic per token and double the
+// resident KV capacity of the same budget.  gfx950 converts natively
+// (v_cvt_pk_f32_fp8 / v_cvt_pk_fp8_f32 — 2 elements per VALU op); all
+// attention math stays f32.  Kernels touching the cache are templated
+// on one of these codecs; a "vec8" is always 8 cache ELEMENTS (16 B
+// bf16 / 8 B fp8), so per-lane geometry is unchanged.
+typedef float f32x2_cvt __attribute__((ext_vector_type(2)));
+
+struct CacheBF16 {
+  using elem = uint16_t;
+  using vec8 = uint4;
+  static DEV_INLINE void to_f32(const vec8& raw, float* out8) {
+    const bf16x8* p = reinterpret_cast<const bf16x8*>(&raw);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) out8[i] = bf16_to_f32(p->s[i]);
+  }
+  static DEV_INLINE vec8 from_f32(const float* in8) {
+    bf16x8 o;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) o.s[i] = f32_to_bf16(in8[i]);
+    return o.u;
+  }
+};
+
+struct CacheFP8 {
+  using elem = uint8_t;
+  using vec8 = uint2;
+  static DEV_INLINE void to_f32(const vec8& raw, float* out8) {
+    const f32x2_cvt a = __builtin_amdgcn_cvt_pk_f32_fp8(raw.x, false);
+    const f32x2_cvt b = __builtin_amdgcn_cvt_pk_f32_fp8(raw.x, true);
+    const f32x2_cvt c = __builtin_amdgcn_cvt_pk_f32_fp8(raw.y, false);
+    const f32x2_cvt d = __builtin_amdgcn_cvt_pk_f32_fp8(raw.y, true);
+    out8[0] = a.x; out8[1] = a.y; out8[2] = b.x; out8[3] = b.y;
+    out8[4] = c.x; out8[5] = c.y; out8[6] = d.x; out8[7] = d.y;
+  }
+  static DEV_INLINE vec8 from_f32(const float* in8) {
+    vec8 o;
+    o.x = __builtin_amdgcn_cvt_pk_fp8_f32(in8[0], in8[1], 0u, false);
+    o.x = __builtin_amdgcn_cvt_pk_fp8_f32(in8[2], in8[3], o.x, true);
+    o.y = __builtin_amdgcn_cvt_pk_fp8_f32(in8[4], in8[5], 0u, false);
+    o.y = __builtin_amdgcn_cvt_pk_fp8_f32(in8[6], in8[7], o.y, true);
+    return o;
+  }
+};
+
 // Wave-wide reductions (64 lanes).
 DEV_INLINE float wave_sum(float v) {
 #pragma unroll

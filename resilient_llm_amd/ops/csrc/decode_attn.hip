@@ -25,6 +25,8 @@
 
 #include "common.h"
 
+#include <type_traits>
+
 namespace rlli {
 
 namespace {
@@ -45,11 +47,12 @@ constexpr float kNegInf = -1e30f;
 // WRONG, never used for real inference): 1 = skip LDS publish/merge,
 // 2 = skip online softmax, 3 = loads only.  Isolates which phase owns
 // the wall time (guide pitfall 8: ablate before optimizing).
-template <int GW, int GROUP, bool FUSED, int BS = 0, int ABL = 0>
+template <int GW, int GROUP, bool FUSED, int BS = 0, int ABL = 0,
+          typename KC = CacheBF16>
 __global__ __launch_bounds__(256)
 void decode_attn_kernel(const uint16_t* __restrict__ q,
-                        const uint16_t* __restrict__ k_cache,
-                        const uint16_t* __restrict__ v_cache,
+                        const typename KC::elem* __restrict__ k_cache,
+                        const typename KC::elem* __restrict__ v_cache,
                         const int32_t* __restrict__ block_table,
                         const int32_t* __restrict__ seq_lens,
                         uint16_t* __restrict__ out,
@@ -60,8 +63,8 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
                         const int32_t* __restrict__ positions,
                         const float* __restrict__ cos_sin,
                         const int32_t* __restrict__ slot_mapping,
-                        uint16_t* __restrict__ k_cache_w,
-                        uint16_t* __restrict__ v_cache_w,
+                        typename KC::elem* __restrict__ k_cache_w,
+                        typename KC::elem* __restrict__ v_cache_w,
                         int n_split,
                         float* __restrict__ part_out,
                         float* __restrict__ part_ml) {
@@ -157,14 +160,18 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
         vf[i] = bf16_to_f32(vr.s[i]);
       }
       const int32_t slot = slot_mapping[seq];
-      bf16x8 kw;
-#pragma unroll
-      for (int i = 0; i < 8; ++i) kw.s[i] = f32_to_bf16(kf[i]);
       const int64_t cbase =
           (int64_t(slot / block_size) * n_kv_heads + kvh) * block_size * D +
           int64_t(slot % block_size) * D + d0;
-      *reinterpret_cast<uint4*>(k_cache_w + cbase) = kw.u;
-      *reinterpret_cast<uint4*>(v_cache_w + cbase) = vr.u;
+      // append at cache precision (bf16 identity / fp8 cvt_pk); the
+      // self-attention term below uses the STORED values so future
+      // steps reproduce this step's math exactly
+      const typename KC::vec8 kw = KC::from_f32(kf);
+      const typename KC::vec8 vw = KC::from_f32(vf);
+      *reinterpret_cast<typename KC::vec8*>(k_cache_w + cbase) = kw;
+      *reinterpret_cast<typename KC::vec8*>(v_cache_w + cbase) = vw;
+      KC::to_f32(kw, kf);
+      KC::to_f32(vw, vf);
 #pragma unroll
       for (int h = 0; h < GROUP; ++h) {
         float s = 0.f;
@@ -198,35 +205,31 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
   auto fetch_phys = [&](int b) -> int {
     return block_table[int64_t(seq) * max_blocks + b];
   };
-  auto issue_block = [&](int phys, uint4 (&kraw)[TPG], uint4 (&vraw)[TPG]) {
+  using cvec8 = typename KC::vec8;
+  auto issue_block = [&](int phys, cvec8 (&kraw)[TPG], cvec8 (&vraw)[TPG]) {
     const int64_t kv_base =
         (int64_t(phys) * n_kv_heads + kvh) * block_size * D;
 #pragma unroll
     for (int i = 0; i < TPG; ++i) {
       if (i < tok_per_grp) {
         const int tok = group + GPW * i;
-        kraw[i] = *reinterpret_cast<const uint4*>(
+        kraw[i] = *reinterpret_cast<const cvec8*>(
             k_cache + kv_base + int64_t(tok) * D + d0);
-        vraw[i] = *reinterpret_cast<const uint4*>(
+        vraw[i] = *reinterpret_cast<const cvec8*>(
             v_cache + kv_base + int64_t(tok) * D + d0);
       }
     }
   };
-  auto consume_block = [&](int b, const uint4 (&kraw)[TPG],
-                           const uint4 (&vraw)[TPG]) {
+  auto consume_block = [&](int b, const cvec8 (&kraw)[TPG],
+                           const cvec8 (&vraw)[TPG]) {
 #pragma unroll
     for (int i = 0; i < TPG; ++i) {
       if (i >= tok_per_grp) break;
       const int tok = group + GPW * i;         // token within the block
       if (b * block_size + tok >= len_cache) continue;   // group-uniform tail
-      bf16x8 kv, vv;
-      kv.u = kraw[i];
-      vv.u = vraw[i];
       float kf[8], vf[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) kf[j] = bf16_to_f32(kv.s[j]);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) vf[j] = bf16_to_f32(vv.s[j]);
+      KC::to_f32(kraw[i], kf);
+      KC::to_f32(vraw[i], vf);
 
       if constexpr (ABL == 3) {              // loads only
 #pragma unroll
@@ -265,8 +268,8 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
   // tokens contribute exp(kNegInf - m_new) == 0 (kNegInf is a finite
   // -1e30, so no -inf minus -inf NaN is reachable; every walked block
   // holds at least one valid token, so m_new is always finite).
-  auto consume_block_batched = [&](int bb, const uint4 (&kraw)[TPG],
-                                   const uint4 (&vraw)[TPG]) {
+  auto consume_block_batched = [&](int bb, const cvec8 (&kraw)[TPG],
+                                   const cvec8 (&vraw)[TPG]) {
     float vfs[TPG][8];
     float svals[GROUP][TPG];
 #pragma unroll
@@ -274,14 +277,9 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
       if (i >= tok_per_grp) break;
       const int tok = group + GPW * i;
       const bool valid = bb * block_size + tok < len_cache;
-      bf16x8 kv, vv;
-      kv.u = kraw[i];
-      vv.u = vraw[i];
       float kf[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) kf[j] = bf16_to_f32(kv.s[j]);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) vfs[i][j] = bf16_to_f32(vv.s[j]);
+      KC::to_f32(kraw[i], kf);
+      KC::to_f32(vraw[i], vfs[i]);
 #pragma unroll
       for (int h = 0; h < GROUP; ++h) {
         float s = 0.f;
@@ -327,7 +325,7 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
     // depth-2 software pipeline over cache blocks: the NEXT block's
     // K/V loads are in flight while this block computes (named A/B
     // buffers, statically indexed — guide §5.4 rule 20)
-    uint4 kA[TPG], vA[TPG], kB[TPG], vB[TPG];
+    cvec8 kA[TPG], vA[TPG], kB[TPG], vB[TPG];
     int b = split * n_waves + wave;
     int physA = 0, physB = 0;
     if (b < n_blocks) {
@@ -335,8 +333,8 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
       if (b + bstride < n_blocks) physB = fetch_phys(b + bstride);
       issue_block(physA, kA, vA);
     }
-    auto consume = [&](int bb, const uint4 (&kraw)[TPG],
-                       const uint4 (&vraw)[TPG]) {
+    auto consume = [&](int bb, const cvec8 (&kraw)[TPG],
+                       const cvec8 (&vraw)[TPG]) {
       if constexpr (ABL == 0)
         consume_block_batched(bb, kraw, vraw);
       else
@@ -366,16 +364,11 @@ void decode_attn_kernel(const uint16_t* __restrict__ q,
       for (int bi = 0; bi < tok_per_grp; ++bi) {
         const int tok = group + GPW * bi;
         if (b * block_size + tok >= len_cache) continue;
-        bf16x8 kv, vv;
-        kv.u = *reinterpret_cast<const uint4*>(
-            k_cache + kv_base + int64_t(tok) * D + d0);
-        vv.u = *reinterpret_cast<const uint4*>(
-            v_cache + kv_base + int64_t(tok) * D + d0);
         float kf[8], vf[8];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) kf[j] = bf16_to_f32(kv.s[j]);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) vf[j] = bf16_to_f32(vv.s[j]);
+        KC::to_f32(*reinterpret_cast<const cvec8*>(
+            k_cache + kv_base + int64_t(tok) * D + d0), kf);
+        KC::to_f32(*reinterpret_cast<const cvec8*>(
+            v_cache + kv_base + int64_t(tok) * D + d0), vf);
 #pragma unroll
         for (int h = 0; h < GROUP; ++h) {
           float s = 0.f;
@@ -517,18 +510,21 @@ struct FusedArgs {
   const int32_t* positions = nullptr;
   const float* cos_sin = nullptr;
   const int32_t* slot_mapping = nullptr;
-  uint16_t* k_cache_w = nullptr;
-  uint16_t* v_cache_w = nullptr;
+  void* k_cache_w = nullptr;     // cache-codec elem type
+  void* v_cache_w = nullptr;
 };
 
-template <int GW, int GROUP>
-void dispatch_decode(const uint16_t* q, const uint16_t* k_cache,
-                     const uint16_t* v_cache, const int32_t* block_table,
+template <int GW, int GROUP, typename KC>
+void dispatch_decode(const uint16_t* q, const void* k_cache_v,
+                     const void* v_cache_v, const int32_t* block_table,
                      const int32_t* seq_lens, uint16_t* out, int batch,
                      int n_kv_heads, int block_size, int max_blocks,
                      float scale, int q_stride, const FusedArgs* fa,
                      int n_split, float* part_out, float* part_ml,
                      hipStream_t stream) {
+  using CElem = typename KC::elem;
+  const CElem* k_cache = static_cast<const CElem*>(k_cache_v);
+  const CElem* v_cache = static_cast<const CElem*>(v_cache_v);
   constexpr int D = GW * 8;
   const int n_part = 4;          // one in-wave-merged partial per wave
   const size_t smem = size_t(n_part) * GROUP * (D + 2) * sizeof(float);
@@ -537,15 +533,17 @@ void dispatch_decode(const uint16_t* q, const uint16_t* k_cache,
     constexpr int BS = decltype(bs_tag)::value;
     constexpr int ABL = decltype(abl_tag)::value;
     if (fa != nullptr) {
-      hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP, true, BS, ABL>),
+      hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP, true, BS, ABL, KC>),
                          grid, dim3(256), smem, stream,
                          q, k_cache, v_cache, block_table, seq_lens, out,
                          n_kv_heads, block_size, max_blocks, scale, q_stride,
                          fa->k_src, fa->v_src, fa->positions, fa->cos_sin,
-                         fa->slot_mapping, fa->k_cache_w, fa->v_cache_w,
+                         fa->slot_mapping,
+                         static_cast<CElem*>(fa->k_cache_w),
+                         static_cast<CElem*>(fa->v_cache_w),
                          n_split, part_out, part_ml);
     } else {
-      hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP, false, BS, ABL>),
+      hipLaunchKernelGGL((decode_attn_kernel<GW, GROUP, false, BS, ABL, KC>),
                          grid, dim3(256), smem, stream,
                          q, k_cache, v_cache, block_table, seq_lens, out,
                          n_kv_heads, block_size, max_blocks, scale, q_stride,
@@ -554,18 +552,23 @@ void dispatch_decode(const uint16_t* q, const uint16_t* k_cache,
     }
   };
   using I0 = std::integral_constant<int, 0>;
-  // RLLI_ATTN_ABLATE: perf-ablation kernels (WRONG results — bench only)
+  // RLLI_ATTN_ABLATE: perf-ablation kernels (WRONG results — bench
+  // only; instantiated for the bf16 cache only)
   static const int abl = [] {
     const char* e = std::getenv("RLLI_ATTN_ABLATE");
     return e ? atoi(e) : 0;
   }();
   if (block_size == 16) {
     using B16 = std::integral_constant<int, 16>;
-    switch (abl) {
-      case 1: launch(B16{}, std::integral_constant<int, 1>{}); break;
-      case 2: launch(B16{}, std::integral_constant<int, 2>{}); break;
-      case 3: launch(B16{}, std::integral_constant<int, 3>{}); break;
-      default: launch(B16{}, I0{});
+    if constexpr (std::is_same_v<KC, CacheBF16>) {
+      switch (abl) {
+        case 1: launch(B16{}, std::integral_constant<int, 1>{}); break;
+        case 2: launch(B16{}, std::integral_constant<int, 2>{}); break;
+        case 3: launch(B16{}, std::integral_constant<int, 3>{}); break;
+        default: launch(B16{}, I0{});
+      }
+    } else {
+      launch(B16{}, I0{});
     }
   } else {
     launch(I0{}, I0{});
@@ -595,23 +598,31 @@ int decode_attn_n_split(int batch, int n_kv_heads) {
   return n > 16 ? 16 : n;
 }
 
-void launch_decode_attn_impl(const uint16_t* q, const uint16_t* k_cache,
-                             const uint16_t* v_cache,
+void launch_decode_attn_impl(const uint16_t* q, const void* k_cache,
+                             const void* v_cache,
                              const int32_t* block_table,
                              const int32_t* seq_lens, uint16_t* out,
                              int batch, int n_q_heads, int n_kv_heads,
                              int head_dim, int block_size, int max_blocks,
                              float scale, int q_stride, const FusedArgs* fa,
                              int n_split, float* part_out, float* part_ml,
-                             hipStream_t stream) {
+                             bool cache_fp8, hipStream_t stream) {
   if (batch == 0) return;
   if (part_out == nullptr) n_split = 1;
   const int group = n_q_heads / n_kv_heads;
   auto run = [&](auto gw_tag, auto group_tag) {
-    dispatch_decode<decltype(gw_tag)::value, decltype(group_tag)::value>(
-        q, k_cache, v_cache, block_table, seq_lens, out, batch, n_kv_heads,
-        block_size, max_blocks, scale, q_stride, fa, n_split, part_out,
-        part_ml, stream);
+    if (cache_fp8)
+      dispatch_decode<decltype(gw_tag)::value, decltype(group_tag)::value,
+                      CacheFP8>(
+          q, k_cache, v_cache, block_table, seq_lens, out, batch, n_kv_heads,
+          block_size, max_blocks, scale, q_stride, fa, n_split, part_out,
+          part_ml, stream);
+    else
+      dispatch_decode<decltype(gw_tag)::value, decltype(group_tag)::value,
+                      CacheBF16>(
+          q, k_cache, v_cache, block_table, seq_lens, out, batch, n_kv_heads,
+          block_size, max_blocks, scale, q_stride, fa, n_split, part_out,
+          part_ml, stream);
   };
   using I8 = std::integral_constant<int, 8>;
   using I16 = std::integral_constant<int, 16>;
@@ -634,27 +645,27 @@ void launch_decode_attn_impl(const uint16_t* q, const uint16_t* k_cache,
   }
 }
 
-void launch_decode_attn(const uint16_t* q, const uint16_t* k_cache,
-                        const uint16_t* v_cache, const int32_t* block_table,
+void launch_decode_attn(const uint16_t* q, const void* k_cache,
+                        const void* v_cache, const int32_t* block_table,
                         const int32_t* seq_lens, uint16_t* out, int batch,
                         int n_q_heads, int n_kv_heads, int head_dim,
                         int block_size, int max_blocks, float scale,
                         int q_stride, int n_split, float* part_out,
-                        float* part_ml, hipStream_t stream) {
+                        float* part_ml, bool cache_fp8, hipStream_t stream) {
   launch_decode_attn_impl(q, k_cache, v_cache, block_table, seq_lens, out,
                           batch, n_q_heads, n_kv_heads, head_dim, block_size,
                           max_blocks, scale, q_stride, nullptr,
-                          n_split, part_out, part_ml, stream);
+                          n_split, part_out, part_ml, cache_fp8, stream);
 }
 
 void launch_decode_attn_fused(
-    const uint16_t* qkv, uint16_t* k_cache, uint16_t* v_cache,
+    const uint16_t* qkv, void* k_cache, void* v_cache,
     const int32_t* block_table, const int32_t* seq_lens,
     const int32_t* positions, const float* cos_sin,
     const int32_t* slot_mapping, uint16_t* out, int batch, int n_q_heads,
     int n_kv_heads, int head_dim, int block_size, int max_blocks,
     float scale, int qkv_stride, int n_split, float* part_out,
-    float* part_ml, hipStream_t stream) {
+    float* part_ml, bool cache_fp8, hipStream_t stream) {
   FusedArgs fa;
   fa.k_src = qkv + n_q_heads * head_dim;
   fa.v_src = qkv + (n_q_heads + n_kv_heads) * head_dim;
@@ -666,7 +677,7 @@ void launch_decode_attn_fused(
   launch_decode_attn_impl(qkv, k_cache, v_cache, block_table, seq_lens, out,
                           batch, n_q_heads, n_kv_heads, head_dim, block_size,
                           max_blocks, scale, qkv_stride, &fa,
-                          n_split, part_out, part_ml, stream);
+                          n_split, part_out, part_ml, cache_fp8, stream);
 }
 
 }  // namespace rlli

@@ -43,6 +43,20 @@ void check_bf16_contig(const Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be on the GPU");
 }
 
+
+// cache tensors: bf16 or fp8-e4m3 (common.h codecs); returns the fp8
+// flag after validating dtype/contiguity/device
+bool check_cache(const Tensor& k_cache, const Tensor& v_cache) {
+  TORCH_CHECK(k_cache.scalar_type() == v_cache.scalar_type(),
+              "k/v cache dtype mismatch");
+  const bool fp8 = k_cache.scalar_type() == at::kFloat8_e4m3fn;
+  TORCH_CHECK(fp8 || k_cache.scalar_type() == at::kBFloat16,
+              "kv cache must be bf16 or float8_e4m3fn");
+  TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous() &&
+              k_cache.is_cuda() && v_cache.is_cuda());
+  return fp8;
+}
+
 // ------------------------------------------------------------- rmsnorm
 Tensor rmsnorm(const Tensor& x, const Tensor& w, double eps) {
   check_bf16_contig(x, "x");
@@ -101,8 +115,7 @@ void rope_kv_append_(Tensor q, Tensor k, const Tensor& v,
   check_bf16_contig(q, "q");
   check_bf16_contig(k, "k");
   check_bf16_contig(v, "v");
-  check_bf16_contig(k_cache, "k_cache");
-  check_bf16_contig(v_cache, "v_cache");
+  const bool cache_fp8 = check_cache(k_cache, v_cache);
   TORCH_CHECK(positions.scalar_type() == at::kInt && positions.is_contiguous());
   TORCH_CHECK(slot_mapping.scalar_type() == at::kInt && slot_mapping.is_contiguous());
   TORCH_CHECK(cos_sin.scalar_type() == at::kFloat && cos_sin.is_contiguous());
@@ -120,8 +133,10 @@ void rope_kv_append_(Tensor q, Tensor k, const Tensor& v,
   rlli::launch_rope_kv_append(
       bf16_ptr(q), bf16_ptr(k), bf16_ptr(v),
       positions.data_ptr<int32_t>(), cos_sin.data_ptr<float>(),
-      bf16_ptr(k_cache), bf16_ptr(v_cache), slot_mapping.data_ptr<int32_t>(),
-      tokens, n_q, n_kv, D, block_size, n_q * D, n_kv * D, current_stream(q));
+      k_cache.data_ptr(), v_cache.data_ptr(),
+      slot_mapping.data_ptr<int32_t>(),
+      tokens, n_q, n_kv, D, block_size, n_q * D, n_kv * D, cache_fp8,
+      current_stream(q));
   check_launch("rope_kv_append_");
 }
 
@@ -132,8 +147,7 @@ void rope_kv_append_qkv_(Tensor qkv, const Tensor& positions,
                          Tensor v_cache, const Tensor& slot_mapping,
                          int64_t n_q) {
   check_bf16_contig(qkv, "qkv");
-  check_bf16_contig(k_cache, "k_cache");
-  check_bf16_contig(v_cache, "v_cache");
+  const bool cache_fp8 = check_cache(k_cache, v_cache);
   const int n_kv = int(k_cache.size(1));
   const int D = int(k_cache.size(3));
   const int block_size = int(k_cache.size(2));
@@ -147,8 +161,9 @@ void rope_kv_append_qkv_(Tensor qkv, const Tensor& positions,
   rlli::launch_rope_kv_append(
       base, base + n_q * D, base + (n_q + n_kv) * D,
       positions.data_ptr<int32_t>(), cos_sin.data_ptr<float>(),
-      bf16_ptr(k_cache), bf16_ptr(v_cache), slot_mapping.data_ptr<int32_t>(),
-      tokens, int(n_q), n_kv, D, block_size, stride, stride,
+      k_cache.data_ptr(), v_cache.data_ptr(),
+      slot_mapping.data_ptr<int32_t>(),
+      tokens, int(n_q), n_kv, D, block_size, stride, stride, cache_fp8,
       current_stream(qkv));
   check_launch("rope_kv_append_qkv_");
 }
@@ -183,8 +198,7 @@ Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
                    const Tensor& v_cache, const Tensor& block_table,
                    const Tensor& seq_lens, double scale) {
   check_bf16_contig(q, "q");
-  check_bf16_contig(k_cache, "k_cache");
-  check_bf16_contig(v_cache, "v_cache");
+  const bool cache_fp8 = check_cache(k_cache, v_cache);
   TORCH_CHECK(block_table.scalar_type() == at::kInt && block_table.is_contiguous());
   TORCH_CHECK(seq_lens.scalar_type() == at::kInt && seq_lens.is_contiguous());
   const int batch = int(q.size(0));
@@ -205,11 +219,11 @@ Tensor decode_attn(const Tensor& q, const Tensor& k_cache,
   c10::hip::HIPGuardMasqueradingAsCUDA guard(q.device());
   SplitScratch ss = make_split_scratch(q, batch, n_kv, group, D);
   rlli::launch_decode_attn(
-      bf16_ptr(q), bf16_ptr(k_cache), bf16_ptr(v_cache),
+      bf16_ptr(q), k_cache.data_ptr(), v_cache.data_ptr(),
       block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       bf16_ptr(out), batch, n_q, n_kv, D, block_size, max_blocks,
       float(scale), n_q * D, ss.n_split, ss.part_ptr(), ss.ml_ptr(),
-      current_stream(q));
+      cache_fp8, current_stream(q));
   check_launch("decode_attn");
   return out;
 }
@@ -219,6 +233,7 @@ Tensor decode_attn_qkv(const Tensor& qkv, const Tensor& k_cache,
                        const Tensor& v_cache, const Tensor& block_table,
                        const Tensor& seq_lens, double scale, int64_t n_q) {
   check_bf16_contig(qkv, "qkv");
+  const bool cache_fp8 = check_cache(k_cache, v_cache);
   const int batch = int(qkv.size(0));
   const int n_kv = int(k_cache.size(1));
   const int D = int(k_cache.size(3));
@@ -233,11 +248,11 @@ Tensor decode_attn_qkv(const Tensor& qkv, const Tensor& k_cache,
   c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
   SplitScratch ss = make_split_scratch(qkv, batch, n_kv, group, D);
   rlli::launch_decode_attn(
-      bf16_ptr(qkv), bf16_ptr(k_cache), bf16_ptr(v_cache),
+      bf16_ptr(qkv), k_cache.data_ptr(), v_cache.data_ptr(),
       block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       bf16_ptr(out), batch, int(n_q), n_kv, D, block_size, max_blocks,
       float(scale), stride, ss.n_split, ss.part_ptr(), ss.ml_ptr(),
-      current_stream(qkv));
+      cache_fp8, current_stream(qkv));
   check_launch("decode_attn_qkv");
   return out;
 }
@@ -249,8 +264,7 @@ Tensor decode_attn_rope_qkv(const Tensor& qkv, const Tensor& positions,
                             const Tensor& block_table, const Tensor& seq_lens,
                             double scale, int64_t n_q) {
   check_bf16_contig(qkv, "qkv");
-  check_bf16_contig(k_cache, "k_cache");
-  check_bf16_contig(v_cache, "v_cache");
+  const bool cache_fp8 = check_cache(k_cache, v_cache);
   TORCH_CHECK(positions.scalar_type() == at::kInt && positions.is_contiguous());
   TORCH_CHECK(slot_mapping.scalar_type() == at::kInt && slot_mapping.is_contiguous());
   TORCH_CHECK(cos_sin.scalar_type() == at::kFloat && cos_sin.is_contiguous());
@@ -271,12 +285,13 @@ Tensor decode_attn_rope_qkv(const Tensor& qkv, const Tensor& positions,
   c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
   SplitScratch ss = make_split_scratch(qkv, batch, n_kv, group, D);
   rlli::launch_decode_attn_fused(
-      bf16_ptr(qkv), bf16_ptr(k_cache), bf16_ptr(v_cache),
+      bf16_ptr(qkv), k_cache.data_ptr(), v_cache.data_ptr(),
       block_table.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
       positions.data_ptr<int32_t>(), cos_sin.data_ptr<float>(),
       slot_mapping.data_ptr<int32_t>(), bf16_ptr(out), batch, int(n_q),
       n_kv, D, block_size, max_blocks, float(scale), stride,
-      ss.n_split, ss.part_ptr(), ss.ml_ptr(), current_stream(qkv));
+      ss.n_split, ss.part_ptr(), ss.ml_ptr(), cache_fp8,
+      current_stream(qkv));
   check_launch("decode_attn_rope_qkv");
   return out;
 }
@@ -356,8 +371,7 @@ Tensor prefill_paged_attn(const Tensor& qkv, const Tensor& k_cache,
                           const Tensor& block_tables, double scale,
                           int64_t n_q) {
   check_bf16_contig(qkv, "qkv");
-  check_bf16_contig(k_cache, "k_cache");
-  check_bf16_contig(v_cache, "v_cache");
+  const bool cache_fp8 = check_cache(k_cache, v_cache);
   for (const Tensor* t : {&chunk_row0, &chunk_pos0, &chunk_nrows,
                           &chunk_btrow, &block_tables}) {
     TORCH_CHECK(t->scalar_type() == at::kInt && t->is_contiguous() &&
@@ -378,11 +392,11 @@ Tensor prefill_paged_attn(const Tensor& qkv, const Tensor& k_cache,
   Tensor out = at::empty({T, n_q * D}, qkv.options());
   c10::hip::HIPGuardMasqueradingAsCUDA guard(qkv.device());
   rlli::launch_prefill_paged(
-      bf16_ptr(qkv), bf16_ptr(k_cache), bf16_ptr(v_cache),
+      bf16_ptr(qkv), k_cache.data_ptr(), v_cache.data_ptr(),
       chunk_row0.data_ptr<int32_t>(), chunk_pos0.data_ptr<int32_t>(),
       chunk_nrows.data_ptr<int32_t>(), chunk_btrow.data_ptr<int32_t>(),
       block_tables.data_ptr<int32_t>(), bf16_ptr(out), n_chunks, n_kv,
-      group, D, stride, max_blocks, block_size, float(scale),
+      group, D, stride, max_blocks, block_size, float(scale), cache_fp8,
       current_stream(qkv));
   check_launch("prefill_paged_attn");
   return out;

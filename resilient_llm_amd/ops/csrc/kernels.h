@@ -22,13 +22,16 @@ void launch_silu_mul(const uint16_t* gate_up, uint16_t* out, int rows,
 // k/v into the paged cache at slot_mapping[t].
 // q/k/v may be strided slices of one fused [T, (n_q+2*n_kv)*D] qkv
 // tensor: *_stride is the per-token element stride of each view.
+// Cache pointers are void*: the paged cache stores bf16 (default) or
+// OCP fp8-e4m3 (cache_fp8) — see common.h cache codecs.
 void launch_rope_kv_append(
     uint16_t* q, uint16_t* k, uint16_t* v,
     const int32_t* positions, const float* cos_sin,   // [max_pos, head_dim]
-    uint16_t* k_cache, uint16_t* v_cache,
+    void* k_cache, void* v_cache,
     const int32_t* slot_mapping,
     int tokens, int n_q_heads, int n_kv_heads, int head_dim,
-    int block_size, int q_stride, int kv_stride, hipStream_t stream);
+    int block_size, int q_stride, int kv_stride, bool cache_fp8,
+    hipStream_t stream);
 
 // Split-K factor for decode attention (flash-decode): >1 when
 // batch*n_kv_heads workgroups cannot fill 256 CUs.  The caller sizes
@@ -40,25 +43,26 @@ int decode_attn_n_split(int batch, int n_kv_heads);
 // Paged GQA decode attention: one new q token per sequence.
 void launch_decode_attn(
     const uint16_t* q,                 // [batch, n_q_heads, head_dim]
-    const uint16_t* k_cache,           // [blocks, n_kv, block_size, head_dim]
-    const uint16_t* v_cache,
+    const void* k_cache,               // [blocks, n_kv, block_size, head_dim]
+    const void* v_cache,
     const int32_t* block_table,        // [batch, max_blocks]
     const int32_t* seq_lens,           // [batch]
     uint16_t* out,                     // [batch, n_q_heads, head_dim]
     int batch, int n_q_heads, int n_kv_heads, int head_dim,
     int block_size, int max_blocks, float scale, int q_stride,
-    int n_split, float* part_out, float* part_ml, hipStream_t stream);
+    int n_split, float* part_out, float* part_ml, bool cache_fp8,
+    hipStream_t stream);
 
 // Fused decode: rope(q,k) + cache append + paged attention in ONE
 // kernel (decode path; qkv is the raw fused GEMM output).
 void launch_decode_attn_fused(
-    const uint16_t* qkv, uint16_t* k_cache, uint16_t* v_cache,
+    const uint16_t* qkv, void* k_cache, void* v_cache,
     const int32_t* block_table, const int32_t* seq_lens,
     const int32_t* positions, const float* cos_sin,
     const int32_t* slot_mapping, uint16_t* out, int batch, int n_q_heads,
     int n_kv_heads, int head_dim, int block_size, int max_blocks,
     float scale, int qkv_stride, int n_split, float* part_out,
-    float* part_ml, hipStream_t stream);
+    float* part_ml, bool cache_fp8, hipStream_t stream);
 
 // Varlen causal prefill attention over in-batch q/k/v.
 void launch_prefill_attn(
@@ -78,15 +82,16 @@ void launch_prefill_mfma(const uint16_t* qkv, const int32_t* chunk_t0,
                          hipStream_t stream);
 
 // MFMA chunked-prefill attention over the paged cache (mixed batches).
-void launch_prefill_paged(const uint16_t* qkv, const uint16_t* k_cache,
-                          const uint16_t* v_cache, const int32_t* chunk_row0,
+void launch_prefill_paged(const uint16_t* qkv, const void* k_cache,
+                          const void* v_cache, const int32_t* chunk_row0,
                           const int32_t* chunk_pos0,
                           const int32_t* chunk_nrows,
                           const int32_t* chunk_btrow,
                           const int32_t* block_tables, uint16_t* out,
                           int n_chunks, int n_kv_heads, int group,
                           int head_dim, int qkv_stride, int max_blocks,
-                          int block_size, float scale, hipStream_t stream);
+                          int block_size, float scale, bool cache_fp8,
+                          hipStream_t stream);
 
 // Skinny-M GEMM (decode projections): out[M,N] = x[M,K] @ W[N,K]^T.
 // ws is a [splitk, M, N] f32 workspace (unused when splitk == 1).

@@ -16,6 +16,8 @@
 
 #include "common.h"
 
+#include <type_traits>
+
 namespace rlli {
 
 namespace {
@@ -28,11 +30,11 @@ constexpr int QBLK = 32;
 constexpr int KVBLK = 64;
 constexpr int PPAD = KVBLK + 8;
 
-template <int D>
+template <int D, typename KC = CacheBF16>
 __global__ __launch_bounds__(256, 2)
 void prefill_paged_kernel(const uint16_t* __restrict__ qkv,
-                          const uint16_t* __restrict__ k_cache,
-                          const uint16_t* __restrict__ v_cache,
+                          const typename KC::elem* __restrict__ k_cache,
+                          const typename KC::elem* __restrict__ v_cache,
                           const int32_t* __restrict__ chunk_row0,
                           const int32_t* __restrict__ chunk_pos0,
                           const int32_t* __restrict__ chunk_nrows,
@@ -115,21 +117,28 @@ void prefill_paged_kernel(const uint16_t* __restrict__ qkv,
         const int kv = idx / (D / 8);
         const int d8 = (idx % (D / 8)) * 8;
         const int p = kv0 + kv;
-        uint4 raw = {0, 0, 0, 0};
-        uint4 kraw = {0, 0, 0, 0};
+        // stage through the cache codec: fp8 converts to bf16 HERE, so
+        // the LDS images and every MFMA fragment below are unchanged
+        typename KC::vec8 raw = {};
+        typename KC::vec8 kraw = {};
         if (p < kv_hi) {
           const int64_t a = kv_addr(p);
-          raw = *reinterpret_cast<const uint4*>(v_cache + a + d8);
-          kraw = *reinterpret_cast<const uint4*>(k_cache + a + d8);
+          raw = *reinterpret_cast<const typename KC::vec8*>(v_cache + a + d8);
+          kraw = *reinterpret_cast<const typename KC::vec8*>(k_cache + a + d8);
         }
-        bf16x8 piece;
-        piece.u = raw;
+        float vf8[8], kf8[8];
+        KC::to_f32(raw, vf8);
+        KC::to_f32(kraw, kf8);
+        bf16x8 piece, kpiece;
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
+        for (int j = 0; j < 8; ++j) {
+          piece.s[j] = f32_to_bf16(vf8[j]);
+          kpiece.s[j] = f32_to_bf16(kf8[j]);
           vt[(d8 + j) * PPAD + kv] = piece.s[j];
+        }
         *reinterpret_cast<uint4*>(
             reinterpret_cast<char*>(ks) +
-            (kv * D * 2 + ((d8 * 2) ^ ((kv & (D / 8 - 1)) << 4)))) = kraw;
+            (kv * D * 2 + ((d8 * 2) ^ ((kv & (D / 8 - 1)) << 4)))) = kpiece.u;
       }
     }
     __syncthreads();
@@ -241,32 +250,37 @@ void prefill_paged_kernel(const uint16_t* __restrict__ qkv,
 
 }  // namespace
 
-void launch_prefill_paged(const uint16_t* qkv, const uint16_t* k_cache,
-                          const uint16_t* v_cache, const int32_t* chunk_row0,
+void launch_prefill_paged(const uint16_t* qkv, const void* k_cache,
+                          const void* v_cache, const int32_t* chunk_row0,
                           const int32_t* chunk_pos0,
                           const int32_t* chunk_nrows,
                           const int32_t* chunk_btrow,
                           const int32_t* block_tables, uint16_t* out,
                           int n_chunks, int n_kv_heads, int group,
                           int head_dim, int qkv_stride, int max_blocks,
-                          int block_size, float scale, hipStream_t stream) {
+                          int block_size, float scale, bool cache_fp8,
+                          hipStream_t stream) {
   const int n_hw = group > 4 ? group / 4 : 1;
   const int blocks = n_chunks * n_kv_heads * n_hw;
   const size_t smem = size_t(head_dim) * PPAD * 2 +
                       size_t(KVBLK) * head_dim * 2 +
                       size_t(4) * QBLK * PPAD * 2;
+  auto go = [&](auto d_tag, auto kc_tag) {
+    using KC = decltype(kc_tag);
+    hipLaunchKernelGGL((prefill_paged_kernel<decltype(d_tag)::value, KC>),
+                       dim3(blocks), dim3(256), smem, stream, qkv,
+                       static_cast<const typename KC::elem*>(k_cache),
+                       static_cast<const typename KC::elem*>(v_cache),
+                       chunk_row0, chunk_pos0, chunk_nrows, chunk_btrow,
+                       block_tables, out, n_kv_heads, group, n_hw,
+                       qkv_stride, max_blocks, block_size, scale);
+  };
+  using D128 = std::integral_constant<int, 128>;
+  using D64 = std::integral_constant<int, 64>;
   if (head_dim == 128) {
-    hipLaunchKernelGGL(prefill_paged_kernel<128>, dim3(blocks), dim3(256),
-                       smem, stream, qkv, k_cache, v_cache, chunk_row0,
-                       chunk_pos0, chunk_nrows, chunk_btrow, block_tables,
-                       out, n_kv_heads, group, n_hw, qkv_stride, max_blocks,
-                       block_size, scale);
+    if (cache_fp8) go(D128{}, CacheFP8{}); else go(D128{}, CacheBF16{});
   } else if (head_dim == 64) {
-    hipLaunchKernelGGL(prefill_paged_kernel<64>, dim3(blocks), dim3(256),
-                       smem, stream, qkv, k_cache, v_cache, chunk_row0,
-                       chunk_pos0, chunk_nrows, chunk_btrow, block_tables,
-                       out, n_kv_heads, group, n_hw, qkv_stride, max_blocks,
-                       block_size, scale);
+    if (cache_fp8) go(D64{}, CacheFP8{}); else go(D64{}, CacheBF16{});
   }
 }
 

@@ -17,13 +17,14 @@ namespace rlli {
 
 namespace {
 
+template <typename KC>
 __global__ void rope_kv_kernel(uint16_t* __restrict__ q,
                                uint16_t* __restrict__ k,
                                uint16_t* __restrict__ v,
                                const int32_t* __restrict__ positions,
                                const float* __restrict__ cos_sin,
-                               uint16_t* __restrict__ k_cache,
-                               uint16_t* __restrict__ v_cache,
+                               typename KC::elem* __restrict__ k_cache,
+                               typename KC::elem* __restrict__ v_cache,
                                const int32_t* __restrict__ slot_mapping,
                                int n_q_heads, int n_kv_heads, int head_dim,
                                int block_size, int q_stride, int kv_stride) {
@@ -59,20 +60,24 @@ __global__ void rope_kv_kernel(uint16_t* __restrict__ q,
     float4 s23 = *reinterpret_cast<const float4*>(cs_row + half + p8 + 4);
     const float cv[8] = {c01.x, c01.y, c01.z, c01.w, c23.x, c23.y, c23.z, c23.w};
     const float sv[8] = {s01.x, s01.y, s01.z, s01.w, s23.x, s23.y, s23.z, s23.w};
+    float rlo[8], rhi[8];
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
       const float x1 = bf16_to_f32(lo.s[i]);
       const float x2 = bf16_to_f32(hi.s[i]);
-      olo.s[i] = f32_to_bf16(x1 * cv[i] - x2 * sv[i]);
-      ohi.s[i] = f32_to_bf16(x2 * cv[i] + x1 * sv[i]);
+      rlo[i] = x1 * cv[i] - x2 * sv[i];
+      rhi[i] = x2 * cv[i] + x1 * sv[i];
+      olo.s[i] = f32_to_bf16(rlo[i]);
+      ohi.s[i] = f32_to_bf16(rhi[i]);
     }
     *reinterpret_cast<uint4*>(base + p8) = olo.u;
     *reinterpret_cast<uint4*>(base + half + p8) = ohi.u;
     if (!is_q && slot >= 0) {
-      uint16_t* dst = k_cache + cache_base +
+      typename KC::elem* dst = k_cache + cache_base +
                       int64_t(h - n_q_heads) * block_size * head_dim;
-      *reinterpret_cast<uint4*>(dst + p8) = olo.u;
-      *reinterpret_cast<uint4*>(dst + half + p8) = ohi.u;
+      *reinterpret_cast<typename KC::vec8*>(dst + p8) = KC::from_f32(rlo);
+      *reinterpret_cast<typename KC::vec8*>(dst + half + p8) =
+          KC::from_f32(rhi);
     }
   }
 
@@ -82,11 +87,15 @@ __global__ void rope_kv_kernel(uint16_t* __restrict__ q,
     for (int w = threadIdx.x; w < total_v; w += blockDim.x) {
       const int h = w / (head_dim / 8);
       const int d8 = (w % (head_dim / 8)) * 8;
-      const uint4 vv = *reinterpret_cast<const uint4*>(
+      bf16x8 vv;
+      vv.u = *reinterpret_cast<const uint4*>(
           v + int64_t(t) * kv_stride + int64_t(h) * head_dim + d8);
-      uint16_t* dst = v_cache + cache_base +
+      float vf[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) vf[i] = bf16_to_f32(vv.s[i]);
+      typename KC::elem* dst = v_cache + cache_base +
                       int64_t(h) * block_size * head_dim + d8;
-      *reinterpret_cast<uint4*>(dst) = vv;
+      *reinterpret_cast<typename KC::vec8*>(dst) = KC::from_f32(vf);
     }
   }
 }
@@ -96,16 +105,27 @@ __global__ void rope_kv_kernel(uint16_t* __restrict__ q,
 void launch_rope_kv_append(
     uint16_t* q, uint16_t* k, uint16_t* v,
     const int32_t* positions, const float* cos_sin,
-    uint16_t* k_cache, uint16_t* v_cache,
+    void* k_cache, void* v_cache,
     const int32_t* slot_mapping,
     int tokens, int n_q_heads, int n_kv_heads, int head_dim,
-    int block_size, int q_stride, int kv_stride, hipStream_t stream) {
+    int block_size, int q_stride, int kv_stride, bool cache_fp8,
+    hipStream_t stream) {
   if (tokens == 0) return;
   const int threads = 256;
-  hipLaunchKernelGGL(rope_kv_kernel, dim3(tokens), dim3(threads), 0, stream,
-                     q, k, v, positions, cos_sin, k_cache, v_cache,
-                     slot_mapping, n_q_heads, n_kv_heads, head_dim, block_size,
-                     q_stride, kv_stride);
+  if (cache_fp8)
+    hipLaunchKernelGGL(rope_kv_kernel<CacheFP8>, dim3(tokens), dim3(threads),
+                       0, stream, q, k, v, positions, cos_sin,
+                       static_cast<CacheFP8::elem*>(k_cache),
+                       static_cast<CacheFP8::elem*>(v_cache),
+                       slot_mapping, n_q_heads, n_kv_heads, head_dim,
+                       block_size, q_stride, kv_stride);
+  else
+    hipLaunchKernelGGL(rope_kv_kernel<CacheBF16>, dim3(tokens), dim3(threads),
+                       0, stream, q, k, v, positions, cos_sin,
+                       static_cast<CacheBF16::elem*>(k_cache),
+                       static_cast<CacheBF16::elem*>(v_cache),
+                       slot_mapping, n_q_heads, n_kv_heads, head_dim,
+                       block_size, q_stride, kv_stride);
 }
 
 }  // namespace rlli

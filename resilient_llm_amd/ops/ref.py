@@ -70,8 +70,10 @@ def rope_kv_append_(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     blocks = idx // block_size
     rows = idx % block_size
     # cache: [num_blocks, n_kv, block_size, D]; k/v: [T, n_kv, D]
-    k_cache[blocks, :, rows] = k[valid]
-    v_cache[blocks, :, rows] = v[valid]
+    # cast through the cache dtype (bf16 identity; fp8-e4m3 quantizes —
+    # mirrors the GPU codec's cvt_pk_fp8 RNE conversion)
+    k_cache[blocks, :, rows] = k[valid].to(k_cache.dtype)
+    v_cache[blocks, :, rows] = v[valid].to(v_cache.dtype)
 
 
 def decode_attn(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,

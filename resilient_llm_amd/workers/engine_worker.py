@@ -77,6 +77,7 @@ class EngineWorker(Worker):
                  target_step_ms: Optional[float] = None,
                  weights: Optional[str] = None,
                  eos_id=None,
+                 kv_dtype: str = "bf16",
                  seed: int = 0) -> None:
         super().__init__(device=device_label or f"gpu:{device}",
                          models={model_name})
@@ -105,10 +106,22 @@ class EngineWorker(Worker):
         else:
             eos_ids = list(getattr(config, "eos_ids", (2,)))
         self.eos_ids = frozenset(eos_ids)
-        nb = num_blocks or default_num_blocks(config, kv_gb, tp_world=tp_world)
+        # opt-in fp8-e4m3 KV cache: half the bytes per token -> double
+        # the resident KV capacity of the same budget (and half the
+        # decode-attention HBM traffic); attention math stays f32 via
+        # the gfx950 cvt_pk conversions (ops/csrc/common.h codecs)
+        if kv_dtype not in ("bf16", "fp8"):
+            raise ValueError(f"kv_dtype must be bf16 or fp8, got {kv_dtype!r}")
+        use_fp8 = kv_dtype == "fp8"
+        kv_torch_dtype = torch.float8_e4m3fn if use_fp8 else torch.bfloat16
+        elem_bytes = 1 if use_fp8 else 2
+        nb = num_blocks or (default_num_blocks(config, kv_gb,
+                                               tp_world=tp_world)
+                            * 2 // elem_bytes // 1)
         kv = PagedKVCache.for_model(config, nb, device=torch_device,
-                                    tp_world=tp_world)
-        if dtype != torch.bfloat16:
+                                    tp_world=tp_world,
+                                    kv_dtype=kv_torch_dtype)
+        if dtype != torch.bfloat16 and not use_fp8:
             kv.k = kv.k.to(dtype)
             kv.v = kv.v.to(dtype)
         self.engine = LLMEngine(self.model, kv, max_batch_size=max_batch_size,

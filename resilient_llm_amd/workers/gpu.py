@@ -33,6 +33,7 @@ def spawn_gpu_worker(device_index: str, model_name: str,
                      target_step_ms: float | None = None,
                      weights: str | None = None,
                      eos_ids: list | None = None,
+                     kv_dtype: str | None = None,
                      extra_env: dict | None = None) -> subprocess.Popen:
     """``device_index`` may be a sub-device replica like ``0.1``: several
     worker processes co-located on physical GPU 0 — 288 GB of HBM3E holds
@@ -57,6 +58,8 @@ def spawn_gpu_worker(device_index: str, model_name: str,
         cmd += ["--weights", str(weights)]
     for e in eos_ids or []:
         cmd += ["--eos-id", str(e)]
+    if kv_dtype:
+        cmd += ["--kv-dtype", str(kv_dtype)]
     return subprocess.Popen(cmd, env=env)
 
 
@@ -105,13 +108,14 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
         slo = config.cluster.target_step_ms
         extra = opts.get(target, {})
         weights = extra.get("weights")
+        kv_dtype = extra.get("kv_dtype")
         eos = extra.get("eos_id")
         eos_list = None if eos is None else (
             [int(eos)] if isinstance(eos, int) else [int(e) for e in eos])
         proc = spawn_gpu_worker(target, model_name, sock, kv_gb=kv_gb,
                                 max_batch=max_batch, use_graphs=use_graphs,
                                 weights=weights, eos_ids=eos_list,
-                                target_step_ms=slo)
+                                kv_dtype=kv_dtype, target_step_ms=slo)
         client = RpcWorkerClient(f"gpu:{target}", {model_name}, sock)
         client.proc = proc
         client.respawn = (lambda t=target, m=model_name, s=sock:
@@ -119,6 +123,7 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
                                            max_batch=max_batch,
                                            use_graphs=use_graphs,
                                            weights=weights, eos_ids=eos_list,
+                                           kv_dtype=kv_dtype,
                                            target_step_ms=slo))
         registry.register("gpu", target, client)
 
@@ -129,14 +134,18 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
         log_with_timestamp(
             f"spawning TP={pool_def.tensor_parallel} pool worker "
             f"pool:{pool_name} on GPUs {pool_def.gpus} ({model_name})", "grey")
-        weights = opts.get(f"pool:{pool_name}", {}).get("weights")
+        pool_opts = opts.get(f"pool:{pool_name}", {})
+        weights = pool_opts.get("weights")
+        pool_kvd = pool_opts.get("kv_dtype")
         procs = spawn_pool_worker(pool_def, model_name, sock, kv_gb=kv_gb,
-                                  max_batch=max_batch, weights=weights)
+                                  max_batch=max_batch, weights=weights,
+                                  kv_dtype=pool_kvd)
         client = RpcWorkerClient(f"pool:{pool_name}", {model_name}, sock)
         client.proc = procs[0]
         client.proc_group = procs
         client.respawn = (lambda pd=pool_def, m=model_name, s=sock:
                           spawn_pool_worker(pd, m, s, kv_gb=kv_gb,
                                             max_batch=max_batch,
-                                            weights=weights))
+                                            weights=weights,
+                                            kv_dtype=pool_kvd))
         registry.register("pool", pool_name, client)

@@ -26,6 +26,9 @@ def main() -> None:
                          "when absent")
     ap.add_argument("--eos-id", type=int, action="append", default=None,
                     help="override EOS token id(s); repeatable")
+    ap.add_argument("--kv-dtype", default="bf16", choices=("bf16", "fp8"),
+                    help="paged-KV-cache element type: fp8-e4m3 halves "
+                         "bytes/token (2x KV capacity per GiB)")
     ap.add_argument("--device", default=None,
                     help="torch device override (tests: cpu)")
     args = ap.parse_args()
@@ -56,6 +59,7 @@ def main() -> None:
                               use_graphs=args.graphs,
                               weights=args.weights,
                               eos_id=args.eos_id,
+                              kv_dtype=args.kv_dtype,
                               target_step_ms=args.target_step_ms)
         log_with_timestamp(
             f"worker {args.device_label} ready: {args.model} on {device}, "

@@ -84,6 +84,7 @@ def _free_port() -> int:
 
 def spawn_pool_worker(pool: PoolDef, model_name: str, socket_path: str,
                       kv_gb: float = 24.0, max_batch: int = 64,
+                      kv_dtype: Optional[str] = None,
                       device_override: Optional[str] = None,
                       weights: Optional[str] = None,
                       tp_backend: str = "nccl") -> list:
@@ -112,5 +113,7 @@ def spawn_pool_worker(pool: PoolDef, model_name: str, socket_path: str,
             cmd += ["--device", device_override]
         if weights:
             cmd += ["--weights", str(weights)]
+        if kv_dtype:
+            cmd += ["--kv-dtype", str(kv_dtype)]
         procs.append(subprocess.Popen(cmd, env=env))
     return procs

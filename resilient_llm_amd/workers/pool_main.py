@@ -22,6 +22,7 @@ def main() -> None:
     ap.add_argument("--tp-backend", default="nccl")
     ap.add_argument("--weights", default=None,
                     help="safetensors dir — each rank loads its TP shard")
+    ap.add_argument("--kv-dtype", default="bf16", choices=("bf16", "fp8"))
     ap.add_argument("--device", default=None, help="override (tests: cpu)")
     args = ap.parse_args()
 
@@ -51,6 +52,7 @@ def main() -> None:
         device=device, model_name=args.model,
         device_label=f"pool:{args.pool_name}",
         kv_gb=args.kv_gb, max_batch_size=args.max_batch,
+        kv_dtype=args.kv_dtype,
         num_blocks=args.num_blocks, weights=args.weights,
         tp_rank=args.rank, tp_world=args.world, tp_group=tp_group,
         tp_control=control)
