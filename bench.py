@@ -89,7 +89,7 @@ class LoopThread:
 
 def start_worker(loopth: LoopThread, device: str, label: str, sock: str,
                  model: str, max_batch: int, num_blocks: int,
-                 use_graphs: bool):
+                 use_graphs: bool, kv_dtype: str = "bf16"):
     from resilient_llm_amd.workers.engine_worker import EngineWorker
     from resilient_llm_amd.workers.rpc import WorkerRpcServer
 
@@ -99,6 +99,7 @@ def start_worker(loopth: LoopThread, device: str, label: str, sock: str,
                               max_batch_size=max_batch,
                               max_queue=max_batch * 4,
                               num_blocks=num_blocks,
+                              kv_dtype=kv_dtype,
                               use_graphs=use_graphs)
         server = WorkerRpcServer(worker, sock)
         await server.start()
@@ -179,6 +180,9 @@ def main() -> None:
     ap.add_argument("--prompt-tokens", type=int, default=PROMPT_TOKENS)
     ap.add_argument("--output-tokens", type=int, default=OUTPUT_TOKENS)
     ap.add_argument("--no-graphs", action="store_true")
+    ap.add_argument("--kv-dtype", default="bf16", choices=("bf16", "fp8"),
+                    help="paged-KV element type (fp8 = opt-in e4m3 cache; "
+                         "the headline metric stays bf16)")
     ap.add_argument("--device", default=None,
                     help="override torch device (tests: cpu)")
     args = ap.parse_args()
@@ -240,7 +244,8 @@ def main() -> None:
                           os.path.join(run_dir, f"w{rank}.sock"),
                           args.model, max_batch=args.concurrency,
                           num_blocks=num_blocks,
-                          use_graphs=on_gpu and not args.no_graphs)
+                          use_graphs=on_gpu and not args.no_graphs,
+                          kv_dtype=args.kv_dtype)
     log(f"rank {rank}: worker up on {device} in {time.time() - t_init:.1f}s "
         f"({num_blocks} KV blocks)")
     sync("post-worker-startup barrier")

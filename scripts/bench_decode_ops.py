@@ -60,6 +60,8 @@ def main():
     ap.add_argument("--len", type=int, default=192, dest="seqlen")
     ap.add_argument("--iters", type=int, default=200)
     ap.add_argument("--only", default="", help="run just one op (attn/norm/silu/sample)")
+    ap.add_argument("--kv-fp8", action="store_true",
+                    help="fp8-e4m3 KV cache for the attention op")
     args = ap.parse_args()
 
     dev = "cuda:0"
@@ -101,6 +103,9 @@ def main():
         kc = torch.randn(total_blocks, n_kv, bs, D, device=dev,
                          dtype=torch.bfloat16)
         vc = torch.randn_like(kc)
+        if args.kv_fp8:
+            kc = kc.to(torch.float8_e4m3fn)
+            vc = vc.to(torch.float8_e4m3fn)
         bt = torch.arange(B * n_blocks, device=dev,
                           dtype=torch.int32).reshape(B, n_blocks)
         seq_lens = torch.full((B,), L, device=dev, dtype=torch.int32)
@@ -117,7 +122,8 @@ def main():
         e, g = time_op(lambda: ops.decode_attn_rope_qkv(
             qkv, positions, cos_sin, kc, vc, slots, bt, seq_lens, scale, n_q),
             args.iters)
-        report("decode_attn_rope_qkv", B * L * n_kv * D * 2 * 2 / 1e6, e, g)
+        ebytes = 1 if args.kv_fp8 else 2
+        report("decode_attn_rope_qkv", B * L * n_kv * D * ebytes * 2 / 1e6, e, g)
 
     # --- sample ---
     if only in ("", "sample"):
