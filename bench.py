@@ -89,7 +89,8 @@ class LoopThread:
 
 def start_worker(loopth: LoopThread, device: str, label: str, sock: str,
                  model: str, max_batch: int, num_blocks: int,
-                 use_graphs: bool, kv_dtype: str = "bf16"):
+                 use_graphs: bool, kv_dtype: str = "bf16",
+                 quant=None):
     from resilient_llm_amd.workers.engine_worker import EngineWorker
     from resilient_llm_amd.workers.rpc import WorkerRpcServer
 
@@ -100,6 +101,7 @@ def start_worker(loopth: LoopThread, device: str, label: str, sock: str,
                               max_queue=max_batch * 4,
                               num_blocks=num_blocks,
                               kv_dtype=kv_dtype,
+                              quant=quant,
                               use_graphs=use_graphs)
         server = WorkerRpcServer(worker, sock)
         await server.start()
@@ -183,6 +185,9 @@ def main() -> None:
     ap.add_argument("--kv-dtype", default="bf16", choices=("bf16", "fp8"),
                     help="paged-KV element type (fp8 = opt-in e4m3 cache; "
                          "the headline metric stays bf16)")
+    ap.add_argument("--quant", default=None, choices=("fp8",),
+                    help="W8A8-fp8 weight quantization (opt-in mode; the "
+                         "headline metric stays bf16)")
     ap.add_argument("--device", default=None,
                     help="override torch device (tests: cpu)")
     args = ap.parse_args()
@@ -245,7 +250,7 @@ def main() -> None:
                           args.model, max_batch=args.concurrency,
                           num_blocks=num_blocks,
                           use_graphs=on_gpu and not args.no_graphs,
-                          kv_dtype=args.kv_dtype)
+                          kv_dtype=args.kv_dtype, quant=args.quant)
     log(f"rank {rank}: worker up on {device} in {time.time() - t_init:.1f}s "
         f"({num_blocks} KV blocks)")
     sync("post-worker-startup barrier")
@@ -327,7 +332,8 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "fp8-w8a8(gate_up/down/lm_head)+bf16"
+                     if args.quant == "fp8" else "bf16",
             "data": "synthetic",
             "config": {
                 "model": args.model,

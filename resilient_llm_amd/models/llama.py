@@ -125,10 +125,21 @@ class LlamaForCausalLM:
     def __init__(self, config: LlamaConfig, device: str = "cpu",
                  dtype: torch.dtype = torch.bfloat16,
                  tp_rank: int = 0, tp_world: int = 1,
-                 tp_group=None, seed: int = 0) -> None:
+                 tp_group=None, seed: int = 0,
+                 quant: Optional[str] = None) -> None:
         self.config = config
         self.device = torch.device(device)
         self.dtype = dtype
+        # quant="fp8": W8A8-fp8 for the three bandwidth-dominant
+        # projections (gate_up, down, lm_head) — per-out-channel weight
+        # scales + per-row dynamic activation scales emitted by the
+        # fused rmsnorm/silu kernels, rowwise torch._scaled_mm compute
+        # (hipBLASLt fp8 MFMA).  qkv/o stay bf16: measured 1.0x in fp8
+        # at decode shapes (ramp-bound), so quantizing them buys
+        # nothing and costs an attention-output quantization step.
+        if quant not in (None, "fp8"):
+            raise ValueError(f"quant must be None or 'fp8', got {quant!r}")
+        self.quant = quant
         self.tp_rank = tp_rank
         self.tp_world = tp_world
         self.tp_group = tp_group
@@ -144,6 +155,8 @@ class LlamaForCausalLM:
         self.scale = 1.0 / math.sqrt(c.head_dim)
         self.params: dict[str, torch.Tensor] = {}
         self._init_weights(seed)
+        if self.quant == "fp8":
+            self._quantize_weights()
         self.cos_sin = ops.build_cos_sin(c.max_position, c.head_dim,
                                          c.rope_theta, device=self.device)
 
@@ -198,6 +211,35 @@ class LlamaForCausalLM:
         else:
             self._mk("lm_head", (c.vocab_size, c.hidden_size), gen)
 
+    def _quantize_weights(self) -> None:
+        """Quantize gate_up/down/lm_head to fp8-e4m3 with per-out-channel
+        scales and FREE the bf16 originals (8B: ~5.5 GiB back).  Called
+        after random init and again after a safetensors load."""
+        from ..ops import ref as _ref
+        names = [f"l{i}.{n}" for i in range(self.config.n_layers)
+                 for n in ("gate_up", "down")] + ["lm_head"]
+        for name in names:
+            w = self.params.get(name)
+            if w is None or w.dtype != self.dtype:
+                continue
+            q, sc = _ref.quantize_fp8_rowwise(w)
+            self.params[name + ".q8"] = q.contiguous()
+            self.params[name + ".s"] = sc.to(torch.float32).contiguous()
+            if name == "lm_head" and w is self.params.get("embed"):
+                pass              # tied: keep the bf16 embed for lookups
+            del self.params[name]
+
+    def _linear_fp8(self, x8: torch.Tensor, xs: torch.Tensor,
+                    name: str) -> torch.Tensor:
+        w8 = self.params[name + ".q8"]
+        ws = self.params[name + ".s"]
+        if x8.is_cuda:
+            return torch._scaled_mm(x8, w8.t(), scale_a=xs.reshape(-1, 1),
+                                    scale_b=ws.reshape(1, -1),
+                                    out_dtype=torch.bfloat16)
+        from ..ops import ref as _ref
+        return _ref.scaled_mm_ref(x8, xs, w8, ws)
+
     def load_safetensors(self, path: str) -> int:
         """Load HF-Llama-named safetensors shards, TP-sharded at load
         (VERDICT r01 #3): each rank takes its column slice of q/k/v and
@@ -251,9 +293,14 @@ class LlamaForCausalLM:
             self.params["lm_head"] = take("lm_head.weight")
         else:
             self.params["lm_head"] = self.params["embed"]
+        if self.quant == "fp8":
+            self._quantize_weights()
         return loaded
 
     def save_safetensors(self, path: str) -> None:
+        if self.quant is not None:
+            raise RuntimeError("cannot save a weight-quantized model "
+                               "(bf16 originals were freed)")
         """Write the FULL (unsharded) weights in HF-Llama naming — the
         round-trip fixture for the TP-sharded-load tests.  TP=1 only."""
         import safetensors.torch as st
@@ -315,6 +362,13 @@ class LlamaForCausalLM:
         # them into the attention kernel itself)
         attn_out = attn_fn(i, qkv, k_cache, v_cache)   # [T, q_size]
         h = self._all_reduce(_linear(attn_out, p[f"l{i}.o"]))
+        if self.quant == "fp8":
+            h2q, h2s = ops.rmsnorm_residual_fp8(h, residual, p[f"l{i}.ln2"],
+                                                c.rms_eps)
+            gu = self._linear_fp8(h2q, h2s, f"l{i}.gate_up")
+            dq, ds = ops.silu_mul_fp8(gu)
+            mlp = self._all_reduce(self._linear_fp8(dq, ds, f"l{i}.down"))
+            return mlp, residual
         h2 = ops.rmsnorm_residual_(h, residual, p[f"l{i}.ln2"], c.rms_eps)
         gu = _linear(h2, p[f"l{i}.gate_up"])
         mlp = self._all_reduce(_linear(ops.silu_mul(gu), p[f"l{i}.down"]))
@@ -377,6 +431,13 @@ class LlamaForCausalLM:
         for i in range(self.config.n_layers):
             x, residual = self._layer(i, x, residual, positions, kv_cache,
                                       slot_mapping, attn_fn)
+        if self.quant == "fp8":
+            h8, hs = ops.rmsnorm_residual_fp8(x, residual, p["final_ln"],
+                                              self.config.rms_eps)
+            if last_idx is not None:
+                h8 = h8[last_idx]
+                hs = hs[last_idx]
+            return self._linear_fp8(h8, hs, "lm_head")
         h = ops.rmsnorm_residual_(x, residual, p["final_ln"],
                                   self.config.rms_eps)
         if last_idx is not None:

@@ -64,6 +64,28 @@ def rmsnorm_residual_(x: torch.Tensor, residual: torch.Tensor, w: torch.Tensor,
     return ref.rmsnorm_residual_(x, residual, w, eps)
 
 
+def rmsnorm_fp8(x, w, eps: float = 1e-5):
+    """Fused rmsnorm + per-row fp8-e4m3 quantization -> (q, scales)."""
+    if x.is_cuda:
+        _gpu()
+        return torch.ops.rlli.rmsnorm_fp8(x, w, eps)
+    return ref.rmsnorm_fp8(x, w, eps)
+
+
+def rmsnorm_residual_fp8(x, residual, w, eps: float = 1e-5):
+    if x.is_cuda:
+        _gpu()
+        return torch.ops.rlli.rmsnorm_residual_fp8(x, residual, w, eps)
+    return ref.rmsnorm_residual_fp8(x, residual, w, eps)
+
+
+def silu_mul_fp8(gate_up):
+    if gate_up.is_cuda:
+        _gpu()
+        return torch.ops.rlli.silu_mul_fp8(gate_up)
+    return ref.silu_mul_fp8(gate_up)
+
+
 def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:
     if gate_up.is_cuda:
         _gpu()

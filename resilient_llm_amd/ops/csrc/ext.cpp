@@ -90,6 +90,66 @@ Tensor rmsnorm_residual_(const Tensor& x, Tensor residual, const Tensor& w,
   return y;
 }
 
+// ---- fp8-activation emitters (quantized-weight GEMM path) ----
+std::tuple<Tensor, Tensor> rmsnorm_fp8(const Tensor& x, const Tensor& w,
+                                       double eps) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(w, "w");
+  const int dim = int(x.size(-1));
+  TORCH_CHECK(dim % 8 == 0 && dim <= 8192 && w.numel() == dim,
+              "bad dim/weight for fp8 rmsnorm");
+  const int rows = int(x.numel() / dim);
+  Tensor y8 = at::empty(x.sizes(), x.options().dtype(at::kFloat8_e4m3fn));
+  Tensor scales = at::empty({rows}, x.options().dtype(at::kFloat));
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  rlli::launch_rmsnorm_fp8(bf16_ptr(x), nullptr, bf16_ptr(w),
+                           reinterpret_cast<uint8_t*>(y8.data_ptr()),
+                           scales.data_ptr<float>(), rows, dim, float(eps),
+                           current_stream(x));
+  check_launch("rmsnorm_fp8");
+  return {y8, scales};
+}
+
+std::tuple<Tensor, Tensor> rmsnorm_residual_fp8(const Tensor& x,
+                                                Tensor residual,
+                                                const Tensor& w, double eps) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(residual, "residual");
+  check_bf16_contig(w, "w");
+  TORCH_CHECK(x.sizes() == residual.sizes(), "x/residual shape mismatch");
+  const int dim = int(x.size(-1));
+  TORCH_CHECK(dim % 8 == 0 && dim <= 8192 && w.numel() == dim,
+              "bad dim/weight for fp8 rmsnorm");
+  const int rows = int(x.numel() / dim);
+  Tensor y8 = at::empty(x.sizes(), x.options().dtype(at::kFloat8_e4m3fn));
+  Tensor scales = at::empty({rows}, x.options().dtype(at::kFloat));
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  rlli::launch_rmsnorm_fp8(bf16_ptr(x), bf16_ptr(residual), bf16_ptr(w),
+                           reinterpret_cast<uint8_t*>(y8.data_ptr()),
+                           scales.data_ptr<float>(), rows, dim, float(eps),
+                           current_stream(x));
+  check_launch("rmsnorm_residual_fp8");
+  return {y8, scales};
+}
+
+std::tuple<Tensor, Tensor> silu_mul_fp8(const Tensor& gate_up) {
+  check_bf16_contig(gate_up, "gate_up");
+  const int64_t inter = gate_up.size(-1) / 2;
+  TORCH_CHECK(inter % 8 == 0, "inter must be a multiple of 8");
+  const int rows = int(gate_up.numel() / (2 * inter));
+  auto sizes = gate_up.sizes().vec();
+  sizes.back() = inter;
+  Tensor y8 = at::empty(sizes, gate_up.options().dtype(at::kFloat8_e4m3fn));
+  Tensor scales = at::empty({rows}, gate_up.options().dtype(at::kFloat));
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(gate_up.device());
+  rlli::launch_silu_mul_fp8(bf16_ptr(gate_up),
+                            reinterpret_cast<uint8_t*>(y8.data_ptr()),
+                            scales.data_ptr<float>(), rows, int(inter),
+                            current_stream(gate_up));
+  check_launch("silu_mul_fp8");
+  return {y8, scales};
+}
+
 // ------------------------------------------------------------ silu_mul
 Tensor silu_mul(const Tensor& gate_up) {
   check_bf16_contig(gate_up, "gate_up");
@@ -545,6 +605,9 @@ TORCH_LIBRARY(rlli, m) {
   m.def("rmsnorm(Tensor x, Tensor w, float eps) -> Tensor");
   m.def("rmsnorm_residual_(Tensor x, Tensor(a!) residual, Tensor w, float eps) -> Tensor");
   m.def("silu_mul(Tensor gate_up) -> Tensor");
+  m.def("rmsnorm_fp8(Tensor x, Tensor w, float eps) -> (Tensor, Tensor)");
+  m.def("rmsnorm_residual_fp8(Tensor x, Tensor(a!) residual, Tensor w, float eps) -> (Tensor, Tensor)");
+  m.def("silu_mul_fp8(Tensor gate_up) -> (Tensor, Tensor)");
   m.def("rope_kv_append_(Tensor(a!) q, Tensor(b!) k, Tensor v, Tensor positions, "
         "Tensor cos_sin, Tensor(c!) k_cache, Tensor(d!) v_cache, "
         "Tensor slot_mapping) -> ()");
@@ -577,6 +640,9 @@ TORCH_LIBRARY_IMPL(rlli, CUDA, m) {
   m.impl("rmsnorm", &rmsnorm);
   m.impl("rmsnorm_residual_", &rmsnorm_residual_);
   m.impl("silu_mul", &silu_mul);
+  m.impl("rmsnorm_fp8", &rmsnorm_fp8);
+  m.impl("rmsnorm_residual_fp8", &rmsnorm_residual_fp8);
+  m.impl("silu_mul_fp8", &silu_mul_fp8);
   m.impl("rope_kv_append_", &rope_kv_append_);
   m.impl("decode_attn", &decode_attn);
   m.impl("rope_kv_append_qkv_", &rope_kv_append_qkv_);

@@ -24,6 +24,16 @@ void launch_silu_mul(const uint16_t* gate_up, uint16_t* out, int rows,
 // tensor: *_stride is the per-token element stride of each view.
 // Cache pointers are void*: the paged cache stores bf16 (default) or
 // OCP fp8-e4m3 (cache_fp8) — see common.h cache codecs.
+// FP8-activation emitters for the quantized-weight GEMM path
+// (rowwise torch._scaled_mm consumers): normed / activated rows out as
+// fp8-e4m3 + one dequant scale per row.
+void launch_rmsnorm_fp8(const uint16_t* x, uint16_t* residual,
+                        const uint16_t* w, uint8_t* y8, float* scales,
+                        int rows, int dim, float eps, hipStream_t stream);
+void launch_silu_mul_fp8(const uint16_t* gate_up, uint8_t* out8,
+                         float* scales, int rows, int inter,
+                         hipStream_t stream);
+
 void launch_rope_kv_append(
     uint16_t* q, uint16_t* k, uint16_t* v,
     const int32_t* positions, const float* cos_sin,   // [max_pos, head_dim]

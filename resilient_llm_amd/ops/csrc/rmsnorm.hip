@@ -210,6 +210,93 @@ void rmsnorm_wave(const uint16_t* __restrict__ x,
   }
 }
 
+// FP8-output variant for the quantized-weight GEMM path: the normed row
+// is emitted as fp8-e4m3 plus ONE per-row dequant scale (amax/448),
+// computed with a second block reduction — the consumer is a rowwise
+// torch._scaled_mm, so the 2-byte->1-byte write also halves the
+// activation traffic.  Residual is still read-modify-written in bf16.
+template <bool FUSE_RESIDUAL>
+__global__ void rmsnorm_one_vec_fp8(const uint16_t* __restrict__ x,
+                                    uint16_t* __restrict__ residual,
+                                    const uint16_t* __restrict__ w,
+                                    uint8_t* __restrict__ y8,
+                                    float* __restrict__ scales,
+                                    int dim, float eps) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* lds = reinterpret_cast<float*>(smem_raw);
+  const int row = blockIdx.x;
+  const int64_t base = int64_t(row) * dim;
+  const int v8 = threadIdx.x;
+  const bool active = v8 * 8 < dim;
+
+  float vals[8];
+  if (active) {
+    bf16x8 xv;
+    xv.u = *reinterpret_cast<const uint4*>(x + base + v8 * 8);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) vals[i] = bf16_to_f32(xv.s[i]);
+    if constexpr (FUSE_RESIDUAL) {
+      bf16x8 rv;
+      rv.u = *reinterpret_cast<const uint4*>(residual + base + v8 * 8);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) vals[i] += bf16_to_f32(rv.s[i]);
+      bf16x8 out;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) out.s[i] = f32_to_bf16(vals[i]);
+      *reinterpret_cast<uint4*>(residual + base + v8 * 8) = out.u;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) vals[i] = bf16_to_f32(out.s[i]);
+    }
+  } else {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) vals[i] = 0.f;
+  }
+
+  float ss = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) ss += vals[i] * vals[i];
+  ss = block_sum<16>(ss, lds);
+  const float inv = rsqrtf(ss / dim + eps);
+
+  float out8[8];
+  float amax = 0.f;
+  if (active) {
+    bf16x8 wv;
+    wv.u = *reinterpret_cast<const uint4*>(w + v8 * 8);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      out8[i] = vals[i] * inv * bf16_to_f32(wv.s[i]);
+      amax = fmaxf(amax, fabsf(out8[i]));
+    }
+  }
+  // block max through the same LDS slots (barrier separates uses)
+  {
+    const int lane = threadIdx.x & (kWave - 1);
+    const int wave = threadIdx.x / kWave;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      amax = fmaxf(amax, __shfl_xor(amax, off, kWave));
+    __syncthreads();
+    if (lane == 0) lds[wave] = amax;
+    __syncthreads();
+    const int n_waves = (blockDim.x + kWave - 1) / kWave;
+    amax = 0.f;
+#pragma unroll
+    for (int wv = 0; wv < 16; ++wv)
+      if (wv < n_waves) amax = fmaxf(amax, lds[wv]);
+  }
+  const float scale = fmaxf(amax, 1e-12f) / 448.f;
+  if (active) {
+    const float rs = 1.f / scale;
+    float q[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) q[i] = out8[i] * rs;
+    *reinterpret_cast<CacheFP8::vec8*>(y8 + base + v8 * 8) =
+        CacheFP8::from_f32(q);
+  }
+  if (threadIdx.x == 0) scales[row] = scale;
+}
+
 template <bool FR>
 void launch_wave(const uint16_t* x, uint16_t* residual, const uint16_t* w,
                  uint16_t* y, int rows, int dim, float eps,
@@ -227,6 +314,24 @@ void launch_wave(const uint16_t* x, uint16_t* residual, const uint16_t* w,
 }
 
 }  // namespace
+
+void launch_rmsnorm_fp8(const uint16_t* x, uint16_t* residual,
+                        const uint16_t* w, uint8_t* y8, float* scales,
+                        int rows, int dim, float eps, hipStream_t stream) {
+  if (rows == 0) return;
+  const int smem = 16 * sizeof(float);
+  const int threads = ((dim / 8 + kWave - 1) / kWave) * kWave;
+  // fp8 activations are for the quantized-GEMM path; dims beyond the
+  // one-vec envelope (8192) would need the two-pass form — none of the
+  // supported models exceed it
+  if (residual)
+    hipLaunchKernelGGL((rmsnorm_one_vec_fp8<true>), dim3(rows), dim3(threads),
+                       smem, stream, x, residual, w, y8, scales, dim, eps);
+  else
+    hipLaunchKernelGGL((rmsnorm_one_vec_fp8<false>), dim3(rows),
+                       dim3(threads), smem, stream, x, nullptr, w, y8,
+                       scales, dim, eps);
+}
 
 void launch_rmsnorm(const uint16_t* x, uint16_t* residual, const uint16_t* w,
                     uint16_t* y, int rows, int dim, float eps,

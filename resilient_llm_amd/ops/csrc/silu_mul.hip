@@ -7,6 +7,8 @@
 
 #include "common.h"
 
+#include <type_traits>
+
 namespace rlli {
 
 namespace {
@@ -65,7 +67,86 @@ void silu_mul_kernel(const uint16_t* __restrict__ gate_up,
   }
 }
 
+// FP8-output variant (quantized-weight down-proj): one WG per row so
+// the row amax is a single block reduction; values are kept in
+// registers between the amax pass and the quantized write (VPL deep
+// per-lane ILP covers the row: 14336/8 = 1792 vecs / 512 lanes -> 4).
+template <int VPL>
+__global__ __launch_bounds__(512)
+void silu_mul_fp8_kernel(const uint16_t* __restrict__ gate_up,
+                         uint8_t* __restrict__ out8,
+                         float* __restrict__ scales, int64_t inter) {
+  __shared__ float lds[8];
+  const int64_t row = blockIdx.x;
+  const int64_t nvec = inter / 8;
+  const uint16_t* g_row = gate_up + row * 2 * inter;
+  float vals[VPL][8];
+  float amax = 0.f;
+#pragma unroll
+  for (int p = 0; p < VPL; ++p) {
+    const int64_t v = threadIdx.x + int64_t(p) * blockDim.x;
+    if (v >= nvec) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) vals[p][i] = 0.f;
+      continue;
+    }
+    bf16x8 g, u;
+    g.u = *reinterpret_cast<const uint4*>(g_row + v * 8);
+    u.u = *reinterpret_cast<const uint4*>(g_row + v * 8 + inter);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      const float gf = bf16_to_f32(g.s[i]);
+      const float uf = bf16_to_f32(u.s[i]);
+      vals[p][i] = gf / (1.f + __expf(-gf)) * uf;
+      amax = fmaxf(amax, fabsf(vals[p][i]));
+    }
+  }
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wave = threadIdx.x / kWave;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    amax = fmaxf(amax, __shfl_xor(amax, off, kWave));
+  if (lane == 0) lds[wave] = amax;
+  __syncthreads();
+  const int n_waves = blockDim.x / kWave;
+  amax = 0.f;
+#pragma unroll
+  for (int w = 0; w < 8; ++w)
+    if (w < n_waves) amax = fmaxf(amax, lds[w]);
+  const float scale = fmaxf(amax, 1e-12f) / 448.f;
+  const float rs = 1.f / scale;
+#pragma unroll
+  for (int p = 0; p < VPL; ++p) {
+    const int64_t v = threadIdx.x + int64_t(p) * blockDim.x;
+    if (v >= nvec) continue;
+    float q[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) q[i] = vals[p][i] * rs;
+    *reinterpret_cast<CacheFP8::vec8*>(out8 + row * inter + v * 8) =
+        CacheFP8::from_f32(q);
+  }
+  if (threadIdx.x == 0) scales[row] = scale;
+}
+
 }  // namespace
+
+void launch_silu_mul_fp8(const uint16_t* gate_up, uint8_t* out8,
+                         float* scales, int rows, int inter,
+                         hipStream_t stream) {
+  if (rows == 0) return;
+  const int threads = 512;
+  const int vpl = (inter / 8 + threads - 1) / threads;
+  auto go = [&](auto tag) {
+    hipLaunchKernelGGL((silu_mul_fp8_kernel<decltype(tag)::value>),
+                       dim3(rows), dim3(threads), 0, stream, gate_up, out8,
+                       scales, inter);
+  };
+  if (vpl <= 1) go(std::integral_constant<int, 1>{});
+  else if (vpl <= 2) go(std::integral_constant<int, 2>{});
+  else if (vpl <= 4) go(std::integral_constant<int, 4>{});
+  else if (vpl <= 8) go(std::integral_constant<int, 8>{});
+  else go(std::integral_constant<int, 16>{});
+}
 
 void launch_silu_mul(const uint16_t* gate_up, uint16_t* out, int rows,
                      int inter, hipStream_t stream) {

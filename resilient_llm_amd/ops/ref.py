@@ -167,3 +167,35 @@ def sample(logits: torch.Tensor, temperatures: torch.Tensor,
         probs = torch.softmax(lf[i] / float(temperatures[i]), dim=-1)
         out[i] = int(torch.multinomial(probs, 1, generator=gen))
     return out
+
+
+# ---- fp8-activation emitters (quantized-weight GEMM path) ----
+def quantize_fp8_rowwise(t: torch.Tensor):
+    """Per-row fp8-e4m3 quantization: (q, scales) with q*scales ~= t."""
+    f = t.float()
+    amax = f.abs().amax(dim=-1).clamp_min(1e-12)
+    scales = amax / 448.0
+    q = (f / scales.unsqueeze(-1)).to(torch.float8_e4m3fn)
+    return q, scales
+
+
+def rmsnorm_fp8(x, w, eps=1e-5):
+    return quantize_fp8_rowwise(rmsnorm(x, w, eps))
+
+
+def rmsnorm_residual_fp8(x, residual, w, eps=1e-5):
+    return quantize_fp8_rowwise(rmsnorm_residual_(x, residual, w, eps))
+
+
+def silu_mul_fp8(gate_up):
+    return quantize_fp8_rowwise(silu_mul(gate_up))
+
+
+def scaled_mm_ref(x8, xs, w8, ws):
+    """Dequantized reference of a rowwise torch._scaled_mm: x8 [M,K] fp8
+    with per-row scales xs [M], w8 [N,K] fp8 with per-row (out-channel)
+    scales ws [N]; bf16 result.  The CPU engine path uses this (no
+    hipBLASLt fp8 GEMM off-GPU)."""
+    xf = x8.float() * xs.reshape(-1, 1)
+    wf = w8.float() * ws.reshape(-1, 1)
+    return (xf @ wf.t()).to(torch.bfloat16)

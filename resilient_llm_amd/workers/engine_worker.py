@@ -78,6 +78,7 @@ class EngineWorker(Worker):
                  weights: Optional[str] = None,
                  eos_id=None,
                  kv_dtype: str = "bf16",
+                 quant=None,
                  seed: int = 0) -> None:
         super().__init__(device=device_label or f"gpu:{device}",
                          models={model_name})
@@ -90,7 +91,8 @@ class EngineWorker(Worker):
         self.config = config
         self.model = LlamaForCausalLM(config, device=torch_device, dtype=dtype,
                                       tp_rank=tp_rank, tp_world=tp_world,
-                                      tp_group=tp_group, seed=seed)
+                                      tp_group=tp_group, seed=seed,
+                                      quant=quant)
         if weights:
             n = self.model.load_safetensors(weights)
             from ..utils.logging import log_with_timestamp

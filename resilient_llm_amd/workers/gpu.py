@@ -34,6 +34,7 @@ def spawn_gpu_worker(device_index: str, model_name: str,
                      weights: str | None = None,
                      eos_ids: list | None = None,
                      kv_dtype: str | None = None,
+                     quant: str | None = None,
                      extra_env: dict | None = None) -> subprocess.Popen:
     """``device_index`` may be a sub-device replica like ``0.1``: several
     worker processes co-located on physical GPU 0 — 288 GB of HBM3E holds
@@ -60,6 +61,8 @@ def spawn_gpu_worker(device_index: str, model_name: str,
         cmd += ["--eos-id", str(e)]
     if kv_dtype:
         cmd += ["--kv-dtype", str(kv_dtype)]
+    if quant:
+        cmd += ["--quant", str(quant)]
     return subprocess.Popen(cmd, env=env)
 
 
@@ -109,13 +112,15 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
         extra = opts.get(target, {})
         weights = extra.get("weights")
         kv_dtype = extra.get("kv_dtype")
+        quant = extra.get("quantization")
         eos = extra.get("eos_id")
         eos_list = None if eos is None else (
             [int(eos)] if isinstance(eos, int) else [int(e) for e in eos])
         proc = spawn_gpu_worker(target, model_name, sock, kv_gb=kv_gb,
                                 max_batch=max_batch, use_graphs=use_graphs,
                                 weights=weights, eos_ids=eos_list,
-                                kv_dtype=kv_dtype, target_step_ms=slo)
+                                kv_dtype=kv_dtype, quant=quant,
+                                target_step_ms=slo)
         client = RpcWorkerClient(f"gpu:{target}", {model_name}, sock)
         client.proc = proc
         client.respawn = (lambda t=target, m=model_name, s=sock:
@@ -123,7 +128,7 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
                                            max_batch=max_batch,
                                            use_graphs=use_graphs,
                                            weights=weights, eos_ids=eos_list,
-                                           kv_dtype=kv_dtype,
+                                           kv_dtype=kv_dtype, quant=quant,
                                            target_step_ms=slo))
         registry.register("gpu", target, client)
 

@@ -26,6 +26,9 @@ def main() -> None:
                          "when absent")
     ap.add_argument("--eos-id", type=int, action="append", default=None,
                     help="override EOS token id(s); repeatable")
+    ap.add_argument("--quant", default=None, choices=(None, "fp8"),
+                    help="W8A8-fp8 weight quantization for gate_up/down/"
+                         "lm_head (rowwise scaled_mm)")
     ap.add_argument("--kv-dtype", default="bf16", choices=("bf16", "fp8"),
                     help="paged-KV-cache element type: fp8-e4m3 halves "
                          "bytes/token (2x KV capacity per GiB)")
@@ -60,6 +63,7 @@ def main() -> None:
                               weights=args.weights,
                               eos_id=args.eos_id,
                               kv_dtype=args.kv_dtype,
+                              quant=args.quant,
                               target_step_ms=args.target_step_ms)
         log_with_timestamp(
             f"worker {args.device_label} ready: {args.model} on {device}, "
