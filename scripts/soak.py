@@ -20,12 +20,23 @@ spec.loader.exec_module(common)
 
 from resilient_llm_amd.client import APIError, OpenAIClient  # noqa: E402
 
+def _mode_opts():
+    """RLLI_SOAK_MODE=fp8 soaks the quantized-weight + fp8-KV modes."""
+    if os.environ.get("RLLI_SOAK_MODE") == "fp8":
+        return {"quantization": "fp8", "kv_dtype": "fp8"}
+    return {}
+
+
 CONFIG = {
     "cluster": {"port": 4999},
     "model_list": [
-        {"model_name": "soak", "litellm_params": {"model": "gpu/0/llama-3-8b"},
+        {"model_name": "soak",
+         "litellm_params": dict({"model": "gpu/0/llama-3-8b"},
+                                **_mode_opts()),
          "model_info": {"id": "gpu0/soak"}},
-        {"model_name": "soak", "litellm_params": {"model": "gpu/0.1/llama-3-8b"},
+        {"model_name": "soak",
+         "litellm_params": dict({"model": "gpu/0.1/llama-3-8b"},
+                                **_mode_opts()),
          "model_info": {"id": "gpu0.1/soak"}},
     ],
     "router_settings": {"routing_strategy": "simple-shuffle",
