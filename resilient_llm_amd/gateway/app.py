@@ -131,6 +131,7 @@ class GatewayApp:
         self.policy = AccessPolicy(config.raw.get("auth"))
         self.health_interval_s = health_interval_s
         self.respawn_cooldown_s = 10.0
+        self._restart_tasks: dict = {}
         self.last_health: dict = {}     # worker key -> last health dict
         self._health_task: Optional[asyncio.Task] = None
         self.started_at = time.time()
@@ -264,6 +265,10 @@ class GatewayApp:
             return Response.json_response({"requests": self.ledger.recent(n)})
         if route == ("POST", "/admin/fault"):
             return await self.admin_fault(req)
+        if route == ("POST", "/admin/drain"):
+            return await self.admin_drain(req)
+        if route == ("POST", "/admin/restart"):
+            return await self.admin_restart(req)
         if route == ("GET", "/metrics"):
             return self.prometheus_metrics()
         return Response.error(404, f"no route {req.method} {req.path}",
@@ -767,6 +772,100 @@ class GatewayApp:
                                        f"command: {e}")
         return Response.json_response({"device": device, "mode": mode})
 
+    # -------------------------------------------- drain / rolling restart
+    def _set_drain(self, worker: Worker, draining: bool) -> None:
+        worker.draining = draining          # spread (`*`) targets consult this
+        for model_id in self._deployments_on(worker):
+            self.router.set_draining(model_id, draining)
+
+    def _worker_in_flight(self, worker: Worker) -> int:
+        n = getattr(worker, "in_flight", 0)
+        for d in self.config.deployments:
+            st = self.router.state_for_id(d.model_id)
+            kind, _, target = worker.device.partition(":")
+            if (st is not None and d.backend_kind == kind
+                    and d.backend_target == target):
+                n = max(n, st.in_flight)
+        return n
+
+    async def admin_drain(self, req: Request) -> Response:
+        """POST {worker, drain=true|false}: stop routing NEW requests to
+        a worker; in-flight requests finish normally.  The building
+        block of zero-downtime maintenance (see /admin/restart)."""
+        try:
+            body = req.json() or {}
+            device = body["worker"]
+        except (KeyError, json.JSONDecodeError):
+            return Response.error(400, "need {worker, drain?}")
+        worker = self.registry.all().get(device)
+        if worker is None:
+            return Response.error(404, f"no worker {device!r}")
+        draining = bool(body.get("drain", True))
+        self._set_drain(worker, draining)
+        return Response.json_response({
+            "worker": device, "draining": draining,
+            "in_flight": self._worker_in_flight(worker)})
+
+    async def admin_restart(self, req: Request) -> Response:
+        """POST {worker, timeout_s?}: zero-downtime rolling restart —
+        drain, wait for in-flight to land, terminate + respawn the
+        worker process, wait healthy, undrain.  Other replicas keep
+        serving throughout; returns immediately with 202."""
+        try:
+            body = req.json() or {}
+            device = body["worker"]
+        except (KeyError, json.JSONDecodeError):
+            return Response.error(400, "need {worker, timeout_s?}")
+        worker = self.registry.all().get(device)
+        if worker is None:
+            return Response.error(404, f"no worker {device!r}")
+        if getattr(worker, "respawn", None) is None:
+            return Response.error(409, f"worker {device} is not "
+                                       "process-backed (no respawn hook)")
+        timeout_s = float(body.get("timeout_s", 120.0))
+        task = asyncio.get_running_loop().create_task(
+            self._rolling_restart(device, worker, timeout_s))
+        self._restart_tasks[device] = task
+        return Response.json_response({"worker": device,
+                                       "status": "restarting"}, status=202)
+
+    async def _rolling_restart(self, device: str, worker: Worker,
+                               timeout_s: float) -> None:
+        from ..utils.logging import log_with_timestamp, sanitize_error
+        log_with_timestamp(f"rolling restart of {device}: draining", "yellow")
+        self._set_drain(worker, True)
+        t0 = time.monotonic()
+        try:
+            while (self._worker_in_flight(worker) > 0
+                   and time.monotonic() - t0 < timeout_s):
+                await asyncio.sleep(0.1)
+            proc = getattr(worker, "proc", None)
+            if proc is not None and proc.poll() is None:
+                proc.terminate()
+                for _ in range(100):
+                    if proc.poll() is not None:
+                        break
+                    await asyncio.sleep(0.1)
+                if proc.poll() is None:
+                    proc.kill()
+            await worker.respawn_now()
+            while time.monotonic() - t0 < timeout_s:
+                try:
+                    h = await asyncio.wait_for(worker.health(), timeout=5.0)
+                    self.last_health[worker.device] = h
+                    break
+                except Exception:
+                    await asyncio.sleep(0.2)
+            for model_id in self._deployments_on(worker):
+                self.router.set_healthy(model_id, True)
+            log_with_timestamp(f"rolling restart of {device}: back online",
+                               "green")
+        except Exception as e:                            # noqa: BLE001
+            log_with_timestamp(f"rolling restart of {device} FAILED: "
+                               f"{sanitize_error(e)}", "red")
+        finally:
+            self._set_drain(worker, False)
+
     def prometheus_metrics(self) -> Response:
         lines = []
         for row in self.router.describe():
@@ -776,6 +875,7 @@ class GatewayApp:
             lines.append(f'gateway_in_flight{{{labels}}} {row["in_flight"]}')
             lines.append(f'gateway_rpm_used{{{labels}}} {row["rpm_used"]}')
             lines.append(f'gateway_healthy{{{labels}}} {int(row["healthy"])}')
+            lines.append(f'gateway_draining{{{labels}}} {int(row["draining"])}')
         st = self.ledger.stats()
         for k in ("total", "ok", "errors", "throttled", "fallbacks"):
             lines.append(f'gateway_ledger_{k} {st[k]}')

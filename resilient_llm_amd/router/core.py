@@ -54,6 +54,7 @@ class DeploymentState:
     fail_times: list = dataclasses.field(default_factory=list)
     in_flight: int = 0
     healthy: bool = True            # worker heartbeat / registration status
+    draining: bool = False          # admin drain: no NEW admissions
     total_requests: int = 0
     total_failures: int = 0
     total_cooldowns: int = 0
@@ -126,7 +127,7 @@ class Router:
 
     # ----------------------------------------------------------- policies
     def _available(self, s: DeploymentState, tokens: int, exclude: set) -> bool:
-        if id(s) in exclude or not s.healthy:
+        if id(s) in exclude or not s.healthy or s.draining:
             return False
         if s.cooldown_until > self._clock():
             return False
@@ -253,6 +254,15 @@ class Router:
                 if s.dep.model_id == model_id:
                     s.healthy = healthy
 
+    def set_draining(self, model_id: str, draining: bool) -> None:
+        """Graceful drain: a draining deployment takes no NEW requests
+        (in-flight ones finish normally) — the building block of the
+        zero-downtime rolling restart (gateway /admin/drain|restart)."""
+        with self._lock:
+            for s in self.states:
+                if s.dep.model_id == model_id:
+                    s.draining = draining
+
     def describe(self) -> list[dict]:
         now = self._clock()
         out = []
@@ -265,6 +275,7 @@ class Router:
                 "rpm": s.dep.rpm, "tpm": s.dep.tpm,
                 "rpm_used": snap.rpm_used, "tpm_used": snap.tpm_used,
                 "healthy": s.healthy,
+                "draining": s.draining,
                 "cooldown_remaining": max(0.0, s.cooldown_until - now),
                 "in_flight": s.in_flight,
                 "total_requests": s.total_requests,

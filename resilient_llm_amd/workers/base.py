@@ -115,6 +115,12 @@ class WorkerRegistry:
                           if k.startswith(f"{kind}:") and model in w.models]
             if not candidates:
                 raise WorkerDead(f"no {kind} worker holds model {model!r}")
+            # admin drain: spread targets skip draining workers while
+            # any alternative exists
+            active = [w for w in candidates
+                      if not getattr(w, "draining", False)]
+            if active:
+                candidates = active
             # capacity-driven spread: least in flight, random among ties —
             # mirrors CRIS's "distribution is capacity-driven, not
             # client-controlled" 

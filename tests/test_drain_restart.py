@@ -1,0 +1,196 @@
+"""Graceful drain + zero-downtime rolling restart (/admin/drain,
+/admin/restart) — beyond-reference ops surface: the reference's only
+way to take a backend out of rotation was starving its quota."""
+
+import asyncio
+import json
+import os
+import subprocess
+import sys
+import tempfile
+import threading
+import time
+import urllib.request
+
+import pytest
+
+from resilient_llm_amd.client import OpenAIClient
+from resilient_llm_amd.config import load_config
+from resilient_llm_amd.gateway.app import GatewayApp
+from resilient_llm_amd.gateway.http import HttpServer
+from resilient_llm_amd.workers.base import WorkerRegistry
+from resilient_llm_amd.workers.rpc import RpcWorkerClient
+from tests.gateway_harness import free_port, run_gateway, stub_config_dict
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+pytestmark = pytest.mark.timeout(180)
+
+
+def _post(base, path, body):
+    req = urllib.request.Request(base + path, method="POST",
+                                 data=json.dumps(body).encode(),
+                                 headers={"content-type": "application/json"})
+    with urllib.request.urlopen(req, timeout=30) as r:
+        return r.status, json.loads(r.read().decode())
+
+
+def test_drain_excludes_deployment_and_undrain_restores():
+    port = free_port()
+    with run_gateway(stub_config_dict(port)):
+        base = f"http://127.0.0.1:{port}"
+        client = OpenAIClient(base)
+
+        def hits(n=12):
+            seen = set()
+            for i in range(n):
+                r = client.chat.completions.create(
+                    model="llama-cris-demo",
+                    messages=[{"role": "user", "content": f"q{i}"}],
+                    max_tokens=4)
+                seen.add(r.headers.get("x-gateway-device"))
+            return seen
+
+        assert len(hits()) >= 2                       # spread target uses both
+        st, body = _post(base, "/admin/drain", {"worker": "stub:0"})
+        assert st == 200 and body["draining"] is True
+        assert hits() == {"stub:1"}                   # drained replica skipped
+        # lb alias (explicit targets) also avoids the drained worker
+        for i in range(6):
+            r = client.chat.completions.create(
+                model="llama-loadbalance-demo",
+                messages=[{"role": "user", "content": f"lb{i}"}],
+                max_tokens=4)
+            assert r.headers.get("x-gateway-device") == "stub:1"
+        st, body = _post(base, "/admin/drain", {"worker": "stub:0",
+                                                "drain": False})
+        assert st == 200 and body["draining"] is False
+        assert len(hits()) >= 2
+
+        # draining an unknown worker is a 404; restart on a stub is 409
+        import urllib.error
+        try:
+            _post(base, "/admin/drain", {"worker": "stub:9"})
+            assert False
+        except urllib.error.HTTPError as e:
+            assert e.code == 404
+        try:
+            _post(base, "/admin/restart", {"worker": "stub:0"})
+            assert False
+        except urllib.error.HTTPError as e:
+            assert e.code == 409
+
+
+def _spawn_cpu_worker(sock, label):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    return subprocess.Popen(
+        [sys.executable, "-m", "resilient_llm_amd.workers.gpu_main",
+         "--device-label", label, "--model", "tiny", "--socket", sock,
+         "--device", "cpu", "--num-blocks", "64"], env=env)
+
+
+def test_rolling_restart_zero_downtime():
+    """Restart worker 0 while load flows: every request succeeds (the
+    other replica carries the traffic), and the restarted worker has a
+    NEW pid and serves again."""
+    run_dir = tempfile.mkdtemp(prefix="rlli-restart-")
+    socks = [os.path.join(run_dir, f"w{i}.sock") for i in range(2)]
+    port = free_port()
+    cfg = load_config(data={
+        "cluster": {"port": port},
+        "model_list": [
+            {"model_name": "tiny-serve",
+             "litellm_params": {"model": "gpu/0/tiny"},
+             "model_info": {"id": "gpu0/tiny"}},
+            {"model_name": "tiny-serve",
+             "litellm_params": {"model": "gpu/1/tiny"},
+             "model_info": {"id": "gpu1/tiny"}}],
+        "router_settings": {"routing_strategy": "simple-shuffle"},
+    })
+    loop = asyncio.new_event_loop()
+    ready = threading.Event()
+    holder: dict = {}
+
+    async def main():
+        holder["stop"] = asyncio.Event()
+        registry = WorkerRegistry()
+        for i in range(2):
+            c = RpcWorkerClient(f"gpu:{i}", {"tiny"}, socks[i])
+            c.proc = _spawn_cpu_worker(socks[i], f"gpu:{i}")
+            c.respawn = (lambda s=socks[i], l=f"gpu:{i}":
+                         _spawn_cpu_worker(s, l))
+            await c.connect(timeout=120)
+            registry.register("gpu", str(i), c)
+            holder[f"w{i}"] = c
+        app = GatewayApp(cfg, registry, health_interval_s=0.3)
+        server = HttpServer(app.handle, host="127.0.0.1", port=port)
+        await server.start()
+        await app.start_background()
+        ready.set()
+        await holder["stop"].wait()
+        await app.stop_background()
+        await server.stop()
+        await registry.close()
+
+    th = threading.Thread(target=lambda: loop.run_until_complete(main()),
+                          daemon=True)
+    th.start()
+    assert ready.wait(timeout=150)
+    base = f"http://127.0.0.1:{port}"
+    client = OpenAIClient(base)
+    old_pid = holder["w0"].proc.pid
+
+    errors = []
+    stop_load = threading.Event()
+
+    def load():
+        i = 0
+        while not stop_load.is_set():
+            try:
+                client.chat.completions.create(
+                    model="tiny-serve",
+                    messages=[{"role": "user", "content": f"r{i}"}],
+                    max_tokens=4, timeout=30)
+            except Exception as e:                    # noqa: BLE001
+                errors.append(repr(e))
+            i += 1
+
+    try:
+        loaders = [threading.Thread(target=load) for _ in range(4)]
+        for t in loaders:
+            t.start()
+        time.sleep(0.5)
+        st, body = _post(base, "/admin/restart",
+                         {"worker": "gpu:0", "timeout_s": 90})
+        assert st == 202 and body["status"] == "restarting"
+        # wait for the new process to come up and serve
+        deadline = time.monotonic() + 120
+        while time.monotonic() < deadline:
+            p = holder["w0"].proc
+            if p is not None and p.pid != old_pid and p.poll() is None:
+                break
+            time.sleep(0.3)
+        time.sleep(1.0)
+        stop_load.set()
+        for t in loaders:
+            t.join(timeout=30)
+        assert holder["w0"].proc.pid != old_pid, "worker was not restarted"
+        assert not errors, f"requests failed during rolling restart: {errors[:3]}"
+        # the restarted worker serves again (drain cleared)
+        devices = set()
+        for i in range(16):
+            r = client.chat.completions.create(
+                model="tiny-serve",
+                messages=[{"role": "user", "content": f"post{i}"}],
+                max_tokens=4, timeout=30)
+            devices.add(r.headers.get("x-gateway-device"))
+        assert "gpu:0" in devices
+    finally:
+        stop_load.set()
+        loop.call_soon_threadsafe(holder["stop"].set)
+        th.join(timeout=30)
+        for i in range(2):
+            p = holder.get(f"w{i}")
+            if p is not None and p.proc is not None and p.proc.poll() is None:
+                p.proc.kill()
