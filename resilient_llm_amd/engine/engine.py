@@ -133,6 +133,13 @@ class LLMEngine:
         self.block_size = kv_cache.block_size
         self._aborted: set[str] = set()
         self._live: set[str] = set()   # rids currently owned by the engine
+        # live migration (SURVEY.md §5.8 state-migration): extraction and
+        # adoption are queued control ops executed BETWEEN steps on the
+        # engine thread — the only thread that may touch KV/seq state
+        self._extract_reqs: set[str] = set()
+        self._extracted: dict[str, dict] = {}
+        self._adopt_queue: list[dict] = []
+        self._adopt_results: dict[str, object] = {}
         self._arrival_counter = 0
         self._batch_dirty = True
         self._graph_loaded = False
@@ -161,7 +168,8 @@ class LLMEngine:
 
     def has_work(self) -> bool:
         return bool(self.waiting or self.prefilling or self.running
-                    or self._pending)
+                    or self._pending or self._adopt_queue
+                    or self._extract_reqs)
 
     def add_request(self, req_id: str, prompt_ids: list,
                     params: Optional[SamplingParams] = None) -> None:
@@ -277,6 +285,8 @@ class LLMEngine:
         self.step_count += 1
         outs = self._process_pending()
         self._drop_aborted()
+        self._do_extracts()
+        self._do_adopts()
         self.prefilling.extend(self._admit())
         if self.prefilling:
             t0 = time.monotonic()
@@ -356,6 +366,124 @@ class LLMEngine:
         # clear it here so _aborted cannot accumulate stale rids
         with self._queue_lock:
             self._aborted &= self._live
+
+    # ------------------------------------------------------ live migration
+    def request_extract(self, req_id: str) -> None:
+        """Ask the engine thread to extract a live request's full state
+        (tokens + sampling identity + KV blocks) at the next step
+        boundary; collect it with :meth:`take_extracted`."""
+        with self._queue_lock:
+            self._extract_reqs.add(req_id)
+
+    def take_extracted(self, req_id: str):
+        with self._queue_lock:
+            return self._extracted.pop(req_id, None)
+
+    def queue_adopt(self, state: dict) -> None:
+        """Hand an extracted state to this engine; adoption happens on
+        the engine thread at the next step boundary.  Outcome via
+        :meth:`take_adopt_result` ("ok" or an exception)."""
+        with self._queue_lock:
+            self._adopt_queue.append(state)
+
+    def take_adopt_result(self, req_id: str):
+        with self._queue_lock:
+            return self._adopt_results.pop(req_id, None)
+
+    def _do_extracts(self) -> None:
+        with self._queue_lock:
+            if not self._extract_reqs:
+                return
+            rids = self._extract_reqs
+            self._extract_reqs = set()
+        for rid in rids:
+            state = None
+            for attr in ("running", "prefilling", "waiting"):
+                pool = getattr(self, attr)
+                for seq in list(pool):
+                    if seq.req_id != rid:
+                        continue
+                    state = {"rid": rid,
+                             "prompt_ids": list(seq.prompt_ids),
+                             "output_ids": list(seq.output_ids),
+                             "params": dataclasses.asdict(seq.params),
+                             "default_seed": seq.default_seed,
+                             "n_cached": seq.n_cached,
+                             "block_size": self.block_size,
+                             "kv": None}
+                    # KV rides along only for decode-steady sequences;
+                    # a mid-prefill extraction re-prefills on the target
+                    # (prefix caching there may still serve it)
+                    if attr == "running" and seq.n_cached > 0:
+                        nb = -(-seq.n_cached // self.block_size)
+                        idx = torch.tensor(seq.blocks[:nb],
+                                           dtype=torch.long,
+                                           device=self.kv.k.device)
+                        state["kv"] = (
+                            self.kv.k.index_select(1, idx).to("cpu"),
+                            self.kv.v.index_select(1, idx).to("cpu"))
+                    else:
+                        state["n_cached"] = 0
+                        state["output_ids"] = []
+                    if seq.blocks:
+                        self.kv.free(seq.blocks)
+                        seq.blocks = []
+                    if isinstance(pool, list):
+                        pool.remove(seq)
+                    else:
+                        pool.remove(seq)   # deque supports remove too
+                    self._batch_dirty = True
+                    break
+                if state is not None:
+                    break
+            with self._queue_lock:
+                self._live.discard(rid)
+                self._extracted[rid] = state   # None = rid not found
+
+    def _do_adopts(self) -> None:
+        with self._queue_lock:
+            if not self._adopt_queue:
+                return
+            batch = self._adopt_queue
+            self._adopt_queue = []
+        for state in batch:
+            rid = state["rid"]
+            try:
+                params = SamplingParams(**state["params"])
+                seq = SeqState(req_id=rid,
+                               prompt_ids=list(state["prompt_ids"]),
+                               params=params,
+                               output_ids=list(state["output_ids"]),
+                               default_seed=state["default_seed"],
+                               n_cached=int(state["n_cached"]))
+                kv_pair = state.get("kv")
+                usable = (kv_pair is not None
+                          and state.get("block_size") == self.block_size
+                          and seq.output_ids
+                          and seq.n_cached >= len(seq.prompt_ids))
+                seq.blocks = self.kv.allocate(
+                    seq.blocks_needed(self.block_size))
+                if usable:
+                    k_src, v_src = kv_pair
+                    nb = k_src.shape[1]
+                    idx = torch.tensor(seq.blocks[:nb], dtype=torch.long,
+                                       device=self.kv.k.device)
+                    self.kv.k.index_copy_(
+                        1, idx, k_src.to(self.kv.k.device, self.kv.k.dtype))
+                    self.kv.v.index_copy_(
+                        1, idx, v_src.to(self.kv.v.device, self.kv.v.dtype))
+                    self.running.append(seq)
+                else:
+                    seq.n_cached = 0
+                    seq.output_ids = []
+                    self.waiting.append(seq)
+                with self._queue_lock:
+                    self._live.add(rid)
+                    self._adopt_results[rid] = "ok"
+                self._batch_dirty = True
+            except Exception as e:                        # noqa: BLE001
+                with self._queue_lock:
+                    self._adopt_results[rid] = e
 
     @staticmethod
     def _seq_seed(s: SeqState) -> int:

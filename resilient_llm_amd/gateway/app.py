@@ -36,7 +36,7 @@ from ..router.core import (
 from ..router.token_bucket import MinuteWindowLimiter
 from ..workers.base import (
     GenerationRequest, GenerationResult, Worker, WorkerError,
-    WorkerRegistry, WorkerThrottled,
+    WorkerRegistry, WorkerMigrated, WorkerThrottled,
 )
 
 from .http import Request, Response
@@ -267,6 +267,8 @@ class GatewayApp:
             return await self.admin_fault(req)
         if route == ("POST", "/admin/drain"):
             return await self.admin_drain(req)
+        if route == ("POST", "/admin/migrate"):
+            return await self.admin_migrate(req)
         if route == ("POST", "/admin/restart"):
             return await self.admin_restart(req)
         if route == ("GET", "/metrics"):
@@ -495,6 +497,14 @@ class GatewayApp:
                 worker = self._worker_for(ticket)
                 result = await asyncio.wait_for(worker.generate(greq),
                                                 timeout=REQUEST_TIMEOUT_S)
+            except WorkerMigrated:
+                # live migration: not a failure — release the ticket
+                # without penalty and re-route; the adopting worker
+                # continues token-exact (any other replica regenerates
+                # identically from the sticky seed)
+                self.router.complete(ticket, actual_tokens=0)
+                exclude.add(id(ticket.state))
+                continue
             except (WorkerError, asyncio.TimeoutError) as e:
                 last_err = e
                 throttled = isinstance(e, WorkerThrottled)
@@ -658,6 +668,21 @@ class GatewayApp:
                             yield f"data: {json.dumps(usage_evt)}\n\n".encode()
                         yield b"data: [DONE]\n\n"
                         return
+                    except WorkerMigrated:
+                        # live migration mid-stream: no failure penalty;
+                        # re-route — the adopting worker replays the
+                        # already-generated prefix (attach) and the
+                        # emitted-token skip below dedupes it for the
+                        # client, then continues token-exact
+                        sent_tokens += n_emitted
+                        app.router.complete(ticket, actual_tokens=0)
+                        exclude.add(id(ticket.state))
+                        ticket = None
+                        try:
+                            await asyncio.wait_for(it.aclose(), timeout=2.0)
+                        except Exception:
+                            pass
+                        continue
                     except (WorkerError, asyncio.TimeoutError) as e:
                         last_err = e
                         sent_tokens += n_emitted
@@ -802,9 +827,57 @@ class GatewayApp:
             return Response.error(404, f"no worker {device!r}")
         draining = bool(body.get("drain", True))
         self._set_drain(worker, draining)
+        migrated: list = []
+        errors: list = []
+        target_dev = body.get("migrate_to")
+        if draining and target_dev:
+            # evacuate live requests to the target: zero-recompute drain
+            target = self.registry.all().get(target_dev)
+            if target is None:
+                return Response.error(404, f"no worker {target_dev!r}")
+            try:
+                ids = await worker.list_requests()
+            except (WorkerError, AttributeError) as e:
+                ids = []
+                errors.append(str(e))
+            for req_id in ids:
+                try:
+                    blob = await worker.migrate_out(req_id)
+                    await target.migrate_in(blob)
+                    migrated.append(req_id)
+                except WorkerError as e:
+                    errors.append(f"{req_id}: {e}")
         return Response.json_response({
             "worker": device, "draining": draining,
+            "migrated": migrated, "migrate_errors": errors,
             "in_flight": self._worker_in_flight(worker)})
+
+    async def admin_migrate(self, req: Request) -> Response:
+        """POST {request_id, from, to}: live-migrate one in-flight
+        request between workers — tokens + sampling identity + KV blocks
+        move, generation continues token-exact on the target, and the
+        client's blocked/streaming call transparently re-routes (the
+        source raises WorkerMigrated into the gateway retry path)."""
+        try:
+            body = req.json() or {}
+            request_id = body["request_id"]
+            src_dev, dst_dev = body["from"], body["to"]
+        except (KeyError, json.JSONDecodeError):
+            return Response.error(400, "need {request_id, from, to}")
+        workers = self.registry.all()
+        src = workers.get(src_dev)
+        dst = workers.get(dst_dev)
+        if src is None or dst is None:
+            return Response.error(404, f"unknown worker in {src_dev!r} -> "
+                                       f"{dst_dev!r} (have {sorted(workers)})")
+        try:
+            blob = await src.migrate_out(request_id)
+            await dst.migrate_in(blob)
+        except WorkerError as e:
+            return Response.error(409, f"migration failed: {e}")
+        return Response.json_response({"request_id": request_id,
+                                       "from": src_dev, "to": dst_dev,
+                                       "state_bytes": len(blob)})
 
     async def admin_restart(self, req: Request) -> Response:
         """POST {worker, timeout_s?}: zero-downtime rolling restart —

@@ -24,6 +24,12 @@ class WorkerError(Exception):
     """Hard failure inside a worker (counts toward cooldown)."""
 
 
+class WorkerMigrated(WorkerError):
+    """The request's state was live-migrated to a peer worker; the
+    caller should re-route (the adopting worker continues token-exact,
+    any other replica regenerates identically from the seed)."""
+
+
 class WorkerThrottled(WorkerError):
     """Backend over capacity — typed Throttled status (X13)."""
 

@@ -28,7 +28,7 @@ from ..models import LlamaForCausalLM, get_config
 from ..utils.tokenizer import load_tokenizer
 from .base import (
     GenerationChunk, GenerationRequest, GenerationResult, Worker,
-    WorkerDead, WorkerError, WorkerThrottled,
+    WorkerDead, WorkerError, WorkerMigrated, WorkerThrottled,
 )
 
 DEFAULT_KV_GB = 24.0
@@ -150,6 +150,12 @@ class EngineWorker(Worker):
         # buffer engine-side, ONE queue item at finish (64x fewer loop
         # wakeups for non-streaming traffic)
         self._sinks: dict[str, list] = {}
+        # live migration: adopted-but-unattached requests (gateway
+        # request_id -> {"rid", "pre", "prompt_len", "buf", "done"});
+        # outputs produced before the client re-attaches buffer here
+        self._adopted: dict[str, dict] = {}
+        self._adopted_by_rid: dict[str, dict] = {}
+        self._adopt_lock = threading.Lock()
         self._work_event = threading.Event()
         self._stop = False
         self._thread = None
@@ -197,7 +203,19 @@ class EngineWorker(Worker):
             for out in outputs:
                 sink = self._sinks.get(out.req_id)
                 if sink is None:
-                    continue
+                    # adopted-but-unattached: buffer under the adopt
+                    # lock; if the attach raced us and registered the
+                    # sink in between, fall through and deliver to it —
+                    # dropping here would lose the token forever
+                    with self._adopt_lock:
+                        sink = self._sinks.get(out.req_id)
+                        if sink is None:
+                            entry = self._adopted_by_rid.get(out.req_id)
+                            if entry is not None:
+                                entry["buf"].append(out)
+                                if out.finished:
+                                    entry["done"] = True
+                            continue
                 q, loop, mode, buf, first_t = sink
                 if sink[4] is None:
                     sink[4] = now
@@ -274,19 +292,134 @@ class EngineWorker(Worker):
         else:
             self.engine.abort(rid)
 
+    # ------------------------------------------------------ live migration
+    def _find_rid(self, request_id: str) -> Optional[str]:
+        for rid in list(self._sinks):
+            if rid.rsplit("-", 1)[0] == request_id:
+                return rid
+        with self._adopt_lock:
+            e = self._adopted.get(request_id)
+        return e["rid"] if e else None
+
+    def list_requests(self) -> list:
+        """Gateway-side request ids of everything live on this worker
+        (attached sinks + adopted-unattached) — the evacuation list for
+        drain-with-migration."""
+        ids = {rid.rsplit("-", 1)[0] for rid in self._sinks}
+        with self._adopt_lock:
+            ids |= set(self._adopted)
+        return sorted(ids)
+
+    async def migrate_out(self, request_id: str) -> bytes:
+        """Extract a live request (tokens + sampling identity + KV) and
+        return it serialized; the blocked generate/stream for it raises
+        WorkerMigrated so the gateway re-routes.  TP pools are excluded
+        (lockstep state lives on every rank)."""
+        if self.tp_control is not None:
+            raise WorkerError("live migration is not supported on TP pools")
+        rid = self._find_rid(request_id)
+        if rid is None:
+            raise WorkerError(f"no live request {request_id!r}")
+        self.engine.request_extract(rid)
+        self._work_event.set()
+        state = None
+        for _ in range(250):
+            state = self.engine.take_extracted(rid)
+            if state is not None or rid not in self._sinks:
+                break
+            await asyncio.sleep(0.02)
+        if state is None:
+            raise WorkerError(f"extraction of {request_id!r} timed out")
+        sink = self._sinks.get(rid)
+        if sink is not None:
+            sink[0].put_nowait(WorkerMigrated(
+                f"request {request_id} migrated off {self.device}"))
+        import io
+        bio = io.BytesIO()
+        torch.save(state, bio)
+        return bio.getvalue()
+
+    async def migrate_in(self, blob: bytes) -> None:
+        """Adopt a migrated request: KV blocks land in this worker's
+        cache and generation continues token-exact; outputs buffer until
+        the client re-attaches (generate/stream with the same
+        request_id)."""
+        if self.tp_control is not None:
+            raise WorkerError("live migration is not supported on TP pools")
+        import io
+        state = torch.load(io.BytesIO(blob), weights_only=False)
+        rid = state["rid"]
+        # register the catch buffer BEFORE the engine can run the
+        # adopted sequence — outputs between adoption and registration
+        # would otherwise drop on the floor (no sink, no entry)
+        entry = {"rid": rid, "pre": list(state["output_ids"]),
+                 "prompt_len": len(state["prompt_ids"]),
+                 "buf": [], "done": False}
+        with self._adopt_lock:
+            self._adopted[rid.rsplit("-", 1)[0]] = entry
+            self._adopted_by_rid[rid] = entry
+        self.engine.queue_adopt(state)
+        self._work_event.set()
+        res = None
+        for _ in range(250):
+            res = self.engine.take_adopt_result(rid)
+            if res is not None:
+                break
+            await asyncio.sleep(0.02)
+        if res is None or isinstance(res, Exception):
+            with self._adopt_lock:
+                self._adopted.pop(rid.rsplit("-", 1)[0], None)
+                self._adopted_by_rid.pop(rid, None)
+            if res is None:
+                raise WorkerError("adoption timed out")
+            raise WorkerError(f"adoption failed: {res}")
+
+    def _attach(self, req: GenerationRequest, mode: str):
+        """Re-attach a client to an adopted request: returns the sink
+        plus the tokens generated BEFORE attachment (migrated 'pre'
+        tokens + anything buffered since adoption), or None when the
+        request_id is not adopted here."""
+        with self._adopt_lock:
+            entry = self._adopted.pop(req.request_id, None)
+            if entry is None:
+                return None
+            self._adopted_by_rid.pop(entry["rid"], None)
+            rid = entry["rid"]
+            q: asyncio.Queue = asyncio.Queue()
+            loop = asyncio.get_running_loop()
+            sink = [q, loop, mode, [], None]
+            self._sinks[rid] = sink
+            buffered = list(entry["buf"])
+        pre = list(entry["pre"])
+        if mode == "stream":
+            for out in buffered:
+                q.put_nowait(out)
+        else:
+            sink[3].extend(buffered)
+            if buffered and buffered[-1].finished:
+                q.put_nowait((list(sink[3]), time.monotonic()))
+        return rid, q, entry["prompt_len"], pre
+
     async def generate(self, req: GenerationRequest) -> GenerationResult:
         t0 = time.monotonic()
-        rid, q, n_prompt = self._enqueue(req, mode="final")
+        attached = self._attach(req, mode="final")
+        if attached is not None:
+            rid, q, n_prompt, pre = attached
+        else:
+            rid, q, n_prompt = self._enqueue(req, mode="final")
+            pre = []
         self._in_flight += 1
         finished = False
         try:
             got = await q.get()
             self._check_fault()
+            if isinstance(got, WorkerError):
+                raise got                       # typed (e.g. WorkerMigrated)
             if isinstance(got, Exception):
                 raise WorkerError(f"engine error: {got}") from got
             outs, first_t = got
             finished = True    # engine freed the sequence itself: no abort
-            token_ids = [o.token_id for o in outs]
+            token_ids = pre + [o.token_id for o in outs]
             finish = outs[-1].finish_reason or "stop"
             text = self.tokenizer.decode(token_ids)
             cut = _earliest_stop(text, req.stop)
@@ -307,12 +440,17 @@ class EngineWorker(Worker):
                 self._abort(rid)
 
     async def _stream_impl(self, req: GenerationRequest) -> AsyncIterator[GenerationChunk]:
-        rid, q, n_prompt = self._enqueue(req)
+        attached = self._attach(req, mode="stream")
+        if attached is not None:
+            rid, q, n_prompt, pre = attached
+        else:
+            rid, q, n_prompt = self._enqueue(req)
+            pre = []
         self._in_flight += 1
         finished = False
         try:
             emitted = ""
-            token_ids: list[int] = []
+            token_ids: list[int] = list(pre)
             stops = req.stop or []
             # hold back enough text that a stop sequence spanning chunk
             # boundaries is never partially emitted
@@ -320,6 +458,8 @@ class EngineWorker(Worker):
             while True:
                 out = await q.get()
                 self._check_fault()
+                if isinstance(out, WorkerError):
+                    raise out                   # typed (e.g. WorkerMigrated)
                 if isinstance(out, Exception):
                     raise WorkerError(f"engine error: {out}") from out
                 token_ids.append(out.token_id)

@@ -17,7 +17,7 @@ import msgpack
 
 from .base import (
     GenerationChunk, GenerationRequest, GenerationResult, Worker,
-    WorkerDead, WorkerError, WorkerThrottled,
+    WorkerDead, WorkerError, WorkerMigrated, WorkerThrottled,
 )
 
 _LEN = struct.Struct("<I")
@@ -107,6 +107,16 @@ class WorkerRpcServer:
             elif op == "health":
                 await reply({"id": mid, "type": "result",
                              "data": await self.worker.health()})
+            elif op == "list_requests":
+                await reply({"id": mid, "type": "result",
+                             "data": {"ids": self.worker.list_requests()}})
+            elif op == "migrate_out":
+                blob = await self.worker.migrate_out(data["request_id"])
+                await reply({"id": mid, "type": "result",
+                             "data": {"state": blob}})
+            elif op == "migrate_in":
+                await self.worker.migrate_in(data["state"])
+                await reply({"id": mid, "type": "result", "data": {}})
             elif op == "fault":
                 mode = data.get("mode", "none")
                 if mode == "kill" and self.on_kill is not None:
@@ -118,7 +128,8 @@ class WorkerRpcServer:
             else:
                 raise ValueError(f"unknown op {op!r}")
         except Exception as e:
-            kind = ("throttled" if isinstance(e, WorkerThrottled) else
+            kind = ("migrated" if isinstance(e, WorkerMigrated) else
+                    "throttled" if isinstance(e, WorkerThrottled) else
                     "dead" if isinstance(e, WorkerDead) else "error")
             try:
                 await reply({"id": mid, "type": "error",
@@ -131,6 +142,8 @@ class WorkerRpcServer:
 def _raise_remote(data: dict) -> None:
     kind = data.get("kind", "error")
     msg = data.get("message", "remote error")
+    if kind == "migrated":
+        raise WorkerMigrated(msg)
     if kind == "throttled":
         raise WorkerThrottled(msg)
     if kind == "dead":
@@ -257,6 +270,26 @@ class RpcWorkerClient(Worker):
         if msg["type"] == "error":
             _raise_remote(msg["data"])
         return msg["data"]
+
+    async def list_requests(self) -> list:
+        _, q = await self._call("list_requests", {})
+        msg = await asyncio.wait_for(q.get(), timeout=30.0)
+        if msg["type"] == "error":
+            _raise_remote(msg["data"])
+        return msg["data"]["ids"]
+
+    async def migrate_out(self, request_id: str) -> bytes:
+        _, q = await self._call("migrate_out", {"request_id": request_id})
+        msg = await asyncio.wait_for(q.get(), timeout=60.0)
+        if msg["type"] == "error":
+            _raise_remote(msg["data"])
+        return msg["data"]["state"]
+
+    async def migrate_in(self, blob: bytes) -> None:
+        _, q = await self._call("migrate_in", {"state": blob})
+        msg = await asyncio.wait_for(q.get(), timeout=60.0)
+        if msg["type"] == "error":
+            _raise_remote(msg["data"])
 
     async def inject_fault(self, mode: str) -> None:
         _, q = await self._call("fault", {"mode": mode})
