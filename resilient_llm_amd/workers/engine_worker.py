@@ -448,15 +448,26 @@ class EngineWorker(Worker):
             pre = []
         self._in_flight += 1
         finished = False
+        # migrated 'pre' tokens replay as ONE CHUNK PER TOKEN so the
+        # gateway's failover replay-skip (which counts chunks == output
+        # tokens from index 0) stays in units with a fresh regeneration
+        import types as _types
+        pre_outs = [_types.SimpleNamespace(token_id=t, finished=False,
+                                           finish_reason=None) for t in pre]
+        pre_idx = 0
         try:
             emitted = ""
-            token_ids: list[int] = list(pre)
+            token_ids: list[int] = []
             stops = req.stop or []
             # hold back enough text that a stop sequence spanning chunk
             # boundaries is never partially emitted
             hold = max((len(s) - 1 for s in stops), default=0)
             while True:
-                out = await q.get()
+                if pre_idx < len(pre_outs):
+                    out = pre_outs[pre_idx]
+                    pre_idx += 1
+                else:
+                    out = await q.get()
                 self._check_fault()
                 if isinstance(out, WorkerError):
                     raise out                   # typed (e.g. WorkerMigrated)

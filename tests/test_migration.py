@@ -433,3 +433,139 @@ def test_gpu_engine_migration_token_exact():
         for o in dst.step():
             got.append(o.token_id)
     assert got == want, f"GPU migrated continuation diverged"
+
+
+def test_gateway_midstream_migration_completes_stream():
+    """Mid-SSE-stream migration: drain-with-evacuation moves the
+    streaming request; the client's stream continues and ends with full
+    usage (chunk-per-token pre replay keeps the gateway skip in sync)."""
+    import json as _json
+    import os
+    import subprocess
+    import sys
+    import tempfile
+    import threading
+    import time
+    import urllib.request
+
+    import asyncio
+    from resilient_llm_amd.client import OpenAIClient
+    from resilient_llm_amd.config import load_config
+    from resilient_llm_amd.gateway.app import GatewayApp
+    from resilient_llm_amd.gateway.http import HttpServer
+    from resilient_llm_amd.workers.base import WorkerRegistry
+    from resilient_llm_amd.workers.rpc import RpcWorkerClient
+    from tests.gateway_harness import free_port
+
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+    def spawn(sock, label):
+        env = dict(os.environ)
+        env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+        return subprocess.Popen(
+            [sys.executable, "-m", "resilient_llm_amd.workers.gpu_main",
+             "--device-label", label, "--model", "tiny", "--socket", sock,
+             "--device", "cpu", "--num-blocks", "96"], env=env)
+
+    run_dir = tempfile.mkdtemp(prefix="rlli-smig-")
+    socks = [os.path.join(run_dir, f"w{i}.sock") for i in range(2)]
+    port = free_port()
+    cfg = load_config(data={
+        "cluster": {"port": port},
+        "model_list": [
+            {"model_name": "tiny-serve",
+             "litellm_params": {"model": "gpu/0/tiny"},
+             "model_info": {"id": "gpu0/tiny"}},
+            {"model_name": "tiny-serve",
+             "litellm_params": {"model": "gpu/1/tiny"},
+             "model_info": {"id": "gpu1/tiny"}}],
+        "router_settings": {"routing_strategy": "round-robin"},
+    })
+    loop = asyncio.new_event_loop()
+    ready = threading.Event()
+    holder: dict = {}
+
+    async def main():
+        holder["stop"] = asyncio.Event()
+        registry = WorkerRegistry()
+        for i in range(2):
+            c = RpcWorkerClient(f"gpu:{i}", {"tiny"}, socks[i])
+            c.proc = spawn(socks[i], f"gpu:{i}")
+            await c.connect(timeout=120)
+            registry.register("gpu", str(i), c)
+            holder[f"w{i}"] = c
+        app = GatewayApp(cfg, registry, health_interval_s=0.5)
+        server = HttpServer(app.handle, host="127.0.0.1", port=port)
+        await server.start()
+        await app.start_background()
+        ready.set()
+        await holder["stop"].wait()
+        await app.stop_background()
+        await server.stop()
+        await registry.close()
+
+    th = threading.Thread(target=lambda: loop.run_until_complete(main()),
+                          daemon=True)
+    th.start()
+    assert ready.wait(timeout=150)
+    base = f"http://127.0.0.1:{port}"
+    client = OpenAIClient(base)
+    msgs = [{"role": "user", "content": "stream and move " * 20}]
+    events: list = []
+    usage: dict = {}
+    done = threading.Event()
+    first_dev: dict = {}
+
+    def consume():
+        stream = client.chat.completions.create(
+            model="tiny-serve", messages=msgs, max_tokens=300, stream=True,
+            timeout=120, stream_options={"include_usage": True},
+            extra_headers={"x-request-id": "smig-1"})
+        first_dev["d"] = stream.headers.get("x-gateway-device") \
+            if hasattr(stream, "headers") else None
+        for evt in stream:
+            if "error" in evt:
+                raise RuntimeError(evt["error"])
+            if evt.get("usage"):
+                usage.update(evt["usage"])
+            events.append(evt)
+        done.set()
+
+    t = threading.Thread(target=consume)
+    t.start()
+    try:
+        for _ in range(600):
+            if len(events) >= 10 or done.is_set():
+                break
+            time.sleep(0.01)
+        assert not done.is_set(), "stream finished before migration window"
+        migrated = None
+        for src, dst in (("gpu:0", "gpu:1"), ("gpu:1", "gpu:0")):
+            req = urllib.request.Request(
+                base + "/admin/drain", method="POST",
+                data=_json.dumps({"worker": src,
+                                  "migrate_to": dst}).encode(),
+                headers={"content-type": "application/json"})
+            with urllib.request.urlopen(req, timeout=60) as r:
+                body = _json.loads(r.read().decode())
+            if body["migrated"]:
+                migrated = body
+                break
+            # undrain the wrong guess
+            urllib.request.urlopen(urllib.request.Request(
+                base + "/admin/drain", method="POST",
+                data=_json.dumps({"worker": src, "drain": False}).encode(),
+                headers={"content-type": "application/json"}), timeout=30)
+        assert migrated and migrated["migrated"] == ["smig-1"], migrated
+        assert not migrated["migrate_errors"]
+        assert done.wait(timeout=120)
+        t.join(timeout=30)
+        assert usage.get("completion_tokens") == 300, usage
+    finally:
+        done.set()
+        loop.call_soon_threadsafe(holder["stop"].set)
+        th.join(timeout=30)
+        for i in range(2):
+            w = holder.get(f"w{i}")
+            if w is not None and w.proc is not None and w.proc.poll() is None:
+                w.proc.kill()
