@@ -90,6 +90,7 @@ class LLMEngine:
                  enable_prefix_caching: bool = True,
                  target_step_ms: Optional[float] = None,
                  eos_ids=None,
+                 spec_lookup: int = 0,
                  seed: int = 0) -> None:
         self.model = model
         self.kv = kv_cache
@@ -127,6 +128,14 @@ class LLMEngine:
         # under TP lockstep — wall-clock decisions diverge across ranks
         # (same rule as admission_window_s).
         self.target_step_ms = target_step_ms
+        # prompt-lookup speculative decoding (opt-in, greedy-only):
+        # propose up to N continuation tokens by n-gram match against
+        # the sequence's own context and verify them in ONE multi-row
+        # forward (the chunked-prefill machinery — proposals attend the
+        # paged prefix + themselves causally).  Rejected proposals cost
+        # nothing to roll back: their KV slots are simply rewritten when
+        # the real tokens are fed.
+        self.spec_lookup = spec_lookup
         self._prefill_budget = max_prefill_tokens
         self._budget_floor = min(max_prefill_tokens, 256)
         self.step_count = 0
@@ -236,6 +245,10 @@ class LLMEngine:
                 max_reuse = (len(seq.prompt_ids) - 1) // self.block_size
                 reused = self.kv.lookup_prefix(seq.block_keys[:max_reuse])
             needed = seq.blocks_needed(self.block_size) - len(reused)
+            if self.spec_lookup:
+                # speculative verification writes KV up to spec_lookup
+                # positions past the committed length
+                needed += -(-self.spec_lookup // self.block_size)
             if needed > self.kv.free_blocks:
                 if reused:
                     self.kv.free(reused)
@@ -312,7 +325,13 @@ class LLMEngine:
             if _ROCTX:
                 torch.cuda.nvtx.range_push(f"decode b{len(self.running)}")
             nb = len(self.running)
-            outs += self._decode_step()
+            spec_outs = None
+            if self.spec_lookup > 0 and self._spec_eligible():
+                spec_outs = self._decode_step_spec()
+            if spec_outs is not None:
+                outs += spec_outs
+            else:
+                outs += self._decode_step()
             if _ROCTX:
                 torch.cuda.nvtx.range_pop()
             self.stats["decode_steps"] += 1
@@ -678,6 +697,110 @@ class LLMEngine:
         self.prefilling = [s for s in self.prefilling if id(s) not in done]
         self._batch_dirty = True
         return n_chunk_tokens, outs
+
+    # -------------------------------------- speculative decoding (lookup)
+    def _spec_eligible(self) -> bool:
+        return self._pending is None and all(
+            s.params.temperature == 0 and not s.params.needs_torch_sampling
+            for s in self.running)
+
+    @staticmethod
+    def _propose_lookup(seq: SeqState, k: int) -> list:
+        """Prompt-lookup: continuation after the most recent prior
+        occurrence of the context's last bigram."""
+        if k <= 0:
+            return []
+        ctx = seq.prompt_ids + seq.output_ids
+        if len(ctx) < 4:
+            return []
+        a, b = ctx[-2], ctx[-1]
+        for i in range(len(ctx) - 3, 0, -1):
+            if ctx[i - 1] == a and ctx[i] == b:
+                return list(ctx[i + 1:i + 1 + k])
+        return []
+
+    def _decode_step_spec(self) -> Optional[list]:
+        seqs = self.running
+        dev = self.device
+        proposals = []
+        any_prop = False
+        for s in seqs:
+            cap = min(self.spec_lookup,
+                      s.params.max_tokens - len(s.output_ids) - 1)
+            prop = self._propose_lookup(s, cap)
+            proposals.append(prop)
+            any_prop = any_prop or bool(prop)
+        if not any_prop:
+            return None                     # caller runs the normal step
+
+        input_ids: list = []
+        positions: list = []
+        slots: list = []
+        c_row0, c_pos0, c_nrows, c_btrow = [], [], [], []
+        bt_rows: list[list] = []
+        metas = []
+        row = 0
+        for s, prop in zip(seqs, proposals):
+            toks = [s.output_ids[-1]] + prop
+            btr = len(bt_rows)
+            bt_rows.append(s.blocks)
+            c_row0.append(row)
+            c_pos0.append(s.n_cached)
+            c_nrows.append(len(toks))
+            c_btrow.append(btr)
+            input_ids.extend(toks)
+            positions.extend(range(s.n_cached, s.n_cached + len(toks)))
+            slots.extend(self._slot(s, s.n_cached + j)
+                         for j in range(len(toks)))
+            metas.append((s, prop, row, len(toks)))
+            row += len(toks)
+
+        def i32(x):
+            return torch.tensor(x, dtype=torch.int32, device=dev)
+
+        def bt_tensor(rows):
+            w = max((len(r) for r in rows), default=1)
+            t = torch.zeros((max(len(rows), 1), w), dtype=torch.int32)
+            for i, r in enumerate(rows):
+                t[i, :len(r)] = torch.tensor(r, dtype=torch.int32)
+            return t.to(dev)
+
+        logits = self.model.forward_mixed(
+            i32(input_ids), i32(positions), self.kv, i32(slots), 0,
+            None, None, i32(c_row0), i32(c_pos0), i32(c_nrows), i32(c_btrow),
+            bt_tensor(bt_rows),
+            torch.arange(row, dtype=torch.long, device=dev))
+        greedy = logits.argmax(-1).tolist()           # host sync (spec mode)
+
+        outs: list[StepOutput] = []
+        still: list[SeqState] = []
+        accepted_total = 0
+        for s, prop, r0, n in metas:
+            g = greedy[r0:r0 + n]
+            a = 0
+            while a < len(prop) and prop[a] == g[a]:
+                a += 1
+            accepted_total += a
+            finished = False
+            for t in prop[:a] + [g[a]]:
+                s.output_ids.append(t)
+                if self._finish(s, outs, t):
+                    finished = True
+                    break
+            # KV is real through the last ACCEPTED fed position; the
+            # rejected proposal's slot gets rewritten when the real
+            # token is fed next step
+            s.n_cached += 1 + a
+            if not finished:
+                still.append(s)
+        self.running = still
+        self.stats["decode_tokens"] += len(outs)
+        self.stats.setdefault("spec_steps", 0)
+        self.stats.setdefault("spec_accepted", 0)
+        self.stats["spec_steps"] += 1
+        self.stats["spec_accepted"] += accepted_total
+        self._batch_dirty = True           # positions advanced multi-token
+        return outs
 
     def _rebuild_batch(self) -> None:
         """(Re)build persistent device tensors for the running batch.

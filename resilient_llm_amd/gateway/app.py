@@ -149,6 +149,14 @@ class GatewayApp:
             except asyncio.CancelledError:
                 pass
             self._health_task = None
+        for t in list(self._restart_tasks.values()):
+            if not t.done():
+                t.cancel()
+                try:
+                    await t
+                except asyncio.CancelledError:
+                    pass
+        self._restart_tasks.clear()
 
     def _worker_for(self, ticket: Ticket) -> Worker:
         d = ticket.deployment

@@ -79,6 +79,7 @@ class EngineWorker(Worker):
                  eos_id=None,
                  kv_dtype: str = "bf16",
                  quant=None,
+                 spec_lookup: int = 0,
                  seed: int = 0) -> None:
         super().__init__(device=device_label or f"gpu:{device}",
                          models={model_name})
@@ -136,7 +137,8 @@ class EngineWorker(Worker):
                                 # wall-clock tuning diverges under TP
                                 # lockstep, like admission_window_s
                                 target_step_ms=None if tp_control is not None
-                                else target_step_ms)
+                                else target_step_ms,
+                                spec_lookup=spec_lookup)
         if use_graphs and torch_device != "cpu":
             from ..engine.graph import install_graph_runner
             install_graph_runner(self.engine)

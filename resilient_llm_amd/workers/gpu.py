@@ -35,6 +35,7 @@ def spawn_gpu_worker(device_index: str, model_name: str,
                      eos_ids: list | None = None,
                      kv_dtype: str | None = None,
                      quant: str | None = None,
+                     spec_lookup: int | None = None,
                      extra_env: dict | None = None) -> subprocess.Popen:
     """``device_index`` may be a sub-device replica like ``0.1``: several
     worker processes co-located on physical GPU 0 — 288 GB of HBM3E holds
@@ -63,6 +64,8 @@ def spawn_gpu_worker(device_index: str, model_name: str,
         cmd += ["--kv-dtype", str(kv_dtype)]
     if quant:
         cmd += ["--quant", str(quant)]
+    if spec_lookup:
+        cmd += ["--spec-lookup", str(spec_lookup)]
     return subprocess.Popen(cmd, env=env)
 
 
@@ -113,6 +116,7 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
         weights = extra.get("weights")
         kv_dtype = extra.get("kv_dtype")
         quant = extra.get("quantization")
+        spec = extra.get("spec_lookup")
         eos = extra.get("eos_id")
         eos_list = None if eos is None else (
             [int(eos)] if isinstance(eos, int) else [int(e) for e in eos])
@@ -120,7 +124,7 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
                                 max_batch=max_batch, use_graphs=use_graphs,
                                 weights=weights, eos_ids=eos_list,
                                 kv_dtype=kv_dtype, quant=quant,
-                                target_step_ms=slo)
+                                spec_lookup=spec, target_step_ms=slo)
         client = RpcWorkerClient(f"gpu:{target}", {model_name}, sock)
         client.proc = proc
         client.respawn = (lambda t=target, m=model_name, s=sock:
@@ -129,6 +133,7 @@ def register_gpu_workers(config: Config, registry: WorkerRegistry,
                                            use_graphs=use_graphs,
                                            weights=weights, eos_ids=eos_list,
                                            kv_dtype=kv_dtype, quant=quant,
+                                           spec_lookup=spec,
                                            target_step_ms=slo))
         registry.register("gpu", target, client)
 

@@ -26,6 +26,9 @@ def main() -> None:
                          "when absent")
     ap.add_argument("--eos-id", type=int, action="append", default=None,
                     help="override EOS token id(s); repeatable")
+    ap.add_argument("--spec-lookup", type=int, default=0,
+                    help="prompt-lookup speculative decoding depth "
+                         "(greedy requests; 0 = off)")
     ap.add_argument("--quant", default=None, choices=(None, "fp8"),
                     help="W8A8-fp8 weight quantization for gate_up/down/"
                          "lm_head (rowwise scaled_mm)")
@@ -64,6 +67,7 @@ def main() -> None:
                               eos_id=args.eos_id,
                               kv_dtype=args.kv_dtype,
                               quant=args.quant,
+                              spec_lookup=args.spec_lookup,
                               target_step_ms=args.target_step_ms)
         log_with_timestamp(
             f"worker {args.device_label} ready: {args.model} on {device}, "

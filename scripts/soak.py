@@ -20,6 +20,9 @@ spec.loader.exec_module(common)
 
 from resilient_llm_amd.client import APIError, OpenAIClient  # noqa: E402
 
+# RLLI_SOAK_MIGRATE=1: every ~15 s, live-migrate EVERYTHING off one
+# replica onto the other (drain with migrate_to) and undrain — layered
+# on top of the kill/respawn chaos
 def _mode_opts():
     """RLLI_SOAK_MODE=fp8 soaks the quantized-weight + fp8-KV modes."""
     if os.environ.get("RLLI_SOAK_MODE") == "fp8":
@@ -94,6 +97,20 @@ def main(duration_s=420, n_threads=24):
             t.start()
         t0 = time.time()
         killed = 0
+        migrations = 0
+        mig_on = os.environ.get("RLLI_SOAK_MIGRATE") == "1"
+        last_mig = 0.0
+        import json as _json
+        import urllib.request as _url
+
+        def _post(path, body):
+            r = _url.Request(
+                f"http://127.0.0.1:{config.cluster.port}{path}",
+                method="POST", data=_json.dumps(body).encode(),
+                headers={"content-type": "application/json"})
+            with _url.urlopen(r, timeout=60) as resp:
+                return _json.loads(resp.read().decode())
+
         while time.time() - t0 < duration_s:
             time.sleep(5)
             el = time.time() - t0
@@ -105,6 +122,22 @@ def main(duration_s=420, n_threads=24):
                 except APIError:
                     pass
                 killed += 1
+            if mig_on and el - last_mig > 15:
+                last_mig = el
+                src = "gpu:0" if migrations % 2 == 0 else "gpu:0.1"
+                dst = "gpu:0.1" if src == "gpu:0" else "gpu:0"
+                try:
+                    body = _post("/admin/drain", {"worker": src,
+                                                  "migrate_to": dst})
+                    _post("/admin/drain", {"worker": src, "drain": False})
+                    migrations += 1
+                    print(f"[soak {el:.0f}s] migrated "
+                          f"{len(body['migrated'])} reqs {src}->{dst} "
+                          f"(errors: {len(body['migrate_errors'])})",
+                          flush=True)
+                except Exception as e:             # noqa: BLE001
+                    print(f"[soak {el:.0f}s] migration sweep failed: {e}",
+                          flush=True)
             with lock:
                 print(f"[soak {el:.0f}s] {dict(stats)}", flush=True)
         stop.set()
@@ -117,7 +150,8 @@ def main(duration_s=420, n_threads=24):
         total = sum(stats.values())
         ok = stats["ok"] + stats["stream_ok"]
         print(f"SOAK DONE: {ok}/{total} ok ({100*ok/max(total,1):.1f}%), "
-              f"2 worker kills+respawns; in_flight={inflight}; "
+              f"{killed} kills+respawns, {migrations} migration sweeps; "
+              f"in_flight={inflight}; "
               f"healthy={healthy}", flush=True)
 
 
