@@ -457,7 +457,13 @@ class LLMEngine:
                     break
             with self._queue_lock:
                 self._live.discard(rid)
-                self._extracted[rid] = state   # None = rid not found
+                # "missing" (not None): a finished/unknown rid must be
+                # distinguishable from not-yet-extracted, or the
+                # worker's poll waits its full timeout for every
+                # request that finished just before the sweep reached
+                # it — which is exactly what a drain under load hits
+                self._extracted[rid] = state if state is not None \
+                    else "missing"
 
     def _do_adopts(self) -> None:
         with self._queue_lock:

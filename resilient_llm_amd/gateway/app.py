@@ -839,16 +839,24 @@ class GatewayApp:
         errors: list = []
         target_dev = body.get("migrate_to")
         if draining and target_dev:
-            # evacuate live requests to the target: zero-recompute drain
+            # evacuate live requests to the target: zero-recompute
+            # drain, BOUNDED by timeout_s — a slow/contended worker must
+            # never hang the admin call (requests left behind simply
+            # finish in place, which a drain tolerates by definition)
             target = self.registry.all().get(target_dev)
             if target is None:
                 return Response.error(404, f"no worker {target_dev!r}")
+            deadline = time.monotonic() + float(body.get("timeout_s", 30.0))
             try:
                 ids = await worker.list_requests()
             except (WorkerError, AttributeError) as e:
                 ids = []
                 errors.append(str(e))
             for req_id in ids:
+                if time.monotonic() > deadline:
+                    errors.append(f"evacuation deadline: "
+                                  f"{len(ids) - len(migrated)} left in place")
+                    break
                 try:
                     blob = await worker.migrate_out(req_id)
                     await target.migrate_in(blob)

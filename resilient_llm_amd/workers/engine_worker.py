@@ -327,11 +327,14 @@ class EngineWorker(Worker):
         state = None
         for _ in range(250):
             state = self.engine.take_extracted(rid)
-            if state is not None or rid not in self._sinks:
+            if state is not None:
                 break
             await asyncio.sleep(0.02)
         if state is None:
             raise WorkerError(f"extraction of {request_id!r} timed out")
+        if state == "missing":
+            raise WorkerError(f"no live request {request_id!r} "
+                              "(finished before extraction)")
         sink = self._sinks.get(rid)
         if sink is not None:
             sink[0].put_nowait(WorkerMigrated(

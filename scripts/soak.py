@@ -127,9 +127,15 @@ def main(duration_s=420, n_threads=24):
                 src = "gpu:0" if migrations % 2 == 0 else "gpu:0.1"
                 dst = "gpu:0.1" if src == "gpu:0" else "gpu:0"
                 try:
-                    body = _post("/admin/drain", {"worker": src,
-                                                  "migrate_to": dst})
-                    _post("/admin/drain", {"worker": src, "drain": False})
+                    try:
+                        body = _post("/admin/drain", {"worker": src,
+                                                      "migrate_to": dst})
+                    finally:
+                        # NEVER leave a replica draining: a failed sweep
+                        # plus a later sweep of the OTHER replica would
+                        # drain the whole pool
+                        _post("/admin/drain", {"worker": src,
+                                               "drain": False})
                     migrations += 1
                     print(f"[soak {el:.0f}s] migrated "
                           f"{len(body['migrated'])} reqs {src}->{dst} "

@@ -84,11 +84,14 @@ def test_extract_waiting_request_re_prefills():
     assert len(toks) == 5
 
 
-def test_extract_unknown_rid_reports_none():
+def test_extract_unknown_rid_reports_missing():
     e = fresh_engine()
     e.request_extract("ghost")
     e.step()
-    assert e.take_extracted("ghost") is None
+    # explicit sentinel: a finished/unknown rid must be distinguishable
+    # from not-yet-extracted (a None would make the worker poll its
+    # full timeout for every request that finished before the sweep)
+    assert e.take_extracted("ghost") == "missing"
 
 
 # ---------------------------------------------------- worker-level
