@@ -849,9 +849,10 @@ class GatewayApp:
             deadline = time.monotonic() + float(body.get("timeout_s", 30.0))
             try:
                 ids = await worker.list_requests()
-            except (WorkerError, AttributeError) as e:
+            except (WorkerError, AttributeError,
+                    asyncio.TimeoutError) as e:
                 ids = []
-                errors.append(str(e))
+                errors.append(str(e) or "list_requests timed out")
             for req_id in ids:
                 if time.monotonic() > deadline:
                     errors.append(f"evacuation deadline: "
@@ -859,10 +860,16 @@ class GatewayApp:
                     break
                 try:
                     blob = await worker.migrate_out(req_id)
-                    await target.migrate_in(blob)
-                    migrated.append(req_id)
-                except WorkerError as e:
-                    errors.append(f"{req_id}: {e}")
+                    try:
+                        await target.migrate_in(blob)
+                        migrated.append(req_id)
+                    finally:
+                        # release the blocked client only once the
+                        # state landed (or demonstrably failed): the
+                        # re-routed attach then always finds it
+                        await worker.release_migrated(req_id)
+                except (WorkerError, asyncio.TimeoutError) as e:
+                    errors.append(f"{req_id}: {e!r}")
         return Response.json_response({
             "worker": device, "draining": draining,
             "migrated": migrated, "migrate_errors": errors,
@@ -888,9 +895,12 @@ class GatewayApp:
                                        f"{dst_dev!r} (have {sorted(workers)})")
         try:
             blob = await src.migrate_out(request_id)
-            await dst.migrate_in(blob)
-        except WorkerError as e:
-            return Response.error(409, f"migration failed: {e}")
+            try:
+                await dst.migrate_in(blob)
+            finally:
+                await src.release_migrated(request_id)
+        except (WorkerError, asyncio.TimeoutError) as e:
+            return Response.error(409, f"migration failed: {e!r}")
         return Response.json_response({"request_id": request_id,
                                        "from": src_dev, "to": dst_dev,
                                        "state_bytes": len(blob)})

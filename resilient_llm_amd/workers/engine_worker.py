@@ -167,6 +167,42 @@ class EngineWorker(Worker):
                                             daemon=True,
                                             name=f"engine-{self.device}")
             self._thread.start()
+        # stall watchdog: if the engine has work but step_count hasn't
+        # moved for RLLI_STEP_WATCHDOG_S, dump every thread's stack to
+        # stderr ONCE — turns a silent GPU/thread wedge into a
+        # diagnosable trace (ops feature; cheap: one counter compare/s)
+        wd_s = float(os.environ.get("RLLI_STEP_WATCHDOG_S", "60"))
+        if wd_s > 0 and (tp_control is None or tp_control.is_leader):
+            def _watchdog():
+                import faulthandler
+                import sys as _sys
+                last = -1
+                stuck_since = None
+                dumped = False
+                while not self._stop:
+                    time.sleep(1.0)
+                    sc = self.engine.step_count
+                    if not self.engine.has_work():
+                        stuck_since = None
+                        dumped = False
+                    elif sc != last:
+                        stuck_since = None
+                        dumped = False
+                    elif stuck_since is None:
+                        stuck_since = time.monotonic()
+                    elif (not dumped
+                          and time.monotonic() - stuck_since > wd_s):
+                        print(f"[watchdog] {self.device}: engine stalled "
+                              f">{wd_s:.0f}s at step {sc} "
+                              f"(waiting={len(self.engine.waiting)} "
+                              f"prefilling={len(self.engine.prefilling)} "
+                              f"running={len(self.engine.running)})",
+                              file=_sys.stderr, flush=True)
+                        faulthandler.dump_traceback(file=_sys.stderr)
+                        dumped = True
+                    last = sc
+            threading.Thread(target=_watchdog, daemon=True,
+                             name=f"watchdog-{self.device}").start()
 
     # --------------------------------------------------------- engine loop
     # TP lockstep heartbeat: followers BLOCK inside the control-group
@@ -199,6 +235,11 @@ class EngineWorker(Worker):
             try:
                 outputs = self.engine.step()
             except Exception as e:  # engine-level failure -> fail all in flight
+                import traceback
+                from ..utils.logging import log_with_timestamp
+                log_with_timestamp(
+                    f"{self.device}: engine step failed: {e!r}", "red")
+                traceback.print_exc()
                 self._broadcast_error(e)
                 continue
             now = time.monotonic()
@@ -335,14 +376,31 @@ class EngineWorker(Worker):
         if state == "missing":
             raise WorkerError(f"no live request {request_id!r} "
                               "(finished before extraction)")
-        sink = self._sinks.get(rid)
+        # NOTE: the blocked client is NOT released here — the caller
+        # releases it via release_migrated() AFTER the target has
+        # adopted, so the re-routed attach always finds the state
+        # (releasing first raced the adoption and caused duplicate
+        # regeneration on the target)
+        import io
+
+        def _pack():
+            bio = io.BytesIO()
+            torch.save(state, bio)
+            return bio.getvalue()
+        # serialize OFF the event loop: a multi-MB torch.save would
+        # stall health/stream RPCs and flap this worker unhealthy
+        return await asyncio.to_thread(_pack)
+
+    async def release_migrated(self, request_id: str) -> None:
+        """Raise WorkerMigrated into the request's blocked
+        generate/stream so the gateway re-routes (called once the
+        target holds the state — or on migration failure, in which
+        case the re-route regenerates identically from the seed)."""
+        rid = self._find_rid(request_id)
+        sink = self._sinks.get(rid) if rid else None
         if sink is not None:
             sink[0].put_nowait(WorkerMigrated(
                 f"request {request_id} migrated off {self.device}"))
-        import io
-        bio = io.BytesIO()
-        torch.save(state, bio)
-        return bio.getvalue()
 
     async def migrate_in(self, blob: bytes) -> None:
         """Adopt a migrated request: KV blocks land in this worker's
@@ -352,7 +410,8 @@ class EngineWorker(Worker):
         if self.tp_control is not None:
             raise WorkerError("live migration is not supported on TP pools")
         import io
-        state = torch.load(io.BytesIO(blob), weights_only=False)
+        state = await asyncio.to_thread(
+            lambda: torch.load(io.BytesIO(blob), weights_only=False))
         rid = state["rid"]
         # register the catch buffer BEFORE the engine can run the
         # adopted sequence — outputs between adoption and registration

@@ -117,6 +117,9 @@ class WorkerRpcServer:
             elif op == "migrate_in":
                 await self.worker.migrate_in(data["state"])
                 await reply({"id": mid, "type": "result", "data": {}})
+            elif op == "release_migrated":
+                await self.worker.release_migrated(data["request_id"])
+                await reply({"id": mid, "type": "result", "data": {}})
             elif op == "fault":
                 mode = data.get("mode", "none")
                 if mode == "kill" and self.on_kill is not None:
@@ -288,6 +291,13 @@ class RpcWorkerClient(Worker):
     async def migrate_in(self, blob: bytes) -> None:
         _, q = await self._call("migrate_in", {"state": blob})
         msg = await asyncio.wait_for(q.get(), timeout=60.0)
+        if msg["type"] == "error":
+            _raise_remote(msg["data"])
+
+    async def release_migrated(self, request_id: str) -> None:
+        _, q = await self._call("release_migrated",
+                                {"request_id": request_id})
+        msg = await asyncio.wait_for(q.get(), timeout=30.0)
         if msg["type"] == "error":
             _raise_remote(msg["data"])
 

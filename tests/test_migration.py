@@ -122,9 +122,10 @@ def test_worker_migrate_out_in_token_exact():
         task = asyncio.create_task(a.generate(dataclasses_replace(req)))
         await asyncio.sleep(0.05)          # let a few tokens generate
         blob = await a.migrate_out("mig-1")
+        await b.migrate_in(blob)
+        await a.release_migrated("mig-1")  # adopt-then-release protocol
         with pytest.raises(WorkerMigrated):
             await task
-        await b.migrate_in(blob)
         await asyncio.sleep(0.15)          # generation continues unattached
         res = await b.generate(dataclasses_replace(req))
         assert res.text == want, f"{res.text!r} != {want!r}"
