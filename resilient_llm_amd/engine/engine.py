@@ -410,6 +410,8 @@ class LLMEngine:
             return self._adopt_results.pop(req_id, None)
 
     def _do_extracts(self) -> None:
+        if not self._extract_reqs:      # racy-read fast path: a newly
+            return                      # queued rid lands next step
         with self._queue_lock:
             if not self._extract_reqs:
                 return
@@ -466,6 +468,8 @@ class LLMEngine:
                     else "missing"
 
     def _do_adopts(self) -> None:
+        if not self._adopt_queue:       # racy-read fast path
+            return
         with self._queue_lock:
             if not self._adopt_queue:
                 return
