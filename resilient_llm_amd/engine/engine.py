@@ -183,6 +183,16 @@ class LLMEngine:
     def add_request(self, req_id: str, prompt_ids: list,
                     params: Optional[SamplingParams] = None) -> None:
         params = params or SamplingParams()
+        # out-of-vocab ids would fault the GPU inside the embedding
+        # gather (hardware exception, not a python error) — reject at
+        # the API boundary instead
+        vocab = getattr(self.model.config, "vocab_size", None)
+        if vocab and prompt_ids:
+            lo, hi = min(prompt_ids), max(prompt_ids)
+            if lo < 0 or hi >= vocab:
+                raise ValueError(
+                    f"prompt token id out of range [0, {vocab}): "
+                    f"min={lo} max={hi}")
         if len(self.waiting) >= self.max_queue:
             raise CapacityExceeded(f"queue full ({self.max_queue})")
         seq = SeqState(req_id=req_id, prompt_ids=list(prompt_ids), params=params)

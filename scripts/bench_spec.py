@@ -48,13 +48,14 @@ def main():
                              seed=0)
     # code-edit-shaped prompt: a block of structured text repeated with
     # small variations — the regime lookup decoding exists for
-    block = list(range(100, 140))
+    v = cfg.vocab_size
+    block = [t % v for t in range(100, 140)]
     prompts = []
     for i in range(args.batch):
         p = []
         for rep in range(6):
             p.extend(block)
-            p.append(500 + (i * 7 + rep) % 100)
+            p.append((400 + i * 7 + rep) % v)
         prompts.append(p)
     base = run(model, cfg, 0, prompts, args.max_tokens, args.batch)
     fast = run(model, cfg, args.spec, prompts, args.max_tokens, args.batch)

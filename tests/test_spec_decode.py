@@ -95,3 +95,38 @@ def test_spec_respects_max_tokens_and_eos():
     outs = drain(e)["a"]
     assert len(outs) == 5
     assert e.kv.free_blocks == e.kv.num_blocks
+
+
+def test_out_of_vocab_prompt_rejected():
+    """OOV ids must raise at the API boundary, not fault the GPU inside
+    the embedding gather (found by a GPU HSA exception)."""
+    e = eng()
+    with pytest.raises(ValueError):
+        e.add_request("bad", [5, 6, _CFG.vocab_size + 3],
+                      SamplingParams(max_tokens=4))
+    with pytest.raises(ValueError):
+        e.add_request("neg", [5, -1], SamplingParams(max_tokens=4))
+
+
+@pytest.mark.gpu
+def test_gpu_spec_token_exact():
+    """GPU spec path (chunked-prefill verification on HIP kernels) must
+    equal the plain greedy run token for token."""
+    cfg = get_config("tiny-128")
+    model = LlamaForCausalLM(cfg, device="cuda:0", dtype=torch.bfloat16,
+                             seed=7)
+
+    def gpu_eng(spec):
+        kv = PagedKVCache.for_model(cfg, 128, device="cuda:0")
+        return LLMEngine(model, kv, max_batch_size=4, spec_lookup=spec)
+
+    prompt = ([7, 8, 9, 10] * 12) + [3, 4]
+    base = gpu_eng(0)
+    base.add_request("a", prompt, SamplingParams(max_tokens=32,
+                                                 stop_on_eos=False))
+    want = drain(base)["a"]
+    spec = gpu_eng(4)
+    spec.add_request("a", prompt, SamplingParams(max_tokens=32,
+                                                 stop_on_eos=False))
+    got = drain(spec)["a"]
+    assert got == want
