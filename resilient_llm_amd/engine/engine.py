@@ -431,6 +431,28 @@ class LLMEngine:
             state = None
             for attr in ("running", "prefilling", "waiting"):
                 pool = getattr(self, attr)
+                if attr == "waiting":
+                    # waiting is the cross-thread surface (add_request
+                    # appends under the queue lock) — mutate it there
+                    with self._queue_lock:
+                        seq = next((x for x in pool if x.req_id == rid),
+                                   None)
+                        if seq is not None:
+                            pool.remove(seq)
+                    if seq is None:
+                        continue
+                    state = {"rid": rid,
+                             "prompt_ids": list(seq.prompt_ids),
+                             "output_ids": [],
+                             "params": dataclasses.asdict(seq.params),
+                             "default_seed": seq.default_seed,
+                             "n_cached": 0,
+                             "block_size": self.block_size,
+                             "kv": None}
+                    if seq.blocks:
+                        self.kv.free(seq.blocks)
+                        seq.blocks = []
+                    break
                 for seq in list(pool):
                     if seq.req_id != rid:
                         continue
@@ -500,8 +522,10 @@ class LLMEngine:
                           and state.get("block_size") == self.block_size
                           and seq.output_ids
                           and seq.n_cached >= len(seq.prompt_ids))
-                seq.blocks = self.kv.allocate(
-                    seq.blocks_needed(self.block_size))
+                nb_need = seq.blocks_needed(self.block_size)
+                if self.spec_lookup:
+                    nb_need += -(-self.spec_lookup // self.block_size)
+                seq.blocks = self.kv.allocate(nb_need)
                 if usable:
                     k_src, v_src = kv_pair
                     nb = k_src.shape[1]

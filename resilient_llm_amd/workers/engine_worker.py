@@ -418,8 +418,18 @@ class EngineWorker(Worker):
         # would otherwise drop on the floor (no sink, no entry)
         entry = {"rid": rid, "pre": list(state["output_ids"]),
                  "prompt_len": len(state["prompt_ids"]),
-                 "buf": [], "done": False}
+                 "buf": [], "done": False, "t": time.monotonic()}
         with self._adopt_lock:
+            # prune abandoned adoptions (client never re-attached):
+            # finished entries older than 10 min, or oldest when large
+            now = time.monotonic()
+            stale = [k for k, e in self._adopted.items()
+                     if (e.get("done") and now - e.get("t", now) > 600)
+                     or len(self._adopted) > 1024]
+            for k in stale:
+                e = self._adopted.pop(k, None)
+                if e:
+                    self._adopted_by_rid.pop(e["rid"], None)
             self._adopted[rid.rsplit("-", 1)[0]] = entry
             self._adopted_by_rid[rid] = entry
         self.engine.queue_adopt(state)
