@@ -922,19 +922,44 @@ class GatewayApp:
             return Response.error(409, f"worker {device} is not "
                                        "process-backed (no respawn hook)")
         timeout_s = float(body.get("timeout_s", 120.0))
+        migrate_to = body.get("migrate_to")
+        if migrate_to and migrate_to not in self.registry.all():
+            return Response.error(404, f"no worker {migrate_to!r}")
         task = asyncio.get_running_loop().create_task(
-            self._rolling_restart(device, worker, timeout_s))
+            self._rolling_restart(device, worker, timeout_s, migrate_to))
         self._restart_tasks[device] = task
         return Response.json_response({"worker": device,
-                                       "status": "restarting"}, status=202)
+                                       "status": "restarting",
+                                       "migrate_to": migrate_to}, status=202)
 
     async def _rolling_restart(self, device: str, worker: Worker,
-                               timeout_s: float) -> None:
+                               timeout_s: float,
+                               migrate_to: Optional[str] = None) -> None:
         from ..utils.logging import log_with_timestamp, sanitize_error
         log_with_timestamp(f"rolling restart of {device}: draining", "yellow")
         self._set_drain(worker, True)
         t0 = time.monotonic()
         try:
+            if migrate_to:
+                # evacuate instead of waiting for long generations:
+                # zero-recompute restart (failures fall back to the
+                # in-flight wait below, then stateless replay)
+                target = self.registry.all().get(migrate_to)
+                try:
+                    ids = await worker.list_requests()
+                except Exception:                      # noqa: BLE001
+                    ids = []
+                for req_id in ids:
+                    if time.monotonic() - t0 > timeout_s / 2:
+                        break
+                    try:
+                        blob = await worker.migrate_out(req_id)
+                        try:
+                            await target.migrate_in(blob)
+                        finally:
+                            await worker.release_migrated(req_id)
+                    except (WorkerError, asyncio.TimeoutError):
+                        pass
             while (self._worker_in_flight(worker) > 0
                    and time.monotonic() - t0 < timeout_s):
                 await asyncio.sleep(0.1)

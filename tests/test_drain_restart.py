@@ -194,3 +194,95 @@ def test_rolling_restart_zero_downtime():
             p = holder.get(f"w{i}")
             if p is not None and p.proc is not None and p.proc.poll() is None:
                 p.proc.kill()
+
+
+def test_rolling_restart_with_evacuation():
+    """/admin/restart {migrate_to}: in-flight requests evacuate to the
+    peer (no drain wait, no recompute) and the worker restarts with a
+    new pid; all requests succeed."""
+    run_dir = tempfile.mkdtemp(prefix="rlli-restart-evac-")
+    socks = [os.path.join(run_dir, f"w{i}.sock") for i in range(2)]
+    port = free_port()
+    cfg = load_config(data={
+        "cluster": {"port": port},
+        "model_list": [
+            {"model_name": "tiny-serve",
+             "litellm_params": {"model": "gpu/0/tiny"},
+             "model_info": {"id": "gpu0/tiny"}},
+            {"model_name": "tiny-serve",
+             "litellm_params": {"model": "gpu/1/tiny"},
+             "model_info": {"id": "gpu1/tiny"}}],
+        "router_settings": {"routing_strategy": "round-robin"},
+    })
+    loop = asyncio.new_event_loop()
+    ready = threading.Event()
+    holder: dict = {}
+
+    async def main():
+        holder["stop"] = asyncio.Event()
+        registry = WorkerRegistry()
+        for i in range(2):
+            c = RpcWorkerClient(f"gpu:{i}", {"tiny"}, socks[i])
+            c.proc = _spawn_cpu_worker(socks[i], f"gpu:{i}")
+            c.respawn = (lambda s=socks[i], l=f"gpu:{i}":
+                         _spawn_cpu_worker(s, l))
+            await c.connect(timeout=120)
+            registry.register("gpu", str(i), c)
+            holder[f"w{i}"] = c
+        app = GatewayApp(cfg, registry, health_interval_s=0.3)
+        server = HttpServer(app.handle, host="127.0.0.1", port=port)
+        await server.start()
+        await app.start_background()
+        ready.set()
+        await holder["stop"].wait()
+        await app.stop_background()
+        await server.stop()
+        await registry.close()
+
+    th = threading.Thread(target=lambda: loop.run_until_complete(main()),
+                          daemon=True)
+    th.start()
+    assert ready.wait(timeout=150)
+    base = f"http://127.0.0.1:{port}"
+    client = OpenAIClient(base)
+    old_pid = holder["w0"].proc.pid
+    results: dict = {}
+    errors: list = []
+
+    def do(i):
+        try:
+            results[i] = client.chat.completions.create(
+                model="tiny-serve",
+                messages=[{"role": "user", "content": "restart me " * 20}],
+                max_tokens=300, timeout=120,
+                extra_headers={"x-request-id": f"rr-{i}"})
+        except Exception as e:                        # noqa: BLE001
+            errors.append(repr(e))
+
+    threads = [threading.Thread(target=do, args=(i,)) for i in range(4)]
+    for t in threads:
+        t.start()
+    time.sleep(0.4)
+    try:
+        st, body = _post(base, "/admin/restart",
+                         {"worker": "gpu:0", "timeout_s": 90,
+                          "migrate_to": "gpu:1"})
+        assert st == 202 and body["migrate_to"] == "gpu:1"
+        for t in threads:
+            t.join(timeout=120)
+        assert not errors, errors
+        assert len(results) == 4
+        deadline = time.monotonic() + 90
+        while time.monotonic() < deadline:
+            p = holder["w0"].proc
+            if p is not None and p.pid != old_pid and p.poll() is None:
+                break
+            time.sleep(0.3)
+        assert holder["w0"].proc.pid != old_pid
+    finally:
+        loop.call_soon_threadsafe(holder["stop"].set)
+        th.join(timeout=30)
+        for i in range(2):
+            w = holder.get(f"w{i}")
+            if w is not None and w.proc is not None and w.proc.poll() is None:
+                w.proc.kill()
