@@ -132,6 +132,8 @@ class GatewayApp:
         self.health_interval_s = health_interval_s
         self.respawn_cooldown_s = 10.0
         self._restart_tasks: dict = {}
+        self.migration_stats = {"migrated": 0, "failed": 0, "sweeps": 0,
+                                "restarts": 0}
         self.last_health: dict = {}     # worker key -> last health dict
         self._health_task: Optional[asyncio.Task] = None
         self.started_at = time.time()
@@ -863,6 +865,7 @@ class GatewayApp:
                     try:
                         await target.migrate_in(blob)
                         migrated.append(req_id)
+                        self.migration_stats["migrated"] += 1
                     finally:
                         # release the blocked client only once the
                         # state landed (or demonstrably failed): the
@@ -870,6 +873,8 @@ class GatewayApp:
                         await worker.release_migrated(req_id)
                 except (WorkerError, asyncio.TimeoutError) as e:
                     errors.append(f"{req_id}: {e!r}")
+                    self.migration_stats["failed"] += 1
+            self.migration_stats["sweeps"] += 1
         return Response.json_response({
             "worker": device, "draining": draining,
             "migrated": migrated, "migrate_errors": errors,
@@ -925,6 +930,7 @@ class GatewayApp:
         migrate_to = body.get("migrate_to")
         if migrate_to and migrate_to not in self.registry.all():
             return Response.error(404, f"no worker {migrate_to!r}")
+        self.migration_stats["restarts"] += 1
         task = asyncio.get_running_loop().create_task(
             self._rolling_restart(device, worker, timeout_s, migrate_to))
         self._restart_tasks[device] = task
@@ -1003,6 +1009,8 @@ class GatewayApp:
         st = self.ledger.stats()
         for k in ("total", "ok", "errors", "throttled", "fallbacks"):
             lines.append(f'gateway_ledger_{k} {st[k]}')
+        for k, v in self.migration_stats.items():
+            lines.append(f'gateway_migration_{k} {v}')
         for key, h in self.last_health.items():
             wl = f'worker="{key}"'
             for field in ("in_flight", "queued", "running",
