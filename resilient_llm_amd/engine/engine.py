@@ -522,11 +522,14 @@ class LLMEngine:
                           and state.get("block_size") == self.block_size
                           and seq.output_ids
                           and seq.n_cached >= len(seq.prompt_ids))
-                nb_need = seq.blocks_needed(self.block_size)
-                if self.spec_lookup:
-                    nb_need += -(-self.spec_lookup // self.block_size)
-                seq.blocks = self.kv.allocate(nb_need)
                 if usable:
+                    # allocate ONLY on the direct-to-running path; the
+                    # waiting path re-allocates in _admit (allocating
+                    # in both places leaked the first reservation)
+                    nb_need = seq.blocks_needed(self.block_size)
+                    if self.spec_lookup:
+                        nb_need += -(-self.spec_lookup // self.block_size)
+                    seq.blocks = self.kv.allocate(nb_need)
                     k_src, v_src = kv_pair
                     nb = k_src.shape[1]
                     idx = torch.tensor(seq.blocks[:nb], dtype=torch.long,
@@ -539,7 +542,11 @@ class LLMEngine:
                 else:
                     seq.n_cached = 0
                     seq.output_ids = []
-                    self.waiting.append(seq)
+                    if self.enable_prefix_caching:
+                        seq.block_keys = block_hash_chain(
+                            seq.prompt_ids, self.block_size)
+                    with self._queue_lock:
+                        self.waiting.append(seq)
                 with self._queue_lock:
                     self._live.add(rid)
                     self._adopt_results[rid] = "ok"
