@@ -161,6 +161,21 @@ def _parse_deployment(entry: Any, idx: int) -> Deployment:
                  f"model_list[{idx}].{label} must be a positive int, got {v!r}")
     weight = entry.get("weight", rpm if isinstance(rpm, int) else 1)
     extra = {k: v for k, v in lp.items() if k != "model"}
+    # validate the per-deployment engine options up front (config-fuzz
+    # tier: a typo'd value must fail at load, not as a dead worker)
+    if "kv_dtype" in extra:
+        _require(extra["kv_dtype"] in ("bf16", "fp8"),
+                 f"model_list[{idx}].litellm_params.kv_dtype must be "
+                 f"bf16|fp8, got {extra['kv_dtype']!r}")
+    if "quantization" in extra:
+        _require(extra["quantization"] in ("fp8",),
+                 f"model_list[{idx}].litellm_params.quantization must be "
+                 f"fp8, got {extra['quantization']!r}")
+    if "spec_lookup" in extra:
+        _require(isinstance(extra["spec_lookup"], int)
+                 and 0 <= extra["spec_lookup"] <= 16,
+                 f"model_list[{idx}].litellm_params.spec_lookup must be "
+                 f"an int in [0, 16], got {extra['spec_lookup']!r}")
     return Deployment(model_name=name, model=model, model_id=model_id,
                       rpm=rpm, tpm=tpm, weight=int(weight), params=extra)
 

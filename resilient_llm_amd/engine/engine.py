@@ -491,6 +491,15 @@ class LLMEngine:
                     break
             with self._queue_lock:
                 self._live.discard(rid)
+                # prune unclaimed extractions (caller timed out before
+                # the state landed): they hold CPU copies of KV
+                now = time.monotonic()
+                for k in [k for k, v in self._extracted.items()
+                          if isinstance(v, dict)
+                          and now - v.get("_t", now) > 120]:
+                    self._extracted.pop(k, None)
+                if isinstance(state, dict):
+                    state["_t"] = now
                 # "missing" (not None): a finished/unknown rid must be
                 # distinguishable from not-yet-extracted, or the
                 # worker's poll waits its full timeout for every
@@ -549,6 +558,8 @@ class LLMEngine:
                         self.waiting.append(seq)
                 with self._queue_lock:
                     self._live.add(rid)
+                    if len(self._adopt_results) > 256:  # unclaimed
+                        self._adopt_results.clear()
                     self._adopt_results[rid] = "ok"
                 self._batch_dirty = True
             except Exception as e:                        # noqa: BLE001

@@ -184,3 +184,21 @@ def test_config_fuzz_always_config_error():
     for data in bad:
         with pytest.raises(ConfigError):
             load_config(data=data)
+
+
+def test_engine_option_validation():
+    import pytest as _pytest
+    from resilient_llm_amd.config import ConfigError, load_config
+
+    def cfg(**lp):
+        return {"cluster": {"port": 4100},
+                "model_list": [{"model_name": "m",
+                                "litellm_params": {"model": "stub/0/tiny",
+                                                   **lp}}],
+                "router_settings": {}}
+
+    load_config(data=cfg(kv_dtype="fp8", quantization="fp8", spec_lookup=4))
+    for bad in (cfg(kv_dtype="int4"), cfg(quantization="awq"),
+                cfg(spec_lookup="four"), cfg(spec_lookup=99)):
+        with _pytest.raises(ConfigError):
+            load_config(data=bad)
