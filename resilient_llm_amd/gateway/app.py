@@ -849,28 +849,51 @@ class GatewayApp:
             if target is None:
                 return Response.error(404, f"no worker {target_dev!r}")
             deadline = time.monotonic() + float(body.get("timeout_s", 30.0))
+            # evacuating to a dead/booting target would burn the whole
+            # deadline on doomed RPCs while the source sits draining —
+            # leave the requests in place instead (a drain tolerates
+            # that by definition)
+            target_ok = True
             try:
-                ids = await worker.list_requests()
-            except (WorkerError, AttributeError,
-                    asyncio.TimeoutError) as e:
-                ids = []
-                errors.append(str(e) or "list_requests timed out")
+                await asyncio.wait_for(target.health(), timeout=5.0)
+            except Exception as e:                        # noqa: BLE001
+                target_ok = False
+                errors.append(f"target {target_dev} unhealthy, evacuation "
+                              f"skipped: {e!r:.120}")
+            ids = []
+            if target_ok:
+                try:
+                    ids = await worker.list_requests()
+                except (WorkerError, AttributeError,
+                        asyncio.TimeoutError) as e:
+                    errors.append(str(e) or "list_requests timed out")
+
+            def _left() -> float:
+                # every awaited migrate op is clamped to the deadline:
+                # ONE hung RPC must not blow past timeout_s (r02 GPU
+                # soak: migrate_in spinning in reconnect held the drain
+                # >60 s while clients saw "no deployment")
+                return max(0.5, deadline - time.monotonic())
             for req_id in ids:
                 if time.monotonic() > deadline:
                     errors.append(f"evacuation deadline: "
                                   f"{len(ids) - len(migrated)} left in place")
                     break
                 try:
-                    blob = await worker.migrate_out(req_id)
+                    blob = await asyncio.wait_for(
+                        worker.migrate_out(req_id), timeout=_left())
                     try:
-                        await target.migrate_in(blob)
+                        await asyncio.wait_for(
+                            target.migrate_in(blob), timeout=_left())
                         migrated.append(req_id)
                         self.migration_stats["migrated"] += 1
                     finally:
                         # release the blocked client only once the
                         # state landed (or demonstrably failed): the
                         # re-routed attach then always finds it
-                        await worker.release_migrated(req_id)
+                        await asyncio.wait_for(
+                            worker.release_migrated(req_id),
+                            timeout=max(2.0, _left()))
                 except (WorkerError, asyncio.TimeoutError) as e:
                     errors.append(f"{req_id}: {e!r}")
                     self.migration_stats["failed"] += 1
@@ -1011,6 +1034,8 @@ class GatewayApp:
             lines.append(f'gateway_ledger_{k} {st[k]}')
         for k, v in self.migration_stats.items():
             lines.append(f'gateway_migration_{k} {v}')
+        lines.append(f'gateway_last_resort_routes_total '
+                     f'{self.router.last_resort_total}')
         for key, h in self.last_health.items():
             wl = f'worker="{key}"'
             for field in ("in_flight", "queued", "running",

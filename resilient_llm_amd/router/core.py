@@ -111,6 +111,7 @@ class Router:
         for s in self.states:
             self._by_alias.setdefault(s.dep.model_name, []).append(s)
         self._rr_counters: dict[str, int] = {}   # alias -> round-robin cursor
+        self.last_resort_total = 0   # acquires that bypassed drain/cooldown
 
     # ------------------------------------------------------------- lookup
     def alias_states(self, alias: str) -> list[DeploymentState]:
@@ -126,11 +127,21 @@ class Router:
         return None
 
     # ----------------------------------------------------------- policies
-    def _available(self, s: DeploymentState, tokens: int, exclude: set) -> bool:
-        if id(s) in exclude or not s.healthy or s.draining:
+    def _available(self, s: DeploymentState, tokens: int, exclude: set,
+                   last_resort: bool = False) -> bool:
+        if id(s) in exclude or not s.healthy:
             return False
-        if s.cooldown_until > self._clock():
-            return False
+        if not last_resort:
+            # drain and cooldown are ADVISORY exclusions: when they are
+            # the only thing standing between a request and a healthy
+            # deployment, the last-resort pass ignores them — a drained
+            # or cooling replica still serves correctly, while "no
+            # deployment" is a guaranteed client-visible failure.  (An
+            # r02 chaos soak collapsed to 2.3% exactly this way: one
+            # replica dead, the other held draining by migration sweeps
+            # -> hours of instant 429s.)
+            if s.draining or s.cooldown_until > self._clock():
+                return False
         if self.settings.enable_pre_call_checks and not s.limiter.would_admit(tokens):
             return False
         return True
@@ -170,9 +181,11 @@ class Router:
     # ------------------------------------------------------------ acquire
     def _try_alias(self, alias: str, tokens: int, exclude: set,
                    attempted: list,
-                   affinity_key: Optional[str] = None) -> Optional[DeploymentState]:
+                   affinity_key: Optional[str] = None,
+                   last_resort: bool = False) -> Optional[DeploymentState]:
         states = self.alias_states(alias)
-        candidates = [s for s in states if self._available(s, tokens, exclude)]
+        candidates = [s for s in states
+                      if self._available(s, tokens, exclude, last_resort)]
         while candidates:
             s = self._shuffle_pick(alias, candidates, affinity_key)
             if s.limiter.try_acquire(tokens):
@@ -203,6 +216,19 @@ class Router:
                     continue
                 if s is not None:
                     return self._issue(s, alias, tokens_estimate, True, attempted)
+            # last-resort pass: primary alias then fallbacks, ignoring
+            # drain/cooldown (still healthy + within rate limits)
+            for lr_alias in [alias] + self.settings.fallbacks.get(alias, []):
+                try:
+                    s = self._try_alias(lr_alias, tokens_estimate, exclude,
+                                        attempted, affinity_key,
+                                        last_resort=True)
+                except UnknownAlias:
+                    continue
+                if s is not None:
+                    self.last_resort_total += 1
+                    return self._issue(s, alias, tokens_estimate,
+                                       lr_alias != alias, attempted)
         raise RouterRateLimit(
             f"no deployment available for {alias!r} "
             f"(rate limits / cooldowns exhausted; tried fallbacks "

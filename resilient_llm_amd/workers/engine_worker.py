@@ -374,8 +374,24 @@ class EngineWorker(Worker):
         if state is None:
             raise WorkerError(f"extraction of {request_id!r} timed out")
         if state == "missing":
+            # finished/unknown in the engine: also drop any stale
+            # adopted-entry so the NEXT sweep's list_requests no longer
+            # offers this id (an r02 soak retried the same ghost id in
+            # every sweep for its full 600 s prune window)
+            with self._adopt_lock:
+                e = self._adopted.pop(request_id, None)
+                if e:
+                    self._adopted_by_rid.pop(e["rid"], None)
             raise WorkerError(f"no live request {request_id!r} "
                               "(finished before extraction)")
+        # the state has LEFT this engine: an adopted-unattached entry
+        # here is now a ghost (its tokens travel in the blob; the target
+        # registers a fresh entry on migrate_in) — drop it so later
+        # sweeps/attaches don't find a dead id
+        with self._adopt_lock:
+            e = self._adopted.pop(request_id, None)
+            if e:
+                self._adopted_by_rid.pop(e["rid"], None)
         # NOTE: the blocked client is NOT released here — the caller
         # releases it via release_migrated() AFTER the target has
         # adopted, so the re-routed attach always finds the state

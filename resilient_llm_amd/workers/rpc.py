@@ -218,7 +218,13 @@ class RpcWorkerClient(Worker):
 
     async def _call(self, op: str, data) -> tuple[int, asyncio.Queue]:
         if self._writer is None:
-            await self.connect(timeout=600.0)
+            # mid-operation reconnect: keep this SHORT.  A dead worker
+            # raises WorkerDead immediately (poll() check in connect);
+            # a BOOTING respawn can take longer than any caller should
+            # block — respawn_now() owns the long reconnect (900 s), and
+            # callers see a fast WorkerDead instead of a hung RPC (an
+            # r02 soak hung /admin/drain for minutes exactly here).
+            await self.connect(timeout=15.0)
         self._next_id += 1
         mid = self._next_id
         q: asyncio.Queue = asyncio.Queue()

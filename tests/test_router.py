@@ -84,11 +84,42 @@ def test_cooldown_after_allowed_fails():
     r.fail(t1)
     t2 = r.acquire("primary", 1)
     r.fail(t2)
-    # two fails -> cooldown; deployment out of rotation
-    with pytest.raises(RouterRateLimit):
-        r.acquire("primary", 1)
+    # two fails -> cooldown.  With no alternative, the LAST-RESORT pass
+    # still serves the cooling (healthy, within-limits) deployment
+    # rather than guaranteeing a client-visible 429 (r02 chaos-soak
+    # lesson: cooling out the last replica turned blips into outages).
+    before = r.last_resort_total
+    t3 = r.acquire("primary", 1)
+    assert t3.deployment.model_id == "gpu0/primary"
+    assert r.last_resort_total == before + 1
+    r.complete(t3)
     clk.t = 16.0
-    assert r.acquire("primary", 1).deployment.model_id == "gpu0/primary"
+    t4 = r.acquire("primary", 1)
+    assert t4.deployment.model_id == "gpu0/primary"
+    assert r.last_resort_total == before + 1  # cooldown over: normal path
+
+
+def test_cooldown_skips_when_peer_available():
+    clk = FakeClock(0.0)
+    r = make_router(clock=clk, fallbacks={}, allowed_fails=1,
+                    cooldown_time=15.0)
+    t = r.acquire("lb", 1)
+    cooled = t.deployment.model_id
+    r.fail(t)
+    # the cooling replica is out of rotation while its peer serves
+    # (2 < the peer's rpm window: stays on the NORMAL path throughout)
+    for _ in range(2):
+        t2 = r.acquire("lb", 1)
+        assert t2.deployment.model_id != cooled
+        r.complete(t2)
+    assert r.last_resort_total == 0
+
+
+def test_unhealthy_still_raises():
+    r = make_router(fallbacks={})
+    r.set_healthy("gpu0/primary", False)
+    with pytest.raises(RouterRateLimit):
+        r.acquire("primary", 1)  # last-resort never routes to unhealthy
 
 
 def test_fail_refunds_rate_window():
