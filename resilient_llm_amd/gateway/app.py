@@ -35,7 +35,7 @@ from ..router.core import (
 )
 from ..router.token_bucket import MinuteWindowLimiter
 from ..workers.base import (
-    GenerationRequest, GenerationResult, Worker, WorkerError,
+    GenerationRequest, GenerationResult, Worker, WorkerDead, WorkerError,
     WorkerRegistry, WorkerMigrated, WorkerThrottled,
 )
 
@@ -186,16 +186,47 @@ class GatewayApp:
         (with a cooldown) and rejoins rotation once healthy
         (SURVEY.md §5.3)."""
         respawning: set = set()
+        fail_streak: dict = {}
         while True:
             await asyncio.sleep(self.health_interval_s)
             for key, worker in self.registry.all().items():
                 try:
-                    h = await asyncio.wait_for(worker.health(),
-                                               timeout=self.health_interval_s)
+                    # probe timeout is NOT the poll interval: a busy
+                    # worker (GIL-heavy engine step, co-located load)
+                    # can miss a sub-second deadline while being
+                    # perfectly alive — r02 chaos soaks flapped healthy
+                    # workers out of rotation exactly this way
+                    h = await asyncio.wait_for(
+                        worker.health(),
+                        timeout=max(5.0, self.health_interval_s))
                     self.last_health[key] = h
                     ok = True
-                except Exception:
-                    ok = False
+                    fail_streak[key] = 0
+                except Exception as e:
+                    streak = fail_streak.get(key, 0) + 1
+                    fail_streak[key] = streak
+                    proc = getattr(worker, "proc", None)
+                    proc_dead = proc is not None and proc.poll() is not None
+                    # an explicit dead signal (WorkerDead) or a dead
+                    # process is immediate; a slow/errored probe with a
+                    # LIVE process is a blip until it repeats
+                    ok = (not (proc_dead or isinstance(e, WorkerDead))
+                          and streak < 2)
+                    if (streak >= 8 and proc is not None
+                            and proc.poll() is None
+                            and getattr(worker, "respawn", None) is not None):
+                        # wedged-but-alive process (hung loop, stuck
+                        # engine): unresponsive for ~8 probes straight —
+                        # terminate it so the dead-proc respawn path
+                        # below recovers it (r02: a hung worker sat
+                        # unhealthy forever because respawn only fired
+                        # for DEAD processes)
+                        from ..utils.logging import log_with_timestamp
+                        log_with_timestamp(
+                            f"worker {key} unresponsive "
+                            f"({streak} probes) — terminating for "
+                            f"respawn", "red")
+                        proc.terminate()
                 if ok and getattr(worker, "proc_group", None):
                     # a dead FOLLOWER strands the collective while the
                     # leader still answers health RPCs: detect it, take

@@ -271,3 +271,80 @@ def test_rpc_server_survives_malformed_frames():
             await w.close()
 
     asyncio.run(run())
+
+
+def test_wedged_worker_terminated_and_respawned():
+    """A worker whose PROCESS is alive but which stops answering health
+    probes (hung loop / wedged engine) is terminated by the health loop
+    after a streak of failed probes and respawned — r02 chaos soaks left
+    such workers unhealthy forever because respawn only fired for dead
+    processes."""
+    sock = os.path.join(tempfile.mkdtemp(prefix="rlli-wedge-"), "w.sock")
+    port = free_port()
+    cfg = load_config(data={
+        "cluster": {"port": port},
+        "model_list": [
+            {"model_name": "tiny-serve",
+             "litellm_params": {"model": "gpu/0/tiny"},
+             "model_info": {"id": "gpu0/tiny"}}],
+        "router_settings": {"routing_strategy": "simple-shuffle"},
+    })
+    loop = asyncio.new_event_loop()
+    ready = threading.Event()
+    holder = {}
+
+    async def main():
+        holder["stop"] = asyncio.Event()
+        registry = WorkerRegistry()
+        client = RpcWorkerClient("gpu:0", {"tiny"}, sock)
+        client.proc = spawn_cpu_worker(sock)
+        client.respawn = lambda: spawn_cpu_worker(sock)
+        await client.connect(timeout=120)
+        registry.register("gpu", "0", client)
+        holder["client"] = client
+        app = GatewayApp(cfg, registry, health_interval_s=0.2)
+        app.respawn_cooldown_s = 0.0
+        server = HttpServer(app.handle, port=port)
+        await server.start()
+        await app.start_background()
+        ready.set()
+        await holder["stop"].wait()
+        await app.stop_background()
+        await server.stop()
+        await registry.close()
+
+    th = threading.Thread(target=lambda: loop.run_until_complete(main()),
+                          daemon=True)
+    th.start()
+    assert ready.wait(120)
+    try:
+        http = OpenAIClient(f"http://127.0.0.1:{port}")
+        msgs = [{"role": "user", "content": "hello"}]
+        r = http.chat.completions.create(model="tiny-serve", messages=msgs,
+                                         max_tokens=3)
+        assert r.usage.completion_tokens == 3
+
+        # 'hang' fault: process stays ALIVE but health probes fail
+        first_proc = holder["client"].proc
+        http.inject_fault("gpu:0", "hang")
+        assert first_proc.poll() is None
+
+        # streak >= 8 -> terminate -> dead-proc respawn path -> serving
+        deadline = time.time() + 90
+        ok = False
+        while time.time() < deadline:
+            try:
+                r = http.chat.completions.create(model="tiny-serve",
+                                                 messages=msgs, max_tokens=3,
+                                                 timeout=10)
+                if holder["client"].proc is not first_proc:
+                    ok = True
+                    break
+            except APIError:
+                pass
+            time.sleep(1.0)
+        assert ok, "wedged worker was never terminated+respawned"
+        assert first_proc.poll() is not None, "old process still alive"
+    finally:
+        loop.call_soon_threadsafe(holder["stop"].set)
+        th.join(timeout=15)
