@@ -54,6 +54,7 @@ def main(duration_s=420, n_threads=24):
         yaml.safe_dump(CONFIG, f)
     args = SimpleNamespace(base_url=None, config=cfgfile, gpu=True)
     stats = {"ok": 0, "err": 0, "stream_ok": 0, "stream_err": 0}
+    err_kinds: dict = {}
     lock = threading.Lock()
     stop = threading.Event()
 
@@ -87,9 +88,15 @@ def main(duration_s=420, n_threads=24):
                             top_p=rng.choice([1.0, 0.9]), timeout=120)
                         with lock:
                             stats["ok"] += 1
-                except Exception:
+                except Exception as e:                 # noqa: BLE001
                     with lock:
                         stats["stream_err" if stream else "err"] += 1
+                        key = f"{type(e).__name__}:{str(e)[:70]}"
+                        err_kinds[key] = err_kinds.get(key, 0) + 1
+                    # real clients back off on failure; without this a
+                    # brief outage turns into a tight error loop that
+                    # swamps the gateway (and the error counts)
+                    time.sleep(0.2)
 
         threads = [threading.Thread(target=worker, args=(i,), daemon=True)
                    for i in range(n_threads)]
@@ -98,6 +105,7 @@ def main(duration_s=420, n_threads=24):
         t0 = time.time()
         killed = 0
         migrations = 0
+        sweeps_ok = 0
         mig_on = os.environ.get("RLLI_SOAK_MIGRATE") == "1"
         last_mig = 0.0
         import json as _json
@@ -129,7 +137,8 @@ def main(duration_s=420, n_threads=24):
                 try:
                     try:
                         body = _post("/admin/drain", {"worker": src,
-                                                      "migrate_to": dst})
+                                                      "migrate_to": dst,
+                                                      "timeout_s": 20})
                     finally:
                         # NEVER leave a replica draining: a failed sweep
                         # plus a later sweep of the OTHER replica would
@@ -137,11 +146,15 @@ def main(duration_s=420, n_threads=24):
                         _post("/admin/drain", {"worker": src,
                                                "drain": False})
                     migrations += 1
+                    sweeps_ok += 1
                     print(f"[soak {el:.0f}s] migrated "
                           f"{len(body['migrated'])} reqs {src}->{dst} "
                           f"(errors: {len(body['migrate_errors'])})",
                           flush=True)
                 except Exception as e:             # noqa: BLE001
+                    # count failed sweeps too so src alternates instead
+                    # of hammering the same (possibly sick) replica
+                    migrations += 1
                     print(f"[soak {el:.0f}s] migration sweep failed: {e}",
                           flush=True)
             with lock:
@@ -156,9 +169,11 @@ def main(duration_s=420, n_threads=24):
         total = sum(stats.values())
         ok = stats["ok"] + stats["stream_ok"]
         print(f"SOAK DONE: {ok}/{total} ok ({100*ok/max(total,1):.1f}%), "
-              f"{killed} kills+respawns, {migrations} migration sweeps; "
+              f"{killed} kills+respawns, {sweeps_ok}/{migrations} migration sweeps ok; "
               f"in_flight={inflight}; "
               f"healthy={healthy}", flush=True)
+        for k, v in sorted(err_kinds.items(), key=lambda kv: -kv[1])[:8]:
+            print(f"  {v:6d}  {k}", flush=True)
 
 
 if __name__ == "__main__":
