@@ -558,8 +558,14 @@ class LLMEngine:
                         self.waiting.append(seq)
                 with self._queue_lock:
                     self._live.add(rid)
-                    if len(self._adopt_results) > 256:  # unclaimed
-                        self._adopt_results.clear()
+                    if len(self._adopt_results) > 256:
+                        # drop the OLDEST unclaimed half (insertion
+                        # order): clearing everything could wipe a
+                        # result a concurrent migrate_in is polling
+                        # for, making a successful adoption report
+                        # "timed out" while the sequence runs headless
+                        for k in list(self._adopt_results)[:128]:
+                            self._adopt_results.pop(k, None)
                     self._adopt_results[rid] = "ok"
                 self._batch_dirty = True
             except Exception as e:                        # noqa: BLE001
