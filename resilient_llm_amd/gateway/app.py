@@ -226,7 +226,13 @@ class GatewayApp:
                             f"worker {key} unresponsive "
                             f"({streak} probes) — terminating for "
                             f"respawn", "red")
-                        proc.terminate()
+                        if streak >= 12:
+                            # SIGTERM can't reach a process wedged in a
+                            # C extension (Python only runs handlers at
+                            # bytecode boundaries) — escalate
+                            proc.kill()
+                        else:
+                            proc.terminate()
                 if ok and getattr(worker, "proc_group", None):
                     # a dead FOLLOWER strands the collective while the
                     # leader still answers health RPCs: detect it, take
