@@ -934,6 +934,16 @@ class GatewayApp:
                 except (WorkerError, asyncio.TimeoutError) as e:
                     errors.append(f"{req_id}: {e!r}")
                     self.migration_stats["failed"] += 1
+                    # best-effort release: a migrate_out cancelled
+                    # mid-extraction leaves the client blocked on a
+                    # request whose state already left the engine —
+                    # releasing re-routes it NOW (it regenerates from
+                    # its sticky seed) instead of hanging to timeout
+                    try:
+                        await asyncio.wait_for(
+                            worker.release_migrated(req_id), timeout=5.0)
+                    except Exception:                 # noqa: BLE001
+                        pass
             self.migration_stats["sweeps"] += 1
         return Response.json_response({
             "worker": device, "draining": draining,
