@@ -747,3 +747,137 @@ def test_drain_skips_evacuation_to_dead_target():
             w = holder.get(f"w{i}")
             if w is not None and w.proc is not None and w.proc.poll() is None:
                 w.proc.kill()
+
+
+def test_drain_survives_source_death_mid_evacuation():
+    """The SOURCE worker dies while its drain-evacuation sweep is in
+    flight: the admin call still returns (bounded), and every client
+    request completes — either migrated, or failed over with stateless
+    replay after the death."""
+    import json as _json
+    import os
+    import subprocess
+    import sys
+    import tempfile
+    import threading
+    import time
+    import urllib.request
+
+    import asyncio
+    from resilient_llm_amd.client import OpenAIClient
+    from resilient_llm_amd.config import load_config
+    from resilient_llm_amd.gateway.app import GatewayApp
+    from resilient_llm_amd.gateway.http import HttpServer
+    from resilient_llm_amd.workers.base import WorkerRegistry
+    from resilient_llm_amd.workers.rpc import RpcWorkerClient
+    from tests.gateway_harness import free_port
+
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+    def spawn(sock, label):
+        env = dict(os.environ)
+        env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+        return subprocess.Popen(
+            [sys.executable, "-m", "resilient_llm_amd.workers.gpu_main",
+             "--device-label", label, "--model", "tiny", "--socket", sock,
+             "--device", "cpu", "--num-blocks", "96"], env=env)
+
+    run_dir = tempfile.mkdtemp(prefix="rlli-evacdie-")
+    socks = [os.path.join(run_dir, f"w{i}.sock") for i in range(2)]
+    port = free_port()
+    cfg = load_config(data={
+        "cluster": {"port": port},
+        "model_list": [
+            {"model_name": "tiny-serve",
+             "litellm_params": {"model": "gpu/0/tiny"},
+             "model_info": {"id": "gpu0/tiny"}},
+            {"model_name": "tiny-serve",
+             "litellm_params": {"model": "gpu/1/tiny"},
+             "model_info": {"id": "gpu1/tiny"}}],
+        "router_settings": {"routing_strategy": "round-robin"},
+    })
+    loop = asyncio.new_event_loop()
+    ready = threading.Event()
+    holder: dict = {}
+
+    async def main():
+        holder["stop"] = asyncio.Event()
+        registry = WorkerRegistry()
+        for i in range(2):
+            c = RpcWorkerClient(f"gpu:{i}", {"tiny"}, socks[i])
+            c.proc = spawn(socks[i], f"gpu:{i}")
+            await c.connect(timeout=120)
+            registry.register("gpu", str(i), c)
+            holder[f"w{i}"] = c
+        app = GatewayApp(cfg, registry, health_interval_s=0.3)
+        server = HttpServer(app.handle, host="127.0.0.1", port=port)
+        await server.start()
+        await app.start_background()
+        ready.set()
+        await holder["stop"].wait()
+        await app.stop_background()
+        await server.stop()
+        await registry.close()
+
+    th = threading.Thread(target=lambda: loop.run_until_complete(main()),
+                          daemon=True)
+    th.start()
+    assert ready.wait(timeout=150)
+    base = f"http://127.0.0.1:{port}"
+    client = OpenAIClient(base)
+    msgs = [{"role": "user", "content": "hold the line " * 15}]
+    results: dict = {}
+    errors: list = []
+
+    def do(i):
+        try:
+            results[i] = client.chat.completions.create(
+                model="tiny-serve", messages=msgs, max_tokens=300,
+                timeout=120, extra_headers={"x-request-id": f"dv-{i}"})
+        except Exception as e:                        # noqa: BLE001
+            errors.append(repr(e))
+
+    threads = [threading.Thread(target=do, args=(i,)) for i in range(6)]
+    for t in threads:
+        t.start()
+    time.sleep(0.4)
+    drain_result: dict = {}
+
+    def drain():
+        req = urllib.request.Request(
+            base + "/admin/drain", method="POST",
+            data=_json.dumps({"worker": "gpu:0", "migrate_to": "gpu:1",
+                              "timeout_s": 20}).encode(),
+            headers={"content-type": "application/json"})
+        try:
+            with urllib.request.urlopen(req, timeout=60) as r:
+                drain_result.update(_json.loads(r.read().decode()))
+        except Exception as e:                        # noqa: BLE001
+            drain_result["exc"] = repr(e)
+
+    try:
+        dt = threading.Thread(target=drain)
+        dt.start()
+        time.sleep(0.1)                # let the sweep get going
+        holder["w0"].proc.kill()       # source dies MID-evacuation
+        dt.join(timeout=90)
+        assert not dt.is_alive(), "drain call never returned"
+        assert "exc" not in drain_result, drain_result
+        for t in threads:
+            t.join(timeout=120)
+        assert len(results) == 6, f"lost requests: {errors}"
+        for i, r in results.items():
+            assert r.usage.completion_tokens == 300, (i, r.usage)
+        # undrain (source will be respawned by the health loop)
+        urllib.request.urlopen(urllib.request.Request(
+            base + "/admin/drain", method="POST",
+            data=_json.dumps({"worker": "gpu:0",
+                              "drain": False}).encode(),
+            headers={"content-type": "application/json"}), timeout=30)
+    finally:
+        loop.call_soon_threadsafe(holder["stop"].set)
+        th.join(timeout=30)
+        for i in range(2):
+            w = holder.get(f"w{i}")
+            if w is not None and w.proc is not None and w.proc.poll() is None:
+                w.proc.kill()
