@@ -199,8 +199,6 @@ def main(duration_s=60, n_threads=8):
     for t in threads:
         t.join(timeout=70)
     time.sleep(1)
-    state = _post("/admin/fault", {"worker": "gpu:0", "mode": "none"}) \
-        if False else None  # noqa: F841
     with urllib.request.urlopen(base + "/admin/router", timeout=15) as r:
         rows = {d["model_id"]: d
                 for d in json.loads(r.read().decode())["deployments"]}
