@@ -231,3 +231,49 @@ def test_prefix_affinity_without_key_falls_back_to_shuffle():
         picks.add(t.state.dep.model_id)
         r.complete(t, 10)
     assert len(picks) == 3
+
+
+def test_property_acquire_never_starves_while_served():
+    """Hypothesis: under ANY sequence of drain/cooldown/heal operations,
+    acquire('lb') succeeds as long as at least one lb replica is healthy
+    and rate-limit-admissible — drain and cooldown alone can never
+    produce a client-visible 'no deployment' (the last-resort
+    invariant)."""
+    from hypothesis import given, settings as hsettings, strategies as st
+
+    ops = st.lists(
+        st.tuples(
+            st.sampled_from(["drain", "undrain", "cool", "heal", "sick",
+                             "acquire"]),
+            st.sampled_from(["gpu0/r0", "gpu1/r1"])),
+        min_size=1, max_size=40)
+
+    @hsettings(max_examples=200, deadline=None)
+    @given(ops)
+    def run(seq):
+        clk = FakeClock(0.0)
+        r = make_router(clock=clk, fallbacks={}, allowed_fails=1,
+                        cooldown_time=30.0)
+        for op, dep in seq:
+            if op == "drain":
+                r.set_draining(dep, True)
+            elif op == "undrain":
+                r.set_draining(dep, False)
+            elif op == "cool":
+                s = r.state_for_id(dep)
+                s.cooldown_until = clk() + 30.0
+            elif op == "heal":
+                r.set_healthy(dep, True)
+            elif op == "sick":
+                r.set_healthy(dep, False)
+            else:
+                healthy = [s for s in r.alias_states("lb")
+                           if s.healthy and s.limiter.would_admit(1)]
+                if healthy:
+                    t = r.acquire("lb", 1)     # must NOT raise
+                    r.complete(t, actual_tokens=0)  # refund: rpm stays open
+                else:
+                    with pytest.raises(RouterRateLimit):
+                        r.acquire("lb", 1)
+
+    run()
