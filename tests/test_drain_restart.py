@@ -177,14 +177,19 @@ def test_rolling_restart_zero_downtime():
             t.join(timeout=30)
         assert holder["w0"].proc.pid != old_pid, "worker was not restarted"
         assert not errors, f"requests failed during rolling restart: {errors[:3]}"
-        # the restarted worker serves again (drain cleared)
+        # the restarted worker returns to rotation once the restart flow
+        # marks it healthy and undrains it — poll rather than sample a
+        # fixed burst (the new pid can appear moments before the undrain)
         devices = set()
-        for i in range(16):
+        deadline = time.monotonic() + 60
+        i = 0
+        while time.monotonic() < deadline and "gpu:0" not in devices:
             r = client.chat.completions.create(
                 model="tiny-serve",
                 messages=[{"role": "user", "content": f"post{i}"}],
                 max_tokens=4, timeout=30)
             devices.add(r.headers.get("x-gateway-device"))
+            i += 1
         assert "gpu:0" in devices
     finally:
         stop_load.set()
