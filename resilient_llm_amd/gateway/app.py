@@ -873,6 +873,7 @@ class GatewayApp:
         if worker is None:
             return Response.error(404, f"no worker {device!r}")
         draining = bool(body.get("drain", True))
+        t_drain0 = time.monotonic()
         self._set_drain(worker, draining)
         migrated: list = []
         errors: list = []
@@ -948,6 +949,7 @@ class GatewayApp:
         return Response.json_response({
             "worker": device, "draining": draining,
             "migrated": migrated, "migrate_errors": errors,
+            "elapsed_s": round(time.monotonic() - t_drain0, 3),
             "in_flight": self._worker_in_flight(worker)})
 
     async def admin_migrate(self, req: Request) -> Response:
