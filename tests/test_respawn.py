@@ -348,3 +348,93 @@ def test_wedged_worker_terminated_and_respawned():
     finally:
         loop.call_soon_threadsafe(holder["stop"].set)
         th.join(timeout=15)
+
+
+def test_full_outage_both_workers_respawn():
+    """BOTH worker processes die at once (total outage): the health
+    loop respawns both, and serving resumes — the cluster self-heals
+    from zero capacity without operator action."""
+    run_dir = tempfile.mkdtemp(prefix="rlli-outage-")
+    socks = [os.path.join(run_dir, f"w{i}.sock") for i in range(2)]
+    port = free_port()
+    cfg = load_config(data={
+        "cluster": {"port": port},
+        "model_list": [
+            {"model_name": "tiny-serve",
+             "litellm_params": {"model": "gpu/0/tiny"},
+             "model_info": {"id": "gpu0/tiny"}},
+            {"model_name": "tiny-serve",
+             "litellm_params": {"model": "gpu/1/tiny"},
+             "model_info": {"id": "gpu1/tiny"}}],
+        "router_settings": {"routing_strategy": "simple-shuffle"},
+    })
+    loop = asyncio.new_event_loop()
+    ready = threading.Event()
+    holder: dict = {}
+
+    def spawn_one(sock, label):
+        env = dict(os.environ)
+        env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+        return subprocess.Popen(
+            [sys.executable, "-m", "resilient_llm_amd.workers.gpu_main",
+             "--device-label", label, "--model", "tiny", "--socket", sock,
+             "--device", "cpu", "--num-blocks", "64"], env=env)
+
+    async def main():
+        holder["stop"] = asyncio.Event()
+        registry = WorkerRegistry()
+        for i in range(2):
+            c = RpcWorkerClient(f"gpu:{i}", {"tiny"}, socks[i])
+            c.proc = spawn_one(socks[i], f"gpu:{i}")
+            c.respawn = (lambda s=socks[i], l=f"gpu:{i}": spawn_one(s, l))
+            await c.connect(timeout=120)
+            registry.register("gpu", str(i), c)
+            holder[f"w{i}"] = c
+        app = GatewayApp(cfg, registry, health_interval_s=0.3)
+        app.respawn_cooldown_s = 0.0
+        server = HttpServer(app.handle, port=port)
+        await server.start()
+        await app.start_background()
+        ready.set()
+        await holder["stop"].wait()
+        await app.stop_background()
+        await server.stop()
+        await registry.close()
+
+    th = threading.Thread(target=lambda: loop.run_until_complete(main()),
+                          daemon=True)
+    th.start()
+    assert ready.wait(120)
+    try:
+        http = OpenAIClient(f"http://127.0.0.1:{port}")
+        msgs = [{"role": "user", "content": "outage"}]
+        r = http.chat.completions.create(model="tiny-serve", messages=msgs,
+                                         max_tokens=3)
+        assert r.usage.completion_tokens == 3
+        old = [holder["w0"].proc, holder["w1"].proc]
+        for p in old:
+            p.kill()
+        for p in old:
+            p.wait(timeout=10)
+        # total outage -> self-heal -> serving resumes
+        deadline = time.time() + 90
+        ok = False
+        while time.time() < deadline:
+            try:
+                r = http.chat.completions.create(
+                    model="tiny-serve", messages=msgs, max_tokens=3,
+                    timeout=10)
+                ok = True
+                break
+            except APIError:
+                time.sleep(0.5)
+        assert ok, "cluster never recovered from total outage"
+        assert holder["w0"].proc is not old[0]
+        assert holder["w1"].proc is not old[1]
+    finally:
+        loop.call_soon_threadsafe(holder["stop"].set)
+        th.join(timeout=15)
+        for i in range(2):
+            w = holder.get(f"w{i}")
+            if w is not None and w.proc is not None and w.proc.poll() is None:
+                w.proc.kill()
